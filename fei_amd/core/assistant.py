@@ -1,0 +1,271 @@
+"""The Assistant: one chat turn = send -> extract tool calls -> execute ->
+continuation.
+
+Parity targets (reference fei/core/assistant.py):
+  - constructor shape ``Assistant(config, api_key, model, provider,
+    tool_registry, mcp_manager)`` (assistant.py:320-358)
+  - ``chat(message, system_prompt)`` does exactly ONE tool round per call
+    (assistant.py:440-489; multi-step behavior comes from TaskExecutor)
+  - conversation role shapes per the Anthropic format with tool_use /
+    tool_result blocks (assistant.py:266-303)
+  - ``ask()`` — documented in the reference README but never implemented
+    (README.md:155 vs assistant.py); here it is real: an iterating turn
+    that keeps executing tool rounds until the model answers in plain text
+    (bounded by max_tool_rounds).
+
+Reference defect NOT replicated: the continuation response's tool calls
+were silently ignored (assistant.py:629-670); ``ask()`` handles them.
+"""
+
+from __future__ import annotations
+
+import time
+import uuid
+from typing import Any, Dict, List, Optional
+
+from fei_amd.core.backends import Backend, BackendResponse, create_backend
+from fei_amd.utils.config import Config, get_config
+from fei_amd.utils.logging import get_logger
+
+logger = get_logger("core.assistant")
+
+DEFAULT_SYSTEM_PROMPT = (
+    "You are Fei, a coding assistant running locally on AMD Instinct MI355X "
+    "GPUs. Use the available tools to inspect and edit files, search code, "
+    "and manage memories. Answer concisely."
+)
+
+
+class ProviderManager:
+    """Resolve provider/model/api-key into a Backend
+    (reference: assistant.py:25-111, re-aimed at local backends)."""
+
+    def __init__(self, config: Optional[Config] = None,
+                 provider: Optional[str] = None,
+                 model: Optional[str] = None,
+                 api_key: Optional[str] = None,
+                 **backend_kwargs):
+        self.config = config or get_config()
+        self.provider = provider or self.config.get("llm.provider", "local")
+        self.model = model or self.config.get("llm.model", "llama3-8b")
+        self.api_key = api_key or self.config.get("llm.api_key", "")
+        self._backend_kwargs = backend_kwargs
+        self._backend: Optional[Backend] = None
+
+    def get_backend(self) -> Backend:
+        if self._backend is None:
+            self._backend = create_backend(self.provider, self.model,
+                                           **self._backend_kwargs)
+        return self._backend
+
+    def set_backend(self, backend: Backend) -> None:
+        self._backend = backend
+
+
+class ToolManager:
+    """Expose registry tools to the backend and execute calls
+    (reference: assistant.py:114-212)."""
+
+    def __init__(self, tool_registry=None):
+        self.registry = tool_registry
+
+    def get_tools(self) -> Optional[List[Dict[str, Any]]]:
+        if self.registry is None:
+            return None
+        schemas = self.registry.get_schemas()
+        return schemas or None
+
+    def execute_tool(self, name: str, args: Dict[str, Any]) -> Dict[str, Any]:
+        if self.registry is None:
+            return {"error": "no tool registry configured"}
+        return self.registry.execute_tool(name, args)
+
+
+class ConversationManager:
+    """Holds the message list in the Anthropic block format
+    (reference: assistant.py:215-303)."""
+
+    def __init__(self):
+        self.messages: List[Dict[str, Any]] = []
+
+    def add_user_message(self, content: str) -> None:
+        self.messages.append({"role": "user", "content": content})
+
+    def add_assistant_message(self, content: str,
+                              tool_calls: Optional[List[Dict[str, Any]]] = None) -> None:
+        if tool_calls:
+            blocks: List[Dict[str, Any]] = []
+            if content:
+                blocks.append({"type": "text", "text": content})
+            for tc in tool_calls:
+                blocks.append({"type": "tool_use", "id": tc["id"],
+                               "name": tc["name"], "input": tc.get("input", {})})
+            self.messages.append({"role": "assistant", "content": blocks})
+        else:
+            self.messages.append({"role": "assistant", "content": content})
+
+    def add_tool_results(self, results: List[Dict[str, Any]]) -> None:
+        """results: [{"tool_use_id", "content"}]"""
+        blocks = [{"type": "tool_result", "tool_use_id": r["tool_use_id"],
+                   "content": r["content"]} for r in results]
+        self.messages.append({"role": "user", "content": blocks})
+
+    def clear(self) -> None:
+        self.messages.clear()
+
+    def last_text(self) -> str:
+        for msg in reversed(self.messages):
+            if msg["role"] == "assistant":
+                content = msg["content"]
+                if isinstance(content, str):
+                    return content
+                return "\n".join(b.get("text", "") for b in content
+                                 if isinstance(b, dict) and b.get("type") == "text")
+        return ""
+
+    def scrape_tool_output(self) -> str:
+        """Recover the latest tool_result text (the reference scraped the
+        conversation tail when the model answered empty, cli.py:240-264)."""
+        for msg in reversed(self.messages):
+            content = msg.get("content")
+            if isinstance(content, list):
+                parts = [str(b.get("content", "")) for b in content
+                         if isinstance(b, dict) and b.get("type") == "tool_result"]
+                if parts:
+                    return "\n".join(parts)
+        return ""
+
+
+class Assistant:
+    """One agent. ``chat()`` is the single-turn primitive (one tool round);
+    ``ask()`` iterates tool rounds until a plain-text answer."""
+
+    def __init__(
+        self,
+        config: Optional[Config] = None,
+        api_key: Optional[str] = None,
+        model: Optional[str] = None,
+        provider: Optional[str] = None,
+        tool_registry=None,
+        mcp_manager=None,
+        backend: Optional[Backend] = None,
+        **backend_kwargs,
+    ):
+        self.config = config or get_config()
+        self.providers = ProviderManager(self.config, provider, model, api_key,
+                                         **backend_kwargs)
+        if backend is not None:
+            self.providers.set_backend(backend)
+        self.tools = ToolManager(tool_registry)
+        self.conversation = ConversationManager()
+        self.mcp_manager = mcp_manager
+        if mcp_manager is not None and tool_registry is not None:
+            try:
+                mcp_manager.attach_registry(tool_registry)
+            except AttributeError:
+                pass
+        self.max_tokens = self.config.get_typed("llm.max_tokens", 4000)
+        self.temperature = self.config.get_typed("llm.temperature", 0.0)
+        # per-turn metrics (prefill/decode tok/s, tool latency) — SURVEY §5
+        self.turn_metrics: List[Dict[str, Any]] = []
+
+    # -- internals -----------------------------------------------------------
+
+    def _send(self, system_prompt: Optional[str]) -> BackendResponse:
+        backend = self.providers.get_backend()
+        return backend.complete(
+            self.conversation.messages,
+            tools=self.tools.get_tools(),
+            system=system_prompt or DEFAULT_SYSTEM_PROMPT,
+            max_tokens=self.max_tokens,
+            temperature=self.temperature,
+        )
+
+    def process_tool_calls(self, tool_calls: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
+        """Execute tool calls, timing each (reference: assistant.py:392-438)."""
+        results = []
+        for tc in tool_calls:
+            t0 = time.perf_counter()
+            result = self.tools.execute_tool(tc["name"], tc.get("input", {}))
+            dt = time.perf_counter() - t0
+            logger.debug("tool %s took %.3fs", tc["name"], dt)
+            content = result if isinstance(result, str) else _stringify_result(result)
+            results.append({"tool_use_id": tc["id"], "content": content,
+                            "latency_s": dt, "name": tc["name"]})
+        return results
+
+    def _one_round(self, system_prompt: Optional[str]) -> BackendResponse:
+        """send -> (maybe) tools -> continuation. Returns the final response
+        of this round."""
+        t0 = time.perf_counter()
+        response = self._send(system_prompt)
+        metrics: Dict[str, Any] = {"llm_s": time.perf_counter() - t0,
+                                   "usage": response.usage, "tools": []}
+        if response.tool_calls:
+            self.conversation.add_assistant_message(response.content, response.tool_calls)
+            results = self.process_tool_calls(response.tool_calls)
+            metrics["tools"] = [{"name": r["name"], "latency_s": r["latency_s"]}
+                                for r in results]
+            self.conversation.add_tool_results(
+                [{"tool_use_id": r["tool_use_id"], "content": r["content"]}
+                 for r in results])
+            t1 = time.perf_counter()
+            response = self._send(system_prompt)
+            metrics["continuation_s"] = time.perf_counter() - t1
+        metrics["total_s"] = time.perf_counter() - t0
+        self.turn_metrics.append(metrics)
+        return response
+
+    # -- public API ----------------------------------------------------------
+
+    def chat(self, message: str, system_prompt: Optional[str] = None) -> str:
+        """One turn with at most ONE tool round (reference: assistant.py:440-489).
+        Tool calls in the continuation are recorded but not executed."""
+        self.conversation.add_user_message(message)
+        response = self._one_round(system_prompt)
+        self.conversation.add_assistant_message(response.content, response.tool_calls)
+        answer = response.content
+        if not answer:
+            answer = self.conversation.scrape_tool_output()
+        return answer
+
+    def ask(self, message: str, system_prompt: Optional[str] = None,
+            max_tool_rounds: int = 8) -> str:
+        """Iterating turn: keep executing tool rounds until the model
+        answers without tool calls (or the round cap is hit)."""
+        self.conversation.add_user_message(message)
+        response = self._send(system_prompt)
+        rounds = 0
+        while response.tool_calls and rounds < max_tool_rounds:
+            self.conversation.add_assistant_message(response.content, response.tool_calls)
+            results = self.process_tool_calls(response.tool_calls)
+            self.conversation.add_tool_results(
+                [{"tool_use_id": r["tool_use_id"], "content": r["content"]}
+                 for r in results])
+            response = self._send(system_prompt)
+            rounds += 1
+        self.conversation.add_assistant_message(response.content, response.tool_calls)
+        answer = response.content
+        if not answer:
+            answer = self.conversation.scrape_tool_output()
+        return answer
+
+    def reset(self) -> None:
+        self.conversation.clear()
+        self.turn_metrics.clear()
+
+    def close(self) -> None:
+        backend = self.providers._backend
+        if backend is not None:
+            backend.close()
+
+
+def _stringify_result(result: Any, limit: int = 20000) -> str:
+    import json
+    try:
+        text = json.dumps(result, default=str)
+    except (TypeError, ValueError):
+        text = str(result)
+    if len(text) > limit:
+        text = text[:limit] + "...[truncated]"
+    return text
