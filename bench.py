@@ -1,0 +1,134 @@
+#!/usr/bin/env python3
+"""fei_amd flagship benchmark — agent decode throughput (BASELINE.json).
+
+Measures the headline metric "agent tok/s": Llama-3-8B bf16 running locally,
+one agent per GPU (weak scaling over N GPUs, matching the reference's
+one-assistant-per-process model promoted to one-agent-per-GPU). A step is
+ONE decode token through the full device-resident step: embedding, 32x
+(RMSNorm, QKV GEMM, fused RoPE+KV-append, split-K decode attention, O GEMM,
+RMSNorm, gate/up GEMM, SwiGLU, down GEMM), final norm, lm_head, sampling
+kernel, position advance — hipGraph-replayed.
+
+Contract: `python bench.py --gpus N --steps K --warmup W`; for N>1 the
+driver launches one rank per GPU via torch.distributed.run. Rank 0 prints
+one JSON line with the whole-job aggregate.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def main() -> int:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=256)
+    p.add_argument("--warmup", type=int, default=32)
+    p.add_argument("--model", default="llama3-8b")
+    p.add_argument("--batch", type=int, default=1, help="agent sessions per GPU")
+    p.add_argument("--prompt-len", type=int, default=512)
+    p.add_argument("--temperature", type=float, default=0.0)
+    p.add_argument("--no-graph", action="store_true")
+    args = p.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    n_gpus = max(args.gpus, world)
+
+    from fei_amd.engine.config import get_spec
+    from fei_amd.engine.engine import LocalEngine
+    from fei_amd.parallel import pg
+
+    has_gpu = torch.cuda.is_available()
+    ctx = pg.init_from_env() if world > 1 else pg.ParallelContext()
+    device = torch.device(f"cuda:{ctx.local_rank}") if has_gpu else torch.device("cpu")
+
+    spec = get_spec(args.model)
+    max_seq = min(spec.max_seq_len,
+                  args.prompt_len + args.warmup + args.steps + 64)
+    engine = LocalEngine(
+        spec, device=device, batch_size=args.batch, max_seq_len=max_seq,
+        use_hip_graph=(has_gpu and not args.no_graph), seed=1234 + rank,
+    )
+    engine.temperature = args.temperature
+
+    # synthetic prompt of the configured shape (random-init weights; no
+    # network for datasets/checkpoints)
+    rng = torch.Generator().manual_seed(99 + rank)
+    prompt_ids = torch.randint(4, 260, (args.prompt_len,), generator=rng).tolist()
+
+    engine.ensure_graph()               # capture outside the prefill timing
+    tp0 = time.perf_counter()
+    engine.prefill(prompt_ids)
+    if has_gpu:
+        torch.cuda.synchronize(device)
+    prefill_tok_s = args.prompt_len / max(time.perf_counter() - tp0, 1e-9)
+
+    def run_steps(n: int) -> None:
+        if engine._graph is not None:
+            for _ in range(n):
+                engine._graph.replay()
+        else:
+            for _ in range(n):
+                engine._decode_step()
+
+    run_steps(args.warmup)
+
+    # timed region: barrier + sync on both sides, exactly K steps
+    pg.barrier(ctx)
+    if has_gpu:
+        torch.cuda.synchronize(device)
+    t0 = time.perf_counter()
+    run_steps(args.steps)
+    if has_gpu:
+        torch.cuda.synchronize(device)
+    t1 = time.perf_counter()
+    pg.barrier(ctx)
+
+    elapsed = torch.tensor([t1 - t0], dtype=torch.float64)
+    if ctx.is_distributed:
+        import torch.distributed as dist
+        elapsed_d = elapsed.to(device) if ctx.backend == "nccl" else elapsed
+        dist.all_reduce(elapsed_d, op=dist.ReduceOp.MAX)
+        elapsed = elapsed_d.cpu()
+    t_max = float(elapsed[0])
+
+    total_tokens = args.steps * args.batch * (world if world > 1 else 1)
+    value = total_tokens / t_max
+    ms_per_step = t_max / args.steps * 1000.0
+
+    if rank == 0:
+        result = {
+            "metric": f"agent tok/s ({args.model} local decode, 1 agent/GPU)",
+            "value": round(value, 2),
+            "unit": "tok/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if has_gpu else "fp32-cpu",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * (world if world > 1 else 1),
+                "seq_len": args.prompt_len,
+                "parallelism": f"dp{n_gpus} (1 agent per GPU, weak scaling)",
+                "hip_graph": engine._graph is not None,
+                "prefill_tok_s": round(prefill_tok_s, 1),
+            },
+        }
+        print(json.dumps(result))
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,233 @@
+"""LocalEngine: the MI355X inference engine behind the Assistant.
+
+The decode step (the agent-tok/s hot path) is fully device-resident:
+embedding -> 32x(RMSNorm, QKV GEMM, fused RoPE+KV-append, split-K decode
+attention, O GEMM, RMSNorm, gate/up GEMM, SwiGLU, down GEMM) -> final norm
+-> lm_head GEMV -> sampling kernel -> position advance — sampling writes the
+next input token and the out_tokens ring ON DEVICE, so the whole step is
+captured once into a hipGraph (torch.cuda.CUDAGraph) and replayed with zero
+host work per token. EOS is checked every `eos_check_every` replays (one
+device->host copy per chunk, not per token).
+
+Replaces the reference's remote-API call site
+(fei/core/assistant.py:527-530) per BASELINE.json's north star.
+"""
+
+from __future__ import annotations
+
+import math
+import time
+from typing import Dict, List, Optional, Union
+
+import torch
+
+from fei_amd import ops
+from fei_amd.engine.config import ModelSpec, get_spec
+from fei_amd.engine.tokenizer import ByteTokenizer
+from fei_amd.models.llama import LlamaModel
+from fei_amd.parallel.pg import ParallelContext
+from fei_amd.utils.logging import get_logger
+
+logger = get_logger("engine.engine")
+
+SAMPLE_CHUNKS = 64       # stage-1 blocks for the sampling kernel
+
+
+class LocalEngine:
+    def __init__(
+        self,
+        spec: ModelSpec,
+        device: Optional[torch.device] = None,
+        batch_size: int = 1,
+        max_seq_len: Optional[int] = None,
+        use_hip_graph: Optional[bool] = None,
+        tp: Optional[ParallelContext] = None,
+        seed: int = 1234,
+        attn_splits: int = 16,
+        dtype: Optional[torch.dtype] = None,
+    ):
+        self.spec = spec
+        if device is None:
+            device = torch.device("cuda:0") if torch.cuda.is_available() else torch.device("cpu")
+        self.device = device
+        self.is_gpu = device.type == "cuda"
+        if dtype is None:
+            dtype = torch.bfloat16 if self.is_gpu else torch.float32
+        self.dtype = dtype
+        self.B = batch_size
+        self.max_seq_len = max_seq_len or spec.max_seq_len
+        self.tp = tp or ParallelContext()
+        if use_hip_graph is None:
+            use_hip_graph = self.is_gpu and not self.tp.is_distributed
+        self.use_graph = use_hip_graph and self.is_gpu
+        self.attn_splits = attn_splits
+        self.seed = seed
+        self.temperature = 0.0       # graph-captured; set before capture
+        self.tokenizer = ByteTokenizer()
+
+        if self.is_gpu and not ops.kernels_available():
+            ops.require_lib()        # fail loudly: no eager fallback on GPU
+
+        t0 = time.perf_counter()
+        self.model = LlamaModel(spec, device, dtype, tp=self.tp, seed=seed,
+                                max_seq_len=self.max_seq_len)
+        self.k_caches, self.v_caches = self.model.new_kv_cache(self.B, self.max_seq_len)
+        logger.info("model %s init in %.1fs (%.2f GB params)", spec.name,
+                    time.perf_counter() - t0, self.model.param_bytes() / 2**30)
+
+        # static decode state (device-resident; graph-stable)
+        self.token = torch.zeros(self.B, dtype=torch.int32, device=device)
+        self.pos = torch.zeros(self.B, dtype=torch.int32, device=device)
+        self.step = torch.zeros(1, dtype=torch.int32, device=device)
+        self.out_tokens = torch.zeros(self.B, self.max_seq_len,
+                                      dtype=torch.int32, device=device)
+        self.sample_ws = torch.zeros(self.B, SAMPLE_CHUNKS, 2,
+                                     dtype=torch.float32, device=device)
+        Hq_l, D = self.model.hq_l, self.model.D
+        self.attn_ws = (
+            torch.zeros(self.B, Hq_l, attn_splits, D, dtype=torch.float32, device=device),
+            torch.zeros(self.B, Hq_l, attn_splits, 2, dtype=torch.float32, device=device),
+        )
+        self._graph: Optional[torch.cuda.CUDAGraph] = None
+        self.last_metrics: Dict[str, float] = {}
+
+    # -- construction helpers ------------------------------------------------
+
+    @classmethod
+    def create(cls, model: Union[str, ModelSpec] = "llama3-8b", **kwargs) -> "LocalEngine":
+        spec = get_spec(model) if isinstance(model, str) else model
+        return cls(spec, **kwargs)
+
+    # -- decode step ---------------------------------------------------------
+
+    def _decode_step(self) -> None:
+        logits = self.model.forward_decode(
+            self.token, self.pos, self.k_caches, self.v_caches,
+            attn_splits=self.attn_splits, workspace=self.attn_ws)
+        ops.sample(logits, self.token, self.step, self.sample_ws.view(self.B, -1),
+                   out_tokens=self.out_tokens, temperature=self.temperature,
+                   seed=self.seed, nchunks=SAMPLE_CHUNKS)
+        ops.advance(self.pos, self.step)
+
+    def _capture_graph(self) -> None:
+        assert self.is_gpu
+        # state must be valid during warmup/capture: pretend one token exists
+        self.pos.fill_(1)
+        self.step.zero_()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._decode_step()
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._decode_step()
+        torch.cuda.synchronize()
+        self._graph = g
+        logger.info("decode step captured into hipGraph")
+
+    def ensure_graph(self) -> None:
+        if self.use_graph and self._graph is None:
+            t0 = time.perf_counter()
+            self._capture_graph()
+            logger.info("graph capture took %.2fs", time.perf_counter() - t0)
+
+    # -- public API ----------------------------------------------------------
+
+    def prefill(self, token_ids: List[int]) -> None:
+        """Reset state and prefill the prompt; samples the first new token."""
+        S = len(token_ids)
+        if S >= self.max_seq_len:
+            token_ids = token_ids[-(self.max_seq_len - 1):]
+            S = len(token_ids)
+        self.ensure_graph()
+        tokens = torch.tensor([token_ids] * self.B, dtype=torch.int64,
+                              device=self.device)
+        pos0 = torch.zeros(self.B, dtype=torch.int32, device=self.device)
+        logits = self.model.forward_prefill(tokens, pos0, self.k_caches,
+                                            self.v_caches)
+        self.step.zero_()
+        self.out_tokens.zero_()
+        ops.sample(logits.to(self.dtype), self.token, self.step,
+                   self.sample_ws.view(self.B, -1), out_tokens=self.out_tokens,
+                   temperature=self.temperature, seed=self.seed,
+                   nchunks=SAMPLE_CHUNKS)
+        self.pos.fill_(S)
+        if self.is_gpu:
+            torch.cuda.synchronize(self.device)
+        else:
+            self.step += 1   # reference sample() does not advance step
+            return
+        # step was advanced by... (sample kernel does not touch step); advance
+        # host-side: first generated token occupies out_tokens[:,0]
+        self.step.fill_(1)
+
+    def decode(self, n_tokens: int, eos_check_every: int = 32,
+               stop_on_eos: bool = True) -> List[List[int]]:
+        """Generate up to n_tokens per sequence; returns new token ids
+        (including the one sampled by prefill)."""
+        done = 1                      # prefill already produced token 0
+        eos = self.tokenizer.eos_id
+        while done < n_tokens:
+            chunk = min(eos_check_every, n_tokens - done)
+            if self._graph is not None:
+                for _ in range(chunk):
+                    self._graph.replay()
+            else:
+                for _ in range(chunk):
+                    self._decode_step()
+            done += chunk
+            if stop_on_eos:
+                toks = self.out_tokens[:, :done].tolist()
+                if all(eos in row for row in toks):
+                    break
+        if self.is_gpu:
+            torch.cuda.synchronize(self.device)
+        rows = self.out_tokens[:, :done].tolist()
+        if stop_on_eos:
+            rows = [row[: row.index(eos) + 1] if eos in row else row
+                    for row in rows]
+        return rows
+
+    def generate(self, prompt: Union[str, List[int]], max_new_tokens: int = 256,
+                 temperature: float = 0.0, stop_on_eos: bool = True
+                 ) -> Dict[str, object]:
+        """Prefill + decode; returns text and timing metrics."""
+        if isinstance(prompt, str):
+            prompt_ids = self.tokenizer.encode(prompt)
+        else:
+            prompt_ids = list(prompt)
+        max_new_tokens = min(max_new_tokens, self.max_seq_len - len(prompt_ids) - 1)
+        if max_new_tokens < 1:
+            max_new_tokens = 1
+        self.temperature = temperature
+        if self.is_gpu:
+            torch.cuda.synchronize(self.device)
+        t0 = time.perf_counter()
+        self.prefill(prompt_ids)
+        t1 = time.perf_counter()
+        rows = self.decode(max_new_tokens, stop_on_eos=stop_on_eos)
+        t2 = time.perf_counter()
+        new_tokens = len(rows[0])
+        decode_s = t2 - t1
+        self.last_metrics = {
+            "prompt_tokens": len(prompt_ids),
+            "new_tokens": new_tokens,
+            "prefill_s": t1 - t0,
+            "decode_s": decode_s,
+            "prefill_tok_s": len(prompt_ids) / max(t1 - t0, 1e-9),
+            "decode_tok_s": (max(new_tokens - 1, 0) * self.B) / max(decode_s, 1e-9),
+        }
+        text = self.tokenizer.decode(rows[0])
+        return {
+            "text": text,
+            "token_ids": rows[0],
+            **self.last_metrics,
+        }
+
+    def shutdown(self) -> None:
+        self._graph = None
+        self.k_caches = self.v_caches = None  # release HBM
+        if self.is_gpu:
+            torch.cuda.empty_cache()

@@ -1,0 +1,225 @@
+"""Llama-3 style decoder built on fei_amd.ops.
+
+Not an nn.Module: weights are plain tensors, the forward passes are
+allocation-light functions designed around the MI355X execution model —
+fused HIP kernels for norm/rope/attention/activation/sampling, rocBLAS
+(hipBLASLt) for the plain GEMMs, one hipGraph-capturable decode step.
+
+Tensor parallel (TP): column-parallel QKV and gate/up, row-parallel O and
+down with one all-reduce each per layer; lm_head column-parallel over the
+vocab with an all-gather of logits.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+import torch.nn.functional as F
+
+from fei_amd import ops
+from fei_amd.engine.config import ModelSpec
+from fei_amd.ops import reference as ref
+from fei_amd.parallel.pg import ParallelContext, all_gather_cat, all_reduce_sum
+
+
+@dataclass
+class LayerWeights:
+    norm_attn: torch.Tensor       # [C]
+    wqkv: torch.Tensor            # [(Hq_l + 2*Hkv_l)*D, C]
+    wo: torch.Tensor              # [C, Hq_l*D]
+    norm_mlp: torch.Tensor        # [C]
+    wgu: torch.Tensor             # [2*I_l, C]   (gate rows then up rows)
+    wdown: torch.Tensor           # [C, I_l]
+
+
+class LlamaModel:
+    def __init__(
+        self,
+        spec: ModelSpec,
+        device: torch.device,
+        dtype: torch.dtype = torch.bfloat16,
+        tp: Optional[ParallelContext] = None,
+        seed: int = 1234,
+        max_seq_len: Optional[int] = None,
+    ):
+        self.spec = spec
+        self.device = device
+        self.dtype = dtype
+        self.tp = tp or ParallelContext()
+        self.max_seq_len = max_seq_len or spec.max_seq_len
+        tp_size = self.tp.world_size if self.tp.is_distributed else 1
+        if spec.num_kv_heads % tp_size or spec.num_heads % tp_size or \
+                spec.intermediate_size % tp_size or spec.vocab_size % tp_size:
+            raise ValueError(f"model {spec.name} not divisible by tp={tp_size}")
+        self.tp_size = tp_size
+        self.tp_rank = self.tp.rank if self.tp.is_distributed else 0
+        self.hq_l = spec.num_heads // tp_size
+        self.hkv_l = spec.num_kv_heads // tp_size
+        self.inter_l = spec.intermediate_size // tp_size
+        self.vocab_l = spec.vocab_size // tp_size
+        self.D = spec.head_dim
+        self._init_weights(seed)
+        self.rope = ref.rope_table(self.max_seq_len, self.D, spec.rope_theta,
+                                   device=device)
+
+    # -- weights -------------------------------------------------------------
+
+    def _rand(self, gen, *shape, scale: float) -> torch.Tensor:
+        w = torch.randn(*shape, generator=gen, device=self.device,
+                        dtype=torch.float32) * scale
+        return w.to(self.dtype)
+
+    def _init_weights(self, seed: int) -> None:
+        """Random init (synthetic weights — BASELINE.json: no network for
+        checkpoints). Each TP rank seeds identically and slices its shard so
+        ranks are consistent without communication."""
+        s = self.spec
+        gen = torch.Generator(device=self.device)
+        gen.manual_seed(seed)
+        C = s.hidden_size
+        sc = 1.0 / math.sqrt(C)
+        # embedding: replicated
+        self.emb = self._rand(gen, s.vocab_size, C, scale=0.02)
+        self.layers: List[LayerWeights] = []
+        r = self.tp_rank
+        for _ in range(s.num_layers):
+            wq = self._rand(gen, s.num_heads * self.D, C, scale=sc)
+            wk = self._rand(gen, s.num_kv_heads * self.D, C, scale=sc)
+            wv = self._rand(gen, s.num_kv_heads * self.D, C, scale=sc)
+            wo = self._rand(gen, C, s.num_heads * self.D, scale=sc / math.sqrt(2 * s.num_layers))
+            wg = self._rand(gen, s.intermediate_size, C, scale=sc)
+            wu = self._rand(gen, s.intermediate_size, C, scale=sc)
+            wd = self._rand(gen, C, s.intermediate_size,
+                            scale=1.0 / math.sqrt(s.intermediate_size) / math.sqrt(2 * s.num_layers))
+            # shard
+            wq = wq.view(s.num_heads, self.D, C)[r * self.hq_l:(r + 1) * self.hq_l].reshape(-1, C)
+            wk = wk.view(s.num_kv_heads, self.D, C)[r * self.hkv_l:(r + 1) * self.hkv_l].reshape(-1, C)
+            wv = wv.view(s.num_kv_heads, self.D, C)[r * self.hkv_l:(r + 1) * self.hkv_l].reshape(-1, C)
+            wo = wo.view(C, s.num_heads, self.D)[:, r * self.hq_l:(r + 1) * self.hq_l].reshape(C, -1)
+            wg = wg[r * self.inter_l:(r + 1) * self.inter_l]
+            wu = wu[r * self.inter_l:(r + 1) * self.inter_l]
+            wd = wd[:, r * self.inter_l:(r + 1) * self.inter_l]
+            self.layers.append(LayerWeights(
+                norm_attn=torch.ones(C, device=self.device, dtype=self.dtype),
+                wqkv=torch.cat([wq, wk, wv], dim=0).contiguous(),
+                wo=wo.contiguous(),
+                norm_mlp=torch.ones(C, device=self.device, dtype=self.dtype),
+                wgu=torch.cat([wg, wu], dim=0).contiguous(),
+                wdown=wd.contiguous(),
+            ))
+        self.norm_f = torch.ones(C, device=self.device, dtype=self.dtype)
+        lm = self._rand(gen, s.vocab_size, C, scale=sc)
+        self.lm_head = lm[r * self.vocab_l:(r + 1) * self.vocab_l].contiguous()
+
+    def param_bytes(self) -> int:
+        total = self.emb.numel() + self.norm_f.numel() + self.lm_head.numel()
+        for lw in self.layers:
+            for t in (lw.norm_attn, lw.wqkv, lw.wo, lw.norm_mlp, lw.wgu, lw.wdown):
+                total += t.numel()
+        return total * self.emb.element_size()
+
+    # -- kv cache ------------------------------------------------------------
+
+    def new_kv_cache(self, batch: int, max_seq: Optional[int] = None):
+        max_seq = max_seq or self.max_seq_len
+        shape = (batch, self.hkv_l, max_seq, self.D)
+        return (
+            [torch.zeros(shape, device=self.device, dtype=self.dtype)
+             for _ in range(self.spec.num_layers)],
+            [torch.zeros(shape, device=self.device, dtype=self.dtype)
+             for _ in range(self.spec.num_layers)],
+        )
+
+    # -- qkv views -----------------------------------------------------------
+
+    def _qkv_views(self, qkv: torch.Tensor, B: int, S: Optional[int] = None):
+        """Strided views into the fused qkv output (no copies; the HIP
+        kernels take batch/token strides)."""
+        D = self.D
+        W = qkv.shape[-1]
+        off_k = self.hq_l * D
+        off_v = off_k + self.hkv_l * D
+        if S is None:
+            q = qkv.as_strided((B, self.hq_l, D), (W, D, 1))
+            k = qkv.as_strided((B, self.hkv_l, D), (W, D, 1), storage_offset=qkv.storage_offset() + off_k)
+            v = qkv.as_strided((B, self.hkv_l, D), (W, D, 1), storage_offset=qkv.storage_offset() + off_v)
+        else:
+            q = qkv.as_strided((B, S, self.hq_l, D), (S * W, W, D, 1))
+            k = qkv.as_strided((B, S, self.hkv_l, D), (S * W, W, D, 1), storage_offset=qkv.storage_offset() + off_k)
+            v = qkv.as_strided((B, S, self.hkv_l, D), (S * W, W, D, 1), storage_offset=qkv.storage_offset() + off_v)
+        return q, k, v
+
+    # -- forward passes ------------------------------------------------------
+
+    def forward_decode(
+        self,
+        token: torch.Tensor,          # [B] int32/int64
+        pos: torch.Tensor,            # [B] int32 (current length; kv written here)
+        k_caches: List[torch.Tensor],
+        v_caches: List[torch.Tensor],
+        attn_splits: int = 16,
+        workspace=None,
+    ) -> torch.Tensor:
+        """One decode step -> logits [B, vocab] (gathered across TP)."""
+        s = self.spec
+        B = token.shape[0]
+        h = F.embedding(token.long(), self.emb)
+        scale = 1.0 / math.sqrt(self.D)
+        for li, lw in enumerate(self.layers):
+            x = ops.rmsnorm(h, lw.norm_attn, s.norm_eps)
+            qkv = F.linear(x, lw.wqkv)
+            q, k, v = self._qkv_views(qkv, B)
+            ops.rope_kv_decode(q, k, v, k_caches[li], v_caches[li], pos, self.rope)
+            att = ops.attn_decode(q, k_caches[li], v_caches[li], pos,
+                                  splits=attn_splits, scale=scale,
+                                  workspace=workspace)
+            o = F.linear(att.reshape(B, -1), lw.wo)
+            all_reduce_sum(o, self.tp)
+            h = h + o
+            x = ops.rmsnorm(h, lw.norm_mlp, s.norm_eps)
+            gu = F.linear(x, lw.wgu)
+            act = ops.swiglu(gu)
+            d = F.linear(act, lw.wdown)
+            all_reduce_sum(d, self.tp)
+            h = h + d
+        x = ops.rmsnorm(h, self.norm_f, s.norm_eps)
+        logits = F.linear(x, self.lm_head)
+        logits = all_gather_cat(logits, dim=-1, ctx=self.tp)
+        return logits
+
+    def forward_prefill(
+        self,
+        tokens: torch.Tensor,         # [B, S]
+        pos0: torch.Tensor,           # [B] int32 (first absolute position)
+        k_caches: List[torch.Tensor],
+        v_caches: List[torch.Tensor],
+    ) -> torch.Tensor:
+        """Prefill S tokens -> logits of the LAST position [B, vocab]."""
+        s = self.spec
+        B, S = tokens.shape
+        h = F.embedding(tokens.long(), self.emb)          # [B,S,C]
+        scale = 1.0 / math.sqrt(self.D)
+        for li, lw in enumerate(self.layers):
+            x = ops.rmsnorm(h, lw.norm_attn, s.norm_eps)
+            qkv = F.linear(x, lw.wqkv)                    # [B,S,W]
+            qkv2 = qkv.view(B * S, -1)
+            q, k, v = self._qkv_views(qkv2, B, S)
+            ops.rope_kv_prefill(q, k, v, k_caches[li], v_caches[li], pos0, self.rope)
+            att = ops.attn_prefill(q, k_caches[li], v_caches[li], pos0, scale=scale)
+            o = F.linear(att.reshape(B, S, -1), lw.wo)
+            all_reduce_sum(o, self.tp)
+            h = h + o
+            x = ops.rmsnorm(h, lw.norm_mlp, s.norm_eps)
+            gu = F.linear(x, lw.wgu)
+            act = ops.swiglu(gu)
+            d = F.linear(act, lw.wdown)
+            all_reduce_sum(d, self.tp)
+            h = h + d
+        last = h[:, -1, :]
+        x = ops.rmsnorm(last, self.norm_f, s.norm_eps)
+        logits = F.linear(x, self.lm_head)
+        logits = all_gather_cat(logits, dim=-1, ctx=self.tp)
+        return logits

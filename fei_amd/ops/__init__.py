@@ -1,0 +1,296 @@
+"""fei_amd.ops — engine ops with a single dispatch rule:
+
+  - CUDA (= ROCm/HIP) tensors -> the in-tree gfx950 kernel library,
+    loaded via ctypes. If the library is missing on a GPU machine, ops
+    FAIL LOUDLY — there is no silent eager fallback on GPU.
+  - CPU tensors -> the fp32-accurate torch reference implementations
+    (fei_amd/ops/reference.py), used by CPU tests and tiny models.
+
+All kernels launch on torch's current CUDA stream, so the decode step can
+be captured into a hipGraph (torch.cuda.CUDAGraph) — see engine/engine.py.
+"""
+
+from __future__ import annotations
+
+import ctypes
+import math
+import os
+from typing import Optional, Tuple
+
+import torch
+
+from fei_amd.ops import reference as ref
+
+_LIB: Optional[ctypes.CDLL] = None
+_LIB_ERR: Optional[str] = None
+_LIB_PATH = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                         "libfei_kernels.so")
+
+_c = ctypes
+_vp, _i, _f, _u64, _l = _c.c_void_p, _c.c_int, _c.c_float, _c.c_ulonglong, _c.c_long
+
+
+def _try_load() -> None:
+    global _LIB, _LIB_ERR
+    if _LIB is not None or _LIB_ERR is not None:
+        return
+    try:
+        lib = ctypes.CDLL(_LIB_PATH)
+    except OSError as e:
+        _LIB_ERR = str(e)
+        return
+    lib.fei_rmsnorm.argtypes = [_vp, _vp, _vp, _i, _i, _f, _vp]
+    lib.fei_add_rmsnorm.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _f, _vp]
+    lib.fei_rope_kv_decode.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp,
+                                       _i, _i, _i, _i, _i, _l, _l, _vp]
+    lib.fei_rope_kv_prefill.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp,
+                                        _i, _i, _i, _i, _i, _i, _l, _l, _vp]
+    lib.fei_attn_decode.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp,
+                                    _i, _i, _i, _i, _i, _i, _f, _l, _vp]
+    lib.fei_attn_decode_combine.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
+    lib.fei_swiglu.argtypes = [_vp, _vp, _l, _i, _vp]
+    lib.fei_sample.argtypes = [_vp, _vp, _vp, _vp, _vp, _i, _i, _i, _f, _u64,
+                               _i, _vp]
+    lib.fei_advance.argtypes = [_vp, _vp, _i, _vp]
+    lib.fei_attn_prefill.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp,
+                                     _i, _i, _i, _i, _i, _i, _f, _i, _l, _vp]
+    lib.fei_mfma_probe.argtypes = [_vp, _vp, _vp, _vp]
+    _LIB = lib
+
+
+def kernels_available() -> bool:
+    _try_load()
+    return _LIB is not None
+
+
+def require_lib() -> ctypes.CDLL:
+    _try_load()
+    if _LIB is None:
+        raise RuntimeError(
+            "fei_amd HIP kernel library not found at "
+            f"{_LIB_PATH} ({_LIB_ERR}). Build it with "
+            "`python -m fei_amd.ops.build` — GPU execution refuses to fall "
+            "back to eager PyTorch.")
+    return _LIB
+
+
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _ptr(t: torch.Tensor) -> int:
+    return t.data_ptr()
+
+
+# ---------------------------------------------------------------------------
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5,
+            out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """RMSNorm over the last dim. x: [..., C] bf16 (GPU) / any float (CPU)."""
+    if not x.is_cuda:
+        return ref.rmsnorm(x, weight, eps)
+    lib = require_lib()
+    x2 = x.contiguous()
+    if out is None:
+        out = torch.empty_like(x2)
+    rows = x2.numel() // x2.shape[-1]
+    lib.fei_rmsnorm(_ptr(out), _ptr(x2), _ptr(weight), rows, x2.shape[-1],
+                    eps, _stream())
+    return out
+
+
+def fused_add_rmsnorm(x: torch.Tensor, residual: torch.Tensor,
+                      weight: torch.Tensor, eps: float = 1e-5,
+                      out: Optional[torch.Tensor] = None
+                      ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """residual += x (in place on GPU); out = rmsnorm(residual)."""
+    if not x.is_cuda:
+        o, res = ref.fused_add_rmsnorm(x, residual, weight, eps)
+        residual.copy_(res)
+        return o, residual
+    lib = require_lib()
+    x2 = x.contiguous()
+    if out is None:
+        out = torch.empty_like(x2)
+    rows = x2.numel() // x2.shape[-1]
+    lib.fei_add_rmsnorm(_ptr(out), _ptr(residual), _ptr(x2), _ptr(weight),
+                        rows, x2.shape[-1], eps, _stream())
+    return out, residual
+
+
+def rope_kv_decode(q, k, v, k_cache, v_cache, pos, table) -> torch.Tensor:
+    """In-place RoPE on q; k roped + v appended into caches at pos[b].
+    q [B,Hq,D], k/v [B,Hkv,D], caches [B,Hkv,max_seq,D], pos [B] int32."""
+    if not q.is_cuda:
+        q_out = ref.rope_kv_decode(q, k, v, k_cache, v_cache, pos, table)
+        q.copy_(q_out)
+        return q
+    lib = require_lib()
+    B, Hq, D = q.shape
+    Hkv = k.shape[1]
+    assert q.stride(2) == 1 and q.stride(1) == D, "head rows must be contiguous"
+    assert k.stride(1) == D and v.stride(1) == D
+    assert k.stride(0) == v.stride(0)
+    lib.fei_rope_kv_decode(_ptr(q), _ptr(k), _ptr(v), _ptr(k_cache),
+                           _ptr(v_cache), _ptr(table), _ptr(pos),
+                           B, Hq, Hkv, D, k_cache.shape[2],
+                           q.stride(0), k.stride(0), _stream())
+    return q
+
+
+def rope_kv_prefill(q, k, v, k_cache, v_cache, pos0, table) -> torch.Tensor:
+    """In-place RoPE on q [B,S,Hq,D]; k roped + v appended at pos0[b]+s."""
+    if not q.is_cuda:
+        q_out = ref.rope_kv_prefill(q, k, v, k_cache, v_cache, pos0, table)
+        q.copy_(q_out)
+        return q
+    lib = require_lib()
+    B, S, Hq, D = q.shape
+    Hkv = k.shape[2]
+    assert q.stride(3) == 1 and q.stride(2) == D
+    assert q.stride(0) == S * q.stride(1), "token rows must be uniform"
+    assert k.stride(2) == D and k.stride(0) == S * k.stride(1)
+    assert k.stride(1) == v.stride(1)
+    lib.fei_rope_kv_prefill(_ptr(q), _ptr(k), _ptr(v), _ptr(k_cache),
+                            _ptr(v_cache), _ptr(table), _ptr(pos0),
+                            B, S, Hq, Hkv, D, k_cache.shape[2],
+                            q.stride(1), k.stride(1), _stream())
+    return q
+
+
+def attn_decode(q, k_cache, v_cache, pos, splits: int = 16,
+                scale: Optional[float] = None,
+                workspace: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
+                out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Single-token GQA attention over n = pos[b]+1 cache entries.
+    q [B,Hq,D] -> out [B,Hq,D]. The length is read on DEVICE (hipGraph)."""
+    B, Hq, D = q.shape
+    scale = scale if scale is not None else 1.0 / math.sqrt(D)
+    if not q.is_cuda:
+        seqlen = pos + 1
+        return ref.attn_decode(q, k_cache, v_cache, seqlen, scale)
+    lib = require_lib()
+    if workspace is None:
+        part_o = torch.empty(B, Hq, splits, D, dtype=torch.float32, device=q.device)
+        part_ml = torch.empty(B, Hq, splits, 2, dtype=torch.float32, device=q.device)
+    else:
+        part_o, part_ml = workspace
+    if out is None:
+        out = torch.empty_like(q)
+    Hkv = k_cache.shape[1]
+    assert q.stride(2) == 1 and q.stride(1) == D
+    lib.fei_attn_decode(_ptr(q), _ptr(k_cache), _ptr(v_cache), _ptr(part_o),
+                        _ptr(part_ml), _ptr(pos), B, Hq, Hkv, D,
+                        k_cache.shape[2], splits, scale, q.stride(0), _stream())
+    lib.fei_attn_decode_combine(_ptr(out), _ptr(part_o), _ptr(part_ml),
+                                B, Hq, D, splits, _stream())
+    return out
+
+
+def attn_prefill(q, k_cache, v_cache, pos0, scale: Optional[float] = None,
+                 causal: bool = True, kv_len: Optional[torch.Tensor] = None,
+                 out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Causal (or bidirectional) GQA attention of q [B,S,Hq,D] over the
+    caches. q token s attends [0, pos0[b]+s] when causal, else [0, kv_len[b])."""
+    B, S, Hq, D = q.shape
+    scale = scale if scale is not None else 1.0 / math.sqrt(D)
+    if not q.is_cuda:
+        if causal:
+            return ref.attn_prefill(q, k_cache, v_cache, pos0, scale)
+        # bidirectional reference: softmax over [0, kv_len)
+        return _ref_bidir(q, k_cache, v_cache, kv_len, scale)
+    lib = require_lib()
+    if out is None:
+        out = torch.empty_like(q)
+    if kv_len is None:
+        kv_len = pos0  # unused when causal
+    Hkv = k_cache.shape[1]
+    assert q.stride(3) == 1 and q.stride(2) == D
+    assert q.stride(0) == S * q.stride(1)
+    assert out.is_contiguous()
+    lib.fei_attn_prefill(_ptr(q), _ptr(k_cache), _ptr(v_cache), _ptr(out),
+                         _ptr(pos0), _ptr(kv_len), B, S, Hq, Hkv, D,
+                         k_cache.shape[2], scale, 1 if causal else 0,
+                         q.stride(1), _stream())
+    return out
+
+
+def _ref_bidir(q, k_cache, v_cache, kv_len, scale):
+    B, S, Hq, D = q.shape
+    Hkv = k_cache.shape[1]
+    G = Hq // Hkv
+    out = torch.empty_like(q)
+    for b in range(B):
+        n = int(kv_len[b])
+        k = k_cache[b, :, :n, :].float()
+        v = v_cache[b, :, :n, :].float()
+        qb = q[b].float().view(S, Hkv, G, D)
+        s = torch.einsum("shgd,hnd->hgsn", qb, k) * scale
+        p = torch.softmax(s, dim=-1)
+        o = torch.einsum("hgsn,hnd->shgd", p, v)
+        out[b] = o.reshape(S, Hq, D).to(q.dtype)
+    return out
+
+
+def swiglu(gate_up: torch.Tensor,
+           out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """gate_up [..., 2*inter] -> [..., inter]: silu(gate) * up."""
+    if not gate_up.is_cuda:
+        return ref.swiglu(gate_up)
+    lib = require_lib()
+    gu = gate_up.contiguous()
+    inter = gu.shape[-1] // 2
+    rows = gu.numel() // gu.shape[-1]
+    if out is None:
+        out = torch.empty(*gu.shape[:-1], inter, dtype=gu.dtype, device=gu.device)
+    lib.fei_swiglu(_ptr(out), _ptr(gu), rows, inter, _stream())
+    return out
+
+
+def sample(logits: torch.Tensor, token: torch.Tensor,
+           step: torch.Tensor, workspace: torch.Tensor,
+           out_tokens: Optional[torch.Tensor] = None,
+           temperature: float = 0.0, seed: int = 0,
+           nchunks: int = 64) -> torch.Tensor:
+    """Greedy (T<=0) or Gumbel-max categorical (T>0) sampling.
+    logits [B,V] bf16 -> token [B] int32 (written in place; also appended to
+    out_tokens[b, step] when given). All state on device (hipGraph-safe)."""
+    B, V = logits.shape
+    if not logits.is_cuda:
+        t = (ref.argmax_sample(logits) if temperature <= 0
+             else ref.gumbel_sample(logits, temperature))
+        token.copy_(t)
+        if out_tokens is not None:
+            st = int(step)
+            for b in range(B):
+                if st < out_tokens.shape[1]:
+                    out_tokens[b, st] = t[b]
+        return token
+    lib = require_lib()
+    max_new = out_tokens.shape[1] if out_tokens is not None else 0
+    lib.fei_sample(_ptr(logits), _ptr(token),
+                   _ptr(out_tokens) if out_tokens is not None else None,
+                   _ptr(step), _ptr(workspace), B, V, nchunks,
+                   temperature, seed, max_new, _stream())
+    return token
+
+
+def advance(pos: torch.Tensor, step: torch.Tensor) -> None:
+    """pos[b] += 1; step += 1 (device-side, graph-capturable)."""
+    if not pos.is_cuda:
+        pos += 1
+        step += 1
+        return
+    lib = require_lib()
+    lib.fei_advance(_ptr(pos), _ptr(step), pos.shape[0], _stream())
+
+
+def mfma_probe(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
+    """GPU-only: 16x32 @ 32x16 through one MFMA with the kernels' fragment
+    maps; falsifies layout assumptions (tests/test_ops_gpu.py)."""
+    lib = require_lib()
+    C = torch.empty(16, 16, dtype=torch.float32, device=A.device)
+    lib.fei_mfma_probe(_ptr(A.contiguous()), _ptr(B.contiguous()), _ptr(C),
+                       _stream())
+    return C

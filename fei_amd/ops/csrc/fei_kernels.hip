@@ -1,0 +1,494 @@
+// fei_amd gfx950 kernels: RMSNorm, RoPE+KV-append, decode attention
+// (split-K online-softmax), SwiGLU, sampling (argmax / Gumbel), position
+// advance. Semantics match fei_amd/ops/reference.py; numerics tests compare
+// against the fp32 torch reference (tests/test_ops_gpu.py).
+//
+// Design notes (per /opt/skills/guides/cdna_hip_programming.md):
+//  - every memory-bound kernel loads bf16 as 16-byte vectors (Guideline 13)
+//  - decode attention reads the length from a DEVICE tensor so the whole
+//    decode step is hipGraph-capturable with a static grid
+//  - block sizes are multiples of 64 (wave64)
+#include "fei_common.h"
+
+extern "C" {
+
+// ---------------------------------------------------------------------------
+// RMSNorm: out[r,:] = x[r,:] * rsqrt(mean(x^2)+eps) * w ; bf16 I/O, f32 acc.
+// One block per row (grid-stride); cols % 8 == 0; cols/8 <= 256*MAX_V8.
+// ---------------------------------------------------------------------------
+#define RMS_MAXV 8   // up to 256 threads * 8 vec8 = 16384 cols
+
+__global__ void __launch_bounds__(256)
+k_rmsnorm(u16* __restrict__ out, const u16* __restrict__ x,
+          const u16* __restrict__ w, int rows, int cols, float eps) {
+  __shared__ float red[4];
+  const int nv = cols >> 3;                       // vec8 per row
+  float vals[RMS_MAXV * 8];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const s16x8* xr = (const s16x8*)(x + (long)row * cols);
+    float ss = 0.f;
+    int cnt = 0;
+    for (int i = threadIdx.x; i < nv; i += blockDim.x, ++cnt) {
+      s16x8 v = xr[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f((u16)v[j]);
+        vals[cnt * 8 + j] = f;
+        ss = fmaf(f, f, ss);
+      }
+    }
+    ss = block_reduce_sum(ss, red);
+    const float inv = rsqrtf(ss / (float)cols + eps);
+    s16x8* orow = (s16x8*)(out + (long)row * cols);
+    const s16x8* wv = (const s16x8*)w;
+    cnt = 0;
+    for (int i = threadIdx.x; i < nv; i += blockDim.x, ++cnt) {
+      s16x8 wvv = wv[i];
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = (short)f2bf(vals[cnt * 8 + j] * inv * bf2f((u16)wvv[j]));
+      orow[i] = o;
+    }
+    __syncthreads();
+  }
+}
+
+void fei_rmsnorm(void* out, const void* x, const void* w, int rows, int cols,
+                 float eps, hipStream_t stream) {
+  int grid = rows < 2048 ? rows : 2048;
+  hipLaunchKernelGGL(k_rmsnorm, dim3(grid), dim3(256), 0, stream,
+                     (u16*)out, (const u16*)x, (const u16*)w, rows, cols, eps);
+}
+
+// ---------------------------------------------------------------------------
+// Fused residual-add + RMSNorm: res = res + x ; out = rmsnorm(res) * w.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256)
+k_add_rmsnorm(u16* __restrict__ out, u16* __restrict__ res,
+              const u16* __restrict__ x, const u16* __restrict__ w,
+              int rows, int cols, float eps) {
+  __shared__ float red[4];
+  const int nv = cols >> 3;
+  float vals[RMS_MAXV * 8];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const s16x8* xr = (const s16x8*)(x + (long)row * cols);
+    s16x8* rr = (s16x8*)(res + (long)row * cols);
+    float ss = 0.f;
+    int cnt = 0;
+    for (int i = threadIdx.x; i < nv; i += blockDim.x, ++cnt) {
+      s16x8 xv = xr[i];
+      s16x8 rv = rr[i];
+      s16x8 nr;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f((u16)rv[j]) + bf2f((u16)xv[j]);
+        u16 b = f2bf(f);
+        nr[j] = (short)b;
+        f = bf2f(b);                                // match reference rounding
+        vals[cnt * 8 + j] = f;
+        ss = fmaf(f, f, ss);
+      }
+      rr[i] = nr;
+    }
+    ss = block_reduce_sum(ss, red);
+    const float inv = rsqrtf(ss / (float)cols + eps);
+    s16x8* orow = (s16x8*)(out + (long)row * cols);
+    const s16x8* wv = (const s16x8*)w;
+    cnt = 0;
+    for (int i = threadIdx.x; i < nv; i += blockDim.x, ++cnt) {
+      s16x8 wvv = wv[i];
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = (short)f2bf(vals[cnt * 8 + j] * inv * bf2f((u16)wvv[j]));
+      orow[i] = o;
+    }
+    __syncthreads();
+  }
+}
+
+void fei_add_rmsnorm(void* out, void* res, const void* x, const void* w,
+                     int rows, int cols, float eps, hipStream_t stream) {
+  int grid = rows < 2048 ? rows : 2048;
+  hipLaunchKernelGGL(k_add_rmsnorm, dim3(grid), dim3(256), 0, stream,
+                     (u16*)out, (u16*)res, (const u16*)x, (const u16*)w,
+                     rows, cols, eps);
+}
+
+// ---------------------------------------------------------------------------
+// RoPE (decode) + KV append. q [B,Hq,D] roped in place at pos[b]; k,v
+// [B,Hkv,D]: k roped and written with v into caches [B,Hkv,max_seq,D].
+// Half-split pairing (Llama): pair (d, d+D/2). cos_sin: [max_seq, D/2, 2] f32.
+// Grid: (Hq+Hkv, B); block D/2 threads (<=64 -> one wave for D=128).
+// ---------------------------------------------------------------------------
+__global__ void k_rope_kv_decode(u16* __restrict__ q, const u16* __restrict__ k,
+                                 const u16* __restrict__ v,
+                                 u16* __restrict__ kc, u16* __restrict__ vc,
+                                 const float* __restrict__ cos_sin,
+                                 const int* __restrict__ pos,
+                                 int B, int Hq, int Hkv, int D, int max_seq,
+                                 long q_bs, long kv_bs) {
+  const int h = blockIdx.x;
+  const int b = blockIdx.y;
+  const int i = threadIdx.x;               // pair index in [0, D/2)
+  const int p = pos[b];
+  const float c = cos_sin[((long)p * (D / 2) + i) * 2 + 0];
+  const float s = cos_sin[((long)p * (D / 2) + i) * 2 + 1];
+  if (h < Hq) {
+    u16* qp = q + (long)b * q_bs + (long)h * D;
+    float x1 = bf2f(qp[i]), x2 = bf2f(qp[i + D / 2]);
+    qp[i] = f2bf(x1 * c - x2 * s);
+    qp[i + D / 2] = f2bf(x2 * c + x1 * s);
+  } else {
+    const int hk = h - Hq;
+    const u16* kp = k + (long)b * kv_bs + (long)hk * D;
+    const u16* vp = v + (long)b * kv_bs + (long)hk * D;
+    u16* kcp = kc + (((long)b * Hkv + hk) * max_seq + p) * D;
+    u16* vcp = vc + (((long)b * Hkv + hk) * max_seq + p) * D;
+    float x1 = bf2f(kp[i]), x2 = bf2f(kp[i + D / 2]);
+    kcp[i] = f2bf(x1 * c - x2 * s);
+    kcp[i + D / 2] = f2bf(x2 * c + x1 * s);
+    vcp[i] = vp[i];
+    vcp[i + D / 2] = vp[i + D / 2];
+  }
+}
+
+void fei_rope_kv_decode(void* q, const void* k, const void* v, void* k_cache,
+                        void* v_cache, const float* cos_sin, const int* pos,
+                        int B, int Hq, int Hkv, int D, int max_seq,
+                        long q_bs, long kv_bs, hipStream_t stream) {
+  hipLaunchKernelGGL(k_rope_kv_decode, dim3(Hq + Hkv, B), dim3(D / 2), 0,
+                     stream, (u16*)q, (const u16*)k, (const u16*)v,
+                     (u16*)k_cache, (u16*)v_cache, cos_sin, pos,
+                     B, Hq, Hkv, D, max_seq, q_bs, kv_bs);
+}
+
+// ---------------------------------------------------------------------------
+// RoPE (prefill) + KV append for S tokens. q [B,S,Hq,D] roped in place;
+// k [B,S,Hkv,D] roped + appended with v at pos0[b]+s.
+// Grid: (S, Hq+Hkv, B); block D/2.
+// ---------------------------------------------------------------------------
+__global__ void k_rope_kv_prefill(u16* __restrict__ q, const u16* __restrict__ k,
+                                  const u16* __restrict__ v,
+                                  u16* __restrict__ kc, u16* __restrict__ vc,
+                                  const float* __restrict__ cos_sin,
+                                  const int* __restrict__ pos0,
+                                  int B, int S, int Hq, int Hkv, int D,
+                                  int max_seq, long q_ts, long kv_ts) {
+  const int s_idx = blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int i = threadIdx.x;
+  const int p = pos0[b] + s_idx;
+  const long tok = (long)b * S + s_idx;
+  const float c = cos_sin[((long)p * (D / 2) + i) * 2 + 0];
+  const float sn = cos_sin[((long)p * (D / 2) + i) * 2 + 1];
+  if (h < Hq) {
+    u16* qp = q + tok * q_ts + (long)h * D;
+    float x1 = bf2f(qp[i]), x2 = bf2f(qp[i + D / 2]);
+    qp[i] = f2bf(x1 * c - x2 * sn);
+    qp[i + D / 2] = f2bf(x2 * c + x1 * sn);
+  } else {
+    const int hk = h - Hq;
+    const u16* kp = k + tok * kv_ts + (long)hk * D;
+    const u16* vp = v + tok * kv_ts + (long)hk * D;
+    u16* kcp = kc + (((long)b * Hkv + hk) * max_seq + p) * D;
+    u16* vcp = vc + (((long)b * Hkv + hk) * max_seq + p) * D;
+    float x1 = bf2f(kp[i]), x2 = bf2f(kp[i + D / 2]);
+    kcp[i] = f2bf(x1 * c - x2 * sn);
+    kcp[i + D / 2] = f2bf(x2 * c + x1 * sn);
+    vcp[i] = vp[i];
+    vcp[i + D / 2] = vp[i + D / 2];
+  }
+}
+
+void fei_rope_kv_prefill(void* q, const void* k, const void* v, void* k_cache,
+                         void* v_cache, const float* cos_sin, const int* pos0,
+                         int B, int S, int Hq, int Hkv, int D, int max_seq,
+                         long q_ts, long kv_ts, hipStream_t stream) {
+  hipLaunchKernelGGL(k_rope_kv_prefill, dim3(S, Hq + Hkv, B), dim3(D / 2), 0,
+                     stream, (u16*)q, (const u16*)k, (const u16*)v,
+                     (u16*)k_cache, (u16*)v_cache, cos_sin, pos0,
+                     B, S, Hq, Hkv, D, max_seq, q_ts, kv_ts);
+}
+
+// ---------------------------------------------------------------------------
+// Decode attention, split-K partials. q [B,Hq,D] bf16 (roped);
+// caches [B,Hkv,max_seq,D]. n = pos[b]+1 keys (length read on DEVICE so the
+// launch shape is static for hipGraph capture).
+// Grid (splits, Hq, B), block 256 = 4 waves.
+//   phase A: thread-per-key scores + block online-softmax state
+//   phase B: threads as (D/2 pairs x key-groups) accumulate P*V
+// Partials: part_o [B,Hq,splits,D] f32; part_ml [B,Hq,splits,2] f32.
+// ---------------------------------------------------------------------------
+#define DEC_TILE 256
+#define DEC_DMAX 128
+
+__global__ void __launch_bounds__(256)
+k_attn_decode(const u16* __restrict__ q, const u16* __restrict__ kc,
+              const u16* __restrict__ vc, float* __restrict__ part_o,
+              float* __restrict__ part_ml, const int* __restrict__ pos,
+              int B, int Hq, int Hkv, int D, int max_seq, int splits,
+              float scale, long q_bs) {
+  const int split = blockIdx.x;
+  const int hq = blockIdx.y;
+  const int b = blockIdx.z;
+  const int G = Hq / Hkv;
+  const int hkv = hq / G;
+  const int tid = threadIdx.x;
+
+  __shared__ float qs[DEC_DMAX];
+  __shared__ float pl[DEC_TILE];
+  __shared__ float red[4];
+  __shared__ float osh[DEC_DMAX / 2][2];
+
+  const int n = pos[b] + 1;
+  const int chunk = (n + splits - 1) / splits;
+  const int start = split * chunk;
+  const int end = min(start + chunk, n);
+
+  float* po = part_o + (((long)b * Hq + hq) * splits + split) * D;
+  float* pml = part_ml + (((long)b * Hq + hq) * splits + split) * 2;
+
+  if (start >= end) {
+    for (int d = tid; d < D; d += blockDim.x) po[d] = 0.f;
+    if (tid == 0) { pml[0] = -1.0f / 0.0f; pml[1] = 0.f; }
+    return;
+  }
+
+  // stage q (scaled) into LDS
+  for (int d = tid; d < D; d += blockDim.x)
+    qs[d] = bf2f(q[(long)b * q_bs + (long)hq * D + d]) * scale;
+  __syncthreads();
+
+  const u16* kbase = kc + ((long)b * Hkv + hkv) * max_seq * D;
+  const u16* vbase = vc + ((long)b * Hkv + hkv) * max_seq * D;
+
+  const int dpairs = D / 2;                       // 64 for D=128
+  const int kgroups = blockDim.x / dpairs;        // 4 for D=128
+  const int keys_per_group = DEC_TILE / kgroups;  // 64
+  const int dp = tid % dpairs;
+  const int kg = tid / dpairs;
+
+  float m = -1.0f / 0.0f;
+  float l = 0.f;
+  float o0 = 0.f, o1 = 0.f;                        // this thread's d-pair acc
+
+  for (int tile = start; tile < end; tile += DEC_TILE) {
+    const int kk = tile + tid;
+    float s = -1.0f / 0.0f;
+    if (kk < end) {
+      const s16x8* krow = (const s16x8*)(kbase + (long)kk * D);
+      float acc = 0.f;
+      for (int i = 0; i < D / 8; ++i) {
+        s16x8 kv8 = krow[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc = fmaf(qs[i * 8 + j], bf2f((u16)kv8[j]), acc);
+      }
+      s = acc;
+    }
+    const float tile_m = block_reduce_max(s, red);
+    const float m_new = fmaxf(m, tile_m);
+    const float alpha = __expf(m - m_new);         // 0 on first tile (m=-inf)
+    const float pv = (kk < end) ? __expf(s - m_new) : 0.f;
+    pl[tid] = pv;
+    const float tile_sum = block_reduce_sum(pv, red);
+    l = l * alpha + tile_sum;
+    m = m_new;
+    __syncthreads();                               // pl visible to all
+
+    o0 *= alpha; o1 *= alpha;
+    const int kbase_local = kg * keys_per_group;
+    const int kmax = min(DEC_TILE, end - tile);
+    for (int j = 0; j < keys_per_group; ++j) {
+      const int kl = kbase_local + j;
+      if (kl >= kmax) break;
+      const float p = pl[kl];
+      const u16* vrow = vbase + (long)(tile + kl) * D + dp * 2;
+      o0 = fmaf(p, bf2f(vrow[0]), o0);
+      o1 = fmaf(p, bf2f(vrow[1]), o1);
+    }
+    __syncthreads();                               // pl reuse next tile
+  }
+
+  // combine the key-groups' partial o through LDS
+  if (kg == 0) { osh[dp][0] = o0; osh[dp][1] = o1; }
+  __syncthreads();
+  for (int g = 1; g < kgroups; ++g) {
+    if (kg == g) { osh[dp][0] += o0; osh[dp][1] += o1; }
+    __syncthreads();
+  }
+  if (kg == 0) {
+    po[dp * 2] = osh[dp][0];
+    po[dp * 2 + 1] = osh[dp][1];
+  }
+  if (tid == 0) { pml[0] = m; pml[1] = l; }
+}
+
+void fei_attn_decode(const void* q, const void* k_cache, const void* v_cache,
+                     float* part_o, float* part_ml, const int* pos,
+                     int B, int Hq, int Hkv, int D, int max_seq, int splits,
+                     float scale, long q_bs, hipStream_t stream) {
+  hipLaunchKernelGGL(k_attn_decode, dim3(splits, Hq, B), dim3(256), 0, stream,
+                     (const u16*)q, (const u16*)k_cache, (const u16*)v_cache,
+                     part_o, part_ml, pos, B, Hq, Hkv, D, max_seq, splits,
+                     scale, q_bs);
+}
+
+// Combine: out[b,hq,:] = sum_s exp(m_s-m*) o_s / (sum_s exp(m_s-m*) l_s).
+// Grid (Hq, B), block D threads.
+__global__ void k_attn_decode_combine(u16* __restrict__ out,
+                                      const float* __restrict__ part_o,
+                                      const float* __restrict__ part_ml,
+                                      int B, int Hq, int D, int splits) {
+  const int hq = blockIdx.x;
+  const int b = blockIdx.y;
+  const int d = threadIdx.x;
+  const float* pml = part_ml + ((long)b * Hq + hq) * splits * 2;
+  float m = -1.0f / 0.0f;
+  for (int s = 0; s < splits; ++s) m = fmaxf(m, pml[s * 2]);
+  float l = 0.f, o = 0.f;
+  const float* po = part_o + ((long)b * Hq + hq) * splits * D;
+  for (int s = 0; s < splits; ++s) {
+    const float w = __expf(pml[s * 2] - m);
+    l += w * pml[s * 2 + 1];
+    o += w * po[s * D + d];
+  }
+  out[((long)b * Hq + hq) * D + d] = f2bf(l > 0.f ? o / l : 0.f);
+}
+
+void fei_attn_decode_combine(void* out, const float* part_o,
+                             const float* part_ml, int B, int Hq, int D,
+                             int splits, hipStream_t stream) {
+  hipLaunchKernelGGL(k_attn_decode_combine, dim3(Hq, B), dim3(D), 0, stream,
+                     (u16*)out, part_o, part_ml, B, Hq, D, splits);
+}
+
+// ---------------------------------------------------------------------------
+// SwiGLU: gate_up [rows, 2*inter] -> out [rows, inter]; silu(g)*u, f32 math.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256)
+k_swiglu(u16* __restrict__ out, const u16* __restrict__ gu, long rows,
+         int inter) {
+  const long nvec = rows * (inter >> 3);
+  const int iv = inter >> 3;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < nvec;
+       idx += (long)gridDim.x * blockDim.x) {
+    const long row = idx / iv;
+    const int col = (int)(idx % iv);
+    const s16x8* gp = (const s16x8*)(gu + row * 2 * inter) + col;
+    const s16x8* up = (const s16x8*)(gu + row * 2 * inter + inter) + col;
+    s16x8 g8 = *gp, u8 = *up;
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float g = bf2f((u16)g8[j]);
+      float u = bf2f((u16)u8[j]);
+      float silu = g / (1.f + __expf(-g));
+      o[j] = (short)f2bf(silu * u);
+    }
+    ((s16x8*)(out + row * inter))[col] = o;
+  }
+}
+
+void fei_swiglu(void* out, const void* gate_up, long rows, int inter,
+                hipStream_t stream) {
+  long nvec = rows * (inter >> 3);
+  int grid = (int)min((nvec + 255) / 256, (long)2048);
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(k_swiglu, dim3(grid), dim3(256), 0, stream,
+                     (u16*)out, (const u16*)gate_up, rows, inter);
+}
+
+// ---------------------------------------------------------------------------
+// Sampling. Stage 1: per (b, chunk) partial argmax of logits/T + Gumbel
+// noise (temperature<=0 -> greedy, no noise). Stage 2: final reduce, write
+// token[b], out_tokens[b, step]. `step` lives on device so the sequence of
+// kernels is hipGraph-replayable.
+// ws: [B, nchunks, 2] f32 (value, idx-as-float-bits via int store).
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256)
+k_sample_partial(const u16* __restrict__ logits, float* __restrict__ ws,
+                 int B, int vocab, int nchunks, float temperature,
+                 u64 seed, const int* __restrict__ step) {
+  const int chunk = blockIdx.x;
+  const int b = blockIdx.y;
+  const int chunk_size = (vocab + nchunks - 1) / nchunks;
+  const int start = chunk * chunk_size;
+  const int end = min(start + chunk_size, vocab);
+  const u64 st = (u64)(*step);
+  float best = -1.0f / 0.0f;
+  int besti = start;
+  const float invT = temperature > 0.f ? 1.f / temperature : 1.f;
+  for (int i = start + threadIdx.x; i < end; i += blockDim.x) {
+    float v = bf2f(logits[(long)b * vocab + i]) * invT;
+    if (temperature > 0.f) {
+      float u = hash_uniform(seed ^ (st * 0x51ed27f1ull) ^ ((u64)b << 40) ^ (u64)i);
+      v += -__logf(-__logf(u));
+    }
+    if (v > best) { best = v; besti = i; }
+  }
+  // wave then block reduce of (best, besti)
+#pragma unroll
+  for (int off = 32; off; off >>= 1) {
+    float ov = __shfl_xor(best, off);
+    int oi = __shfl_xor(besti, off);
+    if (ov > best || (ov == best && oi < besti)) { best = ov; besti = oi; }
+  }
+  __shared__ float rv[4];
+  __shared__ int ri[4];
+  const int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) { rv[wid] = best; ri[wid] = besti; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int i = 1; i < (int)(blockDim.x >> 6); ++i)
+      if (rv[i] > best || (rv[i] == best && ri[i] < besti)) { best = rv[i]; besti = ri[i]; }
+    ws[((long)b * nchunks + chunk) * 2] = best;
+    ((int*)ws)[((long)b * nchunks + chunk) * 2 + 1] = besti;
+  }
+}
+
+__global__ void k_sample_final(const float* __restrict__ ws,
+                               int* __restrict__ token,
+                               int* __restrict__ out_tokens,
+                               const int* __restrict__ step,
+                               int B, int nchunks, int max_new) {
+  const int b = blockIdx.x;
+  if (threadIdx.x != 0) return;
+  float best = -1.0f / 0.0f;
+  int besti = 0;
+  for (int c = 0; c < nchunks; ++c) {
+    float v = ws[((long)b * nchunks + c) * 2];
+    int i = ((const int*)ws)[((long)b * nchunks + c) * 2 + 1];
+    if (v > best || (v == best && i < besti)) { best = v; besti = i; }
+  }
+  token[b] = besti;
+  const int st = *step;
+  if (out_tokens && st < max_new) out_tokens[(long)b * max_new + st] = besti;
+}
+
+void fei_sample(const void* logits, int* token, int* out_tokens,
+                const int* step, float* ws, int B, int vocab, int nchunks,
+                float temperature, u64 seed, int max_new,
+                hipStream_t stream) {
+  hipLaunchKernelGGL(k_sample_partial, dim3(nchunks, B), dim3(256), 0, stream,
+                     (const u16*)logits, ws, B, vocab, nchunks, temperature,
+                     seed, step);
+  hipLaunchKernelGGL(k_sample_final, dim3(B), dim3(64), 0, stream,
+                     ws, token, out_tokens, step, B, nchunks, max_new);
+}
+
+// advance: pos[b]+=1 ; step+=1  (runs after sampling inside the graph)
+__global__ void k_advance(int* pos, int* step, int B) {
+  if ((int)threadIdx.x < B) pos[threadIdx.x] += 1;
+  if (threadIdx.x == 0 && step) *step += 1;
+}
+
+void fei_advance(int* pos, int* step, int B, hipStream_t stream) {
+  hipLaunchKernelGGL(k_advance, dim3(1), dim3(max(B, 64)), 0, stream,
+                     pos, step, B);
+}
+
+}  // extern "C"

@@ -1,0 +1,92 @@
+"""Process-group helpers: one process per GPU over RCCL
+(torch.distributed backend "nccl" IS RCCL on ROCm) or gloo on CPU.
+
+xGMI note: the MI355X node's GPUs are point-to-point
+connected (7 links x ~153 GB/s per GPU); ring collectives are per-link
+bound, so TP keeps message counts low (one all-reduce per layer-half) and
+DP uses bucketed collectives.
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from fei_amd.utils.logging import get_logger
+
+logger = get_logger("parallel.pg")
+
+
+@dataclass
+class ParallelContext:
+    rank: int = 0
+    world_size: int = 1
+    local_rank: int = 0
+    group: Optional[object] = None       # dist.ProcessGroup | None
+    backend: str = "none"
+
+    @property
+    def is_distributed(self) -> bool:
+        return self.world_size > 1
+
+
+def init_from_env(backend: Optional[str] = None,
+                  timeout_s: float = 300.0) -> ParallelContext:
+    """Initialise torch.distributed from torchrun/driver env vars.
+    Single-process when WORLD_SIZE is absent or 1."""
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world <= 1:
+        return ParallelContext()
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if backend == "nccl":
+        torch.cuda.set_device(local_rank)
+    if not dist.is_initialized():
+        dist.init_process_group(
+            backend=backend,
+            timeout=datetime.timedelta(seconds=timeout_s),
+        )
+    logger.info("dist init: rank %d/%d backend %s", rank, world, backend)
+    return ParallelContext(rank=rank, world_size=world, local_rank=local_rank,
+                           group=dist.group.WORLD, backend=backend)
+
+
+def get_world() -> ParallelContext:
+    if dist.is_available() and dist.is_initialized():
+        return ParallelContext(rank=dist.get_rank(),
+                               world_size=dist.get_world_size(),
+                               local_rank=int(os.environ.get("LOCAL_RANK", "0")),
+                               group=dist.group.WORLD,
+                               backend=dist.get_backend())
+    return ParallelContext()
+
+
+def all_reduce_sum(t: torch.Tensor, ctx: Optional[ParallelContext] = None) -> torch.Tensor:
+    ctx = ctx or get_world()
+    if ctx.is_distributed:
+        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=ctx.group)
+    return t
+
+
+def all_gather_cat(t: torch.Tensor, dim: int = -1,
+                   ctx: Optional[ParallelContext] = None) -> torch.Tensor:
+    """All-gather shards and concatenate along ``dim``."""
+    ctx = ctx or get_world()
+    if not ctx.is_distributed:
+        return t
+    shards = [torch.empty_like(t) for _ in range(ctx.world_size)]
+    dist.all_gather(shards, t.contiguous(), group=ctx.group)
+    return torch.cat(shards, dim=dim)
+
+
+def barrier(ctx: Optional[ParallelContext] = None) -> None:
+    ctx = ctx or get_world()
+    if ctx.is_distributed:
+        dist.barrier(group=ctx.group)

@@ -1,0 +1,227 @@
+"""GPU numerics tests: each HIP kernel vs the fp32 torch reference on the
+same bf16 inputs (tests/conftest.py registers the gpu marker)."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+@pytest.fixture(scope="module")
+def lib():
+    from fei_amd import ops
+    ops.require_lib()
+    return ops
+
+
+def randbf(*shape, seed=0, scale=1.0):
+    g = torch.Generator(device=DEV).manual_seed(seed)
+    return (torch.randn(*shape, generator=g, device=DEV) * scale).to(torch.bfloat16)
+
+
+def test_mfma_probe_layouts(lib):
+    """Falsify the assumed A/B/C fragment maps with asymmetric operands
+    (guide §3: symmetric B would hide a transposed C-write)."""
+    A = randbf(16, 32, seed=1)
+    B = (torch.arange(32 * 16, device=DEV).float().view(32, 16) * 0.01 + 0.1).to(torch.bfloat16)
+    C = lib.mfma_probe(A, B)
+    ref = A.float() @ B.float()
+    assert torch.allclose(C, ref, atol=1e-2, rtol=1e-2), \
+        f"max err {(C - ref).abs().max().item()}"
+
+
+def test_rmsnorm(lib):
+    from fei_amd.ops import reference as ref
+    x = randbf(33, 4096, seed=2)
+    w = randbf(4096, seed=3, scale=0.5)
+    out = lib.rmsnorm(x, w, 1e-5)
+    expected = ref.rmsnorm(x.float(), w.float(), 1e-5)
+    assert (out.float() - expected).abs().max() < 2e-2
+
+
+def test_fused_add_rmsnorm(lib):
+    from fei_amd.ops import reference as ref
+    x = randbf(8, 4096, seed=4)
+    res = randbf(8, 4096, seed=5)
+    w = randbf(4096, seed=6, scale=0.5)
+    res_gpu = res.clone()
+    out, new_res = lib.fused_add_rmsnorm(x, res_gpu, w, 1e-5)
+    exp_out, exp_res = ref.fused_add_rmsnorm(x, res.clone(), w, 1e-5)
+    assert (new_res.float() - exp_res.float()).abs().max() < 2e-2
+    assert (out.float() - exp_out.float()).abs().max() < 2e-2
+
+
+def _mk_cache(B, Hkv, max_seq, D, seed=0):
+    return randbf(B, Hkv, max_seq, D, seed=seed)
+
+
+def test_rope_kv_decode(lib):
+    from fei_amd.ops import reference as ref
+    B, Hq, Hkv, D, MS = 2, 8, 2, 128, 64
+    q = randbf(B, Hq, D, seed=7)
+    k = randbf(B, Hkv, D, seed=8)
+    v = randbf(B, Hkv, D, seed=9)
+    kc = torch.zeros(B, Hkv, MS, D, device=DEV, dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    pos = torch.tensor([3, 11], dtype=torch.int32, device=DEV)
+    table = ref.rope_table(MS, D, device=DEV)
+
+    kc_ref = torch.zeros_like(kc)
+    vc_ref = torch.zeros_like(vc)
+    q_ref = ref.rope_kv_decode(q.clone(), k, v, kc_ref, vc_ref, pos, table)
+
+    q_gpu = q.clone()
+    lib.rope_kv_decode(q_gpu, k, v, kc, vc, pos, table)
+    assert (q_gpu.float() - q_ref.float()).abs().max() < 2e-2
+    assert (kc.float() - kc_ref.float()).abs().max() < 2e-2
+    assert torch.equal(vc, vc_ref)
+
+
+def test_rope_kv_prefill_strided(lib):
+    """Exercise the strided-q path: q/k/v as views into a fused qkv buffer."""
+    from fei_amd.ops import reference as ref
+    B, S, Hq, Hkv, D, MS = 2, 9, 4, 2, 128, 64
+    W = (Hq + 2 * Hkv) * D
+    qkv = randbf(B * S, W, seed=10)
+    q = qkv.as_strided((B, S, Hq, D), (S * W, W, D, 1))
+    k = qkv.as_strided((B, S, Hkv, D), (S * W, W, D, 1), storage_offset=Hq * D)
+    v = qkv.as_strided((B, S, Hkv, D), (S * W, W, D, 1), storage_offset=(Hq + Hkv) * D)
+    kc = torch.zeros(B, Hkv, MS, D, device=DEV, dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    pos0 = torch.tensor([0, 5], dtype=torch.int32, device=DEV)
+    table = ref.rope_table(MS, D, device=DEV)
+
+    kc_ref = torch.zeros_like(kc)
+    vc_ref = torch.zeros_like(vc)
+    q_ref = ref.rope_kv_prefill(q.contiguous().clone(), k.contiguous(),
+                                v.contiguous(), kc_ref, vc_ref, pos0, table)
+
+    lib.rope_kv_prefill(q, k, v, kc, vc, pos0, table)
+    assert (q.float() - q_ref.float()).abs().max() < 2e-2
+    assert (kc.float() - kc_ref.float()).abs().max() < 2e-2
+    assert torch.equal(vc, vc_ref)
+
+
+@pytest.mark.parametrize("n,splits", [(1, 1), (7, 4), (300, 16), (1000, 16)])
+def test_attn_decode(lib, n, splits):
+    from fei_amd.ops import reference as ref
+    B, Hq, Hkv, D, MS = 2, 8, 2, 128, 1024
+    q = randbf(B, Hq, D, seed=20 + n)
+    kc = _mk_cache(B, Hkv, MS, D, seed=21 + n)
+    vc = _mk_cache(B, Hkv, MS, D, seed=22 + n)
+    pos = torch.tensor([n - 1, max(n // 2 - 1, 0)], dtype=torch.int32, device=DEV)
+    out = lib.attn_decode(q, kc, vc, pos, splits=splits)
+    expected = ref.attn_decode(q, kc, vc, pos + 1)
+    err = (out.float() - expected.float()).abs().max().item()
+    assert err < 2e-2, f"max err {err}"
+
+
+@pytest.mark.parametrize("D", [64, 128])
+@pytest.mark.parametrize("S,p0", [(16, 0), (64, 0), (129, 0), (64, 37)])
+def test_attn_prefill_causal(lib, D, S, p0):
+    from fei_amd.ops import reference as ref
+    B, Hq, Hkv, MS = 2, 4, 2, 512
+    q = randbf(B, S, Hq, D, seed=30 + S + D)
+    kc = _mk_cache(B, Hkv, MS, D, seed=31 + S)
+    vc = _mk_cache(B, Hkv, MS, D, seed=32 + S)
+    pos0 = torch.tensor([p0, p0], dtype=torch.int32, device=DEV)
+    out = lib.attn_prefill(q, kc, vc, pos0)
+    expected = ref.attn_prefill(q, kc, vc, pos0)
+    err = (out.float() - expected.float()).abs().max().item()
+    assert err < 2e-2, f"max err {err}"
+
+
+def test_attn_prefill_spiked_scores(lib):
+    """Force large max jumps across KV tiles (rule 26: exercise the rescale
+    path of the online softmax with an input that forces it)."""
+    from fei_amd.ops import reference as ref
+    B, S, Hq, Hkv, D, MS = 1, 64, 2, 1, 128, 256
+    q = randbf(B, S, Hq, D, seed=40)
+    kc = _mk_cache(B, Hkv, MS, D, seed=41, )
+    vc = _mk_cache(B, Hkv, MS, D, seed=42)
+    # spike one late key so every row's max jumps at the last tile
+    kc[:, :, 60, :] = (q[0, :, 0, :].mean(0) * 8).to(torch.bfloat16)
+    pos0 = torch.zeros(B, dtype=torch.int32, device=DEV)
+    out = lib.attn_prefill(q, kc, vc, pos0)
+    expected = ref.attn_prefill(q, kc, vc, pos0)
+    err = (out.float() - expected.float()).abs().max().item()
+    assert err < 3e-2, f"max err {err}"
+
+
+def test_attn_prefill_bidirectional(lib):
+    from fei_amd.ops import reference as ref
+    B, S, Hq, Hkv, D, MS = 1, 33, 4, 4, 64, 128
+    q = randbf(B, S, Hq, D, seed=50)
+    kc = _mk_cache(B, Hkv, MS, D, seed=51)
+    vc = _mk_cache(B, Hkv, MS, D, seed=52)
+    pos0 = torch.zeros(B, dtype=torch.int32, device=DEV)
+    kv_len = torch.tensor([S], dtype=torch.int32, device=DEV)
+    out = lib.attn_prefill(q, kc, vc, pos0, causal=False, kv_len=kv_len)
+    expected = lib._ref_bidir(q, kc, vc, kv_len, 1.0 / math.sqrt(D))
+    err = (out.float() - expected.float()).abs().max().item()
+    assert err < 2e-2, f"max err {err}"
+
+
+def test_swiglu(lib):
+    from fei_amd.ops import reference as ref
+    gu = randbf(17, 2 * 14336, seed=60)
+    out = lib.swiglu(gu)
+    expected = ref.swiglu(gu)
+    assert (out.float() - expected.float()).abs().max() < 1e-2
+
+
+def test_sample_greedy(lib):
+    B, V = 2, 128256
+    g = torch.Generator(device=DEV).manual_seed(70)
+    logits = (torch.randn(B, V, generator=g, device=DEV)).to(torch.bfloat16)
+    token = torch.zeros(B, dtype=torch.int32, device=DEV)
+    step = torch.zeros(1, dtype=torch.int32, device=DEV)
+    out_tokens = torch.zeros(B, 8, dtype=torch.int32, device=DEV)
+    ws = torch.zeros(B, 64 * 2, dtype=torch.float32, device=DEV)
+    lib.sample(logits, token, step, ws, out_tokens=out_tokens, temperature=0.0)
+    expected = logits.float().argmax(dim=-1)
+    assert torch.equal(token.long(), expected)
+    assert torch.equal(out_tokens[:, 0].long(), expected)
+
+
+def test_sample_gumbel_prefers_peak(lib):
+    B, V = 1, 1024
+    logits = torch.full((B, V), -10.0, device=DEV).to(torch.bfloat16)
+    logits[0, 123] = 10.0
+    token = torch.zeros(B, dtype=torch.int32, device=DEV)
+    step = torch.zeros(1, dtype=torch.int32, device=DEV)
+    ws = torch.zeros(B, 64 * 2, dtype=torch.float32, device=DEV)
+    hits = 0
+    for s in range(20):
+        step.fill_(s)
+        lib.sample(logits, token, step, ws, temperature=1.0, seed=s)
+        if int(token[0]) == 123:
+            hits += 1
+    assert hits >= 19   # peak is ~20 logits above the rest
+
+
+def test_sample_gumbel_varies_with_step(lib):
+    B, V = 1, 512
+    g = torch.Generator(device=DEV).manual_seed(71)
+    logits = torch.randn(B, V, generator=g, device=DEV).to(torch.bfloat16)
+    token = torch.zeros(B, dtype=torch.int32, device=DEV)
+    step = torch.zeros(1, dtype=torch.int32, device=DEV)
+    ws = torch.zeros(B, 64 * 2, dtype=torch.float32, device=DEV)
+    seen = set()
+    for s in range(16):
+        step.fill_(s)
+        lib.sample(logits, token, step, ws, temperature=2.0, seed=1)
+        seen.add(int(token[0]))
+    assert len(seen) > 3    # high temperature must explore
+
+
+def test_advance(lib):
+    pos = torch.tensor([5, 9], dtype=torch.int32, device=DEV)
+    step = torch.zeros(1, dtype=torch.int32, device=DEV)
+    lib.advance(pos, step)
+    assert pos.tolist() == [6, 10]
+    assert int(step) == 1
