@@ -166,27 +166,27 @@ class LlamaModel:
         """One decode step -> logits [B, vocab] (gathered across TP)."""
         s = self.spec
         B = token.shape[0]
-        h = F.embedding(token.long(), self.emb)
+        h = F.embedding(token.long(), self.emb)          # residual stream
         scale = 1.0 / math.sqrt(self.D)
+        n_layers = len(self.layers)
+        x = ops.rmsnorm(h, self.layers[0].norm_attn, s.norm_eps)
         for li, lw in enumerate(self.layers):
-            x = ops.rmsnorm(h, lw.norm_attn, s.norm_eps)
-            qkv = F.linear(x, lw.wqkv)
+            qkv = ops.linear_decode(x, lw.wqkv)
             q, k, v = self._qkv_views(qkv, B)
             ops.rope_kv_decode(q, k, v, k_caches[li], v_caches[li], pos, self.rope)
             att = ops.attn_decode(q, k_caches[li], v_caches[li], pos,
                                   splits=attn_splits, scale=scale,
                                   workspace=workspace)
-            o = F.linear(att.reshape(B, -1), lw.wo)
+            o = ops.linear_decode(att.reshape(B, -1), lw.wo)
             all_reduce_sum(o, self.tp)
-            h = h + o
-            x = ops.rmsnorm(h, lw.norm_mlp, s.norm_eps)
-            gu = F.linear(x, lw.wgu)
-            act = ops.swiglu(gu)
-            d = F.linear(act, lw.wdown)
+            x, h = ops.fused_add_rmsnorm(o, h, lw.norm_mlp, s.norm_eps)
+            act = ops.gemv_swiglu(x, lw.wgu)
+            d = ops.linear_decode(act, lw.wdown)
             all_reduce_sum(d, self.tp)
-            h = h + d
-        x = ops.rmsnorm(h, self.norm_f, s.norm_eps)
-        logits = F.linear(x, self.lm_head)
+            next_norm = (self.layers[li + 1].norm_attn if li + 1 < n_layers
+                         else self.norm_f)
+            x, h = ops.fused_add_rmsnorm(d, h, next_norm, s.norm_eps)
+        logits = ops.linear_decode(x, self.lm_head)
         logits = all_gather_cat(logits, dim=-1, ctx=self.tp)
         return logits
 
@@ -200,10 +200,11 @@ class LlamaModel:
         """Prefill S tokens -> logits of the LAST position [B, vocab]."""
         s = self.spec
         B, S = tokens.shape
-        h = F.embedding(tokens.long(), self.emb)          # [B,S,C]
+        h = F.embedding(tokens.long(), self.emb)          # [B,S,C] residual
         scale = 1.0 / math.sqrt(self.D)
+        n_layers = len(self.layers)
+        x = ops.rmsnorm(h, self.layers[0].norm_attn, s.norm_eps)
         for li, lw in enumerate(self.layers):
-            x = ops.rmsnorm(h, lw.norm_attn, s.norm_eps)
             qkv = F.linear(x, lw.wqkv)                    # [B,S,W]
             qkv2 = qkv.view(B * S, -1)
             q, k, v = self._qkv_views(qkv2, B, S)
@@ -211,15 +212,15 @@ class LlamaModel:
             att = ops.attn_prefill(q, k_caches[li], v_caches[li], pos0, scale=scale)
             o = F.linear(att.reshape(B, S, -1), lw.wo)
             all_reduce_sum(o, self.tp)
-            h = h + o
-            x = ops.rmsnorm(h, lw.norm_mlp, s.norm_eps)
+            x, h = ops.fused_add_rmsnorm(o, h, lw.norm_mlp, s.norm_eps)
             gu = F.linear(x, lw.wgu)
             act = ops.swiglu(gu)
             d = F.linear(act, lw.wdown)
             all_reduce_sum(d, self.tp)
-            h = h + d
-        last = h[:, -1, :]
-        x = ops.rmsnorm(last, self.norm_f, s.norm_eps)
-        logits = F.linear(x, self.lm_head)
+            next_norm = (self.layers[li + 1].norm_attn if li + 1 < n_layers
+                         else self.norm_f)
+            x, h = ops.fused_add_rmsnorm(d, h, next_norm, s.norm_eps)
+        last = x[:, -1, :]
+        logits = F.linear(last, self.lm_head)
         logits = all_gather_cat(logits, dim=-1, ctx=self.tp)
         return logits

@@ -55,6 +55,8 @@ def _try_load() -> None:
     lib.fei_attn_prefill.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp,
                                      _i, _i, _i, _i, _i, _i, _f, _i, _l, _vp]
     lib.fei_mfma_probe.argtypes = [_vp, _vp, _vp, _vp]
+    lib.fei_gemv.argtypes = [_vp, _vp, _vp, _i, _i, _i, _vp]
+    lib.fei_gemv_swiglu.argtypes = [_vp, _vp, _vp, _i, _i, _i, _vp]
     _LIB = lib
 
 
@@ -294,3 +296,46 @@ def mfma_probe(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
     lib.fei_mfma_probe(_ptr(A.contiguous()), _ptr(B.contiguous()), _ptr(C),
                        _stream())
     return C
+
+
+# -- decode GEMV dispatch ----------------------------------------------------
+
+_GEMV_MAX_LDS = 64 * 1024      # keep >=2 blocks/CU (160 KB LDS per CU)
+
+
+def _gemv_ok(M: int, K: int) -> bool:
+    return M in (1, 2, 4, 8) and K % 8 == 0 and M * K * 2 <= _GEMV_MAX_LDS
+
+
+def linear_decode(x: torch.Tensor, w: torch.Tensor,
+                  out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """F.linear for skinny M: hand-rolled streaming GEMV on GPU
+    (profiles/r01: rocBLAS reaches ~4 TB/s on M=1; streaming reaches the
+    flat-read ceiling). Falls back to torch for unsupported shapes/CPU."""
+    M = x.numel() // x.shape[-1]
+    K = x.shape[-1]
+    N = w.shape[0]
+    if not x.is_cuda or not _gemv_ok(M, K):
+        return torch.nn.functional.linear(x, w)
+    lib = require_lib()
+    x2 = x.contiguous().view(M, K)
+    if out is None:
+        out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
+    lib.fei_gemv(_ptr(out), _ptr(x2), _ptr(w), M, N, K, _stream())
+    return out
+
+
+def gemv_swiglu(x: torch.Tensor, wgu: torch.Tensor,
+                out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Fused gate/up GEMV + SwiGLU: x [M,K] @ [2I,K]^T -> silu(g)*u [M,I]."""
+    M = x.numel() // x.shape[-1]
+    K = x.shape[-1]
+    I = wgu.shape[0] // 2
+    if not x.is_cuda or not _gemv_ok(M, K):
+        return swiglu(torch.nn.functional.linear(x, wgu))
+    lib = require_lib()
+    x2 = x.contiguous().view(M, K)
+    if out is None:
+        out = torch.empty(*x.shape[:-1], I, dtype=x.dtype, device=x.device)
+    lib.fei_gemv_swiglu(_ptr(out), _ptr(x2), _ptr(wgu), M, I, K, _stream())
+    return out

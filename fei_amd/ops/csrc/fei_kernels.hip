@@ -16,24 +16,19 @@ extern "C" {
 // RMSNorm: out[r,:] = x[r,:] * rsqrt(mean(x^2)+eps) * w ; bf16 I/O, f32 acc.
 // One block per row (grid-stride); cols % 8 == 0; cols/8 <= 256*MAX_V8.
 // ---------------------------------------------------------------------------
-#define RMS_MAXV 8   // up to 256 threads * 8 vec8 = 16384 cols
-
 __global__ void __launch_bounds__(256)
 k_rmsnorm(u16* __restrict__ out, const u16* __restrict__ x,
           const u16* __restrict__ w, int rows, int cols, float eps) {
   __shared__ float red[4];
   const int nv = cols >> 3;                       // vec8 per row
-  float vals[RMS_MAXV * 8];
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
     const s16x8* xr = (const s16x8*)(x + (long)row * cols);
     float ss = 0.f;
-    int cnt = 0;
-    for (int i = threadIdx.x; i < nv; i += blockDim.x, ++cnt) {
+    for (int i = threadIdx.x; i < nv; i += blockDim.x) {
       s16x8 v = xr[i];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float f = bf2f((u16)v[j]);
-        vals[cnt * 8 + j] = f;
         ss = fmaf(f, f, ss);
       }
     }
@@ -41,13 +36,15 @@ k_rmsnorm(u16* __restrict__ out, const u16* __restrict__ x,
     const float inv = rsqrtf(ss / (float)cols + eps);
     s16x8* orow = (s16x8*)(out + (long)row * cols);
     const s16x8* wv = (const s16x8*)w;
-    cnt = 0;
-    for (int i = threadIdx.x; i < nv; i += blockDim.x, ++cnt) {
+    // second read hits L1/L2 (a row is tiny vs 32 MB L2) — avoids a
+    // runtime-indexed register array (guide rule 20: scratch).
+    for (int i = threadIdx.x; i < nv; i += blockDim.x) {
+      s16x8 v = xr[i];
       s16x8 wvv = wv[i];
       s16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        o[j] = (short)f2bf(vals[cnt * 8 + j] * inv * bf2f((u16)wvv[j]));
+        o[j] = (short)f2bf(bf2f((u16)v[j]) * inv * bf2f((u16)wvv[j]));
       orow[i] = o;
     }
     __syncthreads();
@@ -70,13 +67,11 @@ k_add_rmsnorm(u16* __restrict__ out, u16* __restrict__ res,
               int rows, int cols, float eps) {
   __shared__ float red[4];
   const int nv = cols >> 3;
-  float vals[RMS_MAXV * 8];
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
     const s16x8* xr = (const s16x8*)(x + (long)row * cols);
     s16x8* rr = (s16x8*)(res + (long)row * cols);
     float ss = 0.f;
-    int cnt = 0;
-    for (int i = threadIdx.x; i < nv; i += blockDim.x, ++cnt) {
+    for (int i = threadIdx.x; i < nv; i += blockDim.x) {
       s16x8 xv = xr[i];
       s16x8 rv = rr[i];
       s16x8 nr;
@@ -86,7 +81,6 @@ k_add_rmsnorm(u16* __restrict__ out, u16* __restrict__ res,
         u16 b = f2bf(f);
         nr[j] = (short)b;
         f = bf2f(b);                                // match reference rounding
-        vals[cnt * 8 + j] = f;
         ss = fmaf(f, f, ss);
       }
       rr[i] = nr;
@@ -95,13 +89,15 @@ k_add_rmsnorm(u16* __restrict__ out, u16* __restrict__ res,
     const float inv = rsqrtf(ss / (float)cols + eps);
     s16x8* orow = (s16x8*)(out + (long)row * cols);
     const s16x8* wv = (const s16x8*)w;
-    cnt = 0;
-    for (int i = threadIdx.x; i < nv; i += blockDim.x, ++cnt) {
+    // re-read the just-written residual (L1/L2-hot; rule 20: no runtime-
+    // indexed register array).
+    for (int i = threadIdx.x; i < nv; i += blockDim.x) {
+      s16x8 rv = rr[i];
       s16x8 wvv = wv[i];
       s16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        o[j] = (short)f2bf(vals[cnt * 8 + j] * inv * bf2f((u16)wvv[j]));
+        o[j] = (short)f2bf(bf2f((u16)rv[j]) * inv * bf2f((u16)wvv[j]));
       orow[i] = o;
     }
     __syncthreads();
@@ -214,17 +210,52 @@ void fei_rope_kv_prefill(void* q, const void* k, const void* v, void* k_cache,
 }
 
 // ---------------------------------------------------------------------------
-// Decode attention, split-K partials. q [B,Hq,D] bf16 (roped);
+// Decode attention, split-K partials, GQA-grouped. q [B,Hq,D] bf16 (roped);
 // caches [B,Hkv,max_seq,D]. n = pos[b]+1 keys (length read on DEVICE so the
 // launch shape is static for hipGraph capture).
-// Grid (splits, Hq, B), block 256 = 4 waves.
-//   phase A: thread-per-key scores + block online-softmax state
-//   phase B: threads as (D/2 pairs x key-groups) accumulate P*V
+// Grid (splits, Hkv, B), block 256 = 4 waves. One block serves ALL G=Hq/Hkv
+// q-heads of its kv-head, so each K/V row is read ONCE per block (the
+// one-block-per-q-head version re-read K/V G times — profiles/r01).
+//   phase A: thread-per-key; dot against the G q-vectors staged in LDS,
+//            G-wide block online-softmax state
+//   phase B: threads as (D/2 pairs x key-groups) accumulate P*V for all G
 // Partials: part_o [B,Hq,splits,D] f32; part_ml [B,Hq,splits,2] f32.
 // ---------------------------------------------------------------------------
 #define DEC_TILE 256
 #define DEC_DMAX 128
+#define DEC_GMAX 8
 
+}  // extern "C" (templates need C++ linkage)
+
+namespace {
+
+// Reduce G values at once (one barrier pair for all heads).
+template <int OP, int G>  // OP: 0 = max, 1 = sum
+__device__ __forceinline__ void block_reduce_vec(float* v,
+                                                 float red[DEC_GMAX][4]) {
+  const int nw = blockDim.x >> 6;
+  const int wid = threadIdx.x >> 6;
+  #pragma unroll
+    for (int g = 0; g < G; ++g) {
+    float x = v[g];
+#pragma unroll
+    for (int off = 32; off; off >>= 1)
+      x = OP == 0 ? fmaxf(x, __shfl_xor(x, off)) : x + __shfl_xor(x, off);
+    if ((threadIdx.x & 63) == 0) red[g][wid] = x;
+    v[g] = x;
+  }
+  __syncthreads();
+  #pragma unroll
+    for (int g = 0; g < G; ++g) {
+    float x = red[g][0];
+    for (int i = 1; i < nw; ++i)
+      x = OP == 0 ? fmaxf(x, red[g][i]) : x + red[g][i];
+    v[g] = x;
+  }
+  __syncthreads();
+}
+
+template <int G>
 __global__ void __launch_bounds__(256)
 k_attn_decode(const u16* __restrict__ q, const u16* __restrict__ kc,
               const u16* __restrict__ vc, float* __restrict__ part_o,
@@ -232,15 +263,13 @@ k_attn_decode(const u16* __restrict__ q, const u16* __restrict__ kc,
               int B, int Hq, int Hkv, int D, int max_seq, int splits,
               float scale, long q_bs) {
   const int split = blockIdx.x;
-  const int hq = blockIdx.y;
+  const int hkv = blockIdx.y;
   const int b = blockIdx.z;
-  const int G = Hq / Hkv;
-  const int hkv = hq / G;
   const int tid = threadIdx.x;
 
-  __shared__ float qs[DEC_DMAX];
-  __shared__ float pl[DEC_TILE];
-  __shared__ float red[4];
+  __shared__ float qs[DEC_GMAX][DEC_DMAX];
+  __shared__ float pl[DEC_GMAX][DEC_TILE];
+  __shared__ float red[DEC_GMAX][4];
   __shared__ float osh[DEC_DMAX / 2][2];
 
   const int n = pos[b] + 1;
@@ -248,18 +277,23 @@ k_attn_decode(const u16* __restrict__ q, const u16* __restrict__ kc,
   const int start = split * chunk;
   const int end = min(start + chunk, n);
 
-  float* po = part_o + (((long)b * Hq + hq) * splits + split) * D;
-  float* pml = part_ml + (((long)b * Hq + hq) * splits + split) * 2;
-
   if (start >= end) {
-    for (int d = tid; d < D; d += blockDim.x) po[d] = 0.f;
-    if (tid == 0) { pml[0] = -1.0f / 0.0f; pml[1] = 0.f; }
+    #pragma unroll
+    for (int g = 0; g < G; ++g) {
+      const int hq = hkv * G + g;
+      float* po = part_o + (((long)b * Hq + hq) * splits + split) * D;
+      float* pml = part_ml + (((long)b * Hq + hq) * splits + split) * 2;
+      for (int d = tid; d < D; d += blockDim.x) po[d] = 0.f;
+      if (tid == 0) { pml[0] = -1.0f / 0.0f; pml[1] = 0.f; }
+    }
     return;
   }
 
-  // stage q (scaled) into LDS
-  for (int d = tid; d < D; d += blockDim.x)
-    qs[d] = bf2f(q[(long)b * q_bs + (long)hq * D + d]) * scale;
+  // stage the group's q vectors (scaled) into LDS
+  for (int i = tid; i < G * D; i += blockDim.x) {
+    const int g = i / D, d = i % D;
+    qs[g][d] = bf2f(q[(long)b * q_bs + (long)(hkv * G + g) * D + d]) * scale;
+  }
   __syncthreads();
 
   const u16* kbase = kc + ((long)b * Hkv + hkv) * max_seq * D;
@@ -271,70 +305,119 @@ k_attn_decode(const u16* __restrict__ q, const u16* __restrict__ kc,
   const int dp = tid % dpairs;
   const int kg = tid / dpairs;
 
-  float m = -1.0f / 0.0f;
-  float l = 0.f;
-  float o0 = 0.f, o1 = 0.f;                        // this thread's d-pair acc
+  float m[DEC_GMAX], l[DEC_GMAX], sc[DEC_GMAX];
+  float o0[DEC_GMAX], o1[DEC_GMAX];
+  #pragma unroll
+    for (int g = 0; g < G; ++g) {
+    m[g] = -1.0f / 0.0f; l[g] = 0.f; o0[g] = 0.f; o1[g] = 0.f;
+  }
 
   for (int tile = start; tile < end; tile += DEC_TILE) {
     const int kk = tile + tid;
-    float s = -1.0f / 0.0f;
+    // scores for this key against every q-head of the group
     if (kk < end) {
       const s16x8* krow = (const s16x8*)(kbase + (long)kk * D);
-      float acc = 0.f;
+      #pragma unroll
+    for (int g = 0; g < G; ++g) sc[g] = 0.f;
       for (int i = 0; i < D / 8; ++i) {
         s16x8 kv8 = krow[i];
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          acc = fmaf(qs[i * 8 + j], bf2f((u16)kv8[j]), acc);
+        for (int j = 0; j < 8; ++j) {
+          const float kf = bf2f((u16)kv8[j]);
+          #pragma unroll
+    for (int g = 0; g < G; ++g)
+            sc[g] = fmaf(qs[g][i * 8 + j], kf, sc[g]);
+        }
       }
-      s = acc;
+    } else {
+      #pragma unroll
+    for (int g = 0; g < G; ++g) sc[g] = -1.0f / 0.0f;
     }
-    const float tile_m = block_reduce_max(s, red);
-    const float m_new = fmaxf(m, tile_m);
-    const float alpha = __expf(m - m_new);         // 0 on first tile (m=-inf)
-    const float pv = (kk < end) ? __expf(s - m_new) : 0.f;
-    pl[tid] = pv;
-    const float tile_sum = block_reduce_sum(pv, red);
-    l = l * alpha + tile_sum;
-    m = m_new;
-    __syncthreads();                               // pl visible to all
+    float tile_m[DEC_GMAX];
+    #pragma unroll
+    for (int g = 0; g < G; ++g) tile_m[g] = sc[g];
+    block_reduce_vec<0, G>(tile_m, red);
+    float alpha[DEC_GMAX], pv[DEC_GMAX];
+    #pragma unroll
+    for (int g = 0; g < G; ++g) {
+      const float m_new = fmaxf(m[g], tile_m[g]);
+      alpha[g] = __expf(m[g] - m_new);             // 0 on first tile
+      pv[g] = (kk < end) ? __expf(sc[g] - m_new) : 0.f;
+      pl[g][tid] = pv[g];
+      m[g] = m_new;
+    }
+    float tile_sum[DEC_GMAX];
+    #pragma unroll
+    for (int g = 0; g < G; ++g) tile_sum[g] = pv[g];
+    block_reduce_vec<1, G>(tile_sum, red);         // also makes pl visible
+    #pragma unroll
+    for (int g = 0; g < G; ++g) l[g] = l[g] * alpha[g] + tile_sum[g];
 
-    o0 *= alpha; o1 *= alpha;
+    // P*V: V row read once, used by every head
+    #pragma unroll
+    for (int g = 0; g < G; ++g) { o0[g] *= alpha[g]; o1[g] *= alpha[g]; }
     const int kbase_local = kg * keys_per_group;
     const int kmax = min(DEC_TILE, end - tile);
     for (int j = 0; j < keys_per_group; ++j) {
       const int kl = kbase_local + j;
       if (kl >= kmax) break;
-      const float p = pl[kl];
       const u16* vrow = vbase + (long)(tile + kl) * D + dp * 2;
-      o0 = fmaf(p, bf2f(vrow[0]), o0);
-      o1 = fmaf(p, bf2f(vrow[1]), o1);
+      const float v0 = bf2f(vrow[0]);
+      const float v1 = bf2f(vrow[1]);
+      #pragma unroll
+    for (int g = 0; g < G; ++g) {
+        const float p = pl[g][kl];
+        o0[g] = fmaf(p, v0, o0[g]);
+        o1[g] = fmaf(p, v1, o1[g]);
+      }
     }
     __syncthreads();                               // pl reuse next tile
   }
 
-  // combine the key-groups' partial o through LDS
-  if (kg == 0) { osh[dp][0] = o0; osh[dp][1] = o1; }
-  __syncthreads();
-  for (int g = 1; g < kgroups; ++g) {
-    if (kg == g) { osh[dp][0] += o0; osh[dp][1] += o1; }
+  // combine the key-groups' partial o through LDS, one head at a time
+  #pragma unroll
+    for (int g = 0; g < G; ++g) {
+    const int hq = hkv * G + g;
+    float* po = part_o + (((long)b * Hq + hq) * splits + split) * D;
+    float* pml = part_ml + (((long)b * Hq + hq) * splits + split) * 2;
+    if (kg == 0) { osh[dp][0] = o0[g]; osh[dp][1] = o1[g]; }
+    __syncthreads();
+    for (int gg = 1; gg < kgroups; ++gg) {
+      if (kg == gg) { osh[dp][0] += o0[g]; osh[dp][1] += o1[g]; }
+      __syncthreads();
+    }
+    if (kg == 0) {
+      po[dp * 2] = osh[dp][0];
+      po[dp * 2 + 1] = osh[dp][1];
+    }
+    if (tid == 0) { pml[0] = m[g]; pml[1] = l[g]; }
     __syncthreads();
   }
-  if (kg == 0) {
-    po[dp * 2] = osh[dp][0];
-    po[dp * 2 + 1] = osh[dp][1];
-  }
-  if (tid == 0) { pml[0] = m; pml[1] = l; }
 }
+
+}  // namespace
+
+extern "C" {
 
 void fei_attn_decode(const void* q, const void* k_cache, const void* v_cache,
                      float* part_o, float* part_ml, const int* pos,
                      int B, int Hq, int Hkv, int D, int max_seq, int splits,
                      float scale, long q_bs, hipStream_t stream) {
-  hipLaunchKernelGGL(k_attn_decode, dim3(splits, Hq, B), dim3(256), 0, stream,
-                     (const u16*)q, (const u16*)k_cache, (const u16*)v_cache,
-                     part_o, part_ml, pos, B, Hq, Hkv, D, max_seq, splits,
-                     scale, q_bs);
+  const int G = Hq / Hkv;
+  dim3 grid(splits, Hkv, B);
+#define LAUNCH_DEC(GV) \
+  hipLaunchKernelGGL(k_attn_decode<GV>, grid, dim3(256), 0, stream, \
+                     (const u16*)q, (const u16*)k_cache, (const u16*)v_cache, \
+                     part_o, part_ml, pos, B, Hq, Hkv, D, max_seq, splits, \
+                     scale, q_bs)
+  switch (G) {
+    case 1: LAUNCH_DEC(1); break;
+    case 2: LAUNCH_DEC(2); break;
+    case 4: LAUNCH_DEC(4); break;
+    case 8: LAUNCH_DEC(8); break;
+    default: break;  // unsupported group size: wrapper validates
+  }
+#undef LAUNCH_DEC
 }
 
 // Combine: out[b,hq,:] = sum_s exp(m_s-m*) o_s / (sum_s exp(m_s-m*) l_s).

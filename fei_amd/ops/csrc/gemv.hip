@@ -1,0 +1,134 @@
+// Skinny-M GEMV kernels for the decode hot path (gfx950).
+// out[M, N] = x[M, K] @ W[N, K]^T  — torch F.linear weight layout (W row-
+// major over K). M <= 8 (decode batch), K % 512 == 0.
+//
+// Design: one W row per wave, streamed with 16 B/lane coalesced loads
+// (64 lanes x 16 B = 2 cachelines per instruction); x staged in LDS and
+// broadcast-read; per-lane f32 dot then wave shfl reduce. This is pure
+// HBM streaming — rocBLAS's tiled GEMM kernels reach only ~4 TB/s on
+// M=1 shapes (profiles/r01), streaming reaches the flat-read ceiling.
+//
+// fei_gemv_swiglu fuses the MLP gate/up pair: wave computes both dots
+// (rows n and n+N) and writes silu(g)*u — kills the separate swiglu
+// kernel and the 2*I-wide intermediate.
+#include "fei_common.h"
+
+namespace {
+
+template <int M>
+__global__ void __launch_bounds__(256)
+k_gemv(u16* __restrict__ out, const u16* __restrict__ x,
+       const u16* __restrict__ w, int N, int K) {
+  // 4 waves per block, one W row per wave
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  u16* xs = (u16*)smem;                       // [M][K]
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int n = blockIdx.x * 4 + wid;
+  for (int i = tid; i < M * (K >> 3); i += blockDim.x)
+    ((s16x8*)xs)[i] = ((const s16x8*)x)[i];
+  __syncthreads();
+  if (n >= N) return;
+
+  const s16x8* wrow = (const s16x8*)(w + (long)n * K);
+  float acc[M];
+#pragma unroll
+  for (int m = 0; m < M; ++m) acc[m] = 0.f;
+  const int nv = K >> 3;                      // vec8 per row
+  for (int i = lane; i < nv; i += 64) {
+    s16x8 wv = wrow[i];
+#pragma unroll
+    for (int m = 0; m < M; ++m) {
+      s16x8 xv = ((const s16x8*)(xs + m * K))[i];
+      acc[m] += dot8_bf16(xv, wv);
+    }
+  }
+#pragma unroll
+  for (int m = 0; m < M; ++m) {
+    float v = wave_reduce_sum(acc[m]);
+    if (lane == 0) out[(long)m * N + n] = f2bf(v);
+  }
+}
+
+// gate/up + SwiGLU fused: W = [gate rows (N) ; up rows (N)] stacked, out[M,N]
+template <int M>
+__global__ void __launch_bounds__(256)
+k_gemv_swiglu(u16* __restrict__ out, const u16* __restrict__ x,
+              const u16* __restrict__ w, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  u16* xs = (u16*)smem;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int n = blockIdx.x * 4 + wid;
+  for (int i = tid; i < M * (K >> 3); i += blockDim.x)
+    ((s16x8*)xs)[i] = ((const s16x8*)x)[i];
+  __syncthreads();
+  if (n >= N) return;
+
+  const s16x8* grow = (const s16x8*)(w + (long)n * K);
+  const s16x8* urow = (const s16x8*)(w + (long)(n + N) * K);
+  float accg[M], accu[M];
+#pragma unroll
+  for (int m = 0; m < M; ++m) { accg[m] = 0.f; accu[m] = 0.f; }
+  const int nv = K >> 3;
+  for (int i = lane; i < nv; i += 64) {
+    s16x8 gv = grow[i];
+    s16x8 uv = urow[i];
+#pragma unroll
+    for (int m = 0; m < M; ++m) {
+      s16x8 xv = ((const s16x8*)(xs + m * K))[i];
+      accg[m] += dot8_bf16(xv, gv);
+      accu[m] += dot8_bf16(xv, uv);
+    }
+  }
+#pragma unroll
+  for (int m = 0; m < M; ++m) {
+    float g = wave_reduce_sum(accg[m]);
+    float u = wave_reduce_sum(accu[m]);
+    if (lane == 0) {
+      const float silu = g / (1.f + __expf(-g));
+      out[(long)m * N + n] = f2bf(silu * u);
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+void fei_gemv(void* out, const void* x, const void* w, int M, int N, int K,
+              hipStream_t stream) {
+  dim3 grid((N + 3) / 4);
+  const size_t lds = (size_t)M * K * 2;
+#define LG(MV) hipLaunchKernelGGL(k_gemv<MV>, grid, dim3(256), lds, stream, \
+                                  (u16*)out, (const u16*)x, (const u16*)w, N, K)
+  switch (M) {
+    case 1: LG(1); break;
+    case 2: LG(2); break;
+    case 4: LG(4); break;
+    case 8: LG(8); break;
+    default: break;   // wrapper validates
+  }
+#undef LG
+}
+
+void fei_gemv_swiglu(void* out, const void* x, const void* w, int M, int N,
+                     int K, hipStream_t stream) {
+  dim3 grid((N + 3) / 4);
+  const size_t lds = (size_t)M * K * 2;
+#define LG(MV) hipLaunchKernelGGL(k_gemv_swiglu<MV>, grid, dim3(256), lds, \
+                                  stream, (u16*)out, (const u16*)x, \
+                                  (const u16*)w, N, K)
+  switch (M) {
+    case 1: LG(1); break;
+    case 2: LG(2); break;
+    case 4: LG(4); break;
+    case 8: LG(8); break;
+    default: break;
+  }
+#undef LG
+}
+
+}  // extern "C"

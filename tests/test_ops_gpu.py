@@ -225,3 +225,28 @@ def test_advance(lib):
     lib.advance(pos, step)
     assert pos.tolist() == [6, 10]
     assert int(step) == 1
+
+
+def test_gemv_matches_linear(lib):
+    import torch.nn.functional as F
+    for M, N, K in [(1, 4096, 4096), (1, 6144, 4096), (2, 512, 1024),
+                    (4, 128256, 4096)]:
+        x = randbf(M, K, seed=80 + M)
+        w = randbf(N, K, seed=81 + N % 97, scale=0.02)
+        out = lib.linear_decode(x, w)
+        ref_out = F.linear(x.float(), w.float())
+        err = (out.float() - ref_out).abs().max().item()
+        tol = 0.02 * max(1.0, K / 4096)
+        assert err < tol, f"M{M} N{N} K{K}: max err {err}"
+
+
+def test_gemv_swiglu_matches(lib):
+    import torch.nn.functional as F
+    from fei_amd.ops import reference as ref
+    M, I, K = 1, 14336, 4096
+    x = randbf(M, K, seed=90)
+    wgu = randbf(2 * I, K, seed=91, scale=0.02)
+    out = lib.gemv_swiglu(x, wgu)
+    expected = ref.swiglu(F.linear(x.float(), wgu.float()).to(torch.bfloat16))
+    err = (out.float() - expected.float()).abs().max().item()
+    assert err < 2e-2, f"max err {err}"
