@@ -43,7 +43,7 @@ class LocalEngine:
         use_hip_graph: Optional[bool] = None,
         tp: Optional[ParallelContext] = None,
         seed: int = 1234,
-        attn_splits: int = 16,
+        attn_splits: int = 4,
         dtype: Optional[torch.dtype] = None,
     ):
         self.spec = spec

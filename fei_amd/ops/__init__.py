@@ -161,7 +161,7 @@ def rope_kv_prefill(q, k, v, k_cache, v_cache, pos0, table) -> torch.Tensor:
     return q
 
 
-def attn_decode(q, k_cache, v_cache, pos, splits: int = 16,
+def attn_decode(q, k_cache, v_cache, pos, splits: int = 4,
                 scale: Optional[float] = None,
                 workspace: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
                 out: Optional[torch.Tensor] = None) -> torch.Tensor:

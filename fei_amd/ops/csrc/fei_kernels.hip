@@ -421,7 +421,11 @@ void fei_attn_decode(const void* q, const void* k_cache, const void* v_cache,
 }
 
 // Combine: out[b,hq,:] = sum_s exp(m_s-m*) o_s / (sum_s exp(m_s-m*) l_s).
-// Grid (Hq, B), block D threads.
+// Grid (Hq, B), block D threads. All split partials are loaded in PARALLEL
+// (LDS-staged m/l, unrolled independent o loads) — a serial per-split loop
+// of dependent global loads cost ~7 us at splits=16 (profiles/r01).
+#define CMB_SMAX 32
+
 __global__ void k_attn_decode_combine(u16* __restrict__ out,
                                       const float* __restrict__ part_o,
                                       const float* __restrict__ part_ml,
@@ -429,14 +433,32 @@ __global__ void k_attn_decode_combine(u16* __restrict__ out,
   const int hq = blockIdx.x;
   const int b = blockIdx.y;
   const int d = threadIdx.x;
+  __shared__ float sml[CMB_SMAX][2];
   const float* pml = part_ml + ((long)b * Hq + hq) * splits * 2;
+  if (d < 2 * splits)
+    ((float*)sml)[d] = pml[d];                    // parallel m/l fetch
+  __syncthreads();
   float m = -1.0f / 0.0f;
-  for (int s = 0; s < splits; ++s) m = fmaxf(m, pml[s * 2]);
-  float l = 0.f, o = 0.f;
+  for (int s = 0; s < splits; ++s) m = fmaxf(m, sml[s][0]);
   const float* po = part_o + ((long)b * Hq + hq) * splits * D;
-  for (int s = 0; s < splits; ++s) {
-    const float w = __expf(pml[s * 2] - m);
-    l += w * pml[s * 2 + 1];
+  float l = 0.f, o = 0.f;
+  int s = 0;
+  for (; s + 4 <= splits; s += 4) {               // independent load quads
+    const float o0 = po[(s + 0) * D + d];
+    const float o1 = po[(s + 1) * D + d];
+    const float o2 = po[(s + 2) * D + d];
+    const float o3 = po[(s + 3) * D + d];
+    const float w0 = __expf(sml[s + 0][0] - m);
+    const float w1 = __expf(sml[s + 1][0] - m);
+    const float w2 = __expf(sml[s + 2][0] - m);
+    const float w3 = __expf(sml[s + 3][0] - m);
+    l += w0 * sml[s + 0][1] + w1 * sml[s + 1][1] +
+         w2 * sml[s + 2][1] + w3 * sml[s + 3][1];
+    o += w0 * o0 + w1 * o1 + w2 * o2 + w3 * o3;
+  }
+  for (; s < splits; ++s) {
+    const float w = __expf(sml[s][0] - m);
+    l += w * sml[s][1];
     o += w * po[s * D + d];
   }
   out[((long)b * Hq + hq) * D + d] = f2bf(l > 0.f ? o / l : 0.f);
@@ -538,18 +560,28 @@ __global__ void k_sample_final(const float* __restrict__ ws,
                                int* __restrict__ out_tokens,
                                const int* __restrict__ step,
                                int B, int nchunks, int max_new) {
+  // one wave, parallel chunk loads + shfl argmax (a serial thread-0 loop
+  // cost ~8 us at nchunks=64)
   const int b = blockIdx.x;
-  if (threadIdx.x != 0) return;
+  const int lane = threadIdx.x & 63;
   float best = -1.0f / 0.0f;
-  int besti = 0;
-  for (int c = 0; c < nchunks; ++c) {
-    float v = ws[((long)b * nchunks + c) * 2];
-    int i = ((const int*)ws)[((long)b * nchunks + c) * 2 + 1];
+  int besti = 0x7fffffff;
+  for (int c = lane; c < nchunks; c += 64) {
+    const float v = ws[((long)b * nchunks + c) * 2];
+    const int i = ((const int*)ws)[((long)b * nchunks + c) * 2 + 1];
     if (v > best || (v == best && i < besti)) { best = v; besti = i; }
   }
-  token[b] = besti;
-  const int st = *step;
-  if (out_tokens && st < max_new) out_tokens[(long)b * max_new + st] = besti;
+#pragma unroll
+  for (int off = 32; off; off >>= 1) {
+    const float ov = __shfl_xor(best, off);
+    const int oi = __shfl_xor(besti, off);
+    if (ov > best || (ov == best && oi < besti)) { best = ov; besti = oi; }
+  }
+  if (lane == 0) {
+    token[b] = besti;
+    const int st = *step;
+    if (out_tokens && st < max_new) out_tokens[(long)b * max_new + st] = besti;
+  }
 }
 
 void fei_sample(const void* logits, int* token, int* out_tokens,

@@ -1,0 +1,85 @@
+"""Memory tool layer tests (direct-FS handlers + MemoryManager facade)."""
+
+import pytest
+
+from fei_amd.tools.memory_tools import MemoryManager, MemoryTools, create_memory_tools
+from fei_amd.tools.registry import ToolRegistry
+
+
+@pytest.fixture
+def tools(memdir_base):
+    return MemoryTools(base=memdir_base)
+
+
+def test_create_view_list_delete(tools):
+    out = tools.create({"subject": "remember this", "tags": "a,b",
+                        "priority": "high", "body": "the body"})
+    assert out["success"]
+    mid = out["memory_id"]
+
+    mem = tools.view({"memory_id": mid})
+    assert mem["headers"]["Subject"] == "remember this"
+    assert mem["content"] == "the body"
+
+    lst = tools.list({})
+    assert lst["count"] == 1
+    assert lst["memories"][0]["memory_id"] == mid
+
+    out = tools.delete({"memory_id": mid})
+    assert out["success"] and out["folder"] == ".Trash"
+    assert tools.list({})["count"] == 0
+
+
+def test_search_and_tag(tools):
+    tools.create({"subject": "gpu kernel notes", "tags": "gpu",
+                  "body": "mfma tiling"})
+    tools.create({"subject": "grocery list", "tags": "life"})
+    out = tools.search({"query": "mfma", "with_content": True})
+    assert out["count"] == 1
+    out = tools.search_by_tag({"tag": "gpu"})
+    assert out["count"] == 1
+
+
+def test_registered_tool_names(memdir_base):
+    reg = ToolRegistry()
+    create_memory_tools(reg, base=memdir_base)
+    names = reg.list_tools()
+    for n in ["memory_search", "memory_semantic_search", "memory_index_build",
+              "memory_create", "memory_view", "memory_list", "memory_delete",
+              "memory_search_by_tag", "memdir_server_start",
+              "memdir_server_stop", "memdir_server_status"]:
+        assert n in names
+    out = reg.execute_tool("memory_create", {"subject": "via registry"})
+    assert out["success"]
+    out = reg.execute_tool("memory_search", {"query": "registry"})
+    assert out["count"] == 1
+
+
+def test_semantic_index_cpu(tools):
+    """CPU path: build a tiny index with the bge encoder on CPU and query it
+    (GPU speed comes from the same code path on MFMA)."""
+    tools.create({"subject": "rocprof kernel profiling", "tags": "gpu",
+                  "body": "per-kernel time breakdown on MI355X"})
+    tools.create({"subject": "pasta recipe", "tags": "cooking",
+                  "body": "boil water, add salt"})
+    out = tools.index_build({})
+    assert out["indexed"] == 2
+    res = tools.semantic_search({"query": "GPU profiling", "topk": 2})
+    assert res["count"] == 2
+    assert all("score" in m for m in res["results"])
+
+
+def test_memory_manager_save_conversation(memdir_base):
+    mgr = MemoryManager(base=memdir_base)
+    messages = [
+        {"role": "user", "content": "how do I profile kernels?"},
+        {"role": "assistant", "content": [
+            {"type": "text", "text": "use rocprofv3"},
+            {"type": "tool_use", "name": "Shell", "id": "1", "input": {}},
+        ]},
+    ]
+    out = mgr.save_conversation(messages, subject="profiling chat")
+    assert out["success"]
+    results = mgr.recall("rocprofv3")
+    assert len(results) == 1
+    assert "use rocprofv3" in results[0]["content"]

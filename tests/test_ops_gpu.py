@@ -247,6 +247,11 @@ def test_gemv_swiglu_matches(lib):
     x = randbf(M, K, seed=90)
     wgu = randbf(2 * I, K, seed=91, scale=0.02)
     out = lib.gemv_swiglu(x, wgu)
-    expected = ref.swiglu(F.linear(x.float(), wgu.float()).to(torch.bfloat16))
-    err = (out.float() - expected.float()).abs().max().item()
+    # the kernel keeps gate/up dots in f32 through silu; compare against the
+    # unrounded f32 reference (bf16-rounding g first shifts silu(g)*u by more
+    # than the kernel's true error)
+    gu32 = F.linear(x.float(), wgu.float())
+    g, u = gu32[..., :I], gu32[..., I:]
+    expected = torch.nn.functional.silu(g) * u
+    err = (out.float() - expected).abs().max().item()
     assert err < 2e-2, f"max err {err}"
