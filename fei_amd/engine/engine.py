@@ -90,6 +90,12 @@ class LocalEngine:
         )
         self._graph: Optional[torch.cuda.CUDAGraph] = None
         self.last_metrics: Dict[str, float] = {}
+        if self.is_gpu:
+            # one-time hipBLASLt init off the first prefill's critical path
+            a = torch.zeros(8, 64, device=device, dtype=self.dtype)
+            w = torch.zeros(64, 64, device=device, dtype=self.dtype)
+            torch.nn.functional.linear(a, w)
+            torch.cuda.synchronize(device)
 
     # -- construction helpers ------------------------------------------------
 
