@@ -253,5 +253,6 @@ def test_gemv_swiglu_matches(lib):
     gu32 = F.linear(x.float(), wgu.float())
     g, u = gu32[..., :I], gu32[..., I:]
     expected = torch.nn.functional.silu(g) * u
-    err = (out.float() - expected).abs().max().item()
-    assert err < 2e-2, f"max err {err}"
+    # relative tolerance: outputs reach |8|, where one bf16 ulp is ~0.03
+    err = ((out.float() - expected).abs() / (1 + expected.abs())).max().item()
+    assert err < 2e-2, f"max rel err {err}"
