@@ -55,7 +55,7 @@ def _try_load() -> None:
     lib.fei_attn_prefill.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp,
                                      _i, _i, _i, _i, _i, _i, _f, _i, _l, _vp]
     lib.fei_mfma_probe.argtypes = [_vp, _vp, _vp, _vp]
-    lib.fei_gemv.argtypes = [_vp, _vp, _vp, _i, _i, _i, _vp]
+    lib.fei_gemv.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
     lib.fei_gemv_swiglu.argtypes = [_vp, _vp, _vp, _i, _i, _i, _vp]
     _LIB = lib
 
@@ -301,6 +301,7 @@ def mfma_probe(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
 # -- decode GEMV dispatch ----------------------------------------------------
 
 _GEMV_MAX_LDS = 64 * 1024      # keep >=2 blocks/CU (160 KB LDS per CU)
+_GEMV_NT = os.environ.get("FEI_GEMV_NT", "1") not in ("0", "false")
 
 
 def _gemv_ok(M: int, K: int) -> bool:
@@ -321,7 +322,8 @@ def linear_decode(x: torch.Tensor, w: torch.Tensor,
     x2 = x.contiguous().view(M, K)
     if out is None:
         out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
-    lib.fei_gemv(_ptr(out), _ptr(x2), _ptr(w), M, N, K, _stream())
+    lib.fei_gemv(_ptr(out), _ptr(x2), _ptr(w), M, N, K,
+                 1 if _GEMV_NT else 0, _stream())
     return out
 
 
