@@ -1,6 +1,8 @@
 """GPU microbench: decode GEMV variants vs rocBLAS on the 8B shapes.
 Within-process interleaved A/B (guide rule 24)."""
-import os, time, torch, torch.nn.functional as F
+import os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch, torch.nn.functional as F
 
 def timeit(fn, iters=200, warmup=20):
     for _ in range(warmup):
