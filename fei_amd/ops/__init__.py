@@ -301,7 +301,7 @@ def mfma_probe(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
 # -- decode GEMV dispatch ----------------------------------------------------
 
 _GEMV_MAX_LDS = 64 * 1024      # keep >=2 blocks/CU (160 KB LDS per CU)
-_GEMV_NT = os.environ.get("FEI_GEMV_NT", "1") not in ("0", "false")
+_GEMV_NT_MIN_BYTES = 512 * 1024 * 1024   # NT pays only on >L3-size streams (lm_head)
 
 
 def _gemv_ok(M: int, K: int) -> bool:
@@ -322,8 +322,8 @@ def linear_decode(x: torch.Tensor, w: torch.Tensor,
     x2 = x.contiguous().view(M, K)
     if out is None:
         out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
-    lib.fei_gemv(_ptr(out), _ptr(x2), _ptr(w), M, N, K,
-                 1 if _GEMV_NT else 0, _stream())
+    nt = 1 if (N * K * 2 >= _GEMV_NT_MIN_BYTES) else 0
+    lib.fei_gemv(_ptr(out), _ptr(x2), _ptr(w), M, N, K, nt, _stream())
     return out
 
 
