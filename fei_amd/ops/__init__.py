@@ -300,12 +300,11 @@ def mfma_probe(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
 
 # -- decode GEMV dispatch ----------------------------------------------------
 
-_GEMV_MAX_LDS = 64 * 1024      # keep >=2 blocks/CU (160 KB LDS per CU)
 _GEMV_NT_MIN_BYTES = 512 * 1024 * 1024   # NT pays only on >L3-size streams (lm_head)
 
 
 def _gemv_ok(M: int, K: int) -> bool:
-    return M in (1, 2, 4, 8) and K % 8 == 0 and M * K * 2 <= _GEMV_MAX_LDS
+    return M in (1, 2, 4, 8) and K % 8 == 0
 
 
 def linear_decode(x: torch.Tensor, w: torch.Tensor,

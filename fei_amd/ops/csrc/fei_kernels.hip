@@ -356,11 +356,15 @@ k_attn_decode(const u16* __restrict__ q, const u16* __restrict__ kc,
     // P*V: V row read once, used by every head
     #pragma unroll
     for (int g = 0; g < G; ++g) { o0[g] *= alpha[g]; o1[g] *= alpha[g]; }
+    // Break-free trip count so hipcc can unroll and keep several V loads
+    // in flight (a data-dependent break serialises the loop into dependent
+    // L2 round trips — measured 41 us/kernel at seq 640).
     const int kbase_local = kg * keys_per_group;
     const int kmax = min(DEC_TILE, end - tile);
-    for (int j = 0; j < keys_per_group; ++j) {
+    const int iters = min(keys_per_group, max(0, kmax - kbase_local));
+#pragma unroll 4
+    for (int j = 0; j < iters; ++j) {
       const int kl = kbase_local + j;
-      if (kl >= kmax) break;
       const u16* vrow = vbase + (long)(tile + kl) * D + dp * 2;
       const float v0 = bf2f(vrow[0]);
       const float v1 = bf2f(vrow[1]);

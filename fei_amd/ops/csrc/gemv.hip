@@ -2,11 +2,13 @@
 // out[M, N] = x[M, K] @ W[N, K]^T  — torch F.linear weight layout (W row-
 // major over K). M <= 8 (decode batch), K % 512 == 0.
 //
-// Design: one W row per wave, streamed with 16 B/lane coalesced loads
-// (64 lanes x 16 B = 2 cachelines per instruction); x staged in LDS and
-// broadcast-read; per-lane f32 dot then wave shfl reduce. This is pure
-// HBM streaming — rocBLAS's tiled GEMM kernels reach only ~4 TB/s on
-// M=1 shapes (profiles/r01), streaming reaches the flat-read ceiling.
+// Design: two W rows per wave, streamed with 16 B/lane coalesced loads
+// (64 lanes x 16 B = 2 cachelines per instruction; 2 rows = 2 independent
+// load streams for memory-level parallelism); x read straight through
+// L1/L2 (8-28 KB, cache-hot — LDS staging only added startup latency);
+// per-lane f32 dot then wave shfl reduce. This is pure HBM streaming —
+// rocBLAS's tiled GEMM kernels reach only ~4 TB/s on M=1 shapes
+// (profiles/r01), streaming reaches the flat-read ceiling.
 //
 // fei_gemv_swiglu fuses the MLP gate/up pair: wave computes both dots
 // (rows n and n+N) and writes silu(g)*u — kills the separate swiglu
@@ -22,16 +24,12 @@ k_gemv(u16* __restrict__ out, const u16* __restrict__ x,
   // 4 waves per block, TWO W rows per wave: two independent load streams
   // per wave double the in-flight memory requests (the fused swiglu kernel
   // streams 2 rows/wave and measures ~25 % more bandwidth than 1 row/wave).
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  u16* xs = (u16*)smem;                       // [M][K]
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
   const int n0 = blockIdx.x * 8 + wid * 2;
-  for (int i = tid; i < M * (K >> 3); i += blockDim.x)
-    ((s16x8*)xs)[i] = ((const s16x8*)x)[i];
-  __syncthreads();
   if (n0 >= N) return;
+  const u16* xs = x;                          // L1/L2-hot, no staging
   const bool two = (n0 + 1) < N;
 
   const s16x8* wrow0 = (const s16x8*)(w + (long)n0 * K);
@@ -72,16 +70,12 @@ template <int M>
 __global__ void __launch_bounds__(256)
 k_gemv_swiglu(u16* __restrict__ out, const u16* __restrict__ x,
               const u16* __restrict__ w, int N, int K) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  u16* xs = (u16*)smem;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
   const int n = blockIdx.x * 4 + wid;
-  for (int i = tid; i < M * (K >> 3); i += blockDim.x)
-    ((s16x8*)xs)[i] = ((const s16x8*)x)[i];
-  __syncthreads();
   if (n >= N) return;
+  const u16* xs = x;                          // L1/L2-hot, no staging
 
   const s16x8* grow = (const s16x8*)(w + (long)n * K);
   const s16x8* urow = (const s16x8*)(w + (long)(n + N) * K);
@@ -117,7 +111,7 @@ extern "C" {
 void fei_gemv(void* out, const void* x, const void* w, int M, int N, int K,
               int nontemporal, hipStream_t stream) {
   dim3 grid((N + 7) / 8);
-  const size_t lds = (size_t)M * K * 2;
+  const size_t lds = 0;
 #define LG(MV, NTV) hipLaunchKernelGGL((k_gemv<MV, NTV>), grid, dim3(256), \
                                        lds, stream, (u16*)out, \
                                        (const u16*)x, (const u16*)w, N, K)
@@ -136,7 +130,7 @@ void fei_gemv(void* out, const void* x, const void* w, int M, int N, int K,
 void fei_gemv_swiglu(void* out, const void* x, const void* w, int M, int N,
                      int K, hipStream_t stream) {
   dim3 grid((N + 3) / 4);
-  const size_t lds = (size_t)M * K * 2;
+  const size_t lds = 0;
 #define LG(MV) hipLaunchKernelGGL(k_gemv_swiglu<MV>, grid, dim3(256), lds, \
                                   stream, (u16*)out, (const u16*)x, \
                                   (const u16*)w, N, K)
