@@ -1,0 +1,110 @@
+"""UI tests: CLI chat loop / single message with the scripted backend,
+history persistence, and TUI construction without running the app
+(reference parity: test_textual.py:20-31)."""
+
+import io
+import json
+import os
+
+import pytest
+
+
+@pytest.fixture
+def home(tmp_path, monkeypatch):
+    monkeypatch.setenv("HOME", str(tmp_path))
+    # ChatHistory resolves ~ at import time; patch the path constant
+    import fei_amd.ui.cli as cli_mod
+    monkeypatch.setattr(cli_mod, "HISTORY_PATH",
+                        str(tmp_path / ".fei" / "history.json"))
+    return tmp_path
+
+
+def test_single_message_scripted(home):
+    from fei_amd.ui.cli import CLI
+    cli = CLI(script=[{"content": "forty-two"}], with_memory=False)
+    out = cli.single_message("meaning of life?")
+    assert out == "forty-two"
+    # history persisted
+    data = json.loads((home / ".fei" / "history.json").read_text())
+    assert data[0]["prompt"] == "meaning of life?"
+
+
+def test_task_mode(home):
+    from fei_amd.ui.cli import CLI
+    cli = CLI(script=[{"content": "step"}, {"content": "done [TASK_COMPLETE]"}],
+              with_memory=False)
+    out = cli.single_message("do it", task=True)
+    assert out == "done"
+
+
+def test_chat_loop_commands(home):
+    from fei_amd.ui.cli import CLI
+    cli = CLI(script=[{"content": "hi there"}], with_memory=False)
+    stdin = io.StringIO("hello\nhistory\nexit\n")
+    stdout = io.StringIO()
+    rc = cli.chat_loop(stdin=stdin, stdout=stdout)
+    assert rc == 0
+    text = stdout.getvalue()
+    assert "fei> hi there" in text
+    assert "> hello" in text          # history echo
+
+
+def test_chat_loop_clear(home):
+    from fei_amd.ui.cli import CLI
+    cli = CLI(script=[{"content": "x"}], with_memory=False)
+    stdin = io.StringIO("one\nclear\nquit\n")
+    stdout = io.StringIO()
+    cli.chat_loop(stdin=stdin, stdout=stdout)
+    assert cli.assistant.conversation.messages == []
+
+
+def test_history_cap(home):
+    from fei_amd.ui.cli import ChatHistory
+    h = ChatHistory(str(home / ".fei" / "history.json"))
+    for i in range(120):
+        h.add(f"p{i}", f"r{i}")
+    assert len(h.entries) == 100
+    assert h.entries[0]["prompt"] == "p20"
+
+
+def test_cli_main_message(home, capsys):
+    from fei_amd.ui.cli import main
+    rc = main(["--provider", "stub", "--no-memory", "-m", "echo this back"])
+    assert rc == 0
+    assert "echo this back" in capsys.readouterr().out
+
+
+def test_cli_search_subcommand(home, tmp_path, monkeypatch, capsys):
+    monkeypatch.setenv("MEMDIR_BASE", str(tmp_path / "Memdir"))
+    from fei_amd.memdir import utils as mu
+    mu.create_memory("", {"Subject": "findme note"}, "", status="cur")
+    from fei_amd.ui.cli import main
+    rc = main(["search", "findme"])
+    assert rc == 0
+    assert "findme note" in capsys.readouterr().out
+
+
+def test_tui_construction(home):
+    """Construct the app + exercise /mem handlers without running it."""
+    from fei_amd.ui.tui import FeiChatApp, MemCommandSuggester
+    from fei_amd.tools.memory_tools import MemoryTools
+
+    tools = MemoryTools(base=str(home / "Memdir"))
+    tools.create({"subject": "tui memory", "tags": "t"})
+    app = FeiChatApp(assistant=None, memory_tools=tools)
+    out = app.handle_memory_command("/mem list")
+    assert "tui memory" in out
+    out = app.handle_memory_command("/mem search tui")
+    assert "tui memory" in out
+    out = app.handle_memory_command("/mem help")
+    assert "/mem save" in out
+    out = app.handle_memory_command("/mem bogus")
+    assert "unknown" in out
+
+
+def test_tui_suggester():
+    import asyncio
+    from fei_amd.ui.tui import MemCommandSuggester
+    s = MemCommandSuggester()
+    assert asyncio.run(s.get_suggestion("/mem se")) == "/mem search "
+    assert asyncio.run(s.get_suggestion("hello")) is None
