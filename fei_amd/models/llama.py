@@ -224,3 +224,77 @@ class LlamaModel:
         logits = F.linear(last, self.lm_head)
         logits = all_gather_cat(logits, dim=-1, ctx=self.tp)
         return logits
+
+    # -- checkpointing (safetensors) -----------------------------------------
+
+    def state_dict_full(self) -> Dict[str, torch.Tensor]:
+        """Unsharded state dict (tp=1 only): HF-free flat naming."""
+        if self.tp_size != 1:
+            raise RuntimeError("state_dict_full requires tp=1 (save from an "
+                               "unsharded model; any-tp models can LOAD it)")
+        s = self.spec
+        out: Dict[str, torch.Tensor] = {
+            "emb": self.emb, "norm_f": self.norm_f, "lm_head": self.lm_head,
+        }
+        for i, lw in enumerate(self.layers):
+            qd = s.num_heads * self.D
+            kvd = s.num_kv_heads * self.D
+            out[f"layers.{i}.wq"] = lw.wqkv[:qd]
+            out[f"layers.{i}.wk"] = lw.wqkv[qd:qd + kvd]
+            out[f"layers.{i}.wv"] = lw.wqkv[qd + kvd:]
+            out[f"layers.{i}.wo"] = lw.wo
+            out[f"layers.{i}.wg"] = lw.wgu[:s.intermediate_size]
+            out[f"layers.{i}.wu"] = lw.wgu[s.intermediate_size:]
+            out[f"layers.{i}.wd"] = lw.wdown
+            out[f"layers.{i}.norm_attn"] = lw.norm_attn
+            out[f"layers.{i}.norm_mlp"] = lw.norm_mlp
+        return out
+
+    def save_weights(self, path: str) -> None:
+        """Save an unsharded checkpoint (tp=1) as safetensors + spec json."""
+        import json as _json
+        import os as _os
+        from dataclasses import asdict
+        from safetensors.torch import save_file
+
+        _os.makedirs(path, exist_ok=True)
+        tensors = {k: v.contiguous().cpu() for k, v in self.state_dict_full().items()}
+        save_file(tensors, _os.path.join(path, "model.safetensors"))
+        with open(_os.path.join(path, "config.json"), "w") as f:
+            _json.dump(asdict(self.spec), f, indent=2)
+
+    def load_weights(self, path: str) -> None:
+        """Load an unsharded checkpoint, slicing this rank's TP shard
+        (safetensors -> per-rank shards)."""
+        import os as _os
+        from safetensors.torch import load_file
+
+        s = self.spec
+        full = load_file(_os.path.join(path, "model.safetensors"))
+        r = self.tp_rank
+        dev, dt = self.device, self.dtype
+
+        def to_dev(t):
+            return t.to(device=dev, dtype=dt)
+
+        self.emb = to_dev(full["emb"])
+        self.norm_f = to_dev(full["norm_f"])
+        lm = full["lm_head"]
+        self.lm_head = to_dev(lm[r * self.vocab_l:(r + 1) * self.vocab_l]).contiguous()
+        C = s.hidden_size
+        for i, lw in enumerate(self.layers):
+            wq = full[f"layers.{i}.wq"].view(s.num_heads, self.D, C)
+            wk = full[f"layers.{i}.wk"].view(s.num_kv_heads, self.D, C)
+            wv = full[f"layers.{i}.wv"].view(s.num_kv_heads, self.D, C)
+            wq = wq[r * self.hq_l:(r + 1) * self.hq_l].reshape(-1, C)
+            wk = wk[r * self.hkv_l:(r + 1) * self.hkv_l].reshape(-1, C)
+            wv = wv[r * self.hkv_l:(r + 1) * self.hkv_l].reshape(-1, C)
+            lw.wqkv = to_dev(torch.cat([wq, wk, wv], dim=0)).contiguous()
+            wo = full[f"layers.{i}.wo"].view(C, s.num_heads, self.D)
+            lw.wo = to_dev(wo[:, r * self.hq_l:(r + 1) * self.hq_l].reshape(C, -1)).contiguous()
+            wg = full[f"layers.{i}.wg"][r * self.inter_l:(r + 1) * self.inter_l]
+            wu = full[f"layers.{i}.wu"][r * self.inter_l:(r + 1) * self.inter_l]
+            lw.wgu = to_dev(torch.cat([wg, wu], dim=0)).contiguous()
+            lw.wdown = to_dev(full[f"layers.{i}.wd"][:, r * self.inter_l:(r + 1) * self.inter_l]).contiguous()
+            lw.norm_attn = to_dev(full[f"layers.{i}.norm_attn"])
+            lw.norm_mlp = to_dev(full[f"layers.{i}.norm_mlp"])

@@ -94,3 +94,26 @@ def test_kv_pool_alloc_release():
     with pytest.raises(MemoryError):
         s3 = pool.new_sequence()
         pool.ensure_capacity(s3, 16 * 100)
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    """save_weights (tp=1) -> load into a differently-seeded model ->
+    identical logits (safetensors checkpoint path)."""
+    spec = get_spec("llama3-tiny")
+    m1 = LlamaModel(spec, torch.device("cpu"), torch.float32, seed=7,
+                    max_seq_len=64)
+    m1.save_weights(str(tmp_path / "ckpt"))
+
+    m2 = LlamaModel(spec, torch.device("cpu"), torch.float32, seed=999,
+                    max_seq_len=64)
+    tokens = torch.tensor([[1, 4, 9]])
+    pos0 = torch.zeros(1, dtype=torch.int32)
+    kc, vc = m2.new_kv_cache(1, 64)
+    before = m2.forward_prefill(tokens, pos0, kc, vc)
+    m2.load_weights(str(tmp_path / "ckpt"))
+    kc, vc = m2.new_kv_cache(1, 64)
+    after = m2.forward_prefill(tokens, pos0, kc, vc)
+    kc, vc = m1.new_kv_cache(1, 64)
+    ref = m1.forward_prefill(tokens, pos0, kc, vc)
+    assert not torch.allclose(before, ref)
+    assert torch.allclose(after, ref, atol=1e-5)
