@@ -61,6 +61,9 @@ class LocalEngine:
             use_hip_graph = self.is_gpu and not self.tp.is_distributed
         self.use_graph = use_hip_graph and self.is_gpu
         self.attn_splits = attn_splits
+        # fused single-pass attention (rope+append+attn in one kernel) for
+        # agent-length contexts; split-K path for long context
+        self.fused_attn = (max_seq_len or spec.max_seq_len) <= 2048
         self.seed = seed
         self.temperature = 0.0       # graph-captured; set before capture
         self.tokenizer = ByteTokenizer()
@@ -88,6 +91,8 @@ class LocalEngine:
             torch.zeros(self.B, Hq_l, attn_splits, D, dtype=torch.float32, device=device),
             torch.zeros(self.B, Hq_l, attn_splits, 2, dtype=torch.float32, device=device),
         )
+        self.attn_out = torch.zeros(self.B, Hq_l, D, dtype=self.dtype,
+                                    device=device)
         self._graph: Optional[torch.cuda.CUDAGraph] = None
         self.last_metrics: Dict[str, float] = {}
         if self.is_gpu:
@@ -109,7 +114,8 @@ class LocalEngine:
     def _decode_step(self) -> None:
         logits = self.model.forward_decode(
             self.token, self.pos, self.k_caches, self.v_caches,
-            attn_splits=self.attn_splits, workspace=self.attn_ws)
+            attn_splits=self.attn_splits, workspace=self.attn_ws,
+            fused_attn=self.fused_attn, attn_out=self.attn_out)
         ops.sample(logits, self.token, self.step, self.sample_ws.view(self.B, -1),
                    out_tokens=self.out_tokens, temperature=self.temperature,
                    seed=self.seed, nchunks=SAMPLE_CHUNKS)

@@ -160,8 +160,10 @@ class LlamaModel:
         pos: torch.Tensor,            # [B] int32 (current length; kv written here)
         k_caches: List[torch.Tensor],
         v_caches: List[torch.Tensor],
-        attn_splits: int = 16,
+        attn_splits: int = 4,
         workspace=None,
+        fused_attn: bool = False,
+        attn_out: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
         """One decode step -> logits [B, vocab] (gathered across TP)."""
         s = self.spec
@@ -173,10 +175,16 @@ class LlamaModel:
         for li, lw in enumerate(self.layers):
             qkv = ops.linear_decode(x, lw.wqkv)
             q, k, v = self._qkv_views(qkv, B)
-            ops.rope_kv_decode(q, k, v, k_caches[li], v_caches[li], pos, self.rope)
-            att = ops.attn_decode(q, k_caches[li], v_caches[li], pos,
-                                  splits=attn_splits, scale=scale,
-                                  workspace=workspace)
+            if fused_attn:
+                att = ops.attn_decode_fused(q, k, v, k_caches[li],
+                                            v_caches[li], pos, self.rope,
+                                            scale=scale, out=attn_out)
+            else:
+                ops.rope_kv_decode(q, k, v, k_caches[li], v_caches[li], pos,
+                                   self.rope)
+                att = ops.attn_decode(q, k_caches[li], v_caches[li], pos,
+                                      splits=attn_splits, scale=scale,
+                                      workspace=workspace)
             o = ops.linear_decode(att.reshape(B, -1), lw.wo)
             all_reduce_sum(o, self.tp)
             x, h = ops.fused_add_rmsnorm(o, h, lw.norm_mlp, s.norm_eps)

@@ -56,6 +56,11 @@ def _try_load() -> None:
                                      _i, _i, _i, _i, _i, _i, _f, _i, _l, _vp]
     lib.fei_mfma_probe.argtypes = [_vp, _vp, _vp, _vp]
     lib.fei_gemv.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
+    lib.fei_attn_decode_fused.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp,
+                                          _vp, _i, _i, _i, _i, _i, _f,
+                                          _l, _l, _vp]
+    lib.fei_sample_onepass.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _f, _u64,
+                                       _i, _vp]
     lib.fei_gemv_swiglu.argtypes = [_vp, _vp, _vp, _i, _i, _i, _vp]
     _LIB = lib
 
@@ -190,6 +195,30 @@ def attn_decode(q, k_cache, v_cache, pos, splits: int = 4,
     return out
 
 
+def attn_decode_fused(q, k, v, k_cache, v_cache, pos, table,
+                      scale: Optional[float] = None,
+                      out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Fused decode attention: RoPE(q,k) + KV-append + GQA attention in one
+    kernel (grid Hkv x B, no split partials). For agent-length sequences;
+    long-context uses rope_kv_decode + attn_decode (split-K)."""
+    B, Hq, D = q.shape
+    Hkv = k.shape[1]
+    scale = scale if scale is not None else 1.0 / math.sqrt(D)
+    if not q.is_cuda:
+        rope_kv_decode(q, k, v, k_cache, v_cache, pos, table)
+        return ref.attn_decode(q, k_cache, v_cache, pos + 1, scale)
+    lib = require_lib()
+    assert q.stride(2) == 1 and q.stride(1) == D
+    assert k.stride(1) == D and k.stride(0) == v.stride(0)
+    if out is None:
+        out = torch.empty(B, Hq, D, dtype=torch.bfloat16, device=q.device)
+    lib.fei_attn_decode_fused(_ptr(q), _ptr(k), _ptr(v), _ptr(k_cache),
+                              _ptr(v_cache), _ptr(out), _ptr(table), _ptr(pos),
+                              B, Hq, Hkv, D, k_cache.shape[2], scale,
+                              q.stride(0), k.stride(0), _stream())
+    return out
+
+
 def attn_prefill(q, k_cache, v_cache, pos0, scale: Optional[float] = None,
                  causal: bool = True, kv_len: Optional[torch.Tensor] = None,
                  out: Optional[torch.Tensor] = None) -> torch.Tensor:
@@ -271,10 +300,14 @@ def sample(logits: torch.Tensor, token: torch.Tensor,
         return token
     lib = require_lib()
     max_new = out_tokens.shape[1] if out_tokens is not None else 0
-    lib.fei_sample(_ptr(logits), _ptr(token),
-                   _ptr(out_tokens) if out_tokens is not None else None,
-                   _ptr(step), _ptr(workspace), B, V, nchunks,
-                   temperature, seed, max_new, _stream())
+    out_ptr = _ptr(out_tokens) if out_tokens is not None else None
+    if B <= 8:
+        lib.fei_sample_onepass(_ptr(logits), _ptr(token), out_ptr, _ptr(step),
+                               B, V, temperature, seed, max_new, _stream())
+    else:
+        lib.fei_sample(_ptr(logits), _ptr(token), out_ptr, _ptr(step),
+                       _ptr(workspace), B, V, nchunks, temperature, seed,
+                       max_new, _stream())
     return token
 
 

@@ -611,3 +611,77 @@ void fei_advance(int* pos, int* step, int B, hipStream_t stream) {
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// One-pass sampler for small B: grid (B), block 1024 = 16 waves streaming
+// the whole vocab row; replaces the partial+final pair (one node, no
+// workspace round trip).
+// ---------------------------------------------------------------------------
+extern "C" {
+
+__global__ void __launch_bounds__(1024)
+k_sample_onepass(const u16* __restrict__ logits, int* __restrict__ token,
+                 int* __restrict__ out_tokens, const int* __restrict__ step,
+                 int B, int vocab, float temperature, u64 seed, int max_new) {
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  const u64 st = (u64)(*step);
+  const float invT = temperature > 0.f ? 1.f / temperature : 1.f;
+  float best = -1.0f / 0.0f;
+  int besti = 0x7fffffff;
+  const s16x8* row = (const s16x8*)(logits + (long)b * vocab);
+  const int nv = vocab >> 3;
+  for (int i = tid; i < nv; i += blockDim.x) {
+    s16x8 v8 = row[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int idx = i * 8 + j;
+      float v = bf2f((u16)v8[j]) * invT;
+      if (temperature > 0.f) {
+        float u = hash_uniform(seed ^ (st * 0x51ed27f1ull) ^ ((u64)b << 40) ^ (u64)idx);
+        v += -__logf(-__logf(u));
+      }
+      if (v > best || (v == best && idx < besti)) { best = v; besti = idx; }
+    }
+  }
+  // tail (vocab % 8)
+  for (int idx = nv * 8 + tid; idx < vocab; idx += blockDim.x) {
+    float v = bf2f(logits[(long)b * vocab + idx]) * invT;
+    if (temperature > 0.f) {
+      float u = hash_uniform(seed ^ (st * 0x51ed27f1ull) ^ ((u64)b << 40) ^ (u64)idx);
+      v += -__logf(-__logf(u));
+    }
+    if (v > best || (v == best && idx < besti)) { best = v; besti = idx; }
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) {
+    const float ov = __shfl_xor(best, off);
+    const int oi = __shfl_xor(besti, off);
+    if (ov > best || (ov == best && oi < besti)) { best = ov; besti = oi; }
+  }
+  __shared__ float rv[16];
+  __shared__ int ri[16];
+  const int wid = tid >> 6;
+  if ((tid & 63) == 0) { rv[wid] = best; ri[wid] = besti; }
+  __syncthreads();
+  if (tid == 0) {
+    const int nw = blockDim.x >> 6;
+    for (int i = 1; i < nw; ++i)
+      if (rv[i] > best || (rv[i] == best && ri[i] < besti)) {
+        best = rv[i]; besti = ri[i];
+      }
+    token[b] = besti;
+    if (out_tokens && (int)st < max_new)
+      out_tokens[(long)b * max_new + (int)st] = besti;
+  }
+}
+
+void fei_sample_onepass(const void* logits, int* token, int* out_tokens,
+                        const int* step, int B, int vocab, float temperature,
+                        u64 seed, int max_new, hipStream_t stream) {
+  hipLaunchKernelGGL(k_sample_onepass, dim3(B), dim3(1024), 0, stream,
+                     (const u16*)logits, token, out_tokens, step, B, vocab,
+                     temperature, seed, max_new);
+}
+
+}  // extern "C"

@@ -256,3 +256,45 @@ def test_gemv_swiglu_matches(lib):
     # relative tolerance: outputs reach |8|, where one bf16 ulp is ~0.03
     err = ((out.float() - expected).abs() / (1 + expected.abs())).max().item()
     assert err < 2e-2, f"max rel err {err}"
+
+
+@pytest.mark.parametrize("n", [1, 63, 640])
+def test_attn_decode_fused(lib, n):
+    """Fused rope+append+attention vs the reference pipeline."""
+    from fei_amd.ops import reference as ref
+    B, Hq, Hkv, D, MS = 2, 8, 2, 128, 1024
+    W = (Hq + 2 * Hkv) * D
+    qkv = randbf(B, W, seed=100 + n, scale=0.5)
+    q = qkv.as_strided((B, Hq, D), (W, D, 1))
+    k = qkv.as_strided((B, Hkv, D), (W, D, 1), storage_offset=Hq * D)
+    v = qkv.as_strided((B, Hkv, D), (W, D, 1), storage_offset=(Hq + Hkv) * D)
+    kc = _mk_cache(B, Hkv, MS, D, seed=101 + n)
+    vc = _mk_cache(B, Hkv, MS, D, seed=102 + n)
+    pos = torch.tensor([n - 1, max(n // 2 - 1, 0)], dtype=torch.int32, device=DEV)
+    table = ref.rope_table(MS, D, device=DEV)
+
+    kc_ref, vc_ref = kc.clone(), vc.clone()
+    q_ref = ref.rope_kv_decode(q.contiguous().clone(), k.contiguous(),
+                               v.contiguous(), kc_ref, vc_ref, pos, table)
+    expected = ref.attn_decode(q_ref, kc_ref, vc_ref, pos + 1)
+
+    out = lib.attn_decode_fused(q, k, v, kc, vc, pos, table)
+    err = (out.float() - expected.float()).abs().max().item()
+    assert err < 2e-2, f"n={n}: max err {err}"
+    # cache rows written identically
+    assert (kc.float() - kc_ref.float()).abs().max() < 2e-2
+    assert torch.equal(vc, vc_ref)
+
+
+def test_sample_onepass_greedy(lib):
+    B, V = 2, 128256
+    g = torch.Generator(device=DEV).manual_seed(170)
+    logits = torch.randn(B, V, generator=g, device=DEV).to(torch.bfloat16)
+    token = torch.zeros(B, dtype=torch.int32, device=DEV)
+    step = torch.zeros(1, dtype=torch.int32, device=DEV)
+    out_tokens = torch.zeros(B, 8, dtype=torch.int32, device=DEV)
+    ws = torch.zeros(B, 64 * 2, dtype=torch.float32, device=DEV)
+    lib.sample(logits, token, step, ws, out_tokens=out_tokens, temperature=0.0)
+    expected = logits.float().argmax(dim=-1)
+    assert torch.equal(token.long(), expected)
+    assert torch.equal(out_tokens[:, 0].long(), expected)
