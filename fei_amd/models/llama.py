@@ -180,11 +180,11 @@ class LlamaModel:
                                             v_caches[li], pos, self.rope,
                                             scale=scale, out=attn_out)
             else:
-                ops.rope_kv_decode(q, k, v, k_caches[li], v_caches[li], pos,
-                                   self.rope)
+                # split-K attention with fused in-kernel RoPE + KV-append
                 att = ops.attn_decode(q, k_caches[li], v_caches[li], pos,
                                       splits=attn_splits, scale=scale,
-                                      workspace=workspace)
+                                      workspace=workspace, out=attn_out,
+                                      k=k, v=v, table=self.rope)
             o = ops.linear_decode(att.reshape(B, -1), lw.wo)
             all_reduce_sum(o, self.tp)
             x, h = ops.fused_add_rmsnorm(o, h, lw.norm_mlp, s.norm_eps)

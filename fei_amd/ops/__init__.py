@@ -46,7 +46,8 @@ def _try_load() -> None:
     lib.fei_rope_kv_prefill.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp,
                                         _i, _i, _i, _i, _i, _i, _l, _l, _vp]
     lib.fei_attn_decode.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp,
-                                    _i, _i, _i, _i, _i, _i, _f, _l, _vp]
+                                    _i, _i, _i, _i, _i, _i, _f, _l,
+                                    _vp, _vp, _vp, _l, _vp]
     lib.fei_attn_decode_combine.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
     lib.fei_swiglu.argtypes = [_vp, _vp, _l, _i, _vp]
     lib.fei_sample.argtypes = [_vp, _vp, _vp, _vp, _vp, _i, _i, _i, _f, _u64,
@@ -169,12 +170,19 @@ def rope_kv_prefill(q, k, v, k_cache, v_cache, pos0, table) -> torch.Tensor:
 def attn_decode(q, k_cache, v_cache, pos, splits: int = 4,
                 scale: Optional[float] = None,
                 workspace: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
-                out: Optional[torch.Tensor] = None) -> torch.Tensor:
+                out: Optional[torch.Tensor] = None,
+                k: Optional[torch.Tensor] = None,
+                v: Optional[torch.Tensor] = None,
+                table: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Single-token GQA attention over n = pos[b]+1 cache entries.
-    q [B,Hq,D] -> out [B,Hq,D]. The length is read on DEVICE (hipGraph)."""
+    q [B,Hq,D] -> out [B,Hq,D]. The length is read on DEVICE (hipGraph).
+    When (k, v, table) are given, the kernel also fuses the step's RoPE +
+    KV-append: q is the RAW qkv view and cache[pos] is written in-kernel."""
     B, Hq, D = q.shape
     scale = scale if scale is not None else 1.0 / math.sqrt(D)
     if not q.is_cuda:
+        if table is not None:
+            rope_kv_decode(q, k, v, k_cache, v_cache, pos, table)
         seqlen = pos + 1
         return ref.attn_decode(q, k_cache, v_cache, seqlen, scale)
     lib = require_lib()
@@ -187,9 +195,17 @@ def attn_decode(q, k_cache, v_cache, pos, splits: int = 4,
         out = torch.empty_like(q)
     Hkv = k_cache.shape[1]
     assert q.stride(2) == 1 and q.stride(1) == D
+    if table is not None:
+        assert k is not None and v is not None
+        assert k.stride(1) == D and k.stride(0) == v.stride(0)
+        kin, vin, cs, kv_bs = _ptr(k), _ptr(v), _ptr(table), k.stride(0)
+    else:
+        kin = vin = cs = None
+        kv_bs = 0
     lib.fei_attn_decode(_ptr(q), _ptr(k_cache), _ptr(v_cache), _ptr(part_o),
                         _ptr(part_ml), _ptr(pos), B, Hq, Hkv, D,
-                        k_cache.shape[2], splits, scale, q.stride(0), _stream())
+                        k_cache.shape[2], splits, scale, q.stride(0),
+                        kin, vin, cs, kv_bs, _stream())
     lib.fei_attn_decode_combine(_ptr(out), _ptr(part_o), _ptr(part_ml),
                                 B, Hq, D, splits, _stream())
     return out

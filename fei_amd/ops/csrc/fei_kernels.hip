@@ -255,13 +255,21 @@ __device__ __forceinline__ void block_reduce_vec(float* v,
   __syncthreads();
 }
 
-template <int G>
+// ROPE=true additionally fuses the per-step RoPE + KV-append: q is the RAW
+// qkv-buffer view (roped into LDS during staging), the new token's k/v are
+// roped in-block from kin/vin (every split computes them — the new key may
+// fall in any split's chunk) and the split that OWNS key n-1 writes the
+// caches. Readers never read cache[n-1]; they use the LDS copy, so there is
+// no cross-workgroup ordering (placement-independent by construction).
+template <int G, bool ROPE>
 __global__ void __launch_bounds__(256)
-k_attn_decode(const u16* __restrict__ q, const u16* __restrict__ kc,
-              const u16* __restrict__ vc, float* __restrict__ part_o,
+k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
+              u16* __restrict__ vc, float* __restrict__ part_o,
               float* __restrict__ part_ml, const int* __restrict__ pos,
               int B, int Hq, int Hkv, int D, int max_seq, int splits,
-              float scale, long q_bs) {
+              float scale, long q_bs,
+              const u16* __restrict__ kin, const u16* __restrict__ vin,
+              const float* __restrict__ cos_sin, long kv_bs) {
   const int split = blockIdx.x;
   const int hkv = blockIdx.y;
   const int b = blockIdx.z;
@@ -271,6 +279,8 @@ k_attn_decode(const u16* __restrict__ q, const u16* __restrict__ kc,
   __shared__ float pl[DEC_GMAX][DEC_TILE];
   __shared__ float red[DEC_GMAX][4];
   __shared__ float osh[DEC_DMAX / 2][2];
+  __shared__ float knf[DEC_DMAX];
+  __shared__ float vnf[DEC_DMAX];
 
   const int n = pos[b] + 1;
   const int chunk = (n + splits - 1) / splits;
@@ -289,10 +299,50 @@ k_attn_decode(const u16* __restrict__ q, const u16* __restrict__ kc,
     return;
   }
 
-  // stage the group's q vectors (scaled) into LDS
-  for (int i = tid; i < G * D; i += blockDim.x) {
-    const int g = i / D, d = i % D;
-    qs[g][d] = bf2f(q[(long)b * q_bs + (long)(hkv * G + g) * D + d]) * scale;
+  const int half = D / 2;
+  const int p_new = n - 1;
+  if (ROPE) {
+    // stage roped+scaled q
+    for (int i = tid; i < G * half; i += blockDim.x) {
+      const int g = i / half, d = i % half;
+      const float c = cos_sin[((long)p_new * half + d) * 2 + 0];
+      const float sn = cos_sin[((long)p_new * half + d) * 2 + 1];
+      const u16* qp = q + (long)b * q_bs + (long)(hkv * G + g) * D;
+      const float x1 = bf2f(qp[d]);
+      const float x2 = bf2f(qp[d + half]);
+      qs[g][d] = (x1 * c - x2 * sn) * scale;
+      qs[g][d + half] = (x2 * c + x1 * sn) * scale;
+    }
+    // rope the new token's k; copy v (LDS copy used by every split)
+    const bool owner = (p_new >= start && p_new < end);
+    u16* kcp = kc + (((long)b * Hkv + hkv) * max_seq + p_new) * D;
+    u16* vcp = vc + (((long)b * Hkv + hkv) * max_seq + p_new) * D;
+    for (int d = tid; d < half; d += blockDim.x) {
+      const float c = cos_sin[((long)p_new * half + d) * 2 + 0];
+      const float sn = cos_sin[((long)p_new * half + d) * 2 + 1];
+      const u16* kp = kin + (long)b * kv_bs + (long)hkv * D;
+      const u16* vp = vin + (long)b * kv_bs + (long)hkv * D;
+      const float x1 = bf2f(kp[d]);
+      const float x2 = bf2f(kp[d + half]);
+      const float k1 = x1 * c - x2 * sn;
+      const float k2 = x2 * c + x1 * sn;
+      knf[d] = k1;
+      knf[d + half] = k2;
+      vnf[d] = bf2f(vp[d]);
+      vnf[d + half] = bf2f(vp[d + half]);
+      if (owner) {
+        kcp[d] = f2bf(k1);
+        kcp[d + half] = f2bf(k2);
+        vcp[d] = vp[d];
+        vcp[d + half] = vp[d + half];
+      }
+    }
+  } else {
+    // q pre-roped; stage scaled
+    for (int i = tid; i < G * D; i += blockDim.x) {
+      const int g = i / D, d = i % D;
+      qs[g][d] = bf2f(q[(long)b * q_bs + (long)(hkv * G + g) * D + d]) * scale;
+    }
   }
   __syncthreads();
 
@@ -365,9 +415,10 @@ k_attn_decode(const u16* __restrict__ q, const u16* __restrict__ kc,
 #pragma unroll 4
     for (int j = 0; j < iters; ++j) {
       const int kl = kbase_local + j;
+      const bool is_new = ROPE && (tile + kl == p_new);
       const u16* vrow = vbase + (long)(tile + kl) * D + dp * 2;
-      const float v0 = bf2f(vrow[0]);
-      const float v1 = bf2f(vrow[1]);
+      const float v0 = is_new ? vnf[dp * 2] : bf2f(vrow[0]);
+      const float v1 = is_new ? vnf[dp * 2 + 1] : bf2f(vrow[1]);
       #pragma unroll
     for (int g = 0; g < G; ++g) {
         const float p = pl[g][kl];
@@ -406,21 +457,28 @@ extern "C" {
 void fei_attn_decode(const void* q, const void* k_cache, const void* v_cache,
                      float* part_o, float* part_ml, const int* pos,
                      int B, int Hq, int Hkv, int D, int max_seq, int splits,
-                     float scale, long q_bs, hipStream_t stream) {
+                     float scale, long q_bs,
+                     const void* kin, const void* vin, const float* cos_sin,
+                     long kv_bs, hipStream_t stream) {
   const int G = Hq / Hkv;
   dim3 grid(splits, Hkv, B);
-#define LAUNCH_DEC(GV) \
-  hipLaunchKernelGGL(k_attn_decode<GV>, grid, dim3(256), 0, stream, \
-                     (const u16*)q, (const u16*)k_cache, (const u16*)v_cache, \
+  const int rope = cos_sin != nullptr;
+#define LAUNCH_DEC(GV, RV) \
+  hipLaunchKernelGGL((k_attn_decode<GV, RV>), grid, dim3(256), 0, stream, \
+                     (const u16*)q, (u16*)k_cache, (u16*)v_cache, \
                      part_o, part_ml, pos, B, Hq, Hkv, D, max_seq, splits, \
-                     scale, q_bs)
+                     scale, q_bs, (const u16*)kin, (const u16*)vin, cos_sin, \
+                     kv_bs)
+#define LAUNCH_DEC_R(GV) do { if (rope) LAUNCH_DEC(GV, true); \
+                              else LAUNCH_DEC(GV, false); } while (0)
   switch (G) {
-    case 1: LAUNCH_DEC(1); break;
-    case 2: LAUNCH_DEC(2); break;
-    case 4: LAUNCH_DEC(4); break;
-    case 8: LAUNCH_DEC(8); break;
+    case 1: LAUNCH_DEC_R(1); break;
+    case 2: LAUNCH_DEC_R(2); break;
+    case 4: LAUNCH_DEC_R(4); break;
+    case 8: LAUNCH_DEC_R(8); break;
     default: break;  // unsupported group size: wrapper validates
   }
+#undef LAUNCH_DEC_R
 #undef LAUNCH_DEC
 }
 

@@ -298,3 +298,31 @@ def test_sample_onepass_greedy(lib):
     expected = logits.float().argmax(dim=-1)
     assert torch.equal(token.long(), expected)
     assert torch.equal(out_tokens[:, 0].long(), expected)
+
+
+@pytest.mark.parametrize("n,splits", [(1, 4), (70, 4), (640, 4), (300, 16)])
+def test_attn_decode_rope_fused_split(lib, n, splits):
+    """Split-K attention with in-kernel RoPE+append vs the reference
+    pipeline (covers owner-split cache writes and LDS new-key path)."""
+    from fei_amd.ops import reference as ref
+    B, Hq, Hkv, D, MS = 2, 8, 2, 128, 1024
+    W = (Hq + 2 * Hkv) * D
+    qkv = randbf(B, W, seed=200 + n, scale=0.5)
+    q = qkv.as_strided((B, Hq, D), (W, D, 1))
+    k = qkv.as_strided((B, Hkv, D), (W, D, 1), storage_offset=Hq * D)
+    v = qkv.as_strided((B, Hkv, D), (W, D, 1), storage_offset=(Hq + Hkv) * D)
+    kc = _mk_cache(B, Hkv, MS, D, seed=201 + n)
+    vc = _mk_cache(B, Hkv, MS, D, seed=202 + n)
+    pos = torch.tensor([n - 1, max(n // 2 - 1, 0)], dtype=torch.int32, device=DEV)
+    table = ref.rope_table(MS, D, device=DEV)
+
+    kc_ref, vc_ref = kc.clone(), vc.clone()
+    q_ref = ref.rope_kv_decode(q.contiguous().clone(), k.contiguous(),
+                               v.contiguous(), kc_ref, vc_ref, pos, table)
+    expected = ref.attn_decode(q_ref, kc_ref, vc_ref, pos + 1)
+
+    out = lib.attn_decode(q, kc, vc, pos, splits=splits, k=k, v=v, table=table)
+    err = (out.float() - expected.float()).abs().max().item()
+    assert err < 2e-2, f"n={n} splits={splits}: max err {err}"
+    assert (kc.float() - kc_ref.float()).abs().max() < 2e-2
+    assert torch.equal(vc, vc_ref)
