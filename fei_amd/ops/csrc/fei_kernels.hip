@@ -279,8 +279,6 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
   __shared__ float pl[DEC_GMAX][DEC_TILE];
   __shared__ float red[DEC_GMAX][4];
   __shared__ float osh[DEC_DMAX / 2][2];
-  __shared__ float knf[DEC_DMAX];
-  __shared__ float vnf[DEC_DMAX];
 
   const int n = pos[b] + 1;
   const int chunk = (n + splits - 1) / splits;
@@ -313,26 +311,22 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
       qs[g][d] = (x1 * c - x2 * sn) * scale;
       qs[g][d + half] = (x2 * c + x1 * sn) * scale;
     }
-    // rope the new token's k; copy v (LDS copy used by every split)
-    const bool owner = (p_new >= start && p_new < end);
-    u16* kcp = kc + (((long)b * Hkv + hkv) * max_seq + p_new) * D;
-    u16* vcp = vc + (((long)b * Hkv + hkv) * max_seq + p_new) * D;
-    for (int d = tid; d < half; d += blockDim.x) {
-      const float c = cos_sin[((long)p_new * half + d) * 2 + 0];
-      const float sn = cos_sin[((long)p_new * half + d) * 2 + 1];
-      const u16* kp = kin + (long)b * kv_bs + (long)hkv * D;
-      const u16* vp = vin + (long)b * kv_bs + (long)hkv * D;
-      const float x1 = bf2f(kp[d]);
-      const float x2 = bf2f(kp[d + half]);
-      const float k1 = x1 * c - x2 * sn;
-      const float k2 = x2 * c + x1 * sn;
-      knf[d] = k1;
-      knf[d + half] = k2;
-      vnf[d] = bf2f(vp[d]);
-      vnf[d + half] = bf2f(vp[d + half]);
-      if (owner) {
-        kcp[d] = f2bf(k1);
-        kcp[d + half] = f2bf(k2);
+    // the split whose chunk contains key p_new ropes + appends the new
+    // token's k/v BEFORE this block's barrier, then reads it back from the
+    // cache like any other key (same-workgroup store->load visibility).
+    // No other split's [start,end) range ever touches key p_new.
+    if (p_new >= start && p_new < end) {
+      u16* kcp = kc + (((long)b * Hkv + hkv) * max_seq + p_new) * D;
+      u16* vcp = vc + (((long)b * Hkv + hkv) * max_seq + p_new) * D;
+      for (int d = tid; d < half; d += blockDim.x) {
+        const float c = cos_sin[((long)p_new * half + d) * 2 + 0];
+        const float sn = cos_sin[((long)p_new * half + d) * 2 + 1];
+        const u16* kp = kin + (long)b * kv_bs + (long)hkv * D;
+        const u16* vp = vin + (long)b * kv_bs + (long)hkv * D;
+        const float x1 = bf2f(kp[d]);
+        const float x2 = bf2f(kp[d + half]);
+        kcp[d] = f2bf(x1 * c - x2 * sn);
+        kcp[d + half] = f2bf(x2 * c + x1 * sn);
         vcp[d] = vp[d];
         vcp[d + half] = vp[d + half];
       }
@@ -415,10 +409,9 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
 #pragma unroll 4
     for (int j = 0; j < iters; ++j) {
       const int kl = kbase_local + j;
-      const bool is_new = ROPE && (tile + kl == p_new);
       const u16* vrow = vbase + (long)(tile + kl) * D + dp * 2;
-      const float v0 = is_new ? vnf[dp * 2] : bf2f(vrow[0]);
-      const float v1 = is_new ? vnf[dp * 2 + 1] : bf2f(vrow[1]);
+      const float v0 = bf2f(vrow[0]);
+      const float v1 = bf2f(vrow[1]);
       #pragma unroll
     for (int g = 0; g < G; ++g) {
         const float p = pl[g][kl];
