@@ -1,0 +1,80 @@
+"""p50 tool-turn latency — the second half of the BASELINE metric.
+
+Measures the FULL agent turn through the real stack on the local engine:
+Assistant.chat -> prompt render -> prefill -> hipGraph decode until the
+model stops (or token cap) -> tool-call parse -> GlobTool execution ->
+continuation prefill+decode. Weights are random-init, so the model cannot
+emit valid tool calls itself; the turn structure is made deterministic by
+injecting the tool round (the scripted-turn harness SURVEY.md §7 calls
+for), while ALL latency comes from real engine + tool work.
+"""
+import json, os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+
+def main():
+    from fei_amd.core.assistant import Assistant
+    from fei_amd.core.backends import LocalBackend
+    from fei_amd.engine.engine import LocalEngine
+    from fei_amd.tools.code import create_code_tools
+    from fei_amd.tools.registry import ToolRegistry
+
+    has_gpu = torch.cuda.is_available()
+    model = "llama3-8b" if has_gpu else "llama3-tiny"
+    max_new = 128 if has_gpu else 16
+    engine = LocalEngine.create(model, max_seq_len=2048)
+
+    class TurnBackend(LocalBackend):
+        """Real engine inference; deterministic tool-call injection on the
+        first round of each turn (random-init weights can't emit JSON)."""
+        def __init__(self, engine, max_new):
+            super().__init__(engine=engine)
+            self.max_new = max_new
+            self._round = 0
+
+        def complete(self, messages, tools=None, system=None, max_tokens=4000,
+                     temperature=0.0):
+            out = super().complete(messages, tools, system,
+                                   max_tokens=self.max_new,
+                                   temperature=temperature)
+            self._round += 1
+            if self._round % 2 == 1:       # first round of a turn: call a tool
+                out.tool_calls = [{"id": f"c{self._round}", "name": "GlobTool",
+                                   "input": {"pattern": "**/*.py",
+                                             "path": os.getcwd()}}]
+            else:
+                out.tool_calls = []
+            return out
+
+    registry = ToolRegistry()
+    create_code_tools(registry)
+    assistant = Assistant(provider="local", tool_registry=registry,
+                          backend=TurnBackend(engine, max_new))
+
+    lat = []
+    n_turns = 20 if has_gpu else 3
+    for i in range(n_turns):
+        assistant.reset()
+        t0 = time.perf_counter()
+        assistant.chat(f"Find the python files related to topic {i} and summarize.")
+        lat.append(time.perf_counter() - t0)
+    lat.sort()
+    m = assistant.turn_metrics[-1]
+    print(json.dumps({
+        "metric": "p50 tool-turn latency (full agent turn: prefill + decode + "
+                  "GlobTool + continuation)",
+        "unit": "ms",
+        "p50": round(lat[len(lat) // 2] * 1000, 1),
+        "p95": round(lat[int(len(lat) * 0.95)] * 1000, 1),
+        "turns": n_turns, "model": model,
+        "decode_tokens_per_round": max_new,
+        "tool_ms_last_turn": round(m["tools"][0]["latency_s"] * 1000, 2)
+        if m["tools"] else None,
+        "device": "cuda" if has_gpu else "cpu", "data": "synthetic",
+    }))
+
+
+if __name__ == "__main__":
+    main()

@@ -67,3 +67,42 @@ def test_native_lib_is_loaded():
     lib = ops.require_lib()
     assert "libfei_kernels.so" in (lib._name or "")
     assert ops._LIB_PATH.startswith(ops.os.path.dirname(ops.os.path.abspath(ops.__file__)))
+
+
+def test_bge_encoder_gpu_matches_cpu():
+    """bge-class encoder: GPU (HIP kernels, bf16) vs CPU (torch ref, fp32)
+    with identical seeds — cosine similarity of embeddings ~1."""
+    import torch
+    import torch.nn.functional as F
+    from fei_amd.engine.config import get_spec
+    from fei_amd.models.bge import BgeEncoder
+
+    spec = get_spec("bge-base")
+    gpu = BgeEncoder(spec, device=torch.device("cuda:0"), seed=5)
+    ids = torch.randint(4, 200, (3, 64), device="cuda:0")
+    e_gpu = gpu.encode_ids(ids).cpu()
+
+    cpu = BgeEncoder(spec, device=torch.device("cpu"), seed=5)
+    e_cpu = cpu.encode_ids(ids.cpu())
+    # seeds differ across devices' RNG streams? BgeEncoder seeds a
+    # device-local generator — weights must match; verify first
+    if not torch.allclose(gpu.tok_emb.float().cpu(), cpu.tok_emb.float(),
+                          atol=1e-2):
+        pytest.skip("device RNG streams differ; cross-device weight "
+                    "reproducibility not guaranteed")
+    cos = F.cosine_similarity(e_gpu, e_cpu, dim=-1)
+    assert cos.min() > 0.98, cos
+
+
+def test_assistant_local_backend_turn():
+    """Full agent turn through the real GPU engine (tiny model)."""
+    from fei_amd.core.assistant import Assistant
+    from fei_amd.core.backends import LocalBackend
+    from fei_amd.engine.engine import LocalEngine
+
+    engine = LocalEngine.create("llama3-tiny", max_seq_len=256)
+    a = Assistant(backend=LocalBackend(engine=engine))
+    a.max_tokens = 16
+    out = a.chat("hello engine")
+    assert isinstance(out, str)
+    assert a.turn_metrics[0]["usage"]["output_tokens"] >= 1
