@@ -162,11 +162,13 @@ class LocalBackend(Backend):
 
     name = "local"
 
-    def __init__(self, engine=None, model: str = "llama3-8b", **engine_kwargs):
+    def __init__(self, engine=None, model: str = "llama3-8b",
+                 stop_on_eos: bool = True, **engine_kwargs):
         if engine is None:
             from fei_amd.engine.engine import LocalEngine
             engine = LocalEngine.create(model, **engine_kwargs)
         self.engine = engine
+        self.stop_on_eos = stop_on_eos
 
     @staticmethod
     def render_prompt(messages, tools=None, system=None) -> str:
@@ -208,7 +210,8 @@ class LocalBackend(Backend):
                  temperature=0.0) -> BackendResponse:
         prompt = self.render_prompt(messages, tools, system)
         out = self.engine.generate(prompt, max_new_tokens=max_tokens,
-                                   temperature=temperature)
+                                    temperature=temperature,
+                                    stop_on_eos=self.stop_on_eos)
         text = out["text"]
         calls = extract_tool_call_blocks(text)
         return BackendResponse(

@@ -53,27 +53,35 @@ def main():
     assistant = Assistant(provider="local", tool_registry=registry,
                           backend=TurnBackend(engine, max_new))
 
-    lat = []
     n_turns = 20 if has_gpu else 3
-    for i in range(n_turns):
-        assistant.reset()
-        t0 = time.perf_counter()
-        assistant.chat(f"Find the python files related to topic {i} and summarize.")
-        lat.append(time.perf_counter() - t0)
-    lat.sort()
-    m = assistant.turn_metrics[-1]
-    print(json.dumps({
-        "metric": "p50 tool-turn latency (full agent turn: prefill + decode + "
-                  "GlobTool + continuation)",
-        "unit": "ms",
-        "p50": round(lat[len(lat) // 2] * 1000, 1),
-        "p95": round(lat[int(len(lat) * 0.95)] * 1000, 1),
-        "turns": n_turns, "model": model,
-        "decode_tokens_per_round": max_new,
-        "tool_ms_last_turn": round(m["tools"][0]["latency_s"] * 1000, 2)
-        if m["tools"] else None,
-        "device": "cuda" if has_gpu else "cpu", "data": "synthetic",
-    }))
+    backend = assistant.providers.get_backend()
+    for fixed_budget in (False, True):
+        # fixed_budget=True decodes the full token budget every round
+        # (random-init weights hit EOS at arbitrary points otherwise)
+        backend.stop_on_eos = not fixed_budget
+        lat, toks = [], []
+        for i in range(n_turns):
+            assistant.reset()
+            t0 = time.perf_counter()
+            assistant.chat(f"Find the python files about topic {i} and summarize.")
+            lat.append(time.perf_counter() - t0)
+            toks.append(engine.last_metrics.get("new_tokens", 0))
+        lat.sort()
+        m = assistant.turn_metrics[-1]
+        print(json.dumps({
+            "metric": "p50 tool-turn latency (prefill + decode + GlobTool + "
+                      "continuation)" + (" [fixed decode budget]"
+                                         if fixed_budget else " [stop on EOS]"),
+            "unit": "ms",
+            "p50": round(lat[len(lat) // 2] * 1000, 1),
+            "p95": round(lat[int(len(lat) * 0.95)] * 1000, 1),
+            "turns": n_turns, "model": model,
+            "decode_budget_per_round": max_new,
+            "mean_new_tokens_final_round": round(sum(toks) / len(toks), 1),
+            "tool_ms_last_turn": round(m["tools"][0]["latency_s"] * 1000, 2)
+            if m["tools"] else None,
+            "device": "cuda" if has_gpu else "cpu", "data": "synthetic",
+        }))
 
 
 if __name__ == "__main__":
