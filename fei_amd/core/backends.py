@@ -175,14 +175,19 @@ class LocalBackend(Backend):
         parts: List[str] = []
         sys_text = system or "You are a helpful coding assistant."
         if tools:
-            tool_desc = json.dumps([{"name": t["name"],
-                                     "description": t.get("description", ""),
-                                     "input_schema": t.get("input_schema", {})}
-                                    for t in tools])
+            # compact tool list: name, one-line description, required args.
+            # (The byte-level tokenizer costs ~1 token/char; full JSON
+            # schemas would eat the context. Argument validation happens in
+            # the registry regardless.)
+            lines = []
+            for t in tools:
+                req = ",".join(t.get("input_schema", {}).get("required", []))
+                desc = t.get("description", "").split(". ")[0][:100]
+                lines.append(f"- {t['name']}({req}): {desc}")
             sys_text += (
                 "\nYou can call tools by emitting "
                 '<tool_call>{"name": "...", "arguments": {...}}</tool_call>.'
-                f"\nAvailable tools: {tool_desc}"
+                "\nAvailable tools:\n" + "\n".join(lines)
             )
         parts.append(f"<|system|>\n{sys_text}\n")
         for msg in messages:

@@ -24,7 +24,7 @@ def main():
     has_gpu = torch.cuda.is_available()
     model = "llama3-8b" if has_gpu else "llama3-tiny"
     max_new = 128 if has_gpu else 16
-    engine = LocalEngine.create(model, max_seq_len=2048)
+    engine = LocalEngine.create(model, max_seq_len=8192 if has_gpu else 512)
 
     class TurnBackend(LocalBackend):
         """Real engine inference; deterministic tool-call injection on the
