@@ -169,6 +169,8 @@ class LocalBackend(Backend):
             engine = LocalEngine.create(model, **engine_kwargs)
         self.engine = engine
         self.stop_on_eos = stop_on_eos
+        # prefix cache: token ids currently materialised in the KV caches
+        self._cached_ids: List[int] = []
 
     @staticmethod
     def render_prompt(messages, tools=None, system=None) -> str:
@@ -214,9 +216,19 @@ class LocalBackend(Backend):
     def complete(self, messages, tools=None, system=None, max_tokens=4000,
                  temperature=0.0) -> BackendResponse:
         prompt = self.render_prompt(messages, tools, system)
-        out = self.engine.generate(prompt, max_new_tokens=max_tokens,
-                                    temperature=temperature,
-                                    stop_on_eos=self.stop_on_eos)
+        ids = self.engine.tokenizer.encode(prompt)
+        # prefix cache: a turn's continuation prompt extends the previous
+        # one, so only the delta needs prefilling (the engine's KV caches
+        # already hold the prefix + its generated tokens are NOT part of
+        # the rendered prompt — cache only the prompt prefix).
+        n_common = 0
+        if self._cached_ids and len(ids) > len(self._cached_ids) and                 ids[:len(self._cached_ids)] == self._cached_ids:
+            n_common = len(self._cached_ids)
+        out = self.engine.generate(ids, max_new_tokens=max_tokens,
+                                   temperature=temperature,
+                                   stop_on_eos=self.stop_on_eos,
+                                   from_pos=n_common)
+        self._cached_ids = ids
         text = out["text"]
         calls = extract_tool_call_blocks(text)
         return BackendResponse(

@@ -154,16 +154,27 @@ class LocalEngine:
 
     # -- public API ----------------------------------------------------------
 
-    def prefill(self, token_ids: List[int]) -> None:
-        """Reset state and prefill the prompt; samples the first new token."""
+    def prefill(self, token_ids: List[int], from_pos: int = 0) -> None:
+        """Prefill ``token_ids`` starting at cache position ``from_pos``
+        (0 = fresh conversation; >0 = prefix-cache extension: the first
+        ``from_pos`` tokens are already in the KV caches) and sample the
+        first new token."""
         S = len(token_ids)
-        if S >= self.max_seq_len:
-            token_ids = token_ids[-(self.max_seq_len - 1):]
+        if from_pos + S >= self.max_seq_len:
+            # drop the oldest suffix tokens; a full-context agent should
+            # summarise instead (UI concern)
+            keep = self.max_seq_len - 1 - from_pos
+            if keep <= 0:
+                from_pos = 0
+                token_ids = token_ids[-(self.max_seq_len - 1):]
+            else:
+                token_ids = token_ids[:keep]
             S = len(token_ids)
         self.ensure_graph()
         tokens = torch.tensor([token_ids] * self.B, dtype=torch.int64,
                               device=self.device)
-        pos0 = torch.zeros(self.B, dtype=torch.int32, device=self.device)
+        pos0 = torch.full((self.B,), from_pos, dtype=torch.int32,
+                          device=self.device)
         logits = self.model.forward_prefill(tokens, pos0, self.k_caches,
                                             self.v_caches)
         self.step.zero_()
@@ -172,7 +183,7 @@ class LocalEngine:
                    self.sample_ws.view(self.B, -1), out_tokens=self.out_tokens,
                    temperature=self.temperature, seed=self.seed,
                    nchunks=SAMPLE_CHUNKS)
-        self.pos.fill_(S)
+        self.pos.fill_(from_pos + S)
         if self.is_gpu:
             torch.cuda.synchronize(self.device)
         else:
@@ -210,13 +221,21 @@ class LocalEngine:
         return rows
 
     def generate(self, prompt: Union[str, List[int]], max_new_tokens: int = 256,
-                 temperature: float = 0.0, stop_on_eos: bool = True
-                 ) -> Dict[str, object]:
-        """Prefill + decode; returns text and timing metrics."""
+                 temperature: float = 0.0, stop_on_eos: bool = True,
+                 from_pos: int = 0) -> Dict[str, object]:
+        """Prefill + decode; returns text and timing metrics. ``from_pos``
+        enables prefix caching: the prompt's first ``from_pos`` tokens are
+        already in the KV caches and only the remainder is prefilled."""
         if isinstance(prompt, str):
             prompt_ids = self.tokenizer.encode(prompt)
         else:
             prompt_ids = list(prompt)
+        if from_pos > len(prompt_ids):
+            from_pos = 0
+        new_ids = prompt_ids[from_pos:]
+        if not new_ids:                      # identical prompt: redo last token
+            from_pos = max(0, len(prompt_ids) - 1)
+            new_ids = prompt_ids[from_pos:]
         max_new_tokens = min(max_new_tokens, self.max_seq_len - len(prompt_ids) - 1)
         if max_new_tokens < 1:
             max_new_tokens = 1
@@ -224,7 +243,7 @@ class LocalEngine:
         if self.is_gpu:
             torch.cuda.synchronize(self.device)
         t0 = time.perf_counter()
-        self.prefill(prompt_ids)
+        self.prefill(new_ids, from_pos=from_pos)
         t1 = time.perf_counter()
         rows = self.decode(max_new_tokens, stop_on_eos=stop_on_eos)
         t2 = time.perf_counter()
@@ -232,10 +251,11 @@ class LocalEngine:
         decode_s = t2 - t1
         self.last_metrics = {
             "prompt_tokens": len(prompt_ids),
+            "cached_prefix": from_pos,
             "new_tokens": new_tokens,
             "prefill_s": t1 - t0,
             "decode_s": decode_s,
-            "prefill_tok_s": len(prompt_ids) / max(t1 - t0, 1e-9),
+            "prefill_tok_s": len(new_ids) / max(t1 - t0, 1e-9),
             "decode_tok_s": (max(new_tokens - 1, 0) * self.B) / max(decode_s, 1e-9),
         }
         text = self.tokenizer.decode(rows[0])

@@ -117,3 +117,32 @@ def test_checkpoint_roundtrip(tmp_path):
     ref = m1.forward_prefill(tokens, pos0, kc, vc)
     assert not torch.allclose(before, ref)
     assert torch.allclose(after, ref, atol=1e-5)
+
+
+def test_prefix_cache_extension_matches_scratch(engine):
+    """generate(full) from scratch == generate(prefix) then
+    generate(full, from_pos=len(prefix)) — the prefix-cache contract."""
+    tok = engine.tokenizer
+    p1 = tok.encode("first user message")
+    suffix = tok.encode("and the continuation", add_bos=False)
+    full = p1 + suffix
+
+    a = engine.generate(full, max_new_tokens=5, stop_on_eos=False)
+
+    engine.generate(p1, max_new_tokens=2, stop_on_eos=False)  # fills cache
+    b = engine.generate(full, max_new_tokens=5, stop_on_eos=False,
+                        from_pos=len(p1))
+    assert b["cached_prefix"] == len(p1)
+    assert a["token_ids"] == b["token_ids"]
+
+
+def test_local_backend_prefix_cache(engine):
+    from fei_amd.core.backends import LocalBackend
+    backend = LocalBackend(engine=engine)
+    msgs = [{"role": "user", "content": "question one"}]
+    backend.complete(msgs, max_tokens=4)
+    assert engine.last_metrics["cached_prefix"] == 0
+    msgs = msgs + [{"role": "assistant", "content": "answer one"},
+                   {"role": "user", "content": "question two"}]
+    backend.complete(msgs, max_tokens=4)
+    assert engine.last_metrics["cached_prefix"] > 0
