@@ -320,6 +320,12 @@ class MCPManager:
             raise MCPError(f"no MCP server matches tool {tool_name}")
 
         registry.register_prefix_hook("mcp_", hook)
+        # the reference special-cases the bare "brave_web_search" tool name
+        # (registry.py:340-467); same surface via a prefix hook:
+        registry.register_prefix_hook(
+            "brave_web_search",
+            lambda name, args: {"results": self.brave_search.search(
+                args.get("query", ""), count=int(args.get("count", 5)))})
 
     def shutdown(self) -> None:
         self.client.procs.stop_all()

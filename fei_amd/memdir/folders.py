@@ -151,3 +151,25 @@ class MemdirFolderManager:
                     f.write(mu.format_memory_content(headers, mem.get("content", "")))
                 n += 1
         return n
+
+    # -- symlinked views (reference: folders.py:382-426) ----------------------
+
+    def link_folder(self, target: str, link_name: str) -> bool:
+        """Create a symlinked folder view (e.g. a project alias)."""
+        if not self.folder_exists(target) or self.folder_exists(link_name):
+            return False
+        link_path = self._folder_path(link_name)
+        os.makedirs(os.path.dirname(link_path) or self.base, exist_ok=True)
+        try:
+            os.symlink(self._folder_path(target), link_path,
+                       target_is_directory=True)
+            return True
+        except OSError:
+            return False
+
+    def unlink_folder(self, link_name: str) -> bool:
+        link_path = self._folder_path(link_name)
+        if not os.path.islink(link_path):
+            return False
+        os.unlink(link_path)
+        return True
