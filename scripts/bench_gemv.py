@@ -4,7 +4,12 @@ import os, sys, time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch, torch.nn.functional as F
 
-def timeit(fn, iters=200, warmup=20):
+ITERS = int(os.environ.get("GEMV_ITERS", "200"))
+
+
+def timeit(fn, iters=None, warmup=None):
+    iters = iters or ITERS
+    warmup = warmup if warmup is not None else max(2, ITERS // 10)
     for _ in range(warmup):
         fn()
     torch.cuda.synchronize()
