@@ -195,6 +195,7 @@ def attn_decode(q, k_cache, v_cache, pos, splits: int = 4,
         out = torch.empty_like(q)
     Hkv = k_cache.shape[1]
     assert q.stride(2) == 1 and q.stride(1) == D
+    assert splits <= 32, "combine kernel stages at most 32 split partials"
     if table is not None:
         assert k is not None and v is not None
         assert k.stride(1) == D and k.stride(0) == v.stride(0)

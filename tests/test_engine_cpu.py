@@ -146,3 +146,13 @@ def test_local_backend_prefix_cache(engine):
                    {"role": "user", "content": "question two"}]
     backend.complete(msgs, max_tokens=4)
     assert engine.last_metrics["cached_prefix"] > 0
+
+
+def test_engine_batch_decode():
+    """batch>1: every sequence decodes the same greedy tokens for the same
+    prompt (replicated prompts; independent agent sessions)."""
+    e = LocalEngine.create("llama3-tiny", max_seq_len=64, batch_size=3, seed=3)
+    e.prefill(e.tokenizer.encode("batch test"))
+    rows = e.decode(4, stop_on_eos=False)
+    assert len(rows) == 3
+    assert rows[0] == rows[1] == rows[2]

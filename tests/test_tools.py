@@ -161,3 +161,16 @@ def test_shell_pipeline_heads_checked():
 def test_shell_timeout():
     out = ShellRunner().run("sleep 5", timeout=0.3)
     assert "error" in out and "timed out" in out["error"]
+
+
+def test_shell_background_mode(tmp_path):
+    runner = ShellRunner()
+    marker = tmp_path / "done.txt"
+    out = runner.run(f"sh -c 'sleep 0.2; touch {marker}'", timeout=10,
+                     background=True)
+    assert out.get("background") and out.get("pid")
+    import time
+    deadline = time.time() + 5
+    while not marker.exists() and time.time() < deadline:
+        time.sleep(0.05)
+    assert marker.exists()
