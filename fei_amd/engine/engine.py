@@ -71,6 +71,9 @@ class LocalEngine:
             self.fused_attn = env not in ("0", "false")
         else:
             self.fused_attn = batch_size * spec.num_kv_heads >= 64
+        env = _os.environ.get("FEI_FUSED_NORM")
+        self.fused_norm = (env not in ("0", "false")) if env is not None else \
+            (self.is_gpu and not self.tp.is_distributed)
         self.seed = seed
         self.temperature = 0.0       # graph-captured; set before capture
         self.tokenizer = ByteTokenizer()
@@ -122,7 +125,8 @@ class LocalEngine:
         logits = self.model.forward_decode(
             self.token, self.pos, self.k_caches, self.v_caches,
             attn_splits=self.attn_splits, workspace=self.attn_ws,
-            fused_attn=self.fused_attn, attn_out=self.attn_out)
+            fused_attn=self.fused_attn, attn_out=self.attn_out,
+            fused_norm=self.fused_norm)
         ops.sample(logits, self.token, self.step, self.sample_ws.view(self.B, -1),
                    out_tokens=self.out_tokens, temperature=self.temperature,
                    seed=self.seed, nchunks=SAMPLE_CHUNKS)

@@ -164,8 +164,16 @@ class LlamaModel:
         workspace=None,
         fused_attn: bool = False,
         attn_out: Optional[torch.Tensor] = None,
+        fused_norm: bool = False,
     ) -> torch.Tensor:
-        """One decode step -> logits [B, vocab] (gathered across TP)."""
+        """One decode step -> logits [B, vocab] (gathered across TP).
+        ``fused_norm`` (tp=1 only): the residual stream stays inside the
+        GEMV kernels — norm-prologue QKV/gate-up, residual-epilogue O/down;
+        5 kernels per layer instead of 9."""
+        if fused_norm and self.tp_size == 1:
+            return self._forward_decode_fused_norm(
+                token, pos, k_caches, v_caches, attn_splits, workspace,
+                fused_attn, attn_out)
         s = self.spec
         B = token.shape[0]
         h = F.embedding(token.long(), self.emb)          # residual stream
@@ -197,6 +205,30 @@ class LlamaModel:
         logits = ops.linear_decode(x, self.lm_head)
         logits = all_gather_cat(logits, dim=-1, ctx=self.tp)
         return logits
+
+    def _forward_decode_fused_norm(self, token, pos, k_caches, v_caches,
+                                   attn_splits, workspace, fused_attn,
+                                   attn_out):
+        s = self.spec
+        B = token.shape[0]
+        h = F.embedding(token.long(), self.emb).contiguous()
+        scale = 1.0 / math.sqrt(self.D)
+        for li, lw in enumerate(self.layers):
+            qkv = ops.gemv_norm(h, lw.norm_attn, lw.wqkv, s.norm_eps)
+            q, k, v = self._qkv_views(qkv, B)
+            if fused_attn:
+                att = ops.attn_decode_fused(q, k, v, k_caches[li],
+                                            v_caches[li], pos, self.rope,
+                                            scale=scale, out=attn_out)
+            else:
+                att = ops.attn_decode(q, k_caches[li], v_caches[li], pos,
+                                      splits=attn_splits, scale=scale,
+                                      workspace=workspace, out=attn_out,
+                                      k=k, v=v, table=self.rope)
+            ops.gemv_res(att.reshape(B, -1), lw.wo, h)
+            act = ops.gemv_swiglu_norm(h, lw.norm_mlp, lw.wgu, s.norm_eps)
+            ops.gemv_res(act, lw.wdown, h)
+        return ops.gemv_norm(h, self.norm_f, self.lm_head, s.norm_eps)
 
     def forward_prefill(
         self,

@@ -63,6 +63,10 @@ def _try_load() -> None:
     lib.fei_sample_onepass.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _f, _u64,
                                        _i, _vp]
     lib.fei_gemv_swiglu.argtypes = [_vp, _vp, _vp, _i, _i, _i, _vp]
+    lib.fei_gemv_res.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
+    lib.fei_gemv_norm.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _i, _f, _i, _vp]
+    lib.fei_gemv_swiglu_norm.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _i, _f,
+                                         _vp]
     _LIB = lib
 
 
@@ -389,4 +393,57 @@ def gemv_swiglu(x: torch.Tensor, wgu: torch.Tensor,
     if out is None:
         out = torch.empty(*x.shape[:-1], I, dtype=x.dtype, device=x.device)
     lib.fei_gemv_swiglu(_ptr(out), _ptr(x2), _ptr(wgu), M, I, K, _stream())
+    return out
+
+
+def gemv_res(x: torch.Tensor, w: torch.Tensor, res: torch.Tensor) -> torch.Tensor:
+    """res += x @ w^T (epilogue residual add; in place on res [M,N])."""
+    M = x.numel() // x.shape[-1]
+    K = x.shape[-1]
+    N = w.shape[0]
+    if not x.is_cuda or not _gemv_ok(M, K):
+        lin = torch.nn.functional.linear(x, w)
+        res.copy_((res.float() + lin.float().view_as(res)).to(res.dtype))
+        return res
+    lib = require_lib()
+    x2 = x.contiguous().view(M, K)
+    nt = 1 if (N * K * 2 >= _GEMV_NT_MIN_BYTES) else 0
+    lib.fei_gemv_res(_ptr(res), _ptr(x2), _ptr(w), M, N, K, nt, _stream())
+    return res
+
+
+def gemv_norm(res: torch.Tensor, wnorm: torch.Tensor, w: torch.Tensor,
+              eps: float = 1e-5,
+              out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """out = rmsnorm(res)*wnorm @ w^T (norm-prologue GEMV)."""
+    M = res.numel() // res.shape[-1]
+    K = res.shape[-1]
+    N = w.shape[0]
+    if not res.is_cuda or not _gemv_ok(M, K):
+        return torch.nn.functional.linear(rmsnorm(res, wnorm, eps), w)
+    lib = require_lib()
+    r2 = res.contiguous().view(M, K)
+    if out is None:
+        out = torch.empty(*res.shape[:-1], N, dtype=res.dtype, device=res.device)
+    nt = 1 if (N * K * 2 >= _GEMV_NT_MIN_BYTES) else 0
+    lib.fei_gemv_norm(_ptr(out), _ptr(r2), _ptr(wnorm), _ptr(w), M, N, K,
+                      eps, nt, _stream())
+    return out
+
+
+def gemv_swiglu_norm(res: torch.Tensor, wnorm: torch.Tensor,
+                     wgu: torch.Tensor, eps: float = 1e-5,
+                     out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """out = silu(g)*u where [g;u] = rmsnorm(res)*wnorm @ wgu^T."""
+    M = res.numel() // res.shape[-1]
+    K = res.shape[-1]
+    I = wgu.shape[0] // 2
+    if not res.is_cuda or not _gemv_ok(M, K):
+        return swiglu(torch.nn.functional.linear(rmsnorm(res, wnorm, eps), wgu))
+    lib = require_lib()
+    r2 = res.contiguous().view(M, K)
+    if out is None:
+        out = torch.empty(*res.shape[:-1], I, dtype=res.dtype, device=res.device)
+    lib.fei_gemv_swiglu_norm(_ptr(out), _ptr(r2), _ptr(wnorm), _ptr(wgu),
+                             M, I, K, eps, _stream())
     return out

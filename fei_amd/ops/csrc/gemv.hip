@@ -145,3 +145,216 @@ void fei_gemv_swiglu(void* out, const void* x, const void* w, int M, int N,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// Fused-chain GEMV variants (tp=1 decode): the residual stream never leaves
+// the GEMV kernels —
+//   k_gemv_res:  res[m,n] += x[m,:] @ W[n,:]      (epilogue residual add)
+//   k_gemv_norm: out = rmsnorm(res)*wn @ W^T      (norm prologue; every wave
+//                recomputes the row's sumsq from L1/L2-hot res — cheaper
+//                than a separate kernel + intermediate round trip)
+//   k_gemv_swiglu_norm: norm prologue + fused gate/up + SwiGLU
+// ---------------------------------------------------------------------------
+
+namespace {
+
+template <int M>
+__device__ __forceinline__ void norm_factors(const u16* __restrict__ res,
+                                             int K, float eps, int lane,
+                                             float inv[M]) {
+#pragma unroll
+  for (int m = 0; m < M; ++m) {
+    const s16x8* row = (const s16x8*)(res + (long)m * K);
+    float ss = 0.f;
+    const int nv = K >> 3;
+    for (int i = lane; i < nv; i += 64) {
+      s16x8 v = row[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = bf2f((u16)v[j]);
+        ss = fmaf(f, f, ss);
+      }
+    }
+    ss = wave_reduce_sum(ss);
+    inv[m] = rsqrtf(ss / (float)K + eps);
+  }
+}
+
+template <int M, bool NT>
+__global__ void __launch_bounds__(256)
+k_gemv_res(u16* __restrict__ res, const u16* __restrict__ x,
+           const u16* __restrict__ w, int N, int K) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int n0 = blockIdx.x * 8 + wid * 2;
+  if (n0 >= N) return;
+  const bool two = (n0 + 1) < N;
+  const s16x8* wrow0 = (const s16x8*)(w + (long)n0 * K);
+  const s16x8* wrow1 = (const s16x8*)(w + (long)(n0 + (two ? 1 : 0)) * K);
+  float acc0[M], acc1[M];
+#pragma unroll
+  for (int m = 0; m < M; ++m) { acc0[m] = 0.f; acc1[m] = 0.f; }
+  const int nv = K >> 3;
+  for (int i = lane; i < nv; i += 64) {
+    s16x8 wv0 = NT ? __builtin_nontemporal_load(&wrow0[i]) : wrow0[i];
+    s16x8 wv1 = NT ? __builtin_nontemporal_load(&wrow1[i]) : wrow1[i];
+#pragma unroll
+    for (int m = 0; m < M; ++m) {
+      s16x8 xv = ((const s16x8*)(x + (long)m * K))[i];
+      acc0[m] += dot8_bf16(xv, wv0);
+      acc1[m] += dot8_bf16(xv, wv1);
+    }
+  }
+#pragma unroll
+  for (int m = 0; m < M; ++m) {
+    const float v0 = wave_reduce_sum(acc0[m]);
+    const float v1 = wave_reduce_sum(acc1[m]);
+    if (lane == 0) {
+      u16* r = res + (long)m * N + n0;
+      r[0] = f2bf(bf2f(r[0]) + v0);
+      if (two) r[1] = f2bf(bf2f(r[1]) + v1);
+    }
+  }
+}
+
+template <int M, bool NT>
+__global__ void __launch_bounds__(256)
+k_gemv_norm(u16* __restrict__ out, const u16* __restrict__ res,
+            const u16* __restrict__ wn, const u16* __restrict__ w,
+            int N, int K, float eps) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int n0 = blockIdx.x * 8 + wid * 2;
+  if (n0 >= N) return;
+  float inv[M];
+  norm_factors<M>(res, K, eps, lane, inv);
+  const bool two = (n0 + 1) < N;
+  const s16x8* wrow0 = (const s16x8*)(w + (long)n0 * K);
+  const s16x8* wrow1 = (const s16x8*)(w + (long)(n0 + (two ? 1 : 0)) * K);
+  float acc0[M], acc1[M];
+#pragma unroll
+  for (int m = 0; m < M; ++m) { acc0[m] = 0.f; acc1[m] = 0.f; }
+  const int nv = K >> 3;
+  for (int i = lane; i < nv; i += 64) {
+    s16x8 wv0 = NT ? __builtin_nontemporal_load(&wrow0[i]) : wrow0[i];
+    s16x8 wv1 = NT ? __builtin_nontemporal_load(&wrow1[i]) : wrow1[i];
+    s16x8 wnv = ((const s16x8*)wn)[i];
+#pragma unroll
+    for (int m = 0; m < M; ++m) {
+      s16x8 xv = ((const s16x8*)(res + (long)m * K))[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        // match k_rmsnorm's rounding: the normed activation is quantised
+        // to bf16 before the dot (reference semantics)
+        const float xn = bf2f(f2bf(bf2f((u16)xv[j]) * inv[m] * bf2f((u16)wnv[j])));
+        acc0[m] = fmaf(xn, bf2f((u16)wv0[j]), acc0[m]);
+        acc1[m] = fmaf(xn, bf2f((u16)wv1[j]), acc1[m]);
+      }
+    }
+  }
+#pragma unroll
+  for (int m = 0; m < M; ++m) {
+    const float v0 = wave_reduce_sum(acc0[m]);
+    const float v1 = wave_reduce_sum(acc1[m]);
+    if (lane == 0) {
+      out[(long)m * N + n0] = f2bf(v0);
+      if (two) out[(long)m * N + n0 + 1] = f2bf(v1);
+    }
+  }
+}
+
+template <int M>
+__global__ void __launch_bounds__(256)
+k_gemv_swiglu_norm(u16* __restrict__ out, const u16* __restrict__ res,
+                   const u16* __restrict__ wn, const u16* __restrict__ w,
+                   int N, int K, float eps) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int n = blockIdx.x * 4 + wid;
+  if (n >= N) return;
+  float inv[M];
+  norm_factors<M>(res, K, eps, lane, inv);
+  const s16x8* grow = (const s16x8*)(w + (long)n * K);
+  const s16x8* urow = (const s16x8*)(w + (long)(n + N) * K);
+  float accg[M], accu[M];
+#pragma unroll
+  for (int m = 0; m < M; ++m) { accg[m] = 0.f; accu[m] = 0.f; }
+  const int nv = K >> 3;
+  for (int i = lane; i < nv; i += 64) {
+    s16x8 gv = grow[i];
+    s16x8 uv = urow[i];
+    s16x8 wnv = ((const s16x8*)wn)[i];
+#pragma unroll
+    for (int m = 0; m < M; ++m) {
+      s16x8 xv = ((const s16x8*)(res + (long)m * K))[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float xn = bf2f(f2bf(bf2f((u16)xv[j]) * inv[m] * bf2f((u16)wnv[j])));
+        accg[m] = fmaf(xn, bf2f((u16)gv[j]), accg[m]);
+        accu[m] = fmaf(xn, bf2f((u16)uv[j]), accu[m]);
+      }
+    }
+  }
+#pragma unroll
+  for (int m = 0; m < M; ++m) {
+    const float g = wave_reduce_sum(accg[m]);
+    const float u = wave_reduce_sum(accu[m]);
+    if (lane == 0)
+      out[(long)m * N + n] = f2bf(g / (1.f + __expf(-g)) * u);
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+#define DISPATCH_M(FN, ...) \
+  switch (M) { \
+    case 1: FN(1, __VA_ARGS__); break; \
+    case 2: FN(2, __VA_ARGS__); break; \
+    case 4: FN(4, __VA_ARGS__); break; \
+    case 8: FN(8, __VA_ARGS__); break; \
+    default: break; \
+  }
+
+void fei_gemv_res(void* res, const void* x, const void* w, int M, int N,
+                  int K, int nontemporal, hipStream_t stream) {
+  dim3 grid((N + 7) / 8);
+#define LR(MV, NTV) hipLaunchKernelGGL((k_gemv_res<MV, NTV>), grid, \
+    dim3(256), 0, stream, (u16*)res, (const u16*)x, (const u16*)w, N, K)
+#define LRD(MV, _ignored) do { if (nontemporal) LR(MV, true); else LR(MV, false); } while (0)
+  DISPATCH_M(LRD, 0)
+#undef LRD
+#undef LR
+}
+
+void fei_gemv_norm(void* out, const void* res, const void* wn, const void* w,
+                   int M, int N, int K, float eps, int nontemporal,
+                   hipStream_t stream) {
+  dim3 grid((N + 7) / 8);
+#define LN(MV, NTV) hipLaunchKernelGGL((k_gemv_norm<MV, NTV>), grid, \
+    dim3(256), 0, stream, (u16*)out, (const u16*)res, (const u16*)wn, \
+    (const u16*)w, N, K, eps)
+#define LND(MV, _ignored) do { if (nontemporal) LN(MV, true); else LN(MV, false); } while (0)
+  DISPATCH_M(LND, 0)
+#undef LND
+#undef LN
+}
+
+void fei_gemv_swiglu_norm(void* out, const void* res, const void* wn,
+                          const void* w, int M, int N, int K, float eps,
+                          hipStream_t stream) {
+  dim3 grid((N + 3) / 4);
+#define LS(MV, _ignored) hipLaunchKernelGGL(k_gemv_swiglu_norm<MV>, grid, \
+    dim3(256), 0, stream, (u16*)out, (const u16*)res, (const u16*)wn, \
+    (const u16*)w, N, K, eps)
+  DISPATCH_M(LS, 0)
+#undef LS
+}
+
+#undef DISPATCH_M
+
+}  // extern "C"

@@ -106,3 +106,28 @@ def test_assistant_local_backend_turn():
     out = a.chat("hello engine")
     assert isinstance(out, str)
     assert a.turn_metrics[0]["usage"]["output_tokens"] >= 1
+
+
+def test_fused_norm_chain_matches_unfused():
+    """FEI_FUSED_NORM chain vs the unfused decode path: same greedy logits
+    within bf16 tolerance on the tiny model."""
+    import torch
+    from fei_amd.engine.config import get_spec
+    from fei_amd.models.llama import LlamaModel
+
+    spec = get_spec("llama3-tiny")
+    dev = torch.device("cuda:0")
+    model = LlamaModel(spec, dev, torch.bfloat16, seed=9, max_seq_len=128)
+    tok = torch.tensor([7], dtype=torch.int32, device=dev)
+    pos = torch.tensor([3], dtype=torch.int32, device=dev)
+
+    kc1, vc1 = model.new_kv_cache(1, 128)
+    for c in kc1 + vc1:
+        torch.nn.init.normal_(c, std=0.5)
+    kc2 = [c.clone() for c in kc1]
+    vc2 = [c.clone() for c in vc1]
+
+    a = model.forward_decode(tok, pos.clone(), kc1, vc1, fused_norm=False)
+    b = model.forward_decode(tok, pos.clone(), kc2, vc2, fused_norm=True)
+    err = ((a.float() - b.float()).abs() / (1 + a.float().abs())).max().item()
+    assert err < 5e-2, f"max rel err {err}"

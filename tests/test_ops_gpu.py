@@ -326,3 +326,42 @@ def test_attn_decode_rope_fused_split(lib, n, splits):
     assert err < 2e-2, f"n={n} splits={splits}: max err {err}"
     assert (kc.float() - kc_ref.float()).abs().max() < 2e-2
     assert torch.equal(vc, vc_ref)
+
+
+def test_gemv_res_matches(lib):
+    import torch.nn.functional as F
+    M, N, K = 1, 4096, 4096
+    x = randbf(M, K, seed=300)
+    w = randbf(N, K, seed=301, scale=0.02)
+    res = randbf(M, N, seed=302)
+    expected = (res.float() + F.linear(x.float(), w.float())).to(torch.bfloat16)
+    lib.gemv_res(x, w, res)
+    err = ((res.float() - expected.float()).abs() /
+           (1 + expected.float().abs())).max().item()
+    assert err < 2e-2, f"max rel err {err}"
+
+
+def test_gemv_norm_matches(lib):
+    M, N, K = 1, 6144, 4096
+    res = randbf(M, K, seed=310)
+    wn = randbf(K, seed=311, scale=0.5)
+    w = randbf(N, K, seed=312, scale=0.02)
+    out = lib.gemv_norm(res, wn, w, 1e-5)
+    x = lib.rmsnorm(res, wn, 1e-5)
+    expected = lib.linear_decode(x, w)
+    err = ((out.float() - expected.float()).abs() /
+           (1 + expected.float().abs())).max().item()
+    assert err < 2e-2, f"max rel err {err}"
+
+
+def test_gemv_swiglu_norm_matches(lib):
+    M, I, K = 1, 14336, 4096
+    res = randbf(M, K, seed=320)
+    wn = randbf(K, seed=321, scale=0.5)
+    wgu = randbf(2 * I, K, seed=322, scale=0.02)
+    out = lib.gemv_swiglu_norm(res, wn, wgu, 1e-5)
+    x = lib.rmsnorm(res, wn, 1e-5)
+    expected = lib.gemv_swiglu(x, wgu)
+    err = ((out.float() - expected.float()).abs() /
+           (1 + expected.float().abs())).max().item()
+    assert err < 2e-2, f"max rel err {err}"
