@@ -39,9 +39,15 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
   const int l15 = lane & 15;
   const int lg = lane >> 4;        // 16-lane group 0..3
 
+  // K/V tiles are stored with a T2 XOR swizzle (guide Guideline 4): a
+  // row-major [32][D] bf16 tile read column-slice-wise with ds_read_b128
+  // puts each 16-lane group on ONE bank (16-way conflict). Swizzling the
+  // 16-byte slot index by (row&7) spreads the group over 8 slots. Both
+  // the staging writes and all reads apply the same XOR.
   __shared__ u16 kt[32][D];
   __shared__ u16 vt[32][D];
   __shared__ u16 p_lds[4][16][32];
+#define SWZ16(row, col8) ((col8) ^ ((row) & 7))
 
   const int p0 = pos0[b];
   const int q_hi = min(qt * 64 + 64, S);           // exclusive rel row bound
@@ -79,13 +85,14 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
         const int key = i / (D / 8);
         const int col8 = i % (D / 8);
         const int kk = t * 32 + key;
+        const int dst = key * (D / 8) + SWZ16(key, col8);
         s16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
         if (kk < kv_end) {
-          ((s16x8*)kt)[i] = *(const s16x8*)(kbase + (long)kk * D + col8 * 8);
-          ((s16x8*)vt)[i] = *(const s16x8*)(vbase + (long)kk * D + col8 * 8);
+          ((s16x8*)kt)[dst] = *(const s16x8*)(kbase + (long)kk * D + col8 * 8);
+          ((s16x8*)vt)[dst] = *(const s16x8*)(vbase + (long)kk * D + col8 * 8);
         } else {
-          ((s16x8*)kt)[i] = z;
-          ((s16x8*)vt)[i] = z;
+          ((s16x8*)kt)[dst] = z;
+          ((s16x8*)vt)[dst] = z;
         }
       }
     }
@@ -98,7 +105,9 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < KS; ++ks) {
-        const s16x8 b_k = *(const s16x8*)(&kt[c * 16 + l15][ks * 32 + 8 * lg]);
+        const int krow = c * 16 + l15;
+        const int kcol8 = SWZ16(krow, ks * 4 + lg);
+        const s16x8 b_k = *(const s16x8*)(&((s16x8*)kt)[krow * (D / 8) + kcol8]);
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[ks], b_k, acc, 0, 0, 0);
       }
       sfrag[c] = acc;
@@ -151,8 +160,12 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
     for (int dc = 0; dc < DC; ++dc) {
       s16x8 b_v;
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        b_v[j] = (short)vt[8 * lg + j][dc * 16 + l15];
+      for (int j = 0; j < 8; ++j) {
+        const int vrow = 8 * lg + j;
+        const int col = dc * 16 + l15;
+        const int vcol8 = SWZ16(vrow, col >> 3);
+        b_v[j] = (short)((u16*)vt)[vrow * D + vcol8 * 8 + (col & 7)];
+      }
       o_acc[dc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p, b_v, o_acc[dc],
                                                           0, 0, 0);
     }
