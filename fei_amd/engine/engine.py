@@ -274,3 +274,53 @@ class LocalEngine:
         self.k_caches = self.v_caches = None  # release HBM
         if self.is_gpu:
             torch.cuda.empty_cache()
+
+    # -- eager sampling variants (top-k / top-p) -----------------------------
+
+    def _decode_step_eager_sampled(self, top_k: int = 0, top_p: float = 1.0,
+                                   generator=None) -> None:
+        """Non-graph decode step with top-k / nucleus filtering (the graph
+        path covers greedy + plain Gumbel; filtered sampling runs eagerly)."""
+        from fei_amd.ops import reference as ref
+
+        logits = self.model.forward_decode(
+            self.token, self.pos, self.k_caches, self.v_caches,
+            attn_splits=self.attn_splits, workspace=self.attn_ws,
+            fused_attn=self.fused_attn, attn_out=self.attn_out,
+            fused_norm=self.fused_norm)
+        masked = logits.float()
+        if top_k:
+            masked = ref.topk_mask(masked, top_k)
+        if top_p < 1.0:
+            masked = ref.topp_mask(masked, top_p)
+        nxt = ref.gumbel_sample(masked, self.temperature, generator)
+        self.token.copy_(nxt)
+        st = int(self.step)
+        if st < self.out_tokens.shape[1]:
+            self.out_tokens[:, st] = nxt
+        self.pos += 1
+        self.step += 1
+
+    def generate_sampled(self, prompt, max_new_tokens: int = 256,
+                         temperature: float = 0.8, top_k: int = 0,
+                         top_p: float = 1.0, seed: int = 0):
+        """generate() with top-k / top-p filtering (eager decode path)."""
+        if isinstance(prompt, str):
+            prompt_ids = self.tokenizer.encode(prompt)
+        else:
+            prompt_ids = list(prompt)
+        self.temperature = temperature
+        gen = torch.Generator(device=self.device)
+        gen.manual_seed(seed or self.seed)
+        torch.manual_seed(seed or self.seed)   # prefill's sampler (CPU path)
+        self.prefill(prompt_ids)
+        n = min(max_new_tokens, self.max_seq_len - len(prompt_ids) - 1)
+        for _ in range(max(n - 1, 0)):
+            self._decode_step_eager_sampled(top_k, top_p, gen)
+        if self.is_gpu:
+            torch.cuda.synchronize(self.device)
+        rows = self.out_tokens[:, :n].tolist()
+        eos = self.tokenizer.eos_id
+        rows = [row[: row.index(eos) + 1] if eos in row else row for row in rows]
+        return {"text": self.tokenizer.decode(rows[0]), "token_ids": rows[0],
+                "new_tokens": len(rows[0])}

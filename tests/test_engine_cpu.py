@@ -156,3 +156,17 @@ def test_engine_batch_decode():
     rows = e.decode(4, stop_on_eos=False)
     assert len(rows) == 3
     assert rows[0] == rows[1] == rows[2]
+
+
+def test_generate_sampled_topk_topp():
+    e = LocalEngine.create("llama3-tiny", max_seq_len=64, seed=5)
+    out = e.generate_sampled("sampling test", max_new_tokens=6,
+                             temperature=1.0, top_k=8, seed=11)
+    assert 1 <= out["new_tokens"] <= 6
+    out2 = e.generate_sampled("sampling test", max_new_tokens=6,
+                              temperature=1.0, top_p=0.9, seed=11)
+    assert 1 <= out2["new_tokens"] <= 6
+    # deterministic under the same seed
+    out3 = e.generate_sampled("sampling test", max_new_tokens=6,
+                              temperature=1.0, top_p=0.9, seed=11)
+    assert out2["token_ids"] == out3["token_ids"]
