@@ -1,13 +1,17 @@
 """LocalEngine: the MI355X inference engine behind the Assistant.
 
-The decode step (the agent-tok/s hot path) is fully device-resident:
-embedding -> 32x(RMSNorm, QKV GEMM, fused RoPE+KV-append, split-K decode
-attention, O GEMM, RMSNorm, gate/up GEMM, SwiGLU, down GEMM) -> final norm
--> lm_head GEMV -> sampling kernel -> position advance — sampling writes the
-next input token and the out_tokens ring ON DEVICE, so the whole step is
-captured once into a hipGraph (torch.cuda.CUDAGraph) and replayed with zero
-host work per token. EOS is checked every `eos_check_every` replays (one
-device->host copy per chunk, not per token).
+The decode step (the agent-tok/s hot path) is fully device-resident. At
+tp=1 with the fused-norm chain it is 5 kernels per layer:
+  norm-prologue QKV GEMV -> split-K decode attention (in-kernel RoPE +
+  KV-append) -> residual-epilogue O GEMV -> norm-prologue gate/up GEMV
+  with fused SwiGLU -> residual-epilogue down GEMV
+then a norm-prologue lm_head GEMV, the sampling kernel (writes the next
+input token + the out_tokens ring ON DEVICE) and the position advance —
+the whole step is captured once into a hipGraph (torch.cuda.CUDAGraph)
+and replayed with zero host work per token. EOS is checked every
+`eos_check_every` replays (one device->host copy per chunk, not per
+token). Prefix caching: prefill can start from any cache position, so
+agent continuation rounds prefill only the delta.
 
 Replaces the reference's remote-API call site
 (fei/core/assistant.py:527-530) per BASELINE.json's north star.
