@@ -266,9 +266,14 @@ class MemoryChain:
             return {"accepted": False, "reason": "self-validation failed"}
         votes = 1                                    # self-vote
         total = 1 + len(peers)
+
+        def safe_vote(peer: str) -> bool:
+            try:
+                return bool(self.vote_transport(peer, proposal))
+            except Exception:                        # noqa: BLE001
+                return False                         # failure = "no" vote
         with ThreadPoolExecutor(max_workers=10) as pool:
-            results = list(pool.map(
-                lambda peer: self.vote_transport(peer, proposal), peers))
+            results = list(pool.map(safe_vote, peers))
         votes += sum(1 for r in results if r)
         accepted = votes * 100 >= MIN_QUORUM_PERCENT * total
         out: Dict[str, Any] = {"accepted": accepted, "votes": votes,
@@ -292,9 +297,13 @@ class MemoryChain:
         peers = list(self.nodes)
         if not peers:
             return 0
+        def safe_update(peer: str) -> bool:
+            try:
+                return bool(self.update_transport(peer, serialized))
+            except Exception:                        # noqa: BLE001
+                return False
         with ThreadPoolExecutor(max_workers=10) as pool:
-            results = list(pool.map(
-                lambda peer: self.update_transport(peer, serialized), peers))
+            results = list(pool.map(safe_update, peers))
         return sum(1 for r in results if r)
 
     def receive_chain_update(self, blocks_data: List[Dict[str, Any]]) -> bool:

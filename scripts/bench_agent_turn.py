@@ -22,7 +22,8 @@ def main():
     from fei_amd.tools.registry import ToolRegistry
 
     has_gpu = torch.cuda.is_available()
-    model = "llama3-8b" if has_gpu else "llama3-tiny"
+    model = os.environ.get("TURN_MODEL",
+                           "llama3-8b" if has_gpu else "llama3-tiny")
     max_new = 128 if has_gpu else 16
     engine = LocalEngine.create(model, max_seq_len=8192 if has_gpu else 512)
 
@@ -53,7 +54,7 @@ def main():
     assistant = Assistant(provider="local", tool_registry=registry,
                           backend=TurnBackend(engine, max_new))
 
-    n_turns = 20 if has_gpu else 3
+    n_turns = int(os.environ.get("TURN_N", "20" if has_gpu else "3"))
     backend = assistant.providers.get_backend()
     for fixed_budget in (False, True):
         # fixed_budget=True decodes the full token budget every round

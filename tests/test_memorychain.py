@@ -158,3 +158,29 @@ def test_search_and_stats(tmp_path):
     assert len(c.search_memories("mfma")) == 1
     s = c.stats()
     assert s["length"] == 3 and s["valid"]
+
+
+def test_federation_quorum_fails_when_peers_unreachable(tmp_path):
+    """Transport failures count as 'no' votes (reference semantics,
+    memorychain.py:988-1001): 1 of 4 votes < 51% -> rejected."""
+    h = Harness(tmp_path, 4)
+    for c in h.chains:
+        c.vote_transport = lambda peer, prop: False      # all peers down
+    out = h.chains[0].propose_memory("mX", {"Subject": "unlucky"})
+    assert not out["accepted"]
+    assert out["votes"] == 1 and out["total"] == 4
+    assert len(h.chains[0].blocks) == 1                   # nothing appended
+
+
+def test_federation_vote_transport_exception_safe(tmp_path):
+    """A raising transport must not break consensus accounting."""
+    h = Harness(tmp_path, 3)
+
+    def boom(peer, prop):
+        raise ConnectionError("down")
+
+    h.chains[0].vote_transport = boom
+    out = h.chains[0].propose_memory("mY", {"Subject": "s"})
+    assert not out["accepted"]           # raising transport = "no" votes
+    assert out["votes"] == 1
+    assert h.chains[0].validate_chain()
