@@ -20,7 +20,6 @@ were silently ignored (assistant.py:629-670); ``ask()`` handles them.
 from __future__ import annotations
 
 import time
-import uuid
 from typing import Any, Dict, List, Optional
 
 from fei_amd.core.backends import Backend, BackendResponse, create_backend

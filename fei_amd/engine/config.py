@@ -8,7 +8,7 @@ embedding encoder (fei_amd/models/bge.py).
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict
 
 

@@ -19,7 +19,6 @@ Replaces the reference's remote-API call site
 
 from __future__ import annotations
 
-import math
 import time
 from typing import Dict, List, Optional, Union
 
@@ -64,7 +63,9 @@ class LocalEngine:
         if use_hip_graph is None:
             use_hip_graph = self.is_gpu and not self.tp.is_distributed
         self.use_graph = use_hip_graph and self.is_gpu
-        self.attn_splits = attn_splits
+        import os as _os
+        env_splits = _os.environ.get("FEI_ATTN_SPLITS")
+        self.attn_splits = int(env_splits) if env_splits else attn_splits
         # Fused single-pass attention (rope+append+attn in one kernel) runs
         # grid (Hkv x B) — too few workgroups to pull KV bandwidth at B=1
         # (measured 194 vs 244 tok/s on 8B); it pays only when B*Hkv fills

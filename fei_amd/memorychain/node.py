@@ -21,7 +21,7 @@ from typing import Any, Dict, List, Optional
 import requests
 from flask import Flask, jsonify, request
 
-from fei_amd.memorychain.chain import MemoryChain, TaskState
+from fei_amd.memorychain.chain import MemoryChain
 from fei_amd.utils.logging import get_logger
 
 logger = get_logger("memorychain.node")

@@ -9,7 +9,7 @@ tool handlers directly, background assistant processing (:1002-1031), and
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional
+from typing import Optional
 
 from textual.app import App, ComposeResult
 from textual.containers import VerticalScroll
