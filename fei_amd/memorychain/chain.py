@@ -198,12 +198,16 @@ class MemoryChain:
         return self.blocks[-1]
 
     def add_memory(self, memory_id: str, memory_data: Dict[str, Any],
-                   proposer: Optional[str] = None, **task_kwargs) -> MemoryBlock:
-        """Append + mine (no consensus — use propose_memory for that)."""
+                   proposer: Optional[str] = None,
+                   timestamp: Optional[float] = None,
+                   **task_kwargs) -> MemoryBlock:
+        """Append + mine (no consensus — use propose_memory for that).
+        A fixed ``timestamp`` makes the block deterministic across ranks
+        (collective federation appends)."""
         with self._lock:
             block = MemoryBlock(
                 index=len(self.blocks),
-                timestamp=time.time(),
+                timestamp=time.time() if timestamp is None else timestamp,
                 memory_id=memory_id,
                 memory_data=memory_data,
                 proposer_node=proposer or self.node_id,

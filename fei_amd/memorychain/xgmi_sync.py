@@ -133,3 +133,46 @@ class XgmiSync:
         return broadcast_tensor(embeddings, src, tuple(int(x) for x in shape),
                                 torch.float16 if embeddings is None
                                 else embeddings.dtype, ctx=ctx)
+
+
+def propose_collective(sync: "XgmiSync", memory_id: str, memory_data: dict,
+                       src: int, timestamp: float) -> dict:
+    """Collective consensus round for co-located federation ranks: the
+    proposal is broadcast from ``src``, every rank votes locally, votes are
+    all-gathered, and on quorum EVERY rank appends the identical block
+    (fixed timestamp + deterministic mining) — no chain gossip, no
+    divergence window, unlike the reference's racy HTTP broadcast
+    (SURVEY.md §7 hard-part 5). All ranks MUST call together.
+    """
+    from fei_amd.memorychain.chain import MIN_QUORUM_PERCENT
+
+    ctx = sync.ctx
+    chain = sync.chain
+    if not ctx.is_distributed:
+        return chain.propose_memory(memory_id, memory_data)
+    payload = None
+    if ctx.rank == src:
+        payload = json.dumps({"memory_id": memory_id,
+                              "memory_data": memory_data,
+                              "proposer_node": chain.node_id,
+                              "timestamp": timestamp}).encode()
+    blob = broadcast_bytes(payload, src=src, ctx=ctx)
+    proposal = json.loads(blob.decode())
+    my_vote = 1 if chain.vote_on_proposal(proposal) else 0
+    dev = _comm_device(ctx)
+    vote = torch.tensor([my_vote], dtype=torch.int64, device=dev)
+    votes = [torch.zeros(1, dtype=torch.int64, device=dev)
+             for _ in range(ctx.world_size)]
+    dist.all_gather(votes, vote, group=ctx.group)
+    total_votes = sum(int(v) for v in votes)
+    accepted = total_votes * 100 >= MIN_QUORUM_PERCENT * ctx.world_size
+    out = {"accepted": accepted, "votes": total_votes,
+           "total": ctx.world_size}
+    if accepted:
+        block = chain.add_memory(proposal["memory_id"],
+                                 proposal["memory_data"],
+                                 proposer=proposal["proposer_node"],
+                                 timestamp=proposal["timestamp"])
+        out["block_index"] = block.index
+        out["block_hash"] = block.hash
+    return out
