@@ -62,6 +62,9 @@ def _try_load() -> None:
                                           _l, _l, _vp]
     lib.fei_sample_onepass.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _f, _u64,
                                        _i, _vp]
+    lib.fei_attn_decode_paged.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp,
+                                          _i, _i, _i, _i, _i, _i, _i, _f,
+                                          _l, _vp]
     lib.fei_gemv_swiglu.argtypes = [_vp, _vp, _vp, _i, _i, _i, _vp]
     lib.fei_gemv_res.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
     lib.fei_gemv_norm.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _i, _f, _i, _vp]
@@ -446,4 +449,48 @@ def gemv_swiglu_norm(res: torch.Tensor, wnorm: torch.Tensor,
         out = torch.empty(*res.shape[:-1], I, dtype=res.dtype, device=res.device)
     lib.fei_gemv_swiglu_norm(_ptr(out), _ptr(r2), _ptr(wnorm), _ptr(wgu),
                              M, I, K, eps, _stream())
+    return out
+
+
+def attn_decode_paged(q, k_pool, v_pool, block_table, pos, splits: int = 4,
+                      scale: Optional[float] = None,
+                      workspace: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
+                      out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Decode attention over a PAGED KV pool (engine/kv_cache.PagedKVPool):
+    pools [num_blocks, Hkv, BS, D], block_table [B, max_blocks] int32 maps
+    logical key blocks to physical pool blocks. n = pos[b]+1 keys."""
+    B, Hq, D = q.shape
+    n_blocks, Hkv, BS, _ = k_pool.shape
+    assert BS & (BS - 1) == 0, "block size must be a power of two"
+    bs_log = BS.bit_length() - 1
+    scale = scale if scale is not None else 1.0 / math.sqrt(D)
+    if not q.is_cuda:
+        # gather logical order into a contiguous cache, then reference
+        max_len = block_table.shape[1] * BS
+        kc = torch.zeros(B, Hkv, max_len, D, dtype=k_pool.dtype)
+        vc = torch.zeros_like(kc)
+        for b in range(B):
+            for j, blk in enumerate(block_table[b].tolist()):
+                if blk < 0:
+                    continue
+                kc[b, :, j * BS:(j + 1) * BS] = k_pool[blk]
+                vc[b, :, j * BS:(j + 1) * BS] = v_pool[blk]
+        return ref.attn_decode(q, kc, vc, pos + 1, scale)
+    lib = require_lib()
+    if workspace is None:
+        part_o = torch.empty(B, Hq, splits, D, dtype=torch.float32, device=q.device)
+        part_ml = torch.empty(B, Hq, splits, 2, dtype=torch.float32, device=q.device)
+    else:
+        part_o, part_ml = workspace
+    if out is None:
+        out = torch.empty_like(q)
+    assert q.stride(2) == 1 and q.stride(1) == D
+    assert splits <= 32
+    lib.fei_attn_decode_paged(_ptr(q), _ptr(k_pool), _ptr(v_pool),
+                              _ptr(block_table), _ptr(part_o), _ptr(part_ml),
+                              _ptr(pos), B, Hq, Hkv, D, bs_log,
+                              block_table.shape[1], splits, scale,
+                              q.stride(0), _stream())
+    lib.fei_attn_decode_combine(_ptr(out), _ptr(part_o), _ptr(part_ml),
+                                B, Hq, D, splits, _stream())
     return out

@@ -736,3 +736,187 @@ void fei_sample_onepass(const void* logits, int* token, int* out_tokens,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// Paged decode attention: K/V live in a block pool [num_blocks, Hkv, BS, D]
+// and a per-sequence block table maps logical key blocks to physical pool
+// blocks (engine/kv_cache.py's PagedKVPool). Same split-K online-softmax
+// structure as k_attn_decode (no in-kernel rope here: serving stacks rope
+// at the scheduler level); writes the same part_o/part_ml partials, reused
+// combine kernel. BS is a power of two (bs_log).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+template <int G>
+__global__ void __launch_bounds__(256)
+k_attn_decode_paged(const u16* __restrict__ q, const u16* __restrict__ kp,
+                    const u16* __restrict__ vp,
+                    const int* __restrict__ block_table,
+                    float* __restrict__ part_o, float* __restrict__ part_ml,
+                    const int* __restrict__ pos,
+                    int B, int Hq, int Hkv, int D, int bs_log,
+                    int max_blocks, int splits, float scale, long q_bs) {
+  const int split = blockIdx.x;
+  const int hkv = blockIdx.y;
+  const int b = blockIdx.z;
+  const int tid = threadIdx.x;
+
+  __shared__ float qs[DEC_GMAX][DEC_DMAX];
+  __shared__ float pl[DEC_GMAX][DEC_TILE];
+  __shared__ float red[DEC_GMAX][4];
+  __shared__ float osh[DEC_DMAX / 2][2];
+
+  const int n = pos[b] + 1;
+  const int chunk = (n + splits - 1) / splits;
+  const int start = split * chunk;
+  const int end = min(start + chunk, n);
+  const int BS = 1 << bs_log;
+  const int* bt = block_table + (long)b * max_blocks;
+
+  if (start >= end) {
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      const int hq = hkv * G + g;
+      float* po = part_o + (((long)b * Hq + hq) * splits + split) * D;
+      float* pml = part_ml + (((long)b * Hq + hq) * splits + split) * 2;
+      for (int d = tid; d < D; d += blockDim.x) po[d] = 0.f;
+      if (tid == 0) { pml[0] = -1.0f / 0.0f; pml[1] = 0.f; }
+    }
+    return;
+  }
+
+  for (int i = tid; i < G * D; i += blockDim.x) {
+    const int g = i / D, d = i % D;
+    qs[g][d] = bf2f(q[(long)b * q_bs + (long)(hkv * G + g) * D + d]) * scale;
+  }
+  __syncthreads();
+
+  auto row_ptr = [&](const u16* pool, int kk) -> const u16* {
+    const int blk = bt[kk >> bs_log];
+    const int within = kk & (BS - 1);
+    return pool + (((long)blk * Hkv + hkv) * BS + within) * D;
+  };
+
+  const int dpairs = D / 2;
+  const int kgroups = blockDim.x / dpairs;
+  const int keys_per_group = DEC_TILE / kgroups;
+  const int dp = tid % dpairs;
+  const int kg = tid / dpairs;
+
+  float m[DEC_GMAX], l[DEC_GMAX], sc[DEC_GMAX];
+  float o0[DEC_GMAX], o1[DEC_GMAX];
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    m[g] = -1.0f / 0.0f; l[g] = 0.f; o0[g] = 0.f; o1[g] = 0.f;
+  }
+
+  for (int tile = start; tile < end; tile += DEC_TILE) {
+    const int kk = tile + tid;
+    if (kk < end) {
+      const s16x8* krow = (const s16x8*)row_ptr(kp, kk);
+#pragma unroll
+      for (int g = 0; g < G; ++g) sc[g] = 0.f;
+#pragma unroll 4
+      for (int i = 0; i < D / 8; ++i) {
+        s16x8 kv8 = krow[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float kf = bf2f((u16)kv8[j]);
+#pragma unroll
+          for (int g = 0; g < G; ++g)
+            sc[g] = fmaf(qs[g][i * 8 + j], kf, sc[g]);
+        }
+      }
+    } else {
+#pragma unroll
+      for (int g = 0; g < G; ++g) sc[g] = -1.0f / 0.0f;
+    }
+    float tile_m[DEC_GMAX];
+#pragma unroll
+    for (int g = 0; g < G; ++g) tile_m[g] = sc[g];
+    block_reduce_vec<0, G>(tile_m, red);
+    float alpha[DEC_GMAX], pv[DEC_GMAX];
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      const float m_new = fmaxf(m[g], tile_m[g]);
+      alpha[g] = __expf(m[g] - m_new);
+      pv[g] = (kk < end) ? __expf(sc[g] - m_new) : 0.f;
+      pl[g][tid] = pv[g];
+      m[g] = m_new;
+    }
+    float tile_sum[DEC_GMAX];
+#pragma unroll
+    for (int g = 0; g < G; ++g) tile_sum[g] = pv[g];
+    block_reduce_vec<1, G>(tile_sum, red);
+#pragma unroll
+    for (int g = 0; g < G; ++g) l[g] = l[g] * alpha[g] + tile_sum[g];
+
+#pragma unroll
+    for (int g = 0; g < G; ++g) { o0[g] *= alpha[g]; o1[g] *= alpha[g]; }
+    const int kbase_local = kg * keys_per_group;
+    const int kmax = min(DEC_TILE, end - tile);
+    const int iters = min(keys_per_group, max(0, kmax - kbase_local));
+#pragma unroll 4
+    for (int j = 0; j < iters; ++j) {
+      const int kl = kbase_local + j;
+      const u16* vrow = row_ptr(vp, tile + kl) + dp * 2;
+      const float v0 = bf2f(vrow[0]);
+      const float v1 = bf2f(vrow[1]);
+#pragma unroll
+      for (int g = 0; g < G; ++g) {
+        const float p = pl[g][kl];
+        o0[g] = fmaf(p, v0, o0[g]);
+        o1[g] = fmaf(p, v1, o1[g]);
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    const int hq = hkv * G + g;
+    float* po = part_o + (((long)b * Hq + hq) * splits + split) * D;
+    float* pml = part_ml + (((long)b * Hq + hq) * splits + split) * 2;
+    if (kg == 0) { osh[dp][0] = o0[g]; osh[dp][1] = o1[g]; }
+    __syncthreads();
+    for (int gg = 1; gg < kgroups; ++gg) {
+      if (kg == gg) { osh[dp][0] += o0[g]; osh[dp][1] += o1[g]; }
+      __syncthreads();
+    }
+    if (kg == 0) {
+      po[dp * 2] = osh[dp][0];
+      po[dp * 2 + 1] = osh[dp][1];
+    }
+    if (tid == 0) { pml[0] = m[g]; pml[1] = l[g]; }
+    __syncthreads();
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+void fei_attn_decode_paged(const void* q, const void* k_pool,
+                           const void* v_pool, const int* block_table,
+                           float* part_o, float* part_ml, const int* pos,
+                           int B, int Hq, int Hkv, int D, int bs_log,
+                           int max_blocks, int splits, float scale,
+                           long q_bs, hipStream_t stream) {
+  const int G = Hq / Hkv;
+  dim3 grid(splits, Hkv, B);
+#define LP(GV) hipLaunchKernelGGL(k_attn_decode_paged<GV>, grid, dim3(256), \
+    0, stream, (const u16*)q, (const u16*)k_pool, (const u16*)v_pool, \
+    block_table, part_o, part_ml, pos, B, Hq, Hkv, D, bs_log, max_blocks, \
+    splits, scale, q_bs)
+  switch (G) {
+    case 1: LP(1); break;
+    case 2: LP(2); break;
+    case 4: LP(4); break;
+    case 8: LP(8); break;
+    default: break;
+  }
+#undef LP
+}
+
+}  // extern "C"

@@ -170,3 +170,27 @@ def test_generate_sampled_topk_topp():
     out3 = e.generate_sampled("sampling test", max_new_tokens=6,
                               temperature=1.0, top_p=0.9, seed=11)
     assert out2["token_ids"] == out3["token_ids"]
+
+
+def test_attn_decode_paged_cpu_reference():
+    """CPU path of the paged-attention op (gather + reference)."""
+    import math
+    from fei_amd import ops
+    from fei_amd.ops import reference as ref
+
+    B, Hq, Hkv, D, BS, max_blocks = 1, 4, 2, 64, 16, 4
+    torch.manual_seed(0)
+    k_pool = torch.randn(8, Hkv, BS, D)
+    v_pool = torch.randn(8, Hkv, BS, D)
+    bt = torch.tensor([[3, 1, 6, 0]], dtype=torch.int32)
+    q = torch.randn(B, Hq, D)
+    pos = torch.tensor([40], dtype=torch.int32)
+    out = ops.attn_decode_paged(q, k_pool, v_pool, bt, pos)
+    # manual gather
+    kc = torch.zeros(B, Hkv, max_blocks * BS, D)
+    vc = torch.zeros_like(kc)
+    for j, blk in enumerate(bt[0].tolist()):
+        kc[0, :, j * BS:(j + 1) * BS] = k_pool[blk]
+        vc[0, :, j * BS:(j + 1) * BS] = v_pool[blk]
+    expected = ref.attn_decode(q, kc, vc, pos + 1)
+    assert torch.allclose(out, expected, atol=1e-5)

@@ -365,3 +365,34 @@ def test_gemv_swiglu_norm_matches(lib):
     err = ((out.float() - expected.float()).abs() /
            (1 + expected.float().abs())).max().item()
     assert err < 2e-2, f"max rel err {err}"
+
+
+@pytest.mark.parametrize("n", [5, 100, 500])
+def test_attn_decode_paged(lib, n):
+    """Paged attention over a scrambled block table == contiguous reference."""
+    from fei_amd.ops import reference as ref
+    B, Hq, Hkv, D, BS = 2, 8, 2, 128, 16
+    max_blocks = 64
+    n_phys = 256
+    g = torch.Generator(device=DEV).manual_seed(400 + n)
+    k_pool = (torch.randn(n_phys, Hkv, BS, D, generator=g, device=DEV)).to(torch.bfloat16)
+    v_pool = (torch.randn(n_phys, Hkv, BS, D, generator=g, device=DEV)).to(torch.bfloat16)
+    # scrambled, non-overlapping physical blocks per sequence
+    perm = torch.randperm(n_phys, generator=g, device=DEV)[: B * max_blocks]
+    bt = perm.view(B, max_blocks).to(torch.int32)
+    q = randbf(B, Hq, D, seed=401 + n)
+    pos = torch.tensor([n - 1, max(n // 2 - 1, 0)], dtype=torch.int32, device=DEV)
+
+    out = lib.attn_decode_paged(q, k_pool, v_pool, bt, pos, splits=4)
+
+    # contiguous gather reference
+    max_len = max_blocks * BS
+    kc = torch.zeros(B, Hkv, max_len, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    for b in range(B):
+        for j, blk in enumerate(bt[b].tolist()):
+            kc[b, :, j * BS:(j + 1) * BS] = k_pool[blk]
+            vc[b, :, j * BS:(j + 1) * BS] = v_pool[blk]
+    expected = ref.attn_decode(q, kc, vc, pos + 1)
+    err = (out.float() - expected.float()).abs().max().item()
+    assert err < 2e-2, f"n={n}: max err {err}"
