@@ -35,6 +35,10 @@ def main() -> int:
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--temperature", type=float, default=0.0)
     p.add_argument("--no-graph", action="store_true")
+    p.add_argument("--tp", type=int, default=1,
+                   help="tensor-parallel degree: the W ranks run ONE agent "
+                        "sharded over RCCL/xGMI instead of W independent "
+                        "agents (requires world_size == tp)")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -52,15 +56,24 @@ def main() -> int:
     spec = get_spec(args.model)
     max_seq = min(spec.max_seq_len,
                   args.prompt_len + args.warmup + args.steps + 64)
-    engine = LocalEngine(
-        spec, device=device, batch_size=args.batch, max_seq_len=max_seq,
-        use_hip_graph=(has_gpu and not args.no_graph), seed=1234 + rank,
-    )
+    tp_mode = args.tp > 1
+    if tp_mode:
+        assert world == args.tp, "--tp requires launching exactly tp ranks"
+        engine = LocalEngine(
+            spec, device=device, batch_size=args.batch, max_seq_len=max_seq,
+            use_hip_graph=False,            # collectives not graph-captured
+            seed=1234, tp=ctx,
+        )
+    else:
+        engine = LocalEngine(
+            spec, device=device, batch_size=args.batch, max_seq_len=max_seq,
+            use_hip_graph=(has_gpu and not args.no_graph), seed=1234 + rank,
+        )
     engine.temperature = args.temperature
 
     # synthetic prompt of the configured shape (random-init weights; no
     # network for datasets/checkpoints)
-    rng = torch.Generator().manual_seed(99 + rank)
+    rng = torch.Generator().manual_seed(99 if tp_mode else 99 + rank)
     prompt_ids = torch.randint(4, 260, (args.prompt_len,), generator=rng).tolist()
 
     engine.ensure_graph()               # capture outside the prefill timing
@@ -99,13 +112,16 @@ def main() -> int:
         elapsed = elapsed_d.cpu()
     t_max = float(elapsed[0])
 
-    total_tokens = args.steps * args.batch * (world if world > 1 else 1)
+    # TP: the ranks together produce ONE agent's tokens; DP: one per rank
+    n_agents = 1 if tp_mode else (world if world > 1 else 1)
+    total_tokens = args.steps * args.batch * n_agents
     value = total_tokens / t_max
     ms_per_step = t_max / args.steps * 1000.0
 
     if rank == 0:
         result = {
-            "metric": f"agent tok/s ({args.model} local decode, 1 agent/GPU)",
+            "metric": (f"agent tok/s ({args.model} local decode, "
+                       + ("tp-sharded agent)" if tp_mode else "1 agent/GPU)")),
             "value": round(value, 2),
             "unit": "tok/s",
             "n_gpus": n_gpus,
@@ -119,9 +135,11 @@ def main() -> int:
             "data": "synthetic",
             "config": {
                 "model": args.model,
-                "global_batch": args.batch * (world if world > 1 else 1),
+                "global_batch": args.batch * n_agents,
                 "seq_len": args.prompt_len,
-                "parallelism": f"dp{n_gpus} (1 agent per GPU, weak scaling)",
+                "parallelism": (f"tp{args.tp} (one agent sharded over xGMI)"
+                                if tp_mode else
+                                f"dp{n_gpus} (1 agent per GPU, weak scaling)"),
                 "hip_graph": engine._graph is not None,
                 "prefill_tok_s": round(prefill_tok_s, 1),
             },
