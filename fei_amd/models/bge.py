@@ -1,12 +1,13 @@
-"""Embedding encoder for memdir semantic search (bge-base-class shape:
-12 layers, 768 hidden, 12 heads, D=64).
+"""Embedding encoder for memdir semantic search — bge-base architecture
+(BERT-style: 12 layers, 768 hidden, 12 heads, post-LayerNorm residual
+blocks, GELU MLP, biased linears).
 
 The compute path is the point (BASELINE.json configs[3]: "bge-base
 embeddings on MFMA"): GEMMs on hipBLASLt/MFMA, bidirectional attention on
-the HIP flash kernel (causal=0), fused RMSNorm + SwiGLU kernels. Weights
-are random-init (no network for checkpoints), so the block uses the same
-pre-norm/SwiGLU idiom as the rest of the engine rather than replicating
-BERT's LayerNorm/GELU bit-for-bit — the shapes and FLOPs match bge-base.
+the HIP flash kernel (causal=0), fused (residual-add+)LayerNorm and GELU
+HIP kernels. Weights are random-init — there is no network for
+checkpoints — but the architecture matches bge-base so real weights could
+be loaded via the same tensors.
 
 Output: mean-pooled last hidden state, L2-normalised — [B, hidden].
 """
@@ -14,7 +15,6 @@ Output: mean-pooled last hidden state, L2-normalised — [B, hidden].
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass
 from typing import List, Optional
 
 import torch
@@ -37,6 +37,7 @@ class BgeEncoder:
             dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
         self.dtype = dtype
         self.max_seq_len = min(max_seq_len, self.spec.max_seq_len)
+        self.eps = 1e-12 if device.type == "cpu" else 1e-6   # BERT uses 1e-12
         self._init_weights(seed)
 
     def _init_weights(self, seed: int) -> None:
@@ -49,53 +50,68 @@ class BgeEncoder:
             return (torch.randn(*shape, generator=g, device=self.device,
                                 dtype=torch.float32) * scale).to(self.dtype)
 
+        def zeros(*shape):
+            return torch.zeros(*shape, device=self.device, dtype=self.dtype)
+
+        def ones(*shape):
+            return torch.ones(*shape, device=self.device, dtype=self.dtype)
+
         self.tok_emb = rand(s.vocab_size, C, scale=0.02)
         self.pos_emb = rand(self.max_seq_len, C, scale=0.02)
+        self.emb_ln_w, self.emb_ln_b = ones(C), zeros(C)
         self.layers = []
         for _ in range(s.num_layers):
             self.layers.append({
-                "norm1": torch.ones(C, device=self.device, dtype=self.dtype),
                 "wqkv": rand(3 * s.num_heads * s.head_dim, C, scale=sc),
+                "bqkv": zeros(3 * s.num_heads * s.head_dim),
                 "wo": rand(C, s.num_heads * s.head_dim, scale=sc),
-                "norm2": torch.ones(C, device=self.device, dtype=self.dtype),
-                "wgu": rand(2 * I, C, scale=sc),
-                "wdown": rand(C, I, scale=1.0 / math.sqrt(I)),
+                "bo": zeros(C),
+                "ln1_w": ones(C), "ln1_b": zeros(C),
+                "wi": rand(I, C, scale=sc),
+                "bi": zeros(I),
+                "wo2": rand(C, I, scale=1.0 / math.sqrt(I)),
+                "bo2": zeros(C),
+                "ln2_w": ones(C), "ln2_b": zeros(C),
             })
-        self.norm_f = torch.ones(C, device=self.device, dtype=self.dtype)
 
     @torch.no_grad()
     def encode_ids(self, token_ids: torch.Tensor,
                    lengths: Optional[torch.Tensor] = None) -> torch.Tensor:
-        """token_ids [B, S] -> embeddings [B, C] (fp32, L2-normalised)."""
+        """token_ids [B, S] -> embeddings [B, C] (fp32, L2-normalised).
+        BERT block: h = LN(h + Attn(h)); h = LN(h + GELU(h Wi + bi) Wo2 + bo2)."""
         s = self.spec
         B, S = token_ids.shape
         H, D = s.num_heads, s.head_dim
         h = F.embedding(token_ids.long(), self.tok_emb) + \
             self.pos_emb[:S].unsqueeze(0)
-        h = h.to(self.dtype)
+        h = ops.layernorm(h.to(self.dtype), self.emb_ln_w, self.emb_ln_b,
+                          self.eps)
         scale = 1.0 / math.sqrt(D)
         kv_len = (lengths.to(torch.int32) if lengths is not None
                   else torch.full((B,), S, dtype=torch.int32, device=self.device))
         pos0 = torch.zeros(B, dtype=torch.int32, device=self.device)
         for lw in self.layers:
-            x = ops.rmsnorm(h, lw["norm1"], s.norm_eps)
-            qkv = F.linear(x, lw["wqkv"]).view(B, S, 3, H, D)
+            qkv = (F.linear(h, lw["wqkv"]) + lw["bqkv"]).view(B, S, 3, H, D)
             q = qkv[:, :, 0].contiguous()
             k = qkv[:, :, 1].transpose(1, 2).contiguous()   # [B,H,S,D] "cache"
             v = qkv[:, :, 2].transpose(1, 2).contiguous()
-            att = ops.attn_prefill(q, k, v, pos0, scale=scale, causal=False,
-                                   kv_len=kv_len)
-            h = h + F.linear(att.reshape(B, S, H * D), lw["wo"])
-            x = ops.rmsnorm(h, lw["norm2"], s.norm_eps)
-            act = ops.swiglu(F.linear(x, lw["wgu"]))
-            h = h + F.linear(act, lw["wdown"])
-        h = ops.rmsnorm(h, self.norm_f, s.norm_eps).float()
+            att = ops.attn_prefill(q.to(self.dtype), k.to(self.dtype),
+                                   v.to(self.dtype), pos0, scale=scale,
+                                   causal=False, kv_len=kv_len)
+            o = F.linear(att.reshape(B, S, H * D), lw["wo"]) + lw["bo"]
+            h = ops.layernorm(o.to(self.dtype), lw["ln1_w"], lw["ln1_b"],
+                              self.eps, residual=h)
+            inner = ops.gelu((F.linear(h, lw["wi"]) + lw["bi"]).to(self.dtype))
+            o2 = F.linear(inner, lw["wo2"]) + lw["bo2"]
+            h = ops.layernorm(o2.to(self.dtype), lw["ln2_w"], lw["ln2_b"],
+                              self.eps, residual=h)
+        hf = h.float()
         if lengths is not None:
             mask = (torch.arange(S, device=self.device).unsqueeze(0)
                     < lengths.unsqueeze(1)).float().unsqueeze(-1)
-            pooled = (h * mask).sum(1) / mask.sum(1).clamp_min(1.0)
+            pooled = (hf * mask).sum(1) / mask.sum(1).clamp_min(1.0)
         else:
-            pooled = h.mean(dim=1)
+            pooled = hf.mean(dim=1)
         return F.normalize(pooled, dim=-1)
 
     @torch.no_grad()

@@ -65,6 +65,9 @@ def _try_load() -> None:
     lib.fei_attn_decode_paged.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp,
                                           _i, _i, _i, _i, _i, _i, _i, _f,
                                           _l, _vp]
+    lib.fei_add_layernorm.argtypes = [_vp, _vp, _vp, _vp, _vp, _i, _i, _f,
+                                      _i, _vp]
+    lib.fei_gelu.argtypes = [_vp, _vp, _l, _vp]
     lib.fei_gemv_swiglu.argtypes = [_vp, _vp, _vp, _i, _i, _i, _vp]
     lib.fei_gemv_res.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
     lib.fei_gemv_norm.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _i, _f, _i, _vp]
@@ -493,4 +496,33 @@ def attn_decode_paged(q, k_pool, v_pool, block_table, pos, splits: int = 4,
                               q.stride(0), _stream())
     lib.fei_attn_decode_combine(_ptr(out), _ptr(part_o), _ptr(part_ml),
                                 B, Hq, D, splits, _stream())
+    return out
+
+
+def layernorm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
+              eps: float = 1e-5, residual: Optional[torch.Tensor] = None,
+              out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """LayerNorm of (x [+ residual]) over the last dim (encoder blocks)."""
+    if not x.is_cuda:
+        return ref.layernorm(x, weight, bias, eps, residual)
+    lib = require_lib()
+    x2 = x.contiguous()
+    rows = x2.numel() // x2.shape[-1]
+    if out is None:
+        out = torch.empty_like(x2)
+    res_ptr = _ptr(residual.contiguous()) if residual is not None else _ptr(x2)
+    lib.fei_add_layernorm(_ptr(out), _ptr(x2), res_ptr, _ptr(weight),
+                          _ptr(bias), rows, x2.shape[-1], eps,
+                          1 if residual is not None else 0, _stream())
+    return out
+
+
+def gelu(x: torch.Tensor, out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if not x.is_cuda:
+        return ref.gelu(x)
+    lib = require_lib()
+    x2 = x.contiguous()
+    if out is None:
+        out = torch.empty_like(x2)
+    lib.fei_gelu(_ptr(out), _ptr(x2), x2.numel(), _stream())
     return out

@@ -920,3 +920,102 @@ void fei_attn_decode_paged(const void* q, const void* k_pool,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// LayerNorm (+ optional fused residual add) and bias+GELU — the encoder-side
+// kernels (BERT/bge architecture: post-LN residual blocks, GELU MLP).
+// bf16 I/O, f32 statistics. One block per row.
+// ---------------------------------------------------------------------------
+extern "C" {
+
+__global__ void __launch_bounds__(256)
+k_add_layernorm(u16* __restrict__ out, const u16* __restrict__ x,
+                const u16* __restrict__ res, const u16* __restrict__ w,
+                const u16* __restrict__ bias, int rows, int cols, float eps,
+                int has_res) {
+  __shared__ float red[4];
+  const int nv = cols >> 3;
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const s16x8* xr = (const s16x8*)(x + (long)row * cols);
+    const s16x8* rr = (const s16x8*)(res + (long)row * cols);
+    float sum = 0.f;
+    for (int i = threadIdx.x; i < nv; i += blockDim.x) {
+      s16x8 xv = xr[i];
+      s16x8 rv = has_res ? rr[i] : xv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f((u16)xv[j]) + (has_res ? bf2f((u16)rv[j]) : 0.f);
+        sum += f;
+      }
+    }
+    sum = block_reduce_sum(sum, red);
+    const float mean = sum / (float)cols;
+    float var = 0.f;
+    for (int i = threadIdx.x; i < nv; i += blockDim.x) {
+      s16x8 xv = xr[i];
+      s16x8 rv = has_res ? rr[i] : xv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f((u16)xv[j]) + (has_res ? bf2f((u16)rv[j]) : 0.f) - mean;
+        var = fmaf(f, f, var);
+      }
+    }
+    var = block_reduce_sum(var, red);
+    const float inv = rsqrtf(var / (float)cols + eps);
+    s16x8* orow = (s16x8*)(out + (long)row * cols);
+    const s16x8* wv8 = (const s16x8*)w;
+    const s16x8* bv8 = (const s16x8*)bias;
+    for (int i = threadIdx.x; i < nv; i += blockDim.x) {
+      s16x8 xv = xr[i];
+      s16x8 rv = has_res ? rr[i] : xv;
+      s16x8 wv = wv8[i];
+      s16x8 bv = bv8[i];
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f((u16)xv[j]) + (has_res ? bf2f((u16)rv[j]) : 0.f);
+        o[j] = (short)f2bf((f - mean) * inv * bf2f((u16)wv[j]) +
+                           bf2f((u16)bv[j]));
+      }
+      orow[i] = o;
+    }
+    __syncthreads();
+  }
+}
+
+void fei_add_layernorm(void* out, const void* x, const void* res,
+                       const void* w, const void* bias, int rows, int cols,
+                       float eps, int has_res, hipStream_t stream) {
+  int grid = rows < 2048 ? rows : 2048;
+  hipLaunchKernelGGL(k_add_layernorm, dim3(grid), dim3(256), 0, stream,
+                     (u16*)out, (const u16*)x, (const u16*)res,
+                     (const u16*)w, (const u16*)bias, rows, cols, eps,
+                     has_res);
+}
+
+// exact GELU (erf form, matching torch's default): x * 0.5 * (1 + erf(x/sqrt(2)))
+__global__ void __launch_bounds__(256)
+k_gelu(u16* __restrict__ out, const u16* __restrict__ x, long n8) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    s16x8 v = ((const s16x8*)x)[i];
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float f = bf2f((u16)v[j]);
+      o[j] = (short)f2bf(f * 0.5f * (1.f + erff(f * 0.70710678f)));
+    }
+    ((s16x8*)out)[i] = o;
+  }
+}
+
+void fei_gelu(void* out, const void* x, long n, hipStream_t stream) {
+  long n8 = n >> 3;
+  int grid = (int)((n8 + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(k_gelu, dim3(grid), dim3(256), 0, stream,
+                     (u16*)out, (const u16*)x, n8);
+}
+
+}  // extern "C"

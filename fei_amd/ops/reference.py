@@ -162,6 +162,26 @@ def attn_prefill(
     return out
 
 
+# -- encoder ops (BERT/bge architecture) --------------------------------------
+
+def layernorm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
+              eps: float = 1e-5,
+              residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """(x [+ residual]) -> LayerNorm over the last dim; f32 stats."""
+    xf = x.float()
+    if residual is not None:
+        xf = xf + residual.float()
+    mean = xf.mean(dim=-1, keepdim=True)
+    var = (xf - mean).pow(2).mean(dim=-1, keepdim=True)
+    out = (xf - mean) * torch.rsqrt(var + eps) * weight.float() + bias.float()
+    return out.to(x.dtype)
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    """Exact (erf) GELU."""
+    return torch.nn.functional.gelu(x.float()).to(x.dtype)
+
+
 # -- MLP ---------------------------------------------------------------------
 
 def swiglu(gate_up: torch.Tensor) -> torch.Tensor:

@@ -396,3 +396,25 @@ def test_attn_decode_paged(lib, n):
     expected = ref.attn_decode(q, kc, vc, pos + 1)
     err = (out.float() - expected.float()).abs().max().item()
     assert err < 2e-2, f"n={n}: max err {err}"
+
+
+def test_layernorm_kernel(lib):
+    from fei_amd.ops import reference as ref
+    x = randbf(9, 768, seed=500)
+    res = randbf(9, 768, seed=501)
+    w = randbf(768, seed=502, scale=0.5)
+    b = randbf(768, seed=503, scale=0.1)
+    out = lib.layernorm(x, w, b, 1e-6)
+    expected = ref.layernorm(x, w, b, 1e-6)
+    assert (out.float() - expected.float()).abs().max() < 2e-2
+    out = lib.layernorm(x, w, b, 1e-6, residual=res)
+    expected = ref.layernorm(x, w, b, 1e-6, residual=res)
+    assert (out.float() - expected.float()).abs().max() < 2e-2
+
+
+def test_gelu_kernel(lib):
+    from fei_amd.ops import reference as ref
+    x = randbf(7, 3072, seed=510, scale=2.0)
+    out = lib.gelu(x)
+    expected = ref.gelu(x)
+    assert (out.float() - expected.float()).abs().max() < 1e-2
