@@ -29,6 +29,9 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
                int causal, long q_ts) {
   constexpr int DC = D / 16;       // d-chunks (8 for D=128)
   constexpr int KS = D / 32;       // K-dim steps per QK^T mfma chain
+  constexpr int KV = 64;           // keys per LDS tile (one barrier per 64)
+  constexpr int NC = KV / 16;      // 16-key score chunks per tile
+  constexpr int KA = KV / 32;      // P A-fragments per tile (K=32 each)
   const int qt = blockIdx.x;       // 64-row q tile
   const int hq = blockIdx.y;
   const int b = blockIdx.z;
@@ -44,9 +47,9 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
   // puts each 16-lane group on ONE bank (16-way conflict). Swizzling the
   // 16-byte slot index by (row&7) spreads the group over 8 slots. Both
   // the staging writes and all reads apply the same XOR.
-  __shared__ u16 kt[32][D];
-  __shared__ u16 vt[32][D];
-  __shared__ u16 p_lds[4][16][32];
+  __shared__ u16 kt[64][D];
+  __shared__ u16 vt[64][D];
+  __shared__ u16 p_lds[4][16][64];
 #define SWZ16(row, col8) ((col8) ^ ((row) & 7))
 
   const int p0 = pos0[b];
@@ -75,16 +78,16 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
   const u16* kbase = kc + ((long)b * Hkv + hkv) * max_seq * D;
   const u16* vbase = vc + ((long)b * Hkv + hkv) * max_seq * D;
 
-  const int ntiles = (kv_end + 31) / 32;
+  const int ntiles = (kv_end + KV - 1) / KV;
   for (int t = 0; t < ntiles; ++t) {
-    // ---- stage K/V tile (32 keys x D), zero-padded past kv_end ----------
+    // ---- stage K/V tile (KV keys x D), zero-padded past kv_end ----------
     __syncthreads();                                // vt/kt reuse protection
     {
-      const int nv8 = 32 * D / 8;                   // vec8 slots in a tile
+      const int nv8 = KV * D / 8;                   // vec8 slots in a tile
       for (int i = tid; i < nv8; i += 256) {
         const int key = i / (D / 8);
         const int col8 = i % (D / 8);
-        const int kk = t * 32 + key;
+        const int kk = t * KV + key;
         const int dst = key * (D / 8) + SWZ16(key, col8);
         s16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
         if (kk < kv_end) {
@@ -98,10 +101,10 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
     }
     __syncthreads();
 
-    // ---- QK^T: two 16-key chunks --------------------------------------
-    f32x4 sfrag[2];
+    // ---- QK^T: NC 16-key chunks ---------------------------------------
+    f32x4 sfrag[NC];
 #pragma unroll
-    for (int c = 0; c < 2; ++c) {
+    for (int c = 0; c < NC; ++c) {
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < KS; ++ks) {
@@ -114,28 +117,31 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
     }
 
     // ---- mask + online softmax ----------------------------------------
-    float p_val[2][4];
+    float p_val[NC][4];
     float alpha[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qrow_abs = p0 + qt * 64 + w * 16 + 4 * lg + r;
-      float s0 = sfrag[0][r] * scale;
-      float s1 = sfrag[1][r] * scale;
-      const int key0 = t * 32 + l15;
-      const int key1 = key0 + 16;
-      if (key0 >= kv_end || (causal && key0 > qrow_abs)) s0 = -1.0f / 0.0f;
-      if (key1 >= kv_end || (causal && key1 > qrow_abs)) s1 = -1.0f / 0.0f;
-      float mx = fmaxf(s0, s1);
+      float sv[NC];
+      float mx = -1.0f / 0.0f;
+#pragma unroll
+      for (int c = 0; c < NC; ++c) {
+        sv[c] = sfrag[c][r] * scale;
+        const int key = t * KV + c * 16 + l15;
+        if (key >= kv_end || (causal && key > qrow_abs)) sv[c] = -1.0f / 0.0f;
+        mx = fmaxf(mx, sv[c]);
+      }
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1) mx = fmaxf(mx, __shfl_xor(mx, off));
       const float m_new = fmaxf(m_row[r], mx);
       alpha[r] = __expf(m_row[r] - m_new);          // 0 on first tile
       m_row[r] = m_new;
-      const float p0v = (s0 == -1.0f / 0.0f) ? 0.f : __expf(s0 - m_new);
-      const float p1v = (s1 == -1.0f / 0.0f) ? 0.f : __expf(s1 - m_new);
-      p_val[0][r] = p0v;
-      p_val[1][r] = p1v;
-      float psum = p0v + p1v;
+      float psum = 0.f;
+#pragma unroll
+      for (int c = 0; c < NC; ++c) {
+        p_val[c][r] = (sv[c] == -1.0f / 0.0f) ? 0.f : __expf(sv[c] - m_new);
+        psum += p_val[c][r];
+      }
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1) psum += __shfl_xor(psum, off);
       l_row[r] = l_row[r] * alpha[r] + psum;
@@ -147,27 +153,33 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
 #pragma unroll
       for (int r = 0; r < 4; ++r) o_acc[dc][r] *= alpha[r];
 #pragma unroll
-    for (int c = 0; c < 2; ++c)
+    for (int c = 0; c < NC; ++c)
 #pragma unroll
       for (int r = 0; r < 4; ++r)
         p_lds[w][4 * lg + r][c * 16 + l15] = f2bf(p_val[c][r]);
     // same-wave LDS RAW: compiler inserts lgkmcnt waits; no barrier needed
     // (p_lds[w] is private to wave w).
 
-    // ---- PV -----------------------------------------------------------
-    const s16x8 a_p = *(const s16x8*)(&p_lds[w][l15][8 * lg]);
+    // ---- PV: KA P-fragments of K=32 each -------------------------------
+    s16x8 a_p[KA];
+#pragma unroll
+    for (int ka = 0; ka < KA; ++ka)
+      a_p[ka] = *(const s16x8*)(&p_lds[w][l15][ka * 32 + 8 * lg]);
 #pragma unroll
     for (int dc = 0; dc < DC; ++dc) {
-      s16x8 b_v;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int vrow = 8 * lg + j;
-        const int col = dc * 16 + l15;
-        const int vcol8 = SWZ16(vrow, col >> 3);
-        b_v[j] = (short)((u16*)vt)[vrow * D + vcol8 * 8 + (col & 7)];
+      for (int ka = 0; ka < KA; ++ka) {
+        s16x8 b_v;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int vrow = ka * 32 + 8 * lg + j;
+          const int col = dc * 16 + l15;
+          const int vcol8 = SWZ16(vrow, col >> 3);
+          b_v[j] = (short)((u16*)vt)[vrow * D + vcol8 * 8 + (col & 7)];
+        }
+        o_acc[dc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p[ka], b_v,
+                                                            o_acc[dc], 0, 0, 0);
       }
-      o_acc[dc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p, b_v, o_acc[dc],
-                                                          0, 0, 0);
     }
   }
 
