@@ -329,3 +329,54 @@ class LocalEngine:
         rows = [row[: row.index(eos) + 1] if eos in row else row for row in rows]
         return {"text": self.tokenizer.decode(rows[0]), "token_ids": rows[0],
                 "new_tokens": len(rows[0])}
+
+    # -- ragged multi-session batch ------------------------------------------
+
+    def prefill_ragged(self, prompts: List[List[int]]) -> None:
+        """Prefill B DIFFERENT prompts (padded to the longest; each row's
+        cache rows past its real length are overwritten by decode before
+        they are ever attended). Samples each row's first token from the
+        logits at its own last real position."""
+        assert len(prompts) == self.B, f"need exactly {self.B} prompts"
+        lens = [min(len(p), self.max_seq_len - 1) for p in prompts]
+        prompts = [p[-n:] if n < len(p) else p for p, n in zip(prompts, lens)]
+        S = max(lens)
+        self.ensure_graph()
+        tok = self.tokenizer
+        padded = torch.full((self.B, S), tok.pad_id, dtype=torch.int64,
+                            device=self.device)
+        for b, p in enumerate(prompts):
+            padded[b, :lens[b]] = torch.tensor(p, dtype=torch.int64,
+                                               device=self.device)
+        pos0 = torch.zeros(self.B, dtype=torch.int32, device=self.device)
+        all_logits = self.model.forward_prefill(
+            padded, pos0, self.k_caches, self.v_caches, all_positions=True)
+        idx = torch.tensor([n - 1 for n in lens], device=self.device)
+        logits = all_logits[torch.arange(self.B, device=self.device), idx]
+        self.step.zero_()
+        self.out_tokens.zero_()
+        ops.sample(logits.to(self.dtype).contiguous(), self.token, self.step,
+                   self.sample_ws.view(self.B, -1), out_tokens=self.out_tokens,
+                   temperature=self.temperature, seed=self.seed,
+                   nchunks=SAMPLE_CHUNKS)
+        self.pos.copy_(torch.tensor(lens, dtype=torch.int32,
+                                    device=self.device))
+        if self.is_gpu:
+            torch.cuda.synchronize(self.device)
+            self.step.fill_(1)
+        else:
+            self.step += 1
+
+    def generate_batch(self, prompts, max_new_tokens: int = 256,
+                       temperature: float = 0.0, stop_on_eos: bool = True):
+        """Independent agent sessions in one batch: B different prompts ->
+        B completions, decoded together (per-row positions/EOS)."""
+        ids = [self.tokenizer.encode(p) if isinstance(p, str) else list(p)
+               for p in prompts]
+        budget = self.max_seq_len - max(len(p) for p in ids) - 1
+        max_new_tokens = max(1, min(max_new_tokens, budget))
+        self.temperature = temperature
+        self.prefill_ragged(ids)
+        rows = self.decode(max_new_tokens, stop_on_eos=stop_on_eos)
+        return [{"text": self.tokenizer.decode(r), "token_ids": r}
+                for r in rows]

@@ -236,8 +236,11 @@ class LlamaModel:
         pos0: torch.Tensor,           # [B] int32 (first absolute position)
         k_caches: List[torch.Tensor],
         v_caches: List[torch.Tensor],
+        all_positions: bool = False,
     ) -> torch.Tensor:
-        """Prefill S tokens -> logits of the LAST position [B, vocab]."""
+        """Prefill S tokens -> logits of the LAST position [B, vocab]
+        (or of every position [B, S, vocab] when ``all_positions`` — the
+        ragged-batch path gathers each row's own last real token)."""
         s = self.spec
         B, S = tokens.shape
         h = F.embedding(tokens.long(), self.emb)          # [B,S,C] residual
@@ -260,7 +263,7 @@ class LlamaModel:
             next_norm = (self.layers[li + 1].norm_attn if li + 1 < n_layers
                          else self.norm_f)
             x, h = ops.fused_add_rmsnorm(d, h, next_norm, s.norm_eps)
-        last = x[:, -1, :]
+        last = x if all_positions else x[:, -1, :]
         logits = F.linear(last, self.lm_head)
         logits = all_gather_cat(logits, dim=-1, ctx=self.tp)
         return logits

@@ -131,3 +131,13 @@ def test_fused_norm_chain_matches_unfused():
     b = model.forward_decode(tok, pos.clone(), kc2, vc2, fused_norm=True)
     err = ((a.float() - b.float()).abs() / (1 + a.float().abs())).max().item()
     assert err < 5e-2, f"max rel err {err}"
+
+
+def test_generate_batch_gpu_smoke():
+    from fei_amd.engine.engine import LocalEngine
+    e = LocalEngine.create("llama3-tiny", max_seq_len=128, batch_size=2,
+                          seed=13)
+    outs = e.generate_batch(["short", "a somewhat longer prompt"],
+                            max_new_tokens=8, stop_on_eos=False)
+    assert len(outs) == 2
+    assert all(len(o["token_ids"]) == 8 for o in outs)
