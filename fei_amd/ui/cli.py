@@ -168,6 +168,8 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--max-iterations", type=int, default=10)
     p.add_argument("--textual", action="store_true", help="launch the TUI")
     p.add_argument("--no-memory", action="store_true")
+    p.add_argument("--stats", action="store_true",
+                   help="print per-turn metrics (LLM/tool latencies) as JSON")
     sub = p.add_subparsers(dest="cmd")
 
     a = sub.add_parser("ask", help="search-augmented one-shot question")
@@ -226,6 +228,8 @@ def main(argv: Optional[List[str]] = None) -> int:
         return 0
     if args.message:
         print(cli.single_message(args.message))
+        if args.stats:
+            print(json.dumps(cli.assistant.turn_metrics, default=str))
         return 0
     return cli.chat_loop()
 
