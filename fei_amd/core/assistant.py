@@ -233,16 +233,27 @@ class Assistant:
         """Iterating turn: keep executing tool rounds until the model
         answers without tool calls (or the round cap is hit)."""
         self.conversation.add_user_message(message)
+        t0 = time.perf_counter()
         response = self._send(system_prompt)
+        metrics: Dict[str, Any] = {"llm_s": time.perf_counter() - t0,
+                                   "usage": response.usage, "tools": []}
         rounds = 0
         while response.tool_calls and rounds < max_tool_rounds:
             self.conversation.add_assistant_message(response.content, response.tool_calls)
             results = self.process_tool_calls(response.tool_calls)
+            metrics["tools"].extend({"name": r["name"],
+                                     "latency_s": r["latency_s"]}
+                                    for r in results)
             self.conversation.add_tool_results(
                 [{"tool_use_id": r["tool_use_id"], "content": r["content"]}
                  for r in results])
+            t1 = time.perf_counter()
             response = self._send(system_prompt)
+            metrics["llm_s"] += time.perf_counter() - t1
             rounds += 1
+        metrics["rounds"] = rounds
+        metrics["total_s"] = time.perf_counter() - t0
+        self.turn_metrics.append(metrics)
         self.conversation.add_assistant_message(response.content, response.tool_calls)
         answer = response.content
         if not answer:
