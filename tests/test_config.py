@@ -65,3 +65,25 @@ def test_bad_value_falls_back(clean_env, tmp_path):
     clean_env.setenv("FEI_ENGINE_TP", "not-a-number")
     cfg = Config(ini_path=str(tmp_path / "fei.ini"), load_dotenv=False)
     assert cfg.get_typed("engine.tp") == 1
+
+
+def test_dotenv_loaded_but_env_wins(clean_env, tmp_path, monkeypatch):
+    """.env values load but never override pre-set environment variables
+    (reference contract: config.py:320-365)."""
+    monkeypatch.chdir(tmp_path)
+    (tmp_path / ".env").write_text("FEI_LLM_MODEL=model-from-dotenv\n"
+                                   "LLM_API_KEY=dotenv-key\n")
+    cfg = Config(ini_path=str(tmp_path / "fei.ini"), load_dotenv=True)
+    assert cfg.get("llm.model") == "model-from-dotenv"
+    assert cfg.get("llm.api_key") == "dotenv-key"
+    clean_env.setenv("FEI_LLM_MODEL", "model-from-env")
+    assert cfg.get("llm.model") == "model-from-env"
+
+
+def test_get_section_and_delete(clean_env, tmp_path):
+    cfg = Config(ini_path=str(tmp_path / "fei.ini"), load_dotenv=False)
+    cfg.set("engine.tp", 4)
+    section = cfg.get_section("engine")
+    assert section["tp"] == 4
+    assert cfg.delete("engine.tp") is True
+    assert cfg.get_typed("engine.tp") == 1

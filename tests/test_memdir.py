@@ -155,3 +155,29 @@ def test_create_samples(memdir_base):
     total = sum(len(mu.list_memories(f, s, base=memdir_base))
                 for f in mu.list_folders(memdir_base) for s in ("cur", "new"))
     assert total == 10
+
+
+def test_archiver_cleanup_rules_and_trash_expiry(memdir_base):
+    import os, time
+    mu.ensure_folder(".ToDoLater", memdir_base)
+    mu.ensure_folder(".Trash", memdir_base)
+    old_ts = int(time.time() - 400 * 86400)
+    for folder in (".ToDoLater", ".Trash"):
+        path = os.path.join(mu.get_memdir_base(memdir_base), folder, "cur",
+                            f"{old_ts}.{'b'*8}.host:2,")
+        with open(path, "w") as f:
+            f.write("Subject: stale\n---\n")
+    arch = MemoryArchiver(memdir_base)
+    assert arch.cleanup_memories() == 1          # ToDoLater -> Trash
+    assert arch.empty_trash() >= 1               # stale trash deleted
+
+
+def test_memory_content_without_separator():
+    headers, body = mu.parse_memory_content("just a plain body line")
+    assert headers == {}
+    assert body == "just a plain body line"
+
+
+def test_move_memory_missing_file(memdir_base):
+    mu.ensure_folder("", memdir_base)
+    assert mu.move_memory("nope", "", ".Trash", base=memdir_base) is False

@@ -61,3 +61,17 @@ def test_json_output(tmp_path):
     mapper = RepoMapper(str(_repo(tmp_path)))
     j = mapper.generate_json()
     assert "pkg/core.py" in j["files"]
+
+
+def test_excluded_dirs_skipped(tmp_path):
+    (tmp_path / "node_modules").mkdir()
+    (tmp_path / "node_modules" / "x.py").write_text("def hidden(): pass\n")
+    (tmp_path / "ok.py").write_text("def visible(): pass\n")
+    files = RepoMapper(str(tmp_path)).source_files()
+    assert len(files) == 1 and files[0].endswith("ok.py")
+
+
+def test_budget_zero_still_returns_something(tmp_path):
+    (tmp_path / "a.py").write_text("def f(): pass\n")
+    out = RepoMapper(str(tmp_path)).generate_map(token_budget=0)
+    assert isinstance(out, str)
