@@ -35,6 +35,10 @@ def main() -> int:
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--temperature", type=float, default=0.0)
     p.add_argument("--no-graph", action="store_true")
+    p.add_argument("--weights-fp8", action="store_true",
+                   help="e4m3fn weight quantization on the decode path "
+                        "(REDUCED weight precision: reported as a separate "
+                        "labeled number, never the bf16 headline)")
     p.add_argument("--tp", type=int, default=1,
                    help="tensor-parallel degree: the W ranks run ONE agent "
                         "sharded over RCCL/xGMI instead of W independent "
@@ -68,6 +72,7 @@ def main() -> int:
         engine = LocalEngine(
             spec, device=device, batch_size=args.batch, max_seq_len=max_seq,
             use_hip_graph=(has_gpu and not args.no_graph), seed=1234 + rank,
+            weight_quant="fp8" if args.weights_fp8 else None,
         )
     engine.temperature = args.temperature
 
@@ -131,7 +136,8 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if has_gpu else "fp32-cpu",
+            "dtype": (("bf16-act+fp8-weight" if args.weights_fp8 else "bf16")
+                      if has_gpu else "fp32-cpu"),
             "data": "synthetic",
             "config": {
                 "model": args.model,

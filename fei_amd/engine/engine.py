@@ -48,6 +48,7 @@ class LocalEngine:
         seed: int = 1234,
         attn_splits: int = 4,
         dtype: Optional[torch.dtype] = None,
+        weight_quant: Optional[str] = None,    # None | "fp8" (decode path)
     ):
         self.spec = spec
         if device is None:
@@ -92,6 +93,12 @@ class LocalEngine:
         self.k_caches, self.v_caches = self.model.new_kv_cache(self.B, self.max_seq_len)
         logger.info("model %s init in %.1fs (%.2f GB params)", spec.name,
                     time.perf_counter() - t0, self.model.param_bytes() / 2**30)
+        self.weight_quant = weight_quant or _os.environ.get("FEI_WEIGHT_QUANT")
+        if self.weight_quant == "fp8":
+            t0 = time.perf_counter()
+            self.model.quantize_fp8()
+            logger.info("fp8 weight quantization in %.1fs",
+                        time.perf_counter() - t0)
 
         # static decode state (device-resident; graph-stable)
         self.token = torch.zeros(self.B, dtype=torch.int32, device=device)

@@ -114,6 +114,20 @@ class LlamaModel:
         lm = self._rand(gen, s.vocab_size, C, scale=sc)
         self.lm_head = lm[r * self.vocab_l:(r + 1) * self.vocab_l].contiguous()
 
+    def quantize_fp8(self) -> None:
+        """Quantize the DECODE-path weights to e4m3fn (per-row scales).
+        bf16 weights stay for prefill; the fused-chain decode then streams
+        half the bytes per token (serving mode; bench label fp8-weight)."""
+        self.fp8 = []
+        for lw in self.layers:
+            self.fp8.append({
+                "wqkv": ops.quant_fp8(lw.wqkv),
+                "wo": ops.quant_fp8(lw.wo),
+                "wgu": ops.quant_fp8(lw.wgu),
+                "wdown": ops.quant_fp8(lw.wdown),
+            })
+        self.fp8_lm_head = ops.quant_fp8(self.lm_head)
+
     def param_bytes(self) -> int:
         total = self.emb.numel() + self.norm_f.numel() + self.lm_head.numel()
         for lw in self.layers:
@@ -211,10 +225,16 @@ class LlamaModel:
                                    attn_out):
         s = self.spec
         B = token.shape[0]
+        fp8 = getattr(self, "fp8", None)
         h = F.embedding(token.long(), self.emb).contiguous()
         scale = 1.0 / math.sqrt(self.D)
         for li, lw in enumerate(self.layers):
-            qkv = ops.gemv_norm(h, lw.norm_attn, lw.wqkv, s.norm_eps)
+            if fp8 is not None:
+                q8 = fp8[li]
+                qkv = ops.gemv_norm_fp8(h, lw.norm_attn, *q8["wqkv"],
+                                        s.norm_eps)
+            else:
+                qkv = ops.gemv_norm(h, lw.norm_attn, lw.wqkv, s.norm_eps)
             q, k, v = self._qkv_views(qkv, B)
             if fused_attn:
                 att = ops.attn_decode_fused(q, k, v, k_caches[li],
@@ -225,9 +245,18 @@ class LlamaModel:
                                       splits=attn_splits, scale=scale,
                                       workspace=workspace, out=attn_out,
                                       k=k, v=v, table=self.rope)
-            ops.gemv_res(att.reshape(B, -1), lw.wo, h)
-            act = ops.gemv_swiglu_norm(h, lw.norm_mlp, lw.wgu, s.norm_eps)
-            ops.gemv_res(act, lw.wdown, h)
+            if fp8 is not None:
+                ops.gemv_res_fp8(att.reshape(B, -1), *q8["wo"], h)
+                act = ops.gemv_swiglu_norm_fp8(h, lw.norm_mlp, *q8["wgu"],
+                                               s.norm_eps)
+                ops.gemv_res_fp8(act, *q8["wdown"], h)
+            else:
+                ops.gemv_res(att.reshape(B, -1), lw.wo, h)
+                act = ops.gemv_swiglu_norm(h, lw.norm_mlp, lw.wgu, s.norm_eps)
+                ops.gemv_res(act, lw.wdown, h)
+        if fp8 is not None:
+            return ops.gemv_norm_fp8(h, self.norm_f, *self.fp8_lm_head,
+                                     s.norm_eps)
         return ops.gemv_norm(h, self.norm_f, self.lm_head, s.norm_eps)
 
     def forward_prefill(

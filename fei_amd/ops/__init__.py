@@ -68,6 +68,12 @@ def _try_load() -> None:
     lib.fei_add_layernorm.argtypes = [_vp, _vp, _vp, _vp, _vp, _i, _i, _f,
                                       _i, _vp]
     lib.fei_gelu.argtypes = [_vp, _vp, _l, _vp]
+    lib.fei_gemv_norm_fp8.argtypes = [_vp, _vp, _vp, _vp, _vp, _i, _i, _i,
+                                      _f, _vp]
+    lib.fei_gemv_res_fp8.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _i, _vp]
+    lib.fei_gemv_swiglu_norm_fp8.argtypes = [_vp, _vp, _vp, _vp, _vp, _i, _i,
+                                             _i, _f, _vp]
+    lib.fei_quant_fp8_rows.argtypes = [_vp, _vp, _vp, _i, _i, _vp]
     lib.fei_gemv_swiglu.argtypes = [_vp, _vp, _vp, _i, _i, _i, _vp]
     lib.fei_gemv_res.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
     lib.fei_gemv_norm.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _i, _f, _i, _vp]
@@ -525,4 +531,71 @@ def gelu(x: torch.Tensor, out: Optional[torch.Tensor] = None) -> torch.Tensor:
     if out is None:
         out = torch.empty_like(x2)
     lib.fei_gelu(_ptr(out), _ptr(x2), x2.numel(), _stream())
+    return out
+
+
+# -- fp8 weight-quantized decode path (OCP e4m3fn; serving mode) -------------
+
+def quant_fp8(w: torch.Tensor):
+    """Quantize a weight matrix [N,K] to e4m3fn + per-row fp32 scales."""
+    if not w.is_cuda:
+        return ref.quant_fp8(w)
+    lib = require_lib()
+    N, K = w.shape
+    w8 = torch.empty(N, K, dtype=torch.uint8, device=w.device)
+    scales = torch.empty(N, dtype=torch.float32, device=w.device)
+    lib.fei_quant_fp8_rows(_ptr(w8), _ptr(scales), _ptr(w.contiguous()),
+                           N, K, _stream())
+    return w8, scales
+
+
+def gemv_norm_fp8(res, wnorm, w8, wscale, eps: float = 1e-5,
+                  out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    M = res.numel() // res.shape[-1]
+    K = res.shape[-1]
+    N = w8.shape[0]
+    if not res.is_cuda:
+        w = ref.dequant_fp8(w8, wscale).to(res.dtype)
+        return torch.nn.functional.linear(rmsnorm(res, wnorm, eps), w)
+    assert M * K <= 8192, "fp8 norm-GEMV stages M*K<=8192 activations in LDS"
+    lib = require_lib()
+    r2 = res.contiguous().view(M, K)
+    if out is None:
+        out = torch.empty(*res.shape[:-1], N, dtype=res.dtype, device=res.device)
+    lib.fei_gemv_norm_fp8(_ptr(out), _ptr(r2), _ptr(wnorm), _ptr(w8),
+                          _ptr(wscale), M, N, K, eps, _stream())
+    return out
+
+
+def gemv_res_fp8(x, w8, wscale, res) -> torch.Tensor:
+    M = x.numel() // x.shape[-1]
+    K = x.shape[-1]
+    N = w8.shape[0]
+    if not x.is_cuda:
+        w = ref.dequant_fp8(w8, wscale).to(x.dtype)
+        lin = torch.nn.functional.linear(x, w)
+        res.copy_((res.float() + lin.float().view_as(res)).to(res.dtype))
+        return res
+    lib = require_lib()
+    x2 = x.contiguous().view(M, K)
+    lib.fei_gemv_res_fp8(_ptr(res), _ptr(x2), _ptr(w8), _ptr(wscale),
+                         M, N, K, _stream())
+    return res
+
+
+def gemv_swiglu_norm_fp8(res, wnorm, w8, wscale, eps: float = 1e-5,
+                         out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    M = res.numel() // res.shape[-1]
+    K = res.shape[-1]
+    I = w8.shape[0] // 2
+    if not res.is_cuda:
+        w = ref.dequant_fp8(w8, wscale).to(res.dtype)
+        return swiglu(torch.nn.functional.linear(rmsnorm(res, wnorm, eps), w))
+    assert M * K <= 8192
+    lib = require_lib()
+    r2 = res.contiguous().view(M, K)
+    if out is None:
+        out = torch.empty(*res.shape[:-1], I, dtype=res.dtype, device=res.device)
+    lib.fei_gemv_swiglu_norm_fp8(_ptr(out), _ptr(r2), _ptr(wnorm), _ptr(w8),
+                                 _ptr(wscale), M, I, K, eps, _stream())
     return out

@@ -1,0 +1,321 @@
+// FP8 (OCP e4m3fn) weight-quantized decode GEMVs for gfx950.
+// Weights stored as u8 with one f32 scale per OUTPUT ROW (absmax/448);
+// activations and outputs stay bf16, dots accumulate f32. Halves the
+// per-token weight traffic (the decode roofline) at ~0.5 % relative
+// weight error. gfx950's FP8 is OCP e4m3fn (NOT MI300X fnuz) — the LDS
+// lookup table below is the e4m3fn decode.
+//
+// Only the fused-chain ops the tp=1 decode path uses get fp8 variants:
+//   k_gemv_norm_fp8        (QKV, lm_head: norm prologue)
+//   k_gemv_res_fp8         (O, down: residual epilogue)
+//   k_gemv_swiglu_norm_fp8 (gate/up + SwiGLU + norm prologue)
+// Dequant via a 256-entry LDS LUT: 1 broadcast-class ds_read per element
+// (~33 % of LDS issue at the 2x-bandwidth target) — an arithmetic decode
+// would cost ~8 VALU/element and become the bottleneck.
+//
+// Quantization itself: k_quant_fp8_rows (one block per row: absmax ->
+// scale -> round-to-nearest-even encode), done once at load time.
+#include "fei_common.h"
+
+namespace {
+
+// e4m3fn decode (OCP): e=0 subnormal m/8 * 2^-6; 0x7f/0xff = NaN.
+__device__ __forceinline__ float e4m3_decode(int b) {
+  const int s = (b >> 7) & 1;
+  const int e = (b >> 3) & 0xF;
+  const int m = b & 7;
+  float f;
+  if (e == 0) {
+    f = (float)m * 0.001953125f;            // m/8 * 2^-6
+  } else if (e == 15 && m == 7) {
+    f = __builtin_nanf("");
+  } else {
+    f = ldexpf(1.0f + (float)m * 0.125f, e - 7);
+  }
+  return s ? -f : f;
+}
+
+__device__ __forceinline__ void lut_init(float* lut) {
+  for (int i = threadIdx.x; i < 256; i += blockDim.x)
+    lut[i] = e4m3_decode(i);
+  __syncthreads();
+}
+
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
+
+// dot of 16 fp8 weights (one 16-B load) against 16 bf16 x elems
+__device__ __forceinline__ float dot16_fp8(const float* lut,
+                                           const u16* xs, long xoff,
+                                           u32x4 w16) {
+  float acc = 0.f;
+  const s16x8 xv0 = ((const s16x8*)(xs + xoff))[0];
+  const s16x8 xv1 = ((const s16x8*)(xs + xoff))[1];
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    const unsigned int w = w16[q];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float wf = lut[(w >> (8 * j)) & 0xFF];
+      const int idx = q * 4 + j;
+      const float xf = bf2f((u16)(idx < 8 ? xv0[idx] : xv1[idx - 8]));
+      acc = fmaf(xf, wf, acc);
+    }
+  }
+  return acc;
+}
+
+template <int M>
+__device__ __forceinline__ void norm_factors8(const u16* __restrict__ res,
+                                              int K, float eps, int lane,
+                                              float inv[M]) {
+#pragma unroll
+  for (int m = 0; m < M; ++m) {
+    const s16x8* row = (const s16x8*)(res + (long)m * K);
+    float ss = 0.f;
+    const int nv = K >> 3;
+    for (int i = lane; i < nv; i += 64) {
+      s16x8 v = row[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = bf2f((u16)v[j]);
+        ss = fmaf(f, f, ss);
+      }
+    }
+    ss = wave_reduce_sum(ss);
+    inv[m] = rsqrtf(ss / (float)K + eps);
+  }
+}
+
+// norm-prologue fp8 GEMV: out = (rmsnorm(res)*wn) @ W8^T * row_scale
+template <int M>
+__global__ void __launch_bounds__(256)
+k_gemv_norm_fp8(u16* __restrict__ out, const u16* __restrict__ res,
+                const u16* __restrict__ wn, const unsigned char* __restrict__ w8,
+                const float* __restrict__ wscale, int N, int K, float eps) {
+  __shared__ float lut[256];
+  __shared__ u16 xs[8192];                  // normed activation row(s), M*K
+  lut_init(lut);
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  float inv[M];
+  norm_factors8<M>(res, K, eps, lane, inv);
+  // stage the normed-x rows once per block (bf16-rounded, k_rmsnorm parity)
+  for (int i = tid; i < M * K; i += blockDim.x) {
+    const int m = i / K, kk = i % K;
+    xs[i] = f2bf(bf2f(res[(long)m * K + kk]) * inv[m] * bf2f(wn[kk]));
+  }
+  __syncthreads();
+
+  const int n0 = blockIdx.x * 8 + wid * 2;
+  if (n0 >= N) return;
+  const bool two = (n0 + 1) < N;
+  const u32x4* wrow0 = (const u32x4*)(w8 + (long)n0 * K);
+  const u32x4* wrow1 = (const u32x4*)(w8 + (long)(n0 + (two ? 1 : 0)) * K);
+  float acc0[M], acc1[M];
+#pragma unroll
+  for (int m = 0; m < M; ++m) { acc0[m] = 0.f; acc1[m] = 0.f; }
+  const int nv = K >> 4;                    // 16 elems per 16-B load
+  for (int i = lane; i < nv; i += 64) {
+    const u32x4 wv0 = wrow0[i];
+    const u32x4 wv1 = wrow1[i];
+#pragma unroll
+    for (int m = 0; m < M; ++m) {
+      acc0[m] += dot16_fp8(lut, xs, (long)m * K + i * 16, wv0);
+      acc1[m] += dot16_fp8(lut, xs, (long)m * K + i * 16, wv1);
+    }
+  }
+#pragma unroll
+  for (int m = 0; m < M; ++m) {
+    const float v0 = wave_reduce_sum(acc0[m]) * wscale[n0];
+    const float v1 = wave_reduce_sum(acc1[m]) * wscale[two ? n0 + 1 : n0];
+    if (lane == 0) {
+      out[(long)m * N + n0] = f2bf(v0);
+      if (two) out[(long)m * N + n0 + 1] = f2bf(v1);
+    }
+  }
+}
+
+// residual-epilogue fp8 GEMV: res += x @ W8^T * row_scale
+template <int M>
+__global__ void __launch_bounds__(256)
+k_gemv_res_fp8(u16* __restrict__ res, const u16* __restrict__ x,
+               const unsigned char* __restrict__ w8,
+               const float* __restrict__ wscale, int N, int K) {
+  __shared__ float lut[256];
+  lut_init(lut);
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int n0 = blockIdx.x * 8 + wid * 2;
+  if (n0 >= N) return;
+  const bool two = (n0 + 1) < N;
+  const u32x4* wrow0 = (const u32x4*)(w8 + (long)n0 * K);
+  const u32x4* wrow1 = (const u32x4*)(w8 + (long)(n0 + (two ? 1 : 0)) * K);
+  float acc0[M], acc1[M];
+#pragma unroll
+  for (int m = 0; m < M; ++m) { acc0[m] = 0.f; acc1[m] = 0.f; }
+  const int nv = K >> 4;
+  for (int i = lane; i < nv; i += 64) {
+    const u32x4 wv0 = wrow0[i];
+    const u32x4 wv1 = wrow1[i];
+#pragma unroll
+    for (int m = 0; m < M; ++m) {
+      acc0[m] += dot16_fp8(lut, x, (long)m * K + i * 16, wv0);
+      acc1[m] += dot16_fp8(lut, x, (long)m * K + i * 16, wv1);
+    }
+  }
+#pragma unroll
+  for (int m = 0; m < M; ++m) {
+    const float v0 = wave_reduce_sum(acc0[m]) * wscale[n0];
+    const float v1 = wave_reduce_sum(acc1[m]) * wscale[two ? n0 + 1 : n0];
+    if (lane == 0) {
+      u16* r = res + (long)m * N + n0;
+      r[0] = f2bf(bf2f(r[0]) + v0);
+      if (two) r[1] = f2bf(bf2f(r[1]) + v1);
+    }
+  }
+}
+
+// norm-prologue + gate/up + SwiGLU, fp8 weights
+template <int M>
+__global__ void __launch_bounds__(256)
+k_gemv_swiglu_norm_fp8(u16* __restrict__ out, const u16* __restrict__ res,
+                       const u16* __restrict__ wn,
+                       const unsigned char* __restrict__ w8,
+                       const float* __restrict__ wscale, int N, int K,
+                       float eps) {
+  __shared__ float lut[256];
+  __shared__ u16 xs[8192];
+  lut_init(lut);
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  float inv[M];
+  norm_factors8<M>(res, K, eps, lane, inv);
+  for (int i = tid; i < M * K; i += blockDim.x) {
+    const int m = i / K, kk = i % K;
+    xs[i] = f2bf(bf2f(res[(long)m * K + kk]) * inv[m] * bf2f(wn[kk]));
+  }
+  __syncthreads();
+
+  const int n = blockIdx.x * 4 + wid;
+  if (n >= N) return;
+  const u32x4* grow = (const u32x4*)(w8 + (long)n * K);
+  const u32x4* urow = (const u32x4*)(w8 + (long)(n + N) * K);
+  float accg[M], accu[M];
+#pragma unroll
+  for (int m = 0; m < M; ++m) { accg[m] = 0.f; accu[m] = 0.f; }
+  const int nv = K >> 4;
+  for (int i = lane; i < nv; i += 64) {
+    const u32x4 gv = grow[i];
+    const u32x4 uv = urow[i];
+#pragma unroll
+    for (int m = 0; m < M; ++m) {
+      accg[m] += dot16_fp8(lut, xs, (long)m * K + i * 16, gv);
+      accu[m] += dot16_fp8(lut, xs, (long)m * K + i * 16, uv);
+    }
+  }
+#pragma unroll
+  for (int m = 0; m < M; ++m) {
+    const float g = wave_reduce_sum(accg[m]) * wscale[n];
+    const float u = wave_reduce_sum(accu[m]) * wscale[n + N];
+    if (lane == 0)
+      out[(long)m * N + n] = f2bf(g / (1.f + __expf(-g)) * u);
+  }
+}
+
+// quantize bf16 rows -> e4m3fn + per-row scale (absmax/448), RNE on the
+// mantissa via float rounding: encode(w / scale).
+__device__ __forceinline__ unsigned char e4m3_encode(float f) {
+  if (f != f) return 0x7F;                  // NaN
+  const unsigned int bits = __float_as_uint(f);
+  const unsigned int s = bits >> 31;
+  float a = fabsf(f);
+  if (a > 448.f) a = 448.f;                 // saturate (no inf in e4m3fn)
+  // round to the e4m3 grid: scale into [1,2), round mantissa to 3 bits
+  if (a < 0.001953125f * 0.5f) return (unsigned char)(s << 7);  // -> 0
+  int e;
+  float mant = frexpf(a, &e);               // a = mant * 2^e, mant in [0.5,1)
+  mant *= 2.f; e -= 1;                      // mant in [1,2)
+  if (e < -6) {                             // subnormal: units of 2^-9
+    const int mi = (int)rintf(a * 512.f);   // a / 2^-9, RNE
+    if (mi > 7)                             // rounds up to the min normal
+      return (unsigned char)((s << 7) | (1 << 3));
+    return (unsigned char)((s << 7) | mi);
+  }
+  int mi = (int)rintf((mant - 1.f) * 8.f);  // 3-bit mantissa, RNE via rintf
+  if (mi == 8) { mi = 0; e += 1; }
+  if (e > 8) { e = 8; mi = 6; }             // clamp to 448 = 1.75*2^8
+  return (unsigned char)((s << 7) | ((e + 7) << 3) | mi);
+}
+
+__global__ void __launch_bounds__(256)
+k_quant_fp8_rows(unsigned char* __restrict__ w8, float* __restrict__ wscale,
+                 const u16* __restrict__ w, int N, int K) {
+  __shared__ float red[4];
+  for (int n = blockIdx.x; n < N; n += gridDim.x) {
+    const u16* row = w + (long)n * K;
+    float amax = 0.f;
+    for (int i = threadIdx.x; i < K; i += blockDim.x)
+      amax = fmaxf(amax, fabsf(bf2f(row[i])));
+    amax = block_reduce_max(amax, red);
+    const float scale = amax > 0.f ? amax / 448.f : 1.f;
+    const float inv = 1.f / scale;
+    if (threadIdx.x == 0) wscale[n] = scale;
+    for (int i = threadIdx.x; i < K; i += blockDim.x)
+      w8[(long)n * K + i] = e4m3_encode(bf2f(row[i]) * inv);
+    __syncthreads();
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+#define DISPATCH_M8(FN) \
+  switch (M) { case 1: FN(1); break; case 2: FN(2); break; \
+               case 4: FN(4); break; case 8: FN(8); break; default: break; }
+
+void fei_gemv_norm_fp8(void* out, const void* res, const void* wn,
+                       const void* w8, const float* wscale, int M, int N,
+                       int K, float eps, hipStream_t stream) {
+  dim3 grid((N + 7) / 8);
+#define LNF(MV) hipLaunchKernelGGL(k_gemv_norm_fp8<MV>, grid, dim3(256), 0, \
+    stream, (u16*)out, (const u16*)res, (const u16*)wn, \
+    (const unsigned char*)w8, wscale, N, K, eps)
+  DISPATCH_M8(LNF)
+#undef LNF
+}
+
+void fei_gemv_res_fp8(void* res, const void* x, const void* w8,
+                      const float* wscale, int M, int N, int K,
+                      hipStream_t stream) {
+  dim3 grid((N + 7) / 8);
+#define LRF(MV) hipLaunchKernelGGL(k_gemv_res_fp8<MV>, grid, dim3(256), 0, \
+    stream, (u16*)res, (const u16*)x, (const unsigned char*)w8, wscale, N, K)
+  DISPATCH_M8(LRF)
+#undef LRF
+}
+
+void fei_gemv_swiglu_norm_fp8(void* out, const void* res, const void* wn,
+                              const void* w8, const float* wscale, int M,
+                              int N, int K, float eps, hipStream_t stream) {
+  dim3 grid((N + 3) / 4);
+#define LSF(MV) hipLaunchKernelGGL(k_gemv_swiglu_norm_fp8<MV>, grid, \
+    dim3(256), 0, stream, (u16*)out, (const u16*)res, (const u16*)wn, \
+    (const unsigned char*)w8, wscale, N, K, eps)
+  DISPATCH_M8(LSF)
+#undef LSF
+}
+
+void fei_quant_fp8_rows(void* w8, float* wscale, const void* w, int N, int K,
+                        hipStream_t stream) {
+  int grid = N < 2048 ? N : 2048;
+  hipLaunchKernelGGL(k_quant_fp8_rows, dim3(grid), dim3(256), 0, stream,
+                     (unsigned char*)w8, wscale, (const u16*)w, N, K);
+}
+
+#undef DISPATCH_M8
+
+}  // extern "C"

@@ -231,3 +231,19 @@ def topp_mask(logits: torch.Tensor, p: float) -> torch.Tensor:
     out = torch.full_like(masked, float("-inf"))
     out.scatter_(-1, idx, masked)
     return out.to(logits.dtype)
+
+
+# -- fp8 weight quantization (OCP e4m3fn; decode-path serving mode) ----------
+
+def quant_fp8(w: torch.Tensor):
+    """Per-output-row absmax/448 scaling + e4m3fn encode.
+    Returns (w8 uint8 [N,K], scales fp32 [N])."""
+    wf = w.float()
+    amax = wf.abs().amax(dim=-1).clamp_min(1e-12)
+    scales = amax / 448.0
+    q = (wf / scales.unsqueeze(-1)).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8), scales
+
+
+def dequant_fp8(w8: torch.Tensor, scales: torch.Tensor) -> torch.Tensor:
+    return w8.view(torch.float8_e4m3fn).float() * scales.unsqueeze(-1).float()

@@ -208,3 +208,22 @@ def test_ragged_batch_matches_single_sessions():
         single = LocalEngine.create("llama3-tiny", max_seq_len=96, seed=21)
         ref_out = single.generate(p, max_new_tokens=6, stop_on_eos=False)
         assert outs[i]["token_ids"] == ref_out["token_ids"], f"prompt {i}"
+
+
+def test_fp8_quant_roundtrip_cpu():
+    from fei_amd.ops import reference as ref
+    w = torch.randn(16, 64) * 0.1
+    w8, sc = ref.quant_fp8(w)
+    deq = ref.dequant_fp8(w8, sc)
+    rel = ((deq - w).abs() / w.abs().clamp_min(1e-6)).median()
+    assert rel < 0.05          # e4m3 3-bit mantissa: ~3 % typical error
+
+
+def test_engine_fp8_weight_quant_cpu():
+    e = LocalEngine.create("llama3-tiny", max_seq_len=64, seed=31,
+                           weight_quant="fp8")
+    assert hasattr(e.model, "fp8")
+    # CPU decode uses the reference fallbacks with dequantized weights;
+    # fused-norm is off on CPU so this just checks quantization happened
+    out = e.generate("fp8 check", max_new_tokens=3, stop_on_eos=False)
+    assert len(out["token_ids"]) == 3

@@ -418,3 +418,63 @@ def test_gelu_kernel(lib):
     out = lib.gelu(x)
     expected = ref.gelu(x)
     assert (out.float() - expected.float()).abs().max() < 1e-2
+
+
+def test_quant_fp8_matches_torch(lib):
+    from fei_amd.ops import reference as ref
+    w = randbf(64, 512, seed=600, scale=0.05)
+    w8_gpu, sc_gpu = lib.quant_fp8(w)
+    w8_ref, sc_ref = ref.quant_fp8(w.cpu())
+    assert torch.allclose(sc_gpu.cpu(), sc_ref, rtol=1e-3)
+    # decoded values must match the torch e4m3fn cast (RNE)
+    deq_gpu = ref.dequant_fp8(w8_gpu.cpu(), sc_gpu.cpu())
+    deq_ref = ref.dequant_fp8(w8_ref, sc_ref)
+    mismatch = (deq_gpu != deq_ref).float().mean().item()
+    assert mismatch < 0.01, f"{mismatch:.4f} of encodings differ"
+
+
+def test_gemv_norm_fp8_matches_dequant(lib):
+    from fei_amd.ops import reference as ref
+    M, N, K = 1, 6144, 4096
+    res = randbf(M, K, seed=610)
+    wn = randbf(K, seed=611, scale=0.5)
+    w = randbf(N, K, seed=612, scale=0.02)
+    w8, sc = lib.quant_fp8(w)
+    out = lib.gemv_norm_fp8(res, wn, w8, sc, 1e-5)
+    wdq = ref.dequant_fp8(w8.cpu(), sc.cpu()).to(DEV).to(torch.bfloat16)
+    expected = lib.gemv_norm(res, wn, wdq, 1e-5)
+    err = ((out.float() - expected.float()).abs() /
+           (1 + expected.float().abs())).max().item()
+    assert err < 2e-2, f"max rel err {err}"
+
+
+def test_gemv_res_fp8_matches_dequant(lib):
+    from fei_amd.ops import reference as ref
+    M, N, K = 1, 4096, 4096
+    x = randbf(M, K, seed=620)
+    w = randbf(N, K, seed=621, scale=0.02)
+    res0 = randbf(M, N, seed=622)
+    w8, sc = lib.quant_fp8(w)
+    res_a = res0.clone()
+    lib.gemv_res_fp8(x, w8, sc, res_a)
+    wdq = ref.dequant_fp8(w8.cpu(), sc.cpu()).to(DEV).to(torch.bfloat16)
+    res_b = res0.clone()
+    lib.gemv_res(x, wdq, res_b)
+    err = ((res_a.float() - res_b.float()).abs() /
+           (1 + res_b.float().abs())).max().item()
+    assert err < 2e-2, f"max rel err {err}"
+
+
+def test_gemv_swiglu_norm_fp8_matches_dequant(lib):
+    from fei_amd.ops import reference as ref
+    M, I, K = 1, 14336, 4096
+    res = randbf(M, K, seed=630)
+    wn = randbf(K, seed=631, scale=0.5)
+    wgu = randbf(2 * I, K, seed=632, scale=0.02)
+    w8, sc = lib.quant_fp8(wgu)
+    out = lib.gemv_swiglu_norm_fp8(res, wn, w8, sc, 1e-5)
+    wdq = ref.dequant_fp8(w8.cpu(), sc.cpu()).to(DEV).to(torch.bfloat16)
+    expected = lib.gemv_swiglu_norm(res, wn, wdq, 1e-5)
+    err = ((out.float() - expected.float()).abs() /
+           (1 + expected.float().abs())).max().item()
+    assert err < 3e-2, f"max rel err {err}"
