@@ -31,8 +31,9 @@ HISTORY_CAP = 100
 
 
 class ChatHistory:
-    def __init__(self, path: str = HISTORY_PATH):
-        self.path = path
+    def __init__(self, path: Optional[str] = None):
+        # resolved at call time so tests (and HOME changes) take effect
+        self.path = path or HISTORY_PATH
         self.entries: List[Dict[str, Any]] = []
         self._load()
 
