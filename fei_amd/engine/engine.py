@@ -252,7 +252,14 @@ class LocalEngine:
         if not new_ids:                      # identical prompt: redo last token
             from_pos = max(0, len(prompt_ids) - 1)
             new_ids = prompt_ids[from_pos:]
-        max_new_tokens = min(max_new_tokens, self.max_seq_len - len(prompt_ids) - 1)
+        # over-long fresh prompts: keep the tail, reserving the requested
+        # decode budget (prefix-cached prompts are left alone)
+        if from_pos == 0:
+            cap = max(1, self.max_seq_len - 1 - max_new_tokens)
+            if len(new_ids) > cap:
+                new_ids = new_ids[-cap:]
+        eff_len = min(from_pos + len(new_ids), self.max_seq_len - 1)
+        max_new_tokens = min(max_new_tokens, self.max_seq_len - eff_len - 1)
         if max_new_tokens < 1:
             max_new_tokens = 1
         self.temperature = temperature
