@@ -557,7 +557,8 @@ def gemv_norm_fp8(res, wnorm, w8, wscale, eps: float = 1e-5,
     if not res.is_cuda:
         w = ref.dequant_fp8(w8, wscale).to(res.dtype)
         return torch.nn.functional.linear(rmsnorm(res, wnorm, eps), w)
-    assert M * K <= 8192, "fp8 norm-GEMV stages M*K<=8192 activations in LDS"
+    assert M * K * 2 <= 64 * 1024, \
+        "fp8 norm-GEMV stages M*K bf16 activations in LDS (<=64 KB)"
     lib = require_lib()
     r2 = res.contiguous().view(M, K)
     if out is None:
@@ -591,7 +592,7 @@ def gemv_swiglu_norm_fp8(res, wnorm, w8, wscale, eps: float = 1e-5,
     if not res.is_cuda:
         w = ref.dequant_fp8(w8, wscale).to(res.dtype)
         return swiglu(torch.nn.functional.linear(rmsnorm(res, wnorm, eps), w))
-    assert M * K <= 8192
+    assert M * K * 2 <= 64 * 1024
     lib = require_lib()
     r2 = res.contiguous().view(M, K)
     if out is None:

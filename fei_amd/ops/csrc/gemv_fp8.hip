@@ -97,7 +97,7 @@ __global__ void __launch_bounds__(256)
 k_gemv_norm_fp8(u16* __restrict__ out, const u16* __restrict__ res,
                 const u16* __restrict__ wn, const unsigned char* __restrict__ w8,
                 const float* __restrict__ wscale, int N, int K, float eps) {
-  __shared__ u16 xs[8192];                  // normed activation row(s), M*K
+  extern __shared__ __attribute__((aligned(16))) u16 xs[];   // M*K bf16
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -186,7 +186,7 @@ k_gemv_swiglu_norm_fp8(u16* __restrict__ out, const u16* __restrict__ res,
                        const unsigned char* __restrict__ w8,
                        const float* __restrict__ wscale, int N, int K,
                        float eps) {
-  __shared__ u16 xs[8192];
+  extern __shared__ __attribute__((aligned(16))) u16 xs[];   // M*K bf16
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -280,7 +280,8 @@ void fei_gemv_norm_fp8(void* out, const void* res, const void* wn,
                        const void* w8, const float* wscale, int M, int N,
                        int K, float eps, hipStream_t stream) {
   dim3 grid((N + 7) / 8);
-#define LNF(MV) hipLaunchKernelGGL(k_gemv_norm_fp8<MV>, grid, dim3(256), 0, \
+  const size_t lds = (size_t)M * K * 2;
+#define LNF(MV) hipLaunchKernelGGL(k_gemv_norm_fp8<MV>, grid, dim3(256), lds, \
     stream, (u16*)out, (const u16*)res, (const u16*)wn, \
     (const unsigned char*)w8, wscale, N, K, eps)
   DISPATCH_M8(LNF)
@@ -301,8 +302,9 @@ void fei_gemv_swiglu_norm_fp8(void* out, const void* res, const void* wn,
                               const void* w8, const float* wscale, int M,
                               int N, int K, float eps, hipStream_t stream) {
   dim3 grid((N + 3) / 4);
+  const size_t lds = (size_t)M * K * 2;
 #define LSF(MV) hipLaunchKernelGGL(k_gemv_swiglu_norm_fp8<MV>, grid, \
-    dim3(256), 0, stream, (u16*)out, (const u16*)res, (const u16*)wn, \
+    dim3(256), lds, stream, (u16*)out, (const u16*)res, (const u16*)wn, \
     (const unsigned char*)w8, wscale, N, K, eps)
   DISPATCH_M8(LSF)
 #undef LSF
