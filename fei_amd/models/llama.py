@@ -226,8 +226,10 @@ class LlamaModel:
         s = self.spec
         B = token.shape[0]
         fp8 = getattr(self, "fp8", None)
-        # the fp8 norm-GEMVs stage B*C bf16 activations in LDS (<=64 KB)
-        if fp8 is not None and B * s.hidden_size * 2 > 64 * 1024:
+        # fp8 norm-GEMVs stage B*C bf16 activations in LDS and recompute
+        # all B norms per wave — measured 3.5x SLOWER than bf16 at B=8
+        # (prologue-dominated). fp8 pays only for small-batch agent decode.
+        if fp8 is not None and B * s.hidden_size * 2 > 16 * 1024:
             fp8 = None
         h = F.embedding(token.long(), self.emb).contiguous()
         scale = 1.0 / math.sqrt(self.D)
