@@ -271,3 +271,17 @@ def test_system_prompt_reaches_backend(tmp_path):
     assert seen["system"] == "CUSTOM SYSTEM"
     a.chat("hi again")
     assert "Fei" in seen["system"]            # default prompt restored
+
+
+def test_fp8_gated_to_small_batch():
+    """fp8 fused-norm must fall back to bf16 weights at large B*C (the LDS
+    stage dominates there — profiles/r01_aux_benchmarks.md)."""
+    import fei_amd.models.llama as lm
+    spec = get_spec("llama3-tiny")           # C=256: 8*256*2 = 4 KB <= 16 KB
+    m = LlamaModel(spec, torch.device("cpu"), torch.float32, seed=3,
+                   max_seq_len=64)
+    m.quantize_fp8()
+    assert hasattr(m, "fp8") and len(m.fp8) == spec.num_layers
+    # big hidden: simulate the gate arithmetic
+    assert 8 * 4096 * 2 > 16 * 1024          # B=8 @ 8B hidden -> gated off
+    assert 1 * 4096 * 2 <= 16 * 1024         # B=1 -> fp8 on
