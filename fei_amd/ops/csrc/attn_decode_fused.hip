@@ -58,7 +58,7 @@ k_attn_decode_fused(const u16* __restrict__ q, const u16* __restrict__ kin,
   __shared__ float qs[DEC_GMAX][DEC_DMAX];
   __shared__ float pl[DEC_GMAX][DEC_TILE];
   __shared__ float red[DEC_GMAX][4];
-  __shared__ float osh[DEC_DMAX / 2][2];
+  __shared__ float osh[8][DEC_DMAX / 2][2];
 
   const int p = pos[b];
   const int n = p + 1;
@@ -173,17 +173,19 @@ k_attn_decode_fused(const u16* __restrict__ q, const u16* __restrict__ kin,
   // ---- epilogue: combine key-groups, write bf16 -------------------------
 #pragma unroll
   for (int g = 0; g < G; ++g) {
-    if (kg == 0) { osh[dp][0] = o0[g]; osh[dp][1] = o1[g]; }
+    osh[kg][dp][0] = o0[g];
+    osh[kg][dp][1] = o1[g];
     __syncthreads();
-    for (int gg = 1; gg < kgroups; ++gg) {
-      if (kg == gg) { osh[dp][0] += o0[g]; osh[dp][1] += o1[g]; }
-      __syncthreads();
-    }
     if (kg == 0) {
+      float s0 = osh[0][dp][0], s1 = osh[0][dp][1];
+      for (int gg = 1; gg < kgroups; ++gg) {
+        s0 += osh[gg][dp][0];
+        s1 += osh[gg][dp][1];
+      }
       const float inv_l = l[g] > 0.f ? 1.f / l[g] : 0.f;
       u16* orow = out + ((long)b * Hq + hkv * G + g) * D;
-      orow[dp * 2] = f2bf(osh[dp][0] * inv_l);
-      orow[dp * 2 + 1] = f2bf(osh[dp][1] * inv_l);
+      orow[dp * 2] = f2bf(s0 * inv_l);
+      orow[dp * 2 + 1] = f2bf(s1 * inv_l);
     }
     __syncthreads();
   }

@@ -278,7 +278,7 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
   __shared__ float qs[DEC_GMAX][DEC_DMAX];
   __shared__ float pl[DEC_GMAX][DEC_TILE];
   __shared__ float red[DEC_GMAX][4];
-  __shared__ float osh[DEC_DMAX / 2][2];
+  __shared__ float osh[8][DEC_DMAX / 2][2];
 
   const int n = pos[b] + 1;
   const int chunk = (n + splits - 1) / splits;
@@ -423,20 +423,24 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
   }
 
   // combine the key-groups' partial o through LDS, one head at a time
+  // epilogue: each key-group writes its own LDS slot; ONE barrier, then
+  // key-group 0 sums (a sequential per-group accumulate cost 16 barriers).
   #pragma unroll
     for (int g = 0; g < G; ++g) {
     const int hq = hkv * G + g;
     float* po = part_o + (((long)b * Hq + hq) * splits + split) * D;
     float* pml = part_ml + (((long)b * Hq + hq) * splits + split) * 2;
-    if (kg == 0) { osh[dp][0] = o0[g]; osh[dp][1] = o1[g]; }
+    osh[kg][dp][0] = o0[g];
+    osh[kg][dp][1] = o1[g];
     __syncthreads();
-    for (int gg = 1; gg < kgroups; ++gg) {
-      if (kg == gg) { osh[dp][0] += o0[g]; osh[dp][1] += o1[g]; }
-      __syncthreads();
-    }
     if (kg == 0) {
-      po[dp * 2] = osh[dp][0];
-      po[dp * 2 + 1] = osh[dp][1];
+      float s0 = osh[0][dp][0], s1 = osh[0][dp][1];
+      for (int gg = 1; gg < kgroups; ++gg) {
+        s0 += osh[gg][dp][0];
+        s1 += osh[gg][dp][1];
+      }
+      po[dp * 2] = s0;
+      po[dp * 2 + 1] = s1;
     }
     if (tid == 0) { pml[0] = m[g]; pml[1] = l[g]; }
     __syncthreads();
@@ -765,7 +769,7 @@ k_attn_decode_paged(const u16* __restrict__ q, const u16* __restrict__ kp,
   __shared__ float qs[DEC_GMAX][DEC_DMAX];
   __shared__ float pl[DEC_GMAX][DEC_TILE];
   __shared__ float red[DEC_GMAX][4];
-  __shared__ float osh[DEC_DMAX / 2][2];
+  __shared__ float osh[8][DEC_DMAX / 2][2];
 
   const int n = pos[b] + 1;
   const int chunk = (n + splits - 1) / splits;
@@ -878,15 +882,17 @@ k_attn_decode_paged(const u16* __restrict__ q, const u16* __restrict__ kp,
     const int hq = hkv * G + g;
     float* po = part_o + (((long)b * Hq + hq) * splits + split) * D;
     float* pml = part_ml + (((long)b * Hq + hq) * splits + split) * 2;
-    if (kg == 0) { osh[dp][0] = o0[g]; osh[dp][1] = o1[g]; }
+    osh[kg][dp][0] = o0[g];
+    osh[kg][dp][1] = o1[g];
     __syncthreads();
-    for (int gg = 1; gg < kgroups; ++gg) {
-      if (kg == gg) { osh[dp][0] += o0[g]; osh[dp][1] += o1[g]; }
-      __syncthreads();
-    }
     if (kg == 0) {
-      po[dp * 2] = osh[dp][0];
-      po[dp * 2 + 1] = osh[dp][1];
+      float s0 = osh[0][dp][0], s1 = osh[0][dp][1];
+      for (int gg = 1; gg < kgroups; ++gg) {
+        s0 += osh[gg][dp][0];
+        s1 += osh[gg][dp][1];
+      }
+      po[dp * 2] = s0;
+      po[dp * 2 + 1] = s1;
     }
     if (tid == 0) { pml[0] = m[g]; pml[1] = l[g]; }
     __syncthreads();
