@@ -57,6 +57,10 @@ class ProviderManager:
                                            **self._backend_kwargs)
         return self._backend
 
+    def peek_backend(self) -> Optional[Backend]:
+        """The backend if already created (no lazy construction)."""
+        return self._backend
+
     def set_backend(self, backend: Backend) -> None:
         self._backend = backend
 
@@ -265,7 +269,7 @@ class Assistant:
         self.turn_metrics.clear()
 
     def close(self) -> None:
-        backend = self.providers._backend
+        backend = self.providers.peek_backend()
         if backend is not None:
             backend.close()
 
