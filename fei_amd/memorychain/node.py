@@ -71,6 +71,12 @@ class MemorychainNode:
             node_id=node_id, path=chain_path, difficulty=difficulty,
             vote_transport=http_vote_transport,
             update_transport=http_update_transport, persist=persist)
+        if chain is not None:
+            # a caller-provided chain serves THIS node: give it the HTTP
+            # transports (callers may still override afterwards, e.g. the
+            # test-client harness or the xGMI collective path)
+            self.chain.vote_transport = http_vote_transport
+            self.chain.update_transport = http_update_transport
         self.app = self._build_app()
         self._server_thread: Optional[threading.Thread] = None
 

@@ -118,3 +118,49 @@ def test_update_status_route(node):
     st = c.get("/memorychain/node_status").get_json()
     assert st["status"] == "working" and st["load"] == 0.7
     assert st["current_task"] == "t9"
+
+
+def test_real_socket_two_node_network(tmp_path):
+    """Two nodes over REAL HTTP sockets: connect_to_network (seed join,
+    peer adoption, chain pull) then consensus through the wire."""
+    import socket
+    import threading
+    import time
+
+    def free_port():
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            return s.getsockname()[1]
+
+    ports = [free_port(), free_port()]
+    nodes = []
+    for i, port in enumerate(ports):
+        chain = MemoryChain(node_id=f"sock{i}", path=str(tmp_path / f"s{i}.json"),
+                            difficulty=1,
+                            wallet=FeiCoinWallet(path=str(tmp_path / f"sw{i}.json")))
+        node = MemorychainNode(node_id=f"sock{i}", port=port, chain=chain)
+        nodes.append(node)
+        threading.Thread(target=node.run, daemon=True).start()
+    time.sleep(0.8)
+
+    # seed (node 0) already has a block; node 1 joins and adopts it
+    nodes[0].chain.add_memory("pre-existing", {"Subject": "before join"})
+    # patch peer addresses to 127.0.0.1 (connect_to_network uses localhost)
+    assert nodes[1].connect_to_network(f"127.0.0.1:{ports[0]}")
+    assert len(nodes[1].chain.blocks) == 2
+    assert f"127.0.0.1:{ports[0]}" in nodes[1].chain.nodes
+
+    # node 0 must know node 1 back (register round-trip)
+    peer_of_0 = [p for p in nodes[0].chain.nodes if str(ports[1]) in p]
+    assert peer_of_0, nodes[0].chain.nodes
+
+    # consensus over real HTTP from node 1
+    out = nodes[1].chain.propose_memory("net-mem", {"Subject": "over sockets"})
+    assert out["accepted"], out
+    assert out["votes"] == 2
+    deadline = time.time() + 5
+    while time.time() < deadline and len(nodes[0].chain.blocks) < 3:
+        time.sleep(0.1)
+    assert len(nodes[0].chain.blocks) == 3        # update propagated
+    assert nodes[0].chain.blocks[2].memory_id == "net-mem"
+    assert nodes[0].chain.validate_chain()
