@@ -78,48 +78,28 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
   const u16* kbase = kc + ((long)b * Hkv + hkv) * max_seq * D;
   const u16* vbase = vc + ((long)b * Hkv + hkv) * max_seq * D;
 
-  // T14 async-stage split (guide Guideline 15): tile t+1's global loads are
-  // ISSUED into registers before tile t's compute (HBM latency hides under
-  // the MFMAs), and written to LDS only after the barrier that closes tile
-  // t's reads. Each thread owns NSLOT vec8 pieces of K and V.
-  constexpr int NSLOT = KV * D / 8 / 256;           // 8 for D=128
   const int ntiles = (kv_end + KV - 1) / KV;
-  s16x8 kreg[NSLOT], vreg[NSLOT];
-
-  auto stage_load = [&](int t) {
-#pragma unroll
-    for (int sslot = 0; sslot < NSLOT; ++sslot) {
-      const int i = tid + sslot * 256;
-      const int key = i / (D / 8);
-      const int col8 = i % (D / 8);
-      const int kk = t * KV + key;
-      const s16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (t < ntiles && kk < kv_end) {
-        kreg[sslot] = *(const s16x8*)(kbase + (long)kk * D + col8 * 8);
-        vreg[sslot] = *(const s16x8*)(vbase + (long)kk * D + col8 * 8);
-      } else {
-        kreg[sslot] = z;
-        vreg[sslot] = z;
+  for (int t = 0; t < ntiles; ++t) {
+    // ---- stage K/V tile (KV keys x D), zero-padded past kv_end ----------
+    __syncthreads();                                // vt/kt reuse protection
+    {
+      const int nv8 = KV * D / 8;                   // vec8 slots in a tile
+      for (int i = tid; i < nv8; i += 256) {
+        const int key = i / (D / 8);
+        const int col8 = i % (D / 8);
+        const int kk = t * KV + key;
+        const int dst = key * (D / 8) + SWZ16(key, col8);
+        s16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (kk < kv_end) {
+          ((s16x8*)kt)[dst] = *(const s16x8*)(kbase + (long)kk * D + col8 * 8);
+          ((s16x8*)vt)[dst] = *(const s16x8*)(vbase + (long)kk * D + col8 * 8);
+        } else {
+          ((s16x8*)kt)[dst] = z;
+          ((s16x8*)vt)[dst] = z;
+        }
       }
     }
-  };
-  auto stage_write = [&]() {
-#pragma unroll
-    for (int sslot = 0; sslot < NSLOT; ++sslot) {
-      const int i = tid + sslot * 256;
-      const int key = i / (D / 8);
-      const int col8 = i % (D / 8);
-      const int dst = key * (D / 8) + SWZ16(key, col8);
-      ((s16x8*)kt)[dst] = kreg[sslot];
-      ((s16x8*)vt)[dst] = vreg[sslot];
-    }
-  };
-
-  stage_load(0);
-  stage_write();
-  __syncthreads();
-  for (int t = 0; t < ntiles; ++t) {
-    stage_load(t + 1);                              // overlap with compute
+    __syncthreads();
 
     // ---- QK^T: NC 16-key chunks ---------------------------------------
     f32x4 sfrag[NC];
@@ -201,9 +181,6 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
                                                             o_acc[dc], 0, 0, 0);
       }
     }
-    __syncthreads();                                // tile t fully read
-    stage_write();                                  // tile t+1 into LDS
-    __syncthreads();                                // visible to all waves
   }
 
   // ---- epilogue: O / l, store ------------------------------------------
