@@ -42,13 +42,16 @@ class BgeEncoder:
 
     def _init_weights(self, seed: int) -> None:
         s = self.spec
-        g = torch.Generator(device=self.device).manual_seed(seed)
+        # CPU generator: identical weights on every device for the same seed
+        # (the encoder is small; cross-device reproducibility matters for
+        # the GPU-vs-CPU numerics test and shared federation indexes)
+        g = torch.Generator().manual_seed(seed)
         C, I = s.hidden_size, s.intermediate_size
         sc = 1.0 / math.sqrt(C)
 
         def rand(*shape, scale):
-            return (torch.randn(*shape, generator=g, device=self.device,
-                                dtype=torch.float32) * scale).to(self.dtype)
+            w = torch.randn(*shape, generator=g, dtype=torch.float32) * scale
+            return w.to(device=self.device, dtype=self.dtype)
 
         def zeros(*shape):
             return torch.zeros(*shape, device=self.device, dtype=self.dtype)
