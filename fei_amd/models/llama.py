@@ -227,9 +227,10 @@ class LlamaModel:
         B = token.shape[0]
         fp8 = getattr(self, "fp8", None)
         # fp8 norm-GEMVs stage B*C bf16 activations in LDS and recompute
-        # all B norms per wave — measured 3.5x SLOWER than bf16 at B=8
-        # (prologue-dominated). fp8 pays only for small-batch agent decode.
-        if fp8 is not None and B * s.hidden_size * 2 > 16 * 1024:
+        # all B norms per wave — measured slower than bf16 already at B=2
+        # (386 vs 410 tok/s) and 3.5x slower at B=8. fp8 pays for B=1
+        # agent decode only.
+        if fp8 is not None and B * s.hidden_size * 2 > 8 * 1024:
             fp8 = None
         h = F.embedding(token.long(), self.emb).contiguous()
         scale = 1.0 / math.sqrt(self.D)
