@@ -282,6 +282,6 @@ def test_fp8_gated_to_small_batch():
                    max_seq_len=64)
     m.quantize_fp8()
     assert hasattr(m, "fp8") and len(m.fp8) == spec.num_layers
-    # big hidden: simulate the gate arithmetic
-    assert 8 * 4096 * 2 > 16 * 1024          # B=8 @ 8B hidden -> gated off
-    assert 1 * 4096 * 2 <= 16 * 1024         # B=1 -> fp8 on
+    # big hidden: simulate the gate arithmetic (gate = 8 KB: batch 1 only)
+    assert 2 * 4096 * 2 > 8 * 1024           # B=2 @ 8B hidden -> gated off
+    assert 1 * 4096 * 2 <= 8 * 1024          # B=1 -> fp8 on
