@@ -56,13 +56,18 @@ def test_agent_edits_code_and_validates(tmp_path):
     reg = ToolRegistry()
     create_code_tools(reg)
 
+    # TaskExecutor drives chat(): ONE tool round per iteration, and tool
+    # calls in a continuation are ignored (reference parity) — so each
+    # iteration is {tool round, then a text continuation}.
     script = [
         {"tool_calls": [{"name": "GrepTool",
                          "input": {"pattern": "return a", "path": str(tmp_path)}}]},
+        {"content": "found the bug"},
         {"tool_calls": [{"name": "Edit",
                          "input": {"file_path": str(target),
                                    "old_string": "return a - b",
                                    "new_string": "return a + b"}}]},
+        {"content": "edited"},
         {"tool_calls": [{"name": "Edit", "input": {"file_path": str(target)}}]},
         {"content": "fixed the bug [TASK_COMPLETE]"},
     ]
