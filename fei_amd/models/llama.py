@@ -184,7 +184,11 @@ class LlamaModel:
         ``fused_norm`` (tp=1 only): the residual stream stays inside the
         GEMV kernels — norm-prologue QKV/gate-up, residual-epilogue O/down;
         5 kernels per layer instead of 9."""
-        if fused_norm and self.tp_size == 1:
+        # the fused-norm chain's per-wave norm prologue scales with B
+        # (recomputes all B rows' sumsq per wave): batch 1 only — B=8
+        # measured 572 vs 1130 tok/s on the plain path (docs/BENCHMARKS.md)
+        if fused_norm and self.tp_size == 1 and \
+                token.shape[0] * self.spec.hidden_size * 2 <= 8 * 1024:
             return self._forward_decode_fused_norm(
                 token, pos, k_caches, v_caches, attn_splits, workspace,
                 fused_attn, attn_out)
