@@ -170,11 +170,27 @@ class LocalEngine:
 
     # -- public API ----------------------------------------------------------
 
+    PREFILL_CHUNK = 2048      # bounds activation memory for long prompts
+
     def prefill(self, token_ids: List[int], from_pos: int = 0) -> None:
         """Prefill ``token_ids`` starting at cache position ``from_pos``
         (0 = fresh conversation; >0 = prefix-cache extension: the first
         ``from_pos`` tokens are already in the KV caches) and sample the
-        first new token."""
+        first new token. Long prompts are processed in PREFILL_CHUNK
+        slices (each slice extends the KV caches; only the last slice's
+        logits are sampled)."""
+        while len(token_ids) > self.PREFILL_CHUNK:
+            head, token_ids = (token_ids[:self.PREFILL_CHUNK],
+                               token_ids[self.PREFILL_CHUNK:])
+            if from_pos + len(head) >= self.max_seq_len:
+                break                      # the tail-truncation below handles it
+            tokens = torch.tensor([head] * self.B, dtype=torch.int64,
+                                  device=self.device)
+            pos0 = torch.full((self.B,), from_pos, dtype=torch.int32,
+                              device=self.device)
+            self.model.forward_prefill(tokens, pos0, self.k_caches,
+                                       self.v_caches)
+            from_pos += len(head)
         S = len(token_ids)
         if from_pos + S >= self.max_seq_len:
             # drop the oldest suffix tokens; a full-context agent should

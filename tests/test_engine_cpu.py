@@ -285,3 +285,16 @@ def test_fp8_gated_to_small_batch():
     # big hidden: simulate the gate arithmetic (gate = 8 KB: batch 1 only)
     assert 2 * 4096 * 2 > 8 * 1024           # B=2 @ 8B hidden -> gated off
     assert 1 * 4096 * 2 <= 8 * 1024          # B=1 -> fp8 on
+
+
+def test_chunked_prefill_matches_single_shot():
+    """Prompts longer than PREFILL_CHUNK are processed in slices; greedy
+    output must match the single-shot prefill."""
+    e1 = LocalEngine.create("llama3-tiny", max_seq_len=128, seed=17)
+    e1.PREFILL_CHUNK = 16                  # force chunking
+    prompt = list(range(4, 64))            # 60 tokens -> 4 chunks
+    a = e1.generate(prompt, max_new_tokens=5, stop_on_eos=False)
+
+    e2 = LocalEngine.create("llama3-tiny", max_seq_len=128, seed=17)
+    b = e2.generate(prompt, max_new_tokens=5, stop_on_eos=False)
+    assert a["token_ids"] == b["token_ids"]
