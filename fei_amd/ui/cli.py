@@ -185,7 +185,56 @@ def build_parser() -> argparse.ArgumentParser:
 
     b = sub.add_parser("serve", help="run the memdir HTTP server")
     b.add_argument("--port", type=int, default=5000)
+
+    sub.add_parser("doctor", help="environment and kernel diagnostics")
     return p
+
+
+def run_doctor() -> int:
+    """Environment diagnostics: torch/ROCm, GPU, kernel library, hipcc,
+    memdir base, config sources."""
+    import shutil
+    import subprocess
+
+    import torch
+
+    from fei_amd import __version__, ops
+    from fei_amd.memdir import utils as mu
+
+    rows = [("fei_amd", __version__),
+            ("python", sys.version.split()[0]),
+            ("torch", torch.__version__),
+            ("gpu", "yes" if torch.cuda.is_available() else "no")]
+    if torch.cuda.is_available():
+        rows.append(("device", torch.cuda.get_device_name(0)))
+        free, total = torch.cuda.mem_get_info(0)
+        rows.append(("hbm", f"{free / 2**30:.0f} / {total / 2**30:.0f} GiB free"))
+    rows.append(("kernel lib", ops._LIB_PATH if ops.kernels_available()
+                 else f"MISSING ({ops._LIB_PATH}) — run python -m fei_amd.ops.build"))
+    hipcc = shutil.which("hipcc")
+    if hipcc:
+        try:
+            ver = subprocess.run([hipcc, "--version"], capture_output=True,
+                                 text=True, timeout=20).stdout.splitlines()[0]
+        except (subprocess.SubprocessError, IndexError, OSError):
+            ver = hipcc
+        rows.append(("hipcc", ver))
+    else:
+        rows.append(("hipcc", "not found (kernel rebuilds unavailable)"))
+    rows.append(("memdir base", mu.get_memdir_base()))
+    cfg = get_config()
+    rows.append(("provider", cfg.get("llm.provider")))
+    rows.append(("model", cfg.get("llm.model")))
+    ok = True
+    for key, value in rows:
+        print(f"{key:12s} {value}")
+        if isinstance(value, str) and value.startswith("MISSING"):
+            ok = False
+    if torch.cuda.is_available() and not ops.kernels_available():
+        print("ERROR: GPU present but kernel library missing — GPU execution "
+              "will refuse to run (no eager fallback).")
+        ok = False
+    return 0 if ok else 1
 
 
 def main(argv: Optional[List[str]] = None) -> int:
@@ -214,6 +263,8 @@ def main(argv: Optional[List[str]] = None) -> int:
     if args.cmd == "serve":
         from fei_amd.memdir.run_server import main as serve_main
         return serve_main(["--port", str(args.port)])
+    if args.cmd == "doctor":
+        return run_doctor()
 
     if args.textual:
         from fei_amd.ui.tui import main as tui_main

@@ -108,3 +108,11 @@ def test_tui_suggester():
     s = MemCommandSuggester()
     assert asyncio.run(s.get_suggestion("/mem se")) == "/mem search "
     assert asyncio.run(s.get_suggestion("hello")) is None
+
+
+def test_doctor(home, capsys):
+    from fei_amd.ui.cli import main
+    rc = main(["doctor"])
+    out = capsys.readouterr().out
+    assert rc == 0
+    assert "fei_amd" in out and "torch" in out and "kernel lib" in out
