@@ -165,7 +165,17 @@ class LocalBackend(Backend):
     def __init__(self, engine=None, model: str = "llama3-8b",
                  stop_on_eos: bool = True, **engine_kwargs):
         if engine is None:
+            import torch
+
+            from fei_amd.engine.config import get_spec
             from fei_amd.engine.engine import LocalEngine
+            spec = get_spec(model)
+            if not torch.cuda.is_available() and spec.hidden_size >= 2048:
+                raise RuntimeError(
+                    f"provider 'local' with {model} needs a GPU (a CPU run "
+                    "would materialise the full model in host RAM at fp32). "
+                    "Use --provider stub for plumbing, or --model "
+                    "llama3-tiny for CPU experiments.")
             engine = LocalEngine.create(model, **engine_kwargs)
         self.engine = engine
         self.stop_on_eos = stop_on_eos

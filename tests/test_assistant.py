@@ -127,3 +127,11 @@ def test_assistant_reset(cfg):
     a.reset()
     assert a.conversation.messages == []
     assert a.turn_metrics == []
+
+
+def test_local_provider_refuses_big_model_on_cpu(cfg):
+    import torch
+    if torch.cuda.is_available():
+        pytest.skip("GPU present")
+    with pytest.raises(RuntimeError, match="needs a GPU"):
+        Assistant(config=cfg, provider="local", model="llama3-8b").chat("hi")
