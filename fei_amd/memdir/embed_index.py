@@ -126,8 +126,8 @@ class EmbeddingIndex:
     def search(self, query: str, topk: int = 10) -> List[Tuple[str, float]]:
         """Cosine top-k over the index: one GEMV (MFMA via rocBLAS on GPU)
         + torch.topk. Returns [(id, score)]."""
-        if not self._loaded and not self.load():
-            return []
+        if self.embeddings is None and not self._loaded:
+            self.load()                    # fall back to the on-disk index
         if self.embeddings is None or not len(self.ids):
             return []
         q = self.encoder.encode_texts([query])[0]          # [C], normalised

@@ -83,3 +83,36 @@ def test_memory_manager_save_conversation(memdir_base):
     results = mgr.recall("rocprofv3")
     assert len(results) == 1
     assert "use rocprofv3" in results[0]["content"]
+
+
+def test_embedding_index_persistence(memdir_base):
+    """save() -> fresh instance load() -> identical search results."""
+    from fei_amd.memdir.embed_index import EmbeddingIndex
+    tools = MemoryTools(base=memdir_base)
+    tools.create({"subject": "hip kernels", "body": "mfma and lds tiling"})
+    tools.create({"subject": "lunch plan", "body": "soup and bread"})
+
+    idx = EmbeddingIndex(base=memdir_base)
+    n = idx.build()
+    assert n == 2
+    want = idx.search("matrix core tiling", topk=2)
+    idx.save()
+
+    idx2 = EmbeddingIndex(base=memdir_base)
+    assert idx2.load()
+    assert idx2.ids == idx.ids
+    got = idx2.search("matrix core tiling", topk=2)
+    # fp16 round trip through the .npy file: same ranking, close scores
+    assert [i for i, _ in got] == [i for i, _ in want]
+    for (_, a), (_, b) in zip(got, want):
+        assert abs(a - b) < 1e-2
+
+
+def test_embedding_index_add_texts(memdir_base):
+    from fei_amd.memdir.embed_index import EmbeddingIndex
+    idx = EmbeddingIndex(base=memdir_base)
+    idx.add_texts(["alpha doc about gpus"], ["id1"])
+    idx.add_texts(["beta doc about cooking"], ["id2"])
+    assert idx.embeddings.shape[0] == 2 and idx.ids == ["id1", "id2"]
+    res = idx.search("gpus", topk=1)
+    assert res[0][0] == "id1"
