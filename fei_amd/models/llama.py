@@ -13,6 +13,7 @@ vocab with an all-gather of logits.
 from __future__ import annotations
 
 import math
+import os
 from dataclasses import dataclass
 from typing import Dict, List, Optional
 
@@ -64,6 +65,13 @@ class LlamaModel:
         self._init_weights(seed)
         self.rope = ref.rope_table(self.max_seq_len, self.D, spec.rope_theta,
                                    device=device)
+        # FEI_NORM_PRECOMP=1: precompute each norm's sumsq in the producing
+        # residual epilogue. Measured NEGATIVE on MI355X (242.5 vs 247.5
+        # tok/s, 8B decode): the norm prologue overlaps the previous
+        # kernel's wave drain and is effectively free, while the epilogue
+        # atomics + block barrier + per-step zeroing cost ~84 us/step.
+        # Kept opt-in for re-measurement on future silicon.
+        self.norm_precomp = os.environ.get("FEI_NORM_PRECOMP", "0") == "1"
 
     # -- weights -------------------------------------------------------------
 
@@ -224,6 +232,18 @@ class LlamaModel:
         logits = all_gather_cat(logits, dim=-1, ctx=self.tp)
         return logits
 
+    def _ssq_slots(self, B: int, device) -> torch.Tensor:
+        """[2L, B] f32 sum-of-squares slots for the norm-precompute chain:
+        slot 2i  = residual after layer i's O projection (feeds the MLP norm),
+        slot 2i+1 = residual after layer i's down projection (feeds layer
+        i+1's attention norm, or the final norm). Zeroed every step."""
+        buf = getattr(self, "_ssq", None)
+        if buf is None or buf.shape[1] != B or buf.device != device:
+            buf = torch.zeros(2 * len(self.layers), B, dtype=torch.float32,
+                              device=device)
+            self._ssq = buf
+        return buf
+
     def _forward_decode_fused_norm(self, token, pos, k_caches, v_caches,
                                    attn_splits, workspace, fused_attn,
                                    attn_out):
@@ -238,13 +258,24 @@ class LlamaModel:
             fp8 = None
         h = F.embedding(token.long(), self.emb).contiguous()
         scale = 1.0 / math.sqrt(self.D)
+        # norm-precompute chain (opt-in FEI_NORM_PRECOMP=1; measured loss,
+        # see __init__): each residual GEMV's epilogue accumulates the next
+        # norm's sumsq so the norm prologues collapse to one scalar read.
+        # The layer-0 attention norm has no producing epilogue (embedding)
+        # and keeps the full prologue.
+        ssq = (self._ssq_slots(B, h.device)
+               if (self.norm_precomp and fp8 is None and h.is_cuda) else None)
+        if ssq is not None:
+            ssq.zero_()
         for li, lw in enumerate(self.layers):
             if fp8 is not None:
                 q8 = fp8[li]
                 qkv = ops.gemv_norm_fp8(h, lw.norm_attn, *q8["wqkv"],
                                         s.norm_eps)
             else:
-                qkv = ops.gemv_norm(h, lw.norm_attn, lw.wqkv, s.norm_eps)
+                qkv = ops.gemv_norm(
+                    h, lw.norm_attn, lw.wqkv, s.norm_eps,
+                    ssq=ssq[2 * li - 1] if (ssq is not None and li > 0) else None)
             q, k, v = self._qkv_views(qkv, B)
             if fused_attn:
                 att = ops.attn_decode_fused(q, k, v, k_caches[li],
@@ -261,13 +292,18 @@ class LlamaModel:
                                                s.norm_eps)
                 ops.gemv_res_fp8(act, *q8["wdown"], h)
             else:
-                ops.gemv_res(att.reshape(B, -1), lw.wo, h)
-                act = ops.gemv_swiglu_norm(h, lw.norm_mlp, lw.wgu, s.norm_eps)
-                ops.gemv_res(act, lw.wdown, h)
+                ops.gemv_res(att.reshape(B, -1), lw.wo, h,
+                             ssq_out=ssq[2 * li] if ssq is not None else None)
+                act = ops.gemv_swiglu_norm(
+                    h, lw.norm_mlp, lw.wgu, s.norm_eps,
+                    ssq=ssq[2 * li] if ssq is not None else None)
+                ops.gemv_res(act, lw.wdown, h,
+                             ssq_out=ssq[2 * li + 1] if ssq is not None else None)
         if fp8 is not None:
             return ops.gemv_norm_fp8(h, self.norm_f, *self.fp8_lm_head,
                                      s.norm_eps)
-        return ops.gemv_norm(h, self.norm_f, self.lm_head, s.norm_eps)
+        return ops.gemv_norm(h, self.norm_f, self.lm_head, s.norm_eps,
+                             ssq=ssq[-1] if ssq is not None else None)
 
     def forward_prefill(
         self,

@@ -75,10 +75,11 @@ def _try_load() -> None:
                                              _i, _f, _vp]
     lib.fei_quant_fp8_rows.argtypes = [_vp, _vp, _vp, _i, _i, _vp]
     lib.fei_gemv_swiglu.argtypes = [_vp, _vp, _vp, _i, _i, _i, _vp]
-    lib.fei_gemv_res.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
-    lib.fei_gemv_norm.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _i, _f, _i, _vp]
+    lib.fei_gemv_res.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp, _vp]
+    lib.fei_gemv_norm.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _i, _f, _i, _vp,
+                                  _vp]
     lib.fei_gemv_swiglu_norm.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _i, _f,
-                                         _vp]
+                                         _vp, _vp]
     _LIB = lib
 
 
@@ -408,8 +409,15 @@ def gemv_swiglu(x: torch.Tensor, wgu: torch.Tensor,
     return out
 
 
-def gemv_res(x: torch.Tensor, w: torch.Tensor, res: torch.Tensor) -> torch.Tensor:
-    """res += x @ w^T (epilogue residual add; in place on res [M,N])."""
+def gemv_res(x: torch.Tensor, w: torch.Tensor, res: torch.Tensor,
+             ssq_out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """res += x @ w^T (epilogue residual add; in place on res [M,N]).
+
+    ``ssq_out`` [M] f32, pre-zeroed: the kernel also accumulates the
+    sum-of-squares of the updated residual rows so a following
+    gemv_norm/gemv_swiglu_norm can skip its sumsq prologue pass (pass the
+    same tensor as their ``ssq``). Ignored on the CPU fallback — consumers
+    recompute exactly there."""
     M = x.numel() // x.shape[-1]
     K = x.shape[-1]
     N = w.shape[0]
@@ -420,14 +428,20 @@ def gemv_res(x: torch.Tensor, w: torch.Tensor, res: torch.Tensor) -> torch.Tenso
     lib = require_lib()
     x2 = x.contiguous().view(M, K)
     nt = 1 if (N * K * 2 >= _GEMV_NT_MIN_BYTES) else 0
-    lib.fei_gemv_res(_ptr(res), _ptr(x2), _ptr(w), M, N, K, nt, _stream())
+    lib.fei_gemv_res(_ptr(res), _ptr(x2), _ptr(w), M, N, K, nt,
+                     _ptr(ssq_out) if ssq_out is not None else None,
+                     _stream())
     return res
 
 
 def gemv_norm(res: torch.Tensor, wnorm: torch.Tensor, w: torch.Tensor,
               eps: float = 1e-5,
-              out: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """out = rmsnorm(res)*wnorm @ w^T (norm-prologue GEMV)."""
+              out: Optional[torch.Tensor] = None,
+              ssq: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """out = rmsnorm(res)*wnorm @ w^T (norm-prologue GEMV).
+
+    ``ssq`` [M] f32: precomputed sum-of-squares of res rows (from a prior
+    gemv_res ``ssq_out``); skips the prologue pass over res."""
     M = res.numel() // res.shape[-1]
     K = res.shape[-1]
     N = w.shape[0]
@@ -439,14 +453,17 @@ def gemv_norm(res: torch.Tensor, wnorm: torch.Tensor, w: torch.Tensor,
         out = torch.empty(*res.shape[:-1], N, dtype=res.dtype, device=res.device)
     nt = 1 if (N * K * 2 >= _GEMV_NT_MIN_BYTES) else 0
     lib.fei_gemv_norm(_ptr(out), _ptr(r2), _ptr(wnorm), _ptr(w), M, N, K,
-                      eps, nt, _stream())
+                      eps, nt, _ptr(ssq) if ssq is not None else None,
+                      _stream())
     return out
 
 
 def gemv_swiglu_norm(res: torch.Tensor, wnorm: torch.Tensor,
                      wgu: torch.Tensor, eps: float = 1e-5,
-                     out: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """out = silu(g)*u where [g;u] = rmsnorm(res)*wnorm @ wgu^T."""
+                     out: Optional[torch.Tensor] = None,
+                     ssq: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """out = silu(g)*u where [g;u] = rmsnorm(res)*wnorm @ wgu^T.
+    ``ssq``: see gemv_norm."""
     M = res.numel() // res.shape[-1]
     K = res.shape[-1]
     I = wgu.shape[0] // 2
@@ -457,7 +474,9 @@ def gemv_swiglu_norm(res: torch.Tensor, wnorm: torch.Tensor,
     if out is None:
         out = torch.empty(*res.shape[:-1], I, dtype=res.dtype, device=res.device)
     lib.fei_gemv_swiglu_norm(_ptr(out), _ptr(r2), _ptr(wnorm), _ptr(wgu),
-                             M, I, K, eps, _stream())
+                             M, I, K, eps,
+                             _ptr(ssq) if ssq is not None else None,
+                             _stream())
     return out
 
 

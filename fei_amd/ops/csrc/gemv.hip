@@ -180,40 +180,71 @@ __device__ __forceinline__ void norm_factors(const u16* __restrict__ res,
   }
 }
 
+// ssq (nullable): per-row sum-of-squares of the UPDATED residual,
+// accumulated across workgroups (LDS reduce + one atomicAdd per WG) so the
+// NEXT norm-prologue GEMV can skip its sumsq pass over res (the prologue
+// stalls every wave on an L2-latency-bound read of the whole row before any
+// W bytes stream — measured ~7-13 us per fused GEMV, profiles/r01).
+// Caller zeroes ssq before the step; kernels on one stream order the
+// producer before the consumer. Squares are taken of the bf16-ROUNDED
+// stored value so the result matches what a separate k_rmsnorm would see.
 template <int M, bool NT>
 __global__ void __launch_bounds__(256)
 k_gemv_res(u16* __restrict__ res, const u16* __restrict__ x,
-           const u16* __restrict__ w, int N, int K) {
+           const u16* __restrict__ w, int N, int K, float* __restrict__ ssq) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
   const int n0 = blockIdx.x * 8 + wid * 2;
-  if (n0 >= N) return;
+  const bool active = n0 < N;   // inactive waves still reach the barrier
   const bool two = (n0 + 1) < N;
-  const s16x8* wrow0 = (const s16x8*)(w + (long)n0 * K);
-  const s16x8* wrow1 = (const s16x8*)(w + (long)(n0 + (two ? 1 : 0)) * K);
-  float acc0[M], acc1[M];
+  const s16x8* wrow0 = (const s16x8*)(w + (long)(active ? n0 : 0) * K);
+  const s16x8* wrow1 = (const s16x8*)(w + (long)(active ? n0 + (two ? 1 : 0) : 0) * K);
+  float acc0[M], acc1[M], mysq[M];
 #pragma unroll
-  for (int m = 0; m < M; ++m) { acc0[m] = 0.f; acc1[m] = 0.f; }
+  for (int m = 0; m < M; ++m) { acc0[m] = 0.f; acc1[m] = 0.f; mysq[m] = 0.f; }
   const int nv = K >> 3;
-  for (int i = lane; i < nv; i += 64) {
-    s16x8 wv0 = NT ? __builtin_nontemporal_load(&wrow0[i]) : wrow0[i];
-    s16x8 wv1 = NT ? __builtin_nontemporal_load(&wrow1[i]) : wrow1[i];
+  if (active) {
+    for (int i = lane; i < nv; i += 64) {
+      s16x8 wv0 = NT ? __builtin_nontemporal_load(&wrow0[i]) : wrow0[i];
+      s16x8 wv1 = NT ? __builtin_nontemporal_load(&wrow1[i]) : wrow1[i];
 #pragma unroll
-    for (int m = 0; m < M; ++m) {
-      s16x8 xv = ((const s16x8*)(x + (long)m * K))[i];
-      acc0[m] += dot8_bf16(xv, wv0);
-      acc1[m] += dot8_bf16(xv, wv1);
+      for (int m = 0; m < M; ++m) {
+        s16x8 xv = ((const s16x8*)(x + (long)m * K))[i];
+        acc0[m] += dot8_bf16(xv, wv0);
+        acc1[m] += dot8_bf16(xv, wv1);
+      }
     }
   }
 #pragma unroll
   for (int m = 0; m < M; ++m) {
     const float v0 = wave_reduce_sum(acc0[m]);
     const float v1 = wave_reduce_sum(acc1[m]);
-    if (lane == 0) {
+    if (lane == 0 && active) {
       u16* r = res + (long)m * N + n0;
-      r[0] = f2bf(bf2f(r[0]) + v0);
-      if (two) r[1] = f2bf(bf2f(r[1]) + v1);
+      const u16 b0 = f2bf(bf2f(r[0]) + v0);
+      r[0] = b0;
+      const float f0 = bf2f(b0);
+      mysq[m] = f0 * f0;
+      if (two) {
+        const u16 b1 = f2bf(bf2f(r[1]) + v1);
+        r[1] = b1;
+        const float f1 = bf2f(b1);
+        mysq[m] += f1 * f1;
+      }
+    }
+  }
+  if (ssq) {                         // uniform across the block
+    __shared__ float sred[4][M];
+    if (lane == 0) {
+#pragma unroll
+      for (int m = 0; m < M; ++m) sred[wid][m] = mysq[m];
+    }
+    __syncthreads();
+    if (tid == 0) {
+#pragma unroll
+      for (int m = 0; m < M; ++m)
+        atomicAdd(&ssq[m], sred[0][m] + sred[1][m] + sred[2][m] + sred[3][m]);
     }
   }
 }
@@ -222,14 +253,20 @@ template <int M, bool NT>
 __global__ void __launch_bounds__(256)
 k_gemv_norm(u16* __restrict__ out, const u16* __restrict__ res,
             const u16* __restrict__ wn, const u16* __restrict__ w,
-            int N, int K, float eps) {
+            int N, int K, float eps, const float* __restrict__ ssq) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
   const int n0 = blockIdx.x * 8 + wid * 2;
   if (n0 >= N) return;
   float inv[M];
-  norm_factors<M>(res, K, eps, lane, inv);
+  if (ssq) {     // sumsq precomputed by the producing k_gemv_res epilogue
+#pragma unroll
+    for (int m = 0; m < M; ++m)
+      inv[m] = rsqrtf(ssq[m] / (float)K + eps);
+  } else {
+    norm_factors<M>(res, K, eps, lane, inv);
+  }
   const bool two = (n0 + 1) < N;
   const s16x8* wrow0 = (const s16x8*)(w + (long)n0 * K);
   const s16x8* wrow1 = (const s16x8*)(w + (long)(n0 + (two ? 1 : 0)) * K);
@@ -269,14 +306,20 @@ template <int M>
 __global__ void __launch_bounds__(256)
 k_gemv_swiglu_norm(u16* __restrict__ out, const u16* __restrict__ res,
                    const u16* __restrict__ wn, const u16* __restrict__ w,
-                   int N, int K, float eps) {
+                   int N, int K, float eps, const float* __restrict__ ssq) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
   const int n = blockIdx.x * 4 + wid;
   if (n >= N) return;
   float inv[M];
-  norm_factors<M>(res, K, eps, lane, inv);
+  if (ssq) {
+#pragma unroll
+    for (int m = 0; m < M; ++m)
+      inv[m] = rsqrtf(ssq[m] / (float)K + eps);
+  } else {
+    norm_factors<M>(res, K, eps, lane, inv);
+  }
   const s16x8* grow = (const s16x8*)(w + (long)n * K);
   const s16x8* urow = (const s16x8*)(w + (long)(n + N) * K);
   float accg[M], accu[M];
@@ -321,10 +364,11 @@ extern "C" {
   }
 
 void fei_gemv_res(void* res, const void* x, const void* w, int M, int N,
-                  int K, int nontemporal, hipStream_t stream) {
+                  int K, int nontemporal, void* ssq, hipStream_t stream) {
   dim3 grid((N + 7) / 8);
 #define LR(MV, NTV) hipLaunchKernelGGL((k_gemv_res<MV, NTV>), grid, \
-    dim3(256), 0, stream, (u16*)res, (const u16*)x, (const u16*)w, N, K)
+    dim3(256), 0, stream, (u16*)res, (const u16*)x, (const u16*)w, N, K, \
+    (float*)ssq)
 #define LRD(MV, _ignored) do { if (nontemporal) LR(MV, true); else LR(MV, false); } while (0)
   DISPATCH_M(LRD, 0)
 #undef LRD
@@ -333,11 +377,11 @@ void fei_gemv_res(void* res, const void* x, const void* w, int M, int N,
 
 void fei_gemv_norm(void* out, const void* res, const void* wn, const void* w,
                    int M, int N, int K, float eps, int nontemporal,
-                   hipStream_t stream) {
+                   const void* ssq, hipStream_t stream) {
   dim3 grid((N + 7) / 8);
 #define LN(MV, NTV) hipLaunchKernelGGL((k_gemv_norm<MV, NTV>), grid, \
     dim3(256), 0, stream, (u16*)out, (const u16*)res, (const u16*)wn, \
-    (const u16*)w, N, K, eps)
+    (const u16*)w, N, K, eps, (const float*)ssq)
 #define LND(MV, _ignored) do { if (nontemporal) LN(MV, true); else LN(MV, false); } while (0)
   DISPATCH_M(LND, 0)
 #undef LND
@@ -346,11 +390,11 @@ void fei_gemv_norm(void* out, const void* res, const void* wn, const void* w,
 
 void fei_gemv_swiglu_norm(void* out, const void* res, const void* wn,
                           const void* w, int M, int N, int K, float eps,
-                          hipStream_t stream) {
+                          const void* ssq, hipStream_t stream) {
   dim3 grid((N + 3) / 4);
 #define LS(MV, _ignored) hipLaunchKernelGGL(k_gemv_swiglu_norm<MV>, grid, \
     dim3(256), 0, stream, (u16*)out, (const u16*)res, (const u16*)wn, \
-    (const u16*)w, N, K, eps)
+    (const u16*)w, N, K, eps, (const float*)ssq)
   DISPATCH_M(LS, 0)
 #undef LS
 }

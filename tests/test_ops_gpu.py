@@ -478,3 +478,40 @@ def test_gemv_swiglu_norm_fp8_matches_dequant(lib):
     err = ((out.float() - expected.float()).abs() /
            (1 + expected.float().abs())).max().item()
     assert err < 3e-2, f"max rel err {err}"
+
+
+@pytest.mark.parametrize("M", [1, 4])
+def test_gemv_ssq_chain(lib, M):
+    """gemv_res(ssq_out) -> gemv_norm/gemv_swiglu_norm(ssq): the epilogue-
+    accumulated sum-of-squares must reproduce the full norm prologue."""
+    N, K, I = 4096, 4096, 14336
+    x = randbf(M, K, seed=330)
+    w = randbf(N, K, seed=331, scale=0.02)
+    res = randbf(M, K, seed=332)
+    wn = randbf(K, seed=333, scale=0.5)
+    w2 = randbf(N, K, seed=334, scale=0.02)
+    wgu = randbf(2 * I, K, seed=335, scale=0.02)
+
+    res_pre = res.clone()
+    ssq = torch.zeros(M, dtype=torch.float32, device=res.device)
+    lib.gemv_res(x, w, res, ssq_out=ssq)
+    # the accumulated ssq equals the sumsq of the updated bf16 residual
+    want_ssq = res.float().pow(2).sum(dim=-1)
+    assert torch.allclose(ssq, want_ssq, rtol=1e-4), (ssq, want_ssq)
+
+    # the two paths sum the squares in different fp32 orders; a ~1e-7
+    # difference in the norm factor can flip bf16 rounding of individual
+    # normalized activations, so compare at the kernel-standard tolerance
+    out = lib.gemv_norm(res, wn, w2, 1e-5, ssq=ssq)
+    ref_out = lib.gemv_norm(res, wn, w2, 1e-5)
+    err = ((out.float() - ref_out.float()).abs() /
+           (1 + ref_out.float().abs())).max().item()
+    assert err < 2e-2, f"gemv_norm ssq path max rel err {err}"
+
+    act = lib.gemv_swiglu_norm(res, wn, wgu, 1e-5, ssq=ssq)
+    ref_act = lib.gemv_swiglu_norm(res, wn, wgu, 1e-5)
+    err = ((act.float() - ref_act.float()).abs() /
+           (1 + ref_act.float().abs())).max().item()
+    assert err < 2e-2, f"gemv_swiglu_norm ssq path max rel err {err}"
+    # sanity: the chain actually changed the residual (res_pre unused warn)
+    assert not torch.equal(res, res_pre)
