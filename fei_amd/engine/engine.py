@@ -109,9 +109,12 @@ class LocalEngine:
         self.sample_ws = torch.zeros(self.B, SAMPLE_CHUNKS, 2,
                                      dtype=torch.float32, device=device)
         Hq_l, D = self.model.hq_l, self.model.D
+        # sized from self.attn_splits (NOT the ctor arg: FEI_ATTN_SPLITS
+        # overrides it, and an undersized workspace is an out-of-bounds
+        # write in the split kernel)
         self.attn_ws = (
-            torch.zeros(self.B, Hq_l, attn_splits, D, dtype=torch.float32, device=device),
-            torch.zeros(self.B, Hq_l, attn_splits, 2, dtype=torch.float32, device=device),
+            torch.zeros(self.B, Hq_l, self.attn_splits, D, dtype=torch.float32, device=device),
+            torch.zeros(self.B, Hq_l, self.attn_splits, 2, dtype=torch.float32, device=device),
         )
         self.attn_out = torch.zeros(self.B, Hq_l, D, dtype=self.dtype,
                                     device=device)

@@ -208,6 +208,8 @@ def attn_decode(q, k_cache, v_cache, pos, splits: int = 4,
         part_ml = torch.empty(B, Hq, splits, 2, dtype=torch.float32, device=q.device)
     else:
         part_o, part_ml = workspace
+        assert part_o.shape[2] == splits and part_ml.shape[2] == splits, \
+            f"workspace sized for {part_o.shape[2]} splits, kernel asked {splits}"
     if out is None:
         out = torch.empty_like(q)
     Hkv = k_cache.shape[1]
@@ -510,6 +512,7 @@ def attn_decode_paged(q, k_pool, v_pool, block_table, pos, splits: int = 4,
         part_ml = torch.empty(B, Hq, splits, 2, dtype=torch.float32, device=q.device)
     else:
         part_o, part_ml = workspace
+        assert part_o.shape[2] == splits and part_ml.shape[2] == splits
     if out is None:
         out = torch.empty_like(q)
     assert q.stride(2) == 1 and q.stride(1) == D

@@ -298,3 +298,14 @@ def test_chunked_prefill_matches_single_shot():
     e2 = LocalEngine.create("llama3-tiny", max_seq_len=128, seed=17)
     b = e2.generate(prompt, max_new_tokens=5, stop_on_eos=False)
     assert a["token_ids"] == b["token_ids"]
+
+
+def test_attn_splits_env_sizes_workspace(monkeypatch):
+    """FEI_ATTN_SPLITS must size the split-K workspace (an undersized
+    workspace is an out-of-bounds write in the kernel)."""
+    monkeypatch.setenv("FEI_ATTN_SPLITS", "16")
+    from fei_amd.engine.engine import LocalEngine
+    eng = LocalEngine.create("llama3-tiny")
+    assert eng.attn_splits == 16
+    assert eng.attn_ws[0].shape[2] == 16 and eng.attn_ws[1].shape[2] == 16
+    eng.generate([5, 6, 7], max_new_tokens=4)  # CPU path still consistent
