@@ -46,7 +46,7 @@ class LocalEngine:
         use_hip_graph: Optional[bool] = None,
         tp: Optional[ParallelContext] = None,
         seed: int = 1234,
-        attn_splits: int = 4,
+        attn_splits: int = 32,
         dtype: Optional[torch.dtype] = None,
         weight_quant: Optional[str] = None,    # None | "fp8" (decode path)
     ):
@@ -65,6 +65,11 @@ class LocalEngine:
             use_hip_graph = self.is_gpu and not self.tp.is_distributed
         self.use_graph = use_hip_graph and self.is_gpu
         import os as _os
+        # split-K decode attention: splits x Hkv x B workgroups. 4 splits
+        # (32 WGs on a 256-CU chip) starved KV bandwidth at agent context
+        # lengths: 4 -> 32 splits measured +3% at seq 512 and +41% at seq
+        # 3400 on 8B (docs/BENCHMARKS.md). Empty splits at short lengths
+        # write -inf partials and cost nothing.
         env_splits = _os.environ.get("FEI_ATTN_SPLITS")
         self.attn_splits = int(env_splits) if env_splits else attn_splits
         # Fused single-pass attention (rope+append+attn in one kernel) runs

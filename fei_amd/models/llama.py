@@ -182,7 +182,7 @@ class LlamaModel:
         pos: torch.Tensor,            # [B] int32 (current length; kv written here)
         k_caches: List[torch.Tensor],
         v_caches: List[torch.Tensor],
-        attn_splits: int = 4,
+        attn_splits: int = 32,
         workspace=None,
         fused_attn: bool = False,
         attn_out: Optional[torch.Tensor] = None,

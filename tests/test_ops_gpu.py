@@ -106,7 +106,7 @@ def test_rope_kv_prefill_strided(lib):
     assert torch.equal(vc, vc_ref)
 
 
-@pytest.mark.parametrize("n,splits", [(1, 1), (7, 4), (300, 16), (1000, 16), (20, 32), (2000, 32)])
+@pytest.mark.parametrize("n,splits", [(1, 1), (7, 4), (300, 16), (1000, 16), (20, 32), (1000, 32)])
 def test_attn_decode(lib, n, splits):
     from fei_amd.ops import reference as ref
     B, Hq, Hkv, D, MS = 2, 8, 2, 128, 1024
@@ -300,7 +300,7 @@ def test_sample_onepass_greedy(lib):
     assert torch.equal(out_tokens[:, 0].long(), expected)
 
 
-@pytest.mark.parametrize("n,splits", [(1, 4), (70, 4), (640, 4), (300, 16), (40, 32), (2000, 32)])
+@pytest.mark.parametrize("n,splits", [(1, 4), (70, 4), (640, 4), (300, 16), (40, 32), (1000, 32)])
 def test_attn_decode_rope_fused_split(lib, n, splits):
     """Split-K attention with in-kernel RoPE+append vs the reference
     pipeline (covers owner-split cache writes and LDS new-key path)."""
