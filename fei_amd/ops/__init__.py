@@ -184,7 +184,7 @@ def rope_kv_prefill(q, k, v, k_cache, v_cache, pos0, table) -> torch.Tensor:
     return q
 
 
-def attn_decode(q, k_cache, v_cache, pos, splits: int = 4,
+def attn_decode(q, k_cache, v_cache, pos, splits: int = 32,
                 scale: Optional[float] = None,
                 workspace: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
                 out: Optional[torch.Tensor] = None,
@@ -482,7 +482,7 @@ def gemv_swiglu_norm(res: torch.Tensor, wnorm: torch.Tensor,
     return out
 
 
-def attn_decode_paged(q, k_pool, v_pool, block_table, pos, splits: int = 4,
+def attn_decode_paged(q, k_pool, v_pool, block_table, pos, splits: int = 32,
                       scale: Optional[float] = None,
                       workspace: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
                       out: Optional[torch.Tensor] = None) -> torch.Tensor:
