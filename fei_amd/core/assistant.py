@@ -264,6 +264,21 @@ class Assistant:
             answer = self.conversation.scrape_tool_output()
         return answer
 
+    async def achat(self, message: str,
+                    system_prompt: Optional[str] = None) -> str:
+        """Async facade over chat() (the reference's chat IS async —
+        assistant.py:440; ours runs the GPU decode in a worker thread so an
+        event-loop UI stays responsive)."""
+        import asyncio
+        return await asyncio.to_thread(self.chat, message, system_prompt)
+
+    async def aask(self, message: str, system_prompt: Optional[str] = None,
+                   max_tool_rounds: int = 8) -> str:
+        """Async facade over ask() (reference parity: async tool loop)."""
+        import asyncio
+        return await asyncio.to_thread(self.ask, message, system_prompt,
+                                       max_tool_rounds)
+
     def reset(self) -> None:
         self.conversation.clear()
         self.turn_metrics.clear()

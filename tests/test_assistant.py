@@ -135,3 +135,27 @@ def test_local_provider_refuses_big_model_on_cpu(cfg):
         pytest.skip("GPU present")
     with pytest.raises(RuntimeError, match="needs a GPU"):
         Assistant(config=cfg, provider="local", model="llama3-8b").chat("hi")
+
+
+def test_async_facades(cfg, tmp_path):
+    import asyncio
+    (tmp_path / "z.txt").write_text("zeta\n")
+    reg = ToolRegistry()
+    create_code_tools(reg)
+    a = Assistant(config=cfg, provider="stub", tool_registry=reg)
+
+    async def run():
+        r1 = await a.achat("hello async")
+        script = [
+            {"tool_calls": [{"name": "GlobTool",
+                             "input": {"pattern": "*.txt", "path": str(tmp_path)}}]},
+            {"content": "found z [TASK_COMPLETE]"},
+        ]
+        b = Assistant(config=cfg, backend=ScriptedBackend(script),
+                      tool_registry=reg)
+        r2 = await b.aask("find txt files")
+        return r1, r2
+
+    r1, r2 = asyncio.run(run())
+    assert "hello async" in r1
+    assert "found z" in r2
