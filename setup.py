@@ -27,6 +27,7 @@ setup(
         "console_scripts": [
             "fei=fei_amd.ui.cli:main",
             "memdir=fei_amd.memdir.cli:main",
+            "memdir-server=fei_amd.memdir.run_server:main",
             "memorychain=fei_amd.memorychain.cli:main",
         ],
     },
