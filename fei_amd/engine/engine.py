@@ -260,12 +260,86 @@ class LocalEngine:
                     for row in rows]
         return rows
 
+    def decode_speculative(self, n_tokens: int, ctx_tokens: List[int],
+                           spec_k: int = 8, stop_on_eos: bool = True,
+                           ) -> List[List[int]]:
+        """Greedy decode with prompt-lookup speculation (engine/speculative.py):
+        propose up to ``spec_k`` tokens from the context's own repeats and
+        verify the block with one prefill forward (all-positions logits).
+        Token-identical to decode() at temperature 0; call after prefill().
+        Batch 1 only."""
+        from fei_amd.engine.speculative import NgramIndex
+
+        assert self.B == 1, "speculative decode is a batch-1 agent path"
+        assert self.temperature == 0.0, "speculative decode is greedy-only"
+        eos = self.tokenizer.eos_id
+        first = int(self.token[0])            # token sampled by prefill
+        index = NgramIndex()
+        index.extend(list(ctx_tokens) + [first])
+        pos_h = int(self.pos[0])              # host mirror of the position
+        done = 1
+        n_blocks = n_proposed = n_accepted = 0
+        while done < n_tokens and not (stop_on_eos and index.ctx[-1] == eos):
+            props = index.propose(spec_k)
+            block = [index.ctx[-1]] + props
+            L = len(block)
+            if pos_h + L >= self.max_seq_len:
+                block = block[: self.max_seq_len - 1 - pos_h]
+                props = block[1:]
+                L = len(block)
+                if L == 0:
+                    break
+            tokens = torch.tensor([block], dtype=torch.int64, device=self.device)
+            pos0 = torch.full((1,), pos_h, dtype=torch.int32, device=self.device)
+            logits = self.model.forward_prefill(tokens, pos0, self.k_caches,
+                                                self.v_caches,
+                                                all_positions=True)
+            greedy = logits[0].argmax(dim=-1).tolist()        # len L
+            n_acc = 0
+            while n_acc < len(props) and props[n_acc] == greedy[n_acc]:
+                n_acc += 1
+            emitted = props[:n_acc] + [greedy[n_acc]]
+            emitted = emitted[: n_tokens - done]
+            if stop_on_eos and eos in emitted:
+                emitted = emitted[: emitted.index(eos) + 1]
+            n_blocks += 1
+            n_proposed += len(props)
+            n_accepted += n_acc
+            index.extend(emitted)
+            self.out_tokens[0, done: done + len(emitted)] = torch.tensor(
+                emitted, dtype=torch.int32, device=self.device)
+            done += len(emitted)
+            pos_h += len(emitted)
+        # re-sync the device-resident decode state so plain decode()/graph
+        # replay can continue from here
+        self.pos.fill_(pos_h)
+        self.step.fill_(done)
+        self.token.fill_(index.ctx[-1])
+        if self.is_gpu:
+            torch.cuda.synchronize(self.device)
+        self.last_metrics.update({
+            "spec_blocks": n_blocks,
+            "spec_proposed": n_proposed,
+            "spec_accepted": n_accepted,
+            "spec_acceptance": n_accepted / max(n_proposed, 1),
+            "spec_tokens_per_block": done / max(n_blocks, 1),
+        })
+        rows = self.out_tokens[:, :done].tolist()
+        if stop_on_eos:
+            rows = [row[: row.index(eos) + 1] if eos in row else row
+                    for row in rows]
+        return rows
+
     def generate(self, prompt: Union[str, List[int]], max_new_tokens: int = 256,
                  temperature: float = 0.0, stop_on_eos: bool = True,
-                 from_pos: int = 0) -> Dict[str, object]:
+                 from_pos: int = 0,
+                 speculative: Optional[bool] = None) -> Dict[str, object]:
         """Prefill + decode; returns text and timing metrics. ``from_pos``
         enables prefix caching: the prompt's first ``from_pos`` tokens are
-        already in the KV caches and only the remainder is prefilled."""
+        already in the KV caches and only the remainder is prefilled.
+        ``speculative`` (default: FEI_SPEC_DECODE env) uses prompt-lookup
+        speculative decoding — greedy/batch-1 only, token-identical output;
+        pays on contexts that repeat themselves (tool output, code)."""
         if isinstance(prompt, str):
             prompt_ids = self.tokenizer.encode(prompt)
         else:
@@ -292,11 +366,23 @@ class LocalEngine:
         t0 = time.perf_counter()
         self.prefill(new_ids, from_pos=from_pos)
         t1 = time.perf_counter()
-        rows = self.decode(max_new_tokens, stop_on_eos=stop_on_eos)
+        if speculative is None:
+            import os as _os
+            speculative = _os.environ.get("FEI_SPEC_DECODE", "0") == "1"
+        spec_metrics: Dict[str, object] = {}
+        if speculative and self.B == 1 and temperature == 0.0:
+            ctx = prompt_ids[:from_pos] + new_ids
+            self.last_metrics = {}
+            rows = self.decode_speculative(max_new_tokens, ctx,
+                                           stop_on_eos=stop_on_eos)
+            spec_metrics = dict(self.last_metrics)
+        else:
+            rows = self.decode(max_new_tokens, stop_on_eos=stop_on_eos)
         t2 = time.perf_counter()
         new_tokens = len(rows[0])
         decode_s = t2 - t1
         self.last_metrics = {
+            **spec_metrics,
             "prompt_tokens": len(prompt_ids),
             "cached_prefix": from_pos,
             "new_tokens": new_tokens,
