@@ -1,0 +1,189 @@
+"""Paged multi-session serving: continuous batching over a PagedKVPool.
+
+The reference serves one remote conversation per Assistant; this engine
+serves MANY resident agent sessions on one GPU: each session's KV lives in
+16-token blocks of a shared pool (`engine/kv_cache.py`), sessions join and
+leave the decode batch at any step, and the attention read path is the HIP
+block-table kernel (`k_attn_decode_paged`). Admission is explicit — a
+session that cannot get blocks raises and the caller can evict.
+
+Control-plane costs stay host-side (per-step RoPE/append scatter is a few
+microseconds of torch indexing per layer); the O(context) work — attention
+over the pool — is the paged HIP kernel, and all GEMVs take the same
+streaming kernels as the single-session path. Greedy decode (serving
+agents at temperature 0); graphs are off (batch membership changes).
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Union
+
+import torch
+import torch.nn.functional as F
+
+from fei_amd import ops
+from fei_amd.engine.kv_cache import PagedKVPool
+from fei_amd.ops import reference as ref
+
+
+@dataclass
+class Session:
+    sid: int
+    prompt_ids: List[int]
+    generated: List[int] = field(default_factory=list)
+    pos: int = 0                     # current context length in the pool
+    max_new_tokens: int = 256
+    done: bool = False
+
+
+class PagedSessionManager:
+    def __init__(self, engine, block_size: int = 16,
+                 num_blocks: Optional[int] = None,
+                 mem_fraction: float = 0.25):
+        self.engine = engine
+        self.model = engine.model
+        m = self.model
+        assert m.tp_size == 1, "paged session serving is single-GPU"
+
+        self.pool = PagedKVPool(
+            num_layers=m.spec.num_layers, num_kv_heads=m.hkv_l,
+            head_dim=m.D, block_size=block_size, num_blocks=num_blocks,
+            device=m.device, dtype=m.dtype, mem_fraction=mem_fraction)
+        self.sessions: Dict[int, Session] = {}
+        self.eos = engine.tokenizer.eos_id
+
+    # -- admission -----------------------------------------------------------
+
+    def open(self, prompt: Union[str, List[int]],
+             max_new_tokens: int = 256) -> int:
+        """Prefill a new session into the pool; returns its id. Raises
+        MemoryError when the pool has no blocks (caller evicts/queues)."""
+        ids = (self.engine.tokenizer.encode(prompt)
+               if isinstance(prompt, str) else list(prompt))
+        m, dev = self.model, self.model.device
+        S = len(ids)
+        sid = self.pool.new_sequence()
+        try:
+            self.pool.ensure_capacity(sid, S + 1)
+        except MemoryError:
+            self.pool.release(sid)
+            raise
+        # prefill through the normal MFMA path into a contiguous scratch,
+        # then scatter each layer's S rows into this session's blocks
+        k_tmp = [torch.zeros(1, m.hkv_l, S, m.D, device=dev, dtype=m.dtype)
+                 for _ in range(m.spec.num_layers)]
+        v_tmp = [torch.zeros_like(k_tmp[0]) for _ in range(m.spec.num_layers)]
+        tokens = torch.tensor([ids], dtype=torch.int64, device=dev)
+        pos0 = torch.zeros(1, dtype=torch.int32, device=dev)
+        logits = m.forward_prefill(tokens, pos0, k_tmp, v_tmp)
+        BS = self.pool.block_size
+        blocks = self.pool.table(sid).blocks
+        for li in range(m.spec.num_layers):
+            for j in range(0, S, BS):
+                blk = blocks[j // BS]
+                n = min(BS, S - j)
+                self.pool.k[li][blk, :, :n, :] = k_tmp[li][0, :, j:j + n, :]
+                self.pool.v[li][blk, :, :n, :] = v_tmp[li][0, :, j:j + n, :]
+        first = int(logits[0].argmax())
+        sess = Session(sid=sid, prompt_ids=ids, generated=[first], pos=S,
+                       max_new_tokens=max_new_tokens)
+        if first == self.eos:
+            sess.done = True
+        self.sessions[sid] = sess
+        return sid
+
+    def close(self, sid: int) -> None:
+        self.pool.release(sid)
+        self.sessions.pop(sid, None)
+
+    # -- decode --------------------------------------------------------------
+
+    @property
+    def active(self) -> List[Session]:
+        return [s for s in self.sessions.values() if not s.done]
+
+    def step(self) -> int:
+        """One greedy decode step for every active session (one batched
+        forward over the paged pool). Returns the number of sessions that
+        advanced."""
+        act = self.active
+        if not act:
+            return 0
+        m, dev = self.model, self.model.device
+        for s in act:
+            self.pool.ensure_capacity(s.sid, s.pos + 1)
+        max_blocks = max(len(self.pool.table(s.sid).blocks) for s in act)
+        table = torch.full((len(act), max_blocks), -1, dtype=torch.int32)
+        for i, s in enumerate(act):
+            blocks = self.pool.table(s.sid).blocks
+            table[i, :len(blocks)] = torch.tensor(blocks, dtype=torch.int32)
+        table = table.to(dev)
+        token = torch.tensor([s.generated[-1] for s in act],
+                             dtype=torch.int64, device=dev)
+        pos = torch.tensor([s.pos for s in act], dtype=torch.int32, device=dev)
+        logits = self._forward_paged(token, pos, table,
+                                     [s.pos for s in act])
+        nxt = logits.argmax(dim=-1).tolist()
+        for s, t in zip(act, nxt):
+            s.generated.append(int(t))
+            s.pos += 1
+            if int(t) == self.eos or len(s.generated) >= s.max_new_tokens:
+                s.done = True
+        return len(act)
+
+    def run(self, max_steps: int = 4096) -> None:
+        for _ in range(max_steps):
+            if self.step() == 0:
+                break
+
+    def result(self, sid: int) -> Dict[str, object]:
+        s = self.sessions[sid]
+        gen = s.generated
+        if self.eos in gen:
+            gen = gen[: gen.index(self.eos) + 1]
+        return {"token_ids": gen,
+                "text": self.engine.tokenizer.decode(gen),
+                "done": s.done}
+
+    # -- model forward over the pool ------------------------------------------
+
+    def _forward_paged(self, token: torch.Tensor, pos: torch.Tensor,
+                       table: torch.Tensor,
+                       pos_host: List[int]) -> torch.Tensor:
+        """Plain decode forward with block-table attention. Same GEMV /
+        norm kernels as LlamaModel.forward_decode (llama.py); RoPE + the
+        single-row KV append are torch-side (control plane — O(Hkv*D) per
+        session, independent of context length)."""
+        m = self.model
+        s = m.spec
+        B = token.shape[0]
+        BS = self.pool.block_size
+        h = F.embedding(token.long(), m.emb)
+        scale = 1.0 / math.sqrt(m.D)
+        n_layers = len(m.layers)
+        x = ops.rmsnorm(h, m.layers[0].norm_attn, s.norm_eps)
+        pos_l = pos.long()
+        blk_of = [int(table[i, p // BS]) for i, p in enumerate(pos_host)]
+        off_of = [p % BS for p in pos_host]
+        for li, lw in enumerate(m.layers):
+            qkv = ops.linear_decode(x, lw.wqkv)
+            q, k, v = m._qkv_views(qkv, B)
+            q_r = ref.apply_rope(q, pos_l, m.rope).contiguous()
+            k_r = ref.apply_rope(k, pos_l, m.rope)
+            kp, vp = self.pool.k[li], self.pool.v[li]
+            for i in range(B):
+                kp[blk_of[i], :, off_of[i], :] = k_r[i]
+                vp[blk_of[i], :, off_of[i], :] = v[i]
+            att = ops.attn_decode_paged(q_r, kp, vp, table, pos,
+                                        splits=self.engine.attn_splits,
+                                        scale=scale)
+            o = ops.linear_decode(att.reshape(B, -1), lw.wo)
+            x, h = ops.fused_add_rmsnorm(o, h, lw.norm_mlp, s.norm_eps)
+            act = ops.gemv_swiglu(x, lw.wgu)
+            d = ops.linear_decode(act, lw.wdown)
+            next_norm = (m.layers[li + 1].norm_attn if li + 1 < n_layers
+                         else m.norm_f)
+            x, h = ops.fused_add_rmsnorm(d, h, next_norm, s.norm_eps)
+        return ops.linear_decode(x, m.lm_head)
