@@ -138,3 +138,37 @@ def test_generate_batch_gpu_smoke():
                             max_new_tokens=8, stop_on_eos=False)
     assert len(outs) == 2
     assert all(len(o["token_ids"]) == 8 for o in outs)
+
+
+def test_paged_sessions_gpu(engine):
+    """Continuous-batching manager on the HIP block-table attention kernel:
+    batched sessions match solo generate() runs (both eager kernel paths,
+    same numerics — greedy equality is required, not just close logits)."""
+    from fei_amd.engine.engine import LocalEngine
+    from fei_amd.engine.sessions import PagedSessionManager
+    eng = LocalEngine.create("llama3-tiny", max_seq_len=256, seed=7,
+                             use_hip_graph=False)
+    eng.fused_norm = False   # solo path == paged path kernels (exact compare)
+    prompts = ["paged one", "paged session two longer prompt"]
+    solo = {p: eng.generate(p, max_new_tokens=12, stop_on_eos=False)["token_ids"]
+            for p in prompts}
+    mgr = PagedSessionManager(eng, block_size=16, num_blocks=64)
+    sids = {p: mgr.open(p, max_new_tokens=12) for p in prompts}
+    mgr.run()
+    for p, sid in sids.items():
+        got = mgr.result(sid)["token_ids"]
+        assert got == solo[p][: len(got)], p
+
+
+def test_speculative_gpu(engine):
+    """Speculative greedy decode runs on GPU and emits a sane stream; the
+    verify forward is the prefill kernel path so cross-path equality is
+    not asserted (docs/ENGINE.md), only machinery invariants."""
+    from fei_amd.engine.engine import LocalEngine
+    eng = LocalEngine.create("llama3-tiny", max_seq_len=256, seed=7,
+                             use_hip_graph=False)
+    out = eng.generate("spec spec spec spec spec", max_new_tokens=24,
+                       speculative=True, stop_on_eos=False)
+    assert len(out["token_ids"]) == 24
+    assert out["spec_blocks"] >= 1
+    assert out["spec_tokens_per_block"] >= 1.0
