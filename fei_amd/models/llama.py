@@ -65,6 +65,13 @@ class LlamaModel:
         self._init_weights(seed)
         self.rope = ref.rope_table(self.max_seq_len, self.D, spec.rope_theta,
                                    device=device)
+        # FEI_PREFETCH=<MB>: during each decode layer's attention window
+        # (low HBM traffic) a side-stream kernel pre-reads the first MB of
+        # that layer's gate/up weights into L2/L3 so the big GEMV partially
+        # hits cache (experiment; 0 = off)
+        self.prefetch_mb = int(os.environ.get("FEI_PREFETCH", "0") or 0)
+        self._pf_stream = None
+        self._pf_sink = None
         # FEI_NORM_PRECOMP=1: precompute each norm's sumsq in the producing
         # residual epilogue. Measured NEGATIVE on MI355X (242.5 vs 247.5
         # tok/s, 8B decode): the norm prologue overlaps the previous
@@ -267,6 +274,11 @@ class LlamaModel:
                if (self.norm_precomp and fp8 is None and h.is_cuda) else None)
         if ssq is not None:
             ssq.zero_()
+        pf = self.prefetch_mb > 0 and fp8 is None and h.is_cuda
+        if pf and self._pf_stream is None:
+            self._pf_stream = torch.cuda.Stream()
+            self._pf_sink = torch.zeros(1, dtype=torch.float32,
+                                        device=h.device)
         for li, lw in enumerate(self.layers):
             if fp8 is not None:
                 q8 = fp8[li]
@@ -276,6 +288,15 @@ class LlamaModel:
                 qkv = ops.gemv_norm(
                     h, lw.norm_attn, lw.wqkv, s.norm_eps,
                     ssq=ssq[2 * li - 1] if (ssq is not None and li > 0) else None)
+            if pf:
+                # warm L2/L3 with this layer's gate/up weights while the
+                # attention chain (low HBM traffic) runs on the main stream
+                ev = torch.cuda.Event()
+                ev.record()
+                with torch.cuda.stream(self._pf_stream):
+                    self._pf_stream.wait_event(ev)
+                    ops.prefetch(lw.wgu, self.prefetch_mb << 20,
+                                 self._pf_sink)
             q, k, v = self._qkv_views(qkv, B)
             if fused_attn:
                 att = ops.attn_decode_fused(q, k, v, k_caches[li],
@@ -299,6 +320,10 @@ class LlamaModel:
                     ssq=ssq[2 * li] if ssq is not None else None)
                 ops.gemv_res(act, lw.wdown, h,
                              ssq_out=ssq[2 * li + 1] if ssq is not None else None)
+        if pf:
+            # rejoin the prefetch stream so graph capture sees all work
+            # ordered (pure cache warming: no data dependency otherwise)
+            torch.cuda.current_stream().wait_stream(self._pf_stream)
         if fp8 is not None:
             return ops.gemv_norm_fp8(h, self.norm_f, *self.fp8_lm_head,
                                      s.norm_eps)

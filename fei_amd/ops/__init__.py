@@ -75,6 +75,7 @@ def _try_load() -> None:
                                              _i, _f, _vp]
     lib.fei_quant_fp8_rows.argtypes = [_vp, _vp, _vp, _i, _i, _vp]
     lib.fei_gemv_swiglu.argtypes = [_vp, _vp, _vp, _i, _i, _i, _vp]
+    lib.fei_prefetch.argtypes = [_vp, _l, _vp, _i, _vp]
     lib.fei_gemv_res.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp, _vp]
     lib.fei_gemv_norm.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _i, _f, _i, _vp,
                                   _vp]
@@ -480,6 +481,19 @@ def gemv_swiglu_norm(res: torch.Tensor, wnorm: torch.Tensor,
                              _ptr(ssq) if ssq is not None else None,
                              _stream())
     return out
+
+
+def prefetch(w: torch.Tensor, max_bytes: int, sink: torch.Tensor,
+             n_blocks: int = 64, stream=None) -> None:
+    """Stream the first ``max_bytes`` of ``w`` through cache-filling loads
+    (L2/L3 prefill for a later kernel's reads). GPU-only no-op helper for
+    the decode prefetch experiment; launched on a side stream."""
+    if not w.is_cuda:
+        return
+    lib = require_lib()
+    bytes_ = min(max_bytes, w.numel() * w.element_size())
+    sp = stream if stream is not None else _stream()
+    lib.fei_prefetch(_ptr(w), bytes_, _ptr(sink), n_blocks, sp)
 
 
 def attn_decode_paged(q, k_pool, v_pool, block_table, pos, splits: int = 32,

@@ -350,9 +350,33 @@ k_gemv_swiglu_norm(u16* __restrict__ out, const u16* __restrict__ res,
   }
 }
 
+// Cache prefetch: stream a weight region through normal (cache-filling)
+// loads so a LATER kernel's reads hit L2/L3 instead of HBM. Launched on a
+// side stream during low-bandwidth phases (decode attention) to convert
+// idle HBM cycles into prefill of the next GEMV's operand. The impossible
+// store defeats dead-code elimination; nothing is written in practice.
+__global__ void __launch_bounds__(256)
+k_prefetch(const u16* __restrict__ w, long n_vec8, float* __restrict__ sink) {
+  float acc = 0.f;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_vec8;
+       i += (long)gridDim.x * blockDim.x) {
+    const s16x8 v = ((const s16x8*)w)[i];
+    acc += (float)(short)v[0];
+  }
+  if (acc == 1.0f / 0.0f) sink[0] = acc;
+}
+
 }  // namespace
 
 extern "C" {
+
+void fei_prefetch(const void* w, long bytes, void* sink, int n_blocks,
+                  hipStream_t stream) {
+  const long n_vec8 = bytes / 16;
+  if (n_vec8 <= 0) return;
+  hipLaunchKernelGGL(k_prefetch, dim3(n_blocks), dim3(256), 0, stream,
+                     (const u16*)w, n_vec8, (float*)sink);
+}
 
 #define DISPATCH_M(FN, ...) \
   switch (M) { \
