@@ -484,7 +484,7 @@ def gemv_swiglu_norm(res: torch.Tensor, wnorm: torch.Tensor,
 
 
 def prefetch(w: torch.Tensor, max_bytes: int, sink: torch.Tensor,
-             n_blocks: int = 64, stream=None) -> None:
+             n_blocks: int = 512, stream=None) -> None:
     """Stream the first ``max_bytes`` of ``w`` through cache-filling loads
     (L2/L3 prefill for a later kernel's reads). GPU-only no-op helper for
     the decode prefetch experiment; launched on a side stream."""
