@@ -1,0 +1,157 @@
+"""OpenAI-style HTTP serving front-end for the local engine.
+
+POST /v1/completions        {model?, prompt, max_tokens, temperature, top_k,
+                             top_p, stop_on_eos}
+POST /v1/chat/completions   {model?, messages=[{role, content}], ...}
+GET  /v1/models             available ModelSpecs
+GET  /health                engine + device status
+
+One engine instance serves requests serially under a lock (the decode hot
+path owns the GPU; concurrent multi-session serving is the
+PagedSessionManager API, engine/sessions.py — an HTTP multiplexer over it
+is a natural extension). The reference has no serving surface at all (its
+"server" is the memdir HTTP API); this is engine-native.
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from typing import List, Optional
+
+from pydantic import BaseModel
+
+from fei_amd.engine.config import MODEL_SPECS
+from fei_amd.engine.engine import LocalEngine
+
+
+class CompletionRequest(BaseModel):
+    prompt: str
+    model: Optional[str] = None          # informational; engine is fixed
+    max_tokens: int = 256
+    temperature: float = 0.0
+    top_k: int = 0
+    top_p: float = 1.0
+    stop_on_eos: bool = True
+    speculative: Optional[bool] = None
+
+
+class ChatMessage(BaseModel):
+    role: str
+    content: str
+
+
+class ChatRequest(BaseModel):
+    messages: List[ChatMessage]
+    model: Optional[str] = None
+    max_tokens: int = 256
+    temperature: float = 0.0
+    top_k: int = 0
+    top_p: float = 1.0
+
+
+def flatten_chat(messages: List[ChatMessage]) -> str:
+    """Minimal chat template for the byte tokenizer (a real checkpoint
+    would bring its own template)."""
+    parts = [f"{m.role}: {m.content}" for m in messages]
+    parts.append("assistant:")
+    return "\n".join(parts)
+
+
+def create_app(engine: Optional[LocalEngine] = None,
+               model: str = "llama3-tiny", **engine_kwargs):
+    from fastapi import FastAPI
+
+    app = FastAPI(title="fei_amd", version="0.1.0")
+    eng = engine or LocalEngine.create(model, **engine_kwargs)
+    lock = threading.Lock()
+    model_name = eng.spec.name
+
+    def _generate(prompt: str, req) -> dict:
+        with lock:
+            if req.temperature > 0 and (req.top_k or req.top_p < 1.0):
+                out = eng.generate_sampled(
+                    prompt, max_new_tokens=req.max_tokens,
+                    temperature=req.temperature, top_k=req.top_k,
+                    top_p=req.top_p)
+            else:
+                out = eng.generate(
+                    prompt, max_new_tokens=req.max_tokens,
+                    temperature=req.temperature,
+                    stop_on_eos=getattr(req, "stop_on_eos", True),
+                    speculative=getattr(req, "speculative", None))
+        return out
+
+    @app.get("/health")
+    def health():
+        return {"status": "ok", "model": model_name,
+                "device": str(eng.device),
+                "gpu": eng.is_gpu, "hip_graph": eng.use_graph}
+
+    @app.get("/v1/models")
+    def models():
+        return {"object": "list",
+                "data": [{"id": name, "object": "model",
+                          "owned_by": "fei_amd",
+                          "active": name == model_name}
+                         for name in sorted(MODEL_SPECS)]}
+
+    @app.post("/v1/completions")
+    def completions(req: CompletionRequest):
+        t0 = time.time()
+        out = _generate(req.prompt, req)
+        n_new = len(out["token_ids"])
+        return {
+            "id": f"cmpl-{int(t0 * 1000)}",
+            "object": "text_completion",
+            "created": int(t0),
+            "model": model_name,
+            "choices": [{"index": 0, "text": out["text"],
+                         "finish_reason": "stop"}],
+            "usage": {"prompt_tokens": out.get("prompt_tokens", 0),
+                      "completion_tokens": n_new,
+                      "total_tokens": out.get("prompt_tokens", 0) + n_new},
+            "metrics": {k: v for k, v in eng.last_metrics.items()
+                        if isinstance(v, (int, float))},
+        }
+
+    @app.post("/v1/chat/completions")
+    def chat(req: ChatRequest):
+        t0 = time.time()
+        out = _generate(flatten_chat(req.messages), req)
+        return {
+            "id": f"chatcmpl-{int(t0 * 1000)}",
+            "object": "chat.completion",
+            "created": int(t0),
+            "model": model_name,
+            "choices": [{"index": 0,
+                         "message": {"role": "assistant",
+                                     "content": out["text"]},
+                         "finish_reason": "stop"}],
+            "usage": {"prompt_tokens": out.get("prompt_tokens", 0),
+                      "completion_tokens": len(out["token_ids"]),
+                      "total_tokens": (out.get("prompt_tokens", 0) +
+                                       len(out["token_ids"]))},
+        }
+
+    app.state.engine = eng
+    return app
+
+
+def main(argv=None) -> int:
+    import argparse
+
+    import uvicorn
+
+    p = argparse.ArgumentParser("fei-api",
+                                description="OpenAI-style local serving API")
+    p.add_argument("--model", default="llama3-8b")
+    p.add_argument("--host", default="127.0.0.1")
+    p.add_argument("--port", type=int, default=8123)
+    args = p.parse_args(argv)
+    uvicorn.run(create_app(model=args.model), host=args.host, port=args.port)
+    return 0
+
+
+if __name__ == "__main__":
+    main()

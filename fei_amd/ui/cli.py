@@ -183,6 +183,10 @@ def build_parser() -> argparse.ArgumentParser:
     s = sub.add_parser("search", help="search memdir memories")
     s.add_argument("query", nargs="+")
 
+    api = sub.add_parser("api", help="OpenAI-style local serving API")
+    api.add_argument("--model", default="llama3-8b")
+    api.add_argument("--host", default="127.0.0.1")
+    api.add_argument("--api-port", type=int, default=8123)
     b = sub.add_parser("serve", help="run the memdir HTTP server")
     b.add_argument("--port", type=int, default=5000)
 
@@ -260,6 +264,10 @@ def main(argv: Optional[List[str]] = None) -> int:
         from fei_amd.memdir.search import format_results, search
         print(format_results(search(" ".join(args.query)), "compact"))
         return 0
+    if args.cmd == "api":
+        from fei_amd.serve.api import main as api_main
+        return api_main(["--model", args.model, "--host", args.host,
+                         "--port", str(args.api_port)])
     if args.cmd == "serve":
         from fei_amd.memdir.run_server import main as serve_main
         return serve_main(["--port", str(args.port)])

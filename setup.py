@@ -29,6 +29,7 @@ setup(
             "memdir=fei_amd.memdir.cli:main",
             "memdir-server=fei_amd.memdir.run_server:main",
             "memorychain=fei_amd.memorychain.cli:main",
+            "fei-api=fei_amd.serve.api:main",
         ],
     },
     package_data={"fei_amd.ops": ["csrc/*.hip", "csrc/*.h", "*.so"]},
