@@ -134,6 +134,57 @@ def create_app(engine: Optional[LocalEngine] = None,
                                        len(out["token_ids"]))},
         }
 
+    # -- multi-session serving (paged continuous batching) -------------------
+
+    state = {"mgr": None}
+
+    def _mgr():
+        if state["mgr"] is None:
+            from fei_amd.engine.sessions import PagedSessionManager
+            state["mgr"] = PagedSessionManager(eng)
+        return state["mgr"]
+
+    class SessionOpen(BaseModel):
+        prompt: str
+        max_tokens: int = 256
+
+    @app.post("/v1/sessions")
+    def session_open(req: SessionOpen):
+        from fastapi import HTTPException
+        with lock:
+            try:
+                sid = _mgr().open(req.prompt, max_new_tokens=req.max_tokens)
+            except MemoryError as e:
+                raise HTTPException(status_code=503, detail=str(e))
+        return {"session_id": sid}
+
+    @app.post("/v1/sessions/step")
+    def session_step(n: int = 1):
+        with lock:
+            mgr = _mgr()
+            advanced = 0
+            for _ in range(max(1, n)):
+                advanced = mgr.step()
+                if advanced == 0:
+                    break
+        return {"active": advanced,
+                "open": len(mgr.sessions),
+                "free_blocks": mgr.pool.free_blocks()}
+
+    @app.get("/v1/sessions/{sid}")
+    def session_get(sid: int):
+        from fastapi import HTTPException
+        mgr = _mgr()
+        if sid not in mgr.sessions:
+            raise HTTPException(status_code=404, detail="no such session")
+        return {"session_id": sid, **mgr.result(sid)}
+
+    @app.delete("/v1/sessions/{sid}")
+    def session_close(sid: int):
+        with lock:
+            _mgr().close(sid)
+        return {"closed": sid}
+
     app.state.engine = eng
     return app
 

@@ -56,3 +56,20 @@ def test_chat_completions(client):
 def test_validation_error(client):
     r = client.post("/v1/completions", json={"max_tokens": 4})
     assert r.status_code == 422          # prompt is required
+
+
+def test_sessions_endpoints(client):
+    a = client.post("/v1/sessions", json={"prompt": "abc abc", "max_tokens": 6}).json()
+    b = client.post("/v1/sessions", json={"prompt": "xyz xyz", "max_tokens": 6}).json()
+    sid_a, sid_b = a["session_id"], b["session_id"]
+    for _ in range(8):
+        st = client.post("/v1/sessions/step?n=2").json()
+        if st["active"] == 0:
+            break
+    ra = client.get(f"/v1/sessions/{sid_a}").json()
+    rb = client.get(f"/v1/sessions/{sid_b}").json()
+    assert ra["done"] and rb["done"]
+    assert len(ra["token_ids"]) >= 1 and len(rb["token_ids"]) >= 1
+    assert client.delete(f"/v1/sessions/{sid_a}").json()["closed"] == sid_a
+    client.delete(f"/v1/sessions/{sid_b}")
+    assert client.get(f"/v1/sessions/{sid_a}").status_code == 404
