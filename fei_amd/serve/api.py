@@ -41,6 +41,11 @@ class ChatMessage(BaseModel):
     content: str
 
 
+class SessionOpen(BaseModel):
+    prompt: str
+    max_tokens: int = 256
+
+
 class ChatRequest(BaseModel):
     messages: List[ChatMessage]
     model: Optional[str] = None
@@ -143,10 +148,6 @@ def create_app(engine: Optional[LocalEngine] = None,
             from fei_amd.engine.sessions import PagedSessionManager
             state["mgr"] = PagedSessionManager(eng)
         return state["mgr"]
-
-    class SessionOpen(BaseModel):
-        prompt: str
-        max_tokens: int = 256
 
     @app.post("/v1/sessions")
     def session_open(req: SessionOpen):
