@@ -340,26 +340,8 @@ class LocalEngine:
         ``speculative`` (default: FEI_SPEC_DECODE env) uses prompt-lookup
         speculative decoding — greedy/batch-1 only, token-identical output;
         pays on contexts that repeat themselves (tool output, code)."""
-        if isinstance(prompt, str):
-            prompt_ids = self.tokenizer.encode(prompt)
-        else:
-            prompt_ids = list(prompt)
-        if from_pos > len(prompt_ids):
-            from_pos = 0
-        new_ids = prompt_ids[from_pos:]
-        if not new_ids:                      # identical prompt: redo last token
-            from_pos = max(0, len(prompt_ids) - 1)
-            new_ids = prompt_ids[from_pos:]
-        # over-long fresh prompts: keep the tail, reserving the requested
-        # decode budget (prefix-cached prompts are left alone)
-        if from_pos == 0:
-            cap = max(1, self.max_seq_len - 1 - max_new_tokens)
-            if len(new_ids) > cap:
-                new_ids = new_ids[-cap:]
-        eff_len = min(from_pos + len(new_ids), self.max_seq_len - 1)
-        max_new_tokens = min(max_new_tokens, self.max_seq_len - eff_len - 1)
-        if max_new_tokens < 1:
-            max_new_tokens = 1
+        prompt_ids, new_ids, from_pos, max_new_tokens = self._prep_prompt(
+            prompt, max_new_tokens, from_pos)
         self.temperature = temperature
         if self.is_gpu:
             torch.cuda.synchronize(self.device)
@@ -397,6 +379,92 @@ class LocalEngine:
             "token_ids": rows[0],
             **self.last_metrics,
         }
+
+    def _prep_prompt(self, prompt: Union[str, List[int]],
+                     max_new_tokens: int, from_pos: int):
+        """Shared prompt handling for generate()/generate_stream(): encode,
+        validate the prefix-cache offset, and truncate over-long fresh
+        prompts while reserving the decode budget."""
+        if isinstance(prompt, str):
+            prompt_ids = self.tokenizer.encode(prompt)
+        else:
+            prompt_ids = list(prompt)
+        if from_pos > len(prompt_ids):
+            from_pos = 0
+        new_ids = prompt_ids[from_pos:]
+        if not new_ids:                      # identical prompt: redo last token
+            from_pos = max(0, len(prompt_ids) - 1)
+            new_ids = prompt_ids[from_pos:]
+        # over-long fresh prompts: keep the tail, reserving the requested
+        # decode budget (prefix-cached prompts are left alone)
+        if from_pos == 0:
+            cap = max(1, self.max_seq_len - 1 - max_new_tokens)
+            if len(new_ids) > cap:
+                new_ids = new_ids[-cap:]
+        eff_len = min(from_pos + len(new_ids), self.max_seq_len - 1)
+        max_new_tokens = min(max_new_tokens, self.max_seq_len - eff_len - 1)
+        if max_new_tokens < 1:
+            max_new_tokens = 1
+        return prompt_ids, new_ids, from_pos, max_new_tokens
+
+    def generate_stream(self, prompt: Union[str, List[int]],
+                        max_new_tokens: int = 256,
+                        temperature: float = 0.0, stop_on_eos: bool = True,
+                        from_pos: int = 0, chunk: int = 16):
+        """Streaming generate (batch 1): yields a dict per decoded chunk —
+        {"new_token_ids", "text", "done"} where ``text`` is the cumulative
+        decode (byte tokens can split multi-byte characters, so deltas are
+        only stable on the cumulative string). The final chunk carries the
+        full metrics of generate()."""
+        assert self.B == 1, "streaming is a batch-1 interactive path"
+        prompt_ids, new_ids, from_pos, max_new_tokens = self._prep_prompt(
+            prompt, max_new_tokens, from_pos)
+        self.temperature = temperature
+        if self.is_gpu:
+            torch.cuda.synchronize(self.device)
+        t0 = time.perf_counter()
+        self.prefill(new_ids, from_pos=from_pos)
+        t1 = time.perf_counter()
+        eos = self.tokenizer.eos_id
+        done = 1
+        emitted = 0
+        finished = False
+        while not finished:
+            n_chunk = min(chunk, max_new_tokens - done)
+            if self._graph is not None:
+                for _ in range(n_chunk):
+                    self._graph.replay()
+            else:
+                for _ in range(n_chunk):
+                    self._decode_step()
+            done += n_chunk
+            if self.is_gpu:
+                torch.cuda.synchronize(self.device)
+            row = self.out_tokens[0, :done].tolist()
+            if stop_on_eos and eos in row:
+                row = row[: row.index(eos) + 1]
+                finished = True
+            if done >= max_new_tokens:
+                finished = True
+            new = row[emitted:]
+            emitted = len(row)
+            out = {"new_token_ids": new,
+                   "text": self.tokenizer.decode(row),
+                   "done": finished}
+            if finished:
+                decode_s = time.perf_counter() - t1
+                self.last_metrics = {
+                    "prompt_tokens": len(prompt_ids),
+                    "cached_prefix": from_pos,
+                    "new_tokens": len(row),
+                    "prefill_s": t1 - t0,
+                    "decode_s": decode_s,
+                    "prefill_tok_s": len(new_ids) / max(t1 - t0, 1e-9),
+                    "decode_tok_s": max(len(row) - 1, 0) / max(decode_s, 1e-9),
+                }
+                out["token_ids"] = row
+                out.update(self.last_metrics)
+            yield out
 
     def shutdown(self) -> None:
         self._graph = None

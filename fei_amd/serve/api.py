@@ -34,6 +34,7 @@ class CompletionRequest(BaseModel):
     top_p: float = 1.0
     stop_on_eos: bool = True
     speculative: Optional[bool] = None
+    stream: bool = False
 
 
 class ChatMessage(BaseModel):
@@ -104,6 +105,29 @@ def create_app(engine: Optional[LocalEngine] = None,
     @app.post("/v1/completions")
     def completions(req: CompletionRequest):
         t0 = time.time()
+        if req.stream:
+            import json
+
+            from fastapi.responses import StreamingResponse
+
+            def sse():
+                with lock:
+                    prev = ""
+                    for c in eng.generate_stream(
+                            req.prompt, max_new_tokens=req.max_tokens,
+                            temperature=req.temperature,
+                            stop_on_eos=req.stop_on_eos):
+                        delta = c["text"][len(prev):]
+                        prev = c["text"]
+                        payload = {"object": "text_completion.chunk",
+                                   "model": model_name,
+                                   "choices": [{"index": 0, "text": delta,
+                                                "finish_reason":
+                                                "stop" if c["done"] else None}]}
+                        yield f"data: {json.dumps(payload)}\n\n"
+                yield "data: [DONE]\n\n"
+
+            return StreamingResponse(sse(), media_type="text/event-stream")
         out = _generate(req.prompt, req)
         n_new = len(out["token_ids"])
         return {

@@ -309,3 +309,19 @@ def test_attn_splits_env_sizes_workspace(monkeypatch):
     assert eng.attn_splits == 16
     assert eng.attn_ws[0].shape[2] == 16 and eng.attn_ws[1].shape[2] == 16
     eng.generate([5, 6, 7], max_new_tokens=4)  # CPU path still consistent
+
+
+def test_generate_stream_matches_generate():
+    """Chunked streaming must reassemble to exactly the non-streamed
+    greedy output, and every chunk must be non-empty until done."""
+    from fei_amd.engine.engine import LocalEngine
+    eng = LocalEngine.create("llama3-tiny")
+    plain = eng.generate("stream me a poem", max_new_tokens=20,
+                         stop_on_eos=False)
+    chunks = list(eng.generate_stream("stream me a poem", max_new_tokens=20,
+                                      stop_on_eos=False, chunk=7))
+    ids = [t for c in chunks for t in c["new_token_ids"]]
+    assert ids == plain["token_ids"]
+    assert chunks[-1]["done"] and chunks[-1]["text"] == plain["text"]
+    assert "decode_tok_s" in chunks[-1]
+    assert all(not c["done"] for c in chunks[:-1])

@@ -73,3 +73,18 @@ def test_sessions_endpoints(client):
     assert client.delete(f"/v1/sessions/{sid_a}").json()["closed"] == sid_a
     client.delete(f"/v1/sessions/{sid_b}")
     assert client.get(f"/v1/sessions/{sid_a}").status_code == 404
+
+
+def test_completions_stream_sse(client):
+    with client.stream("POST", "/v1/completions", json={
+            "prompt": "stream abc", "max_tokens": 12, "stream": True,
+            "stop_on_eos": False}) as r:
+        assert r.status_code == 200
+        assert r.headers["content-type"].startswith("text/event-stream")
+        lines = [l for l in r.iter_lines() if l.startswith("data: ")]
+    assert lines[-1] == "data: [DONE]"
+    import json
+    chunks = [json.loads(l[6:]) for l in lines[:-1]]
+    assert chunks and chunks[-1]["choices"][0]["finish_reason"] == "stop"
+    text = "".join(c["choices"][0]["text"] for c in chunks)
+    assert len(text) >= 1
