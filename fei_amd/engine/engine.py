@@ -49,6 +49,8 @@ class LocalEngine:
         attn_splits: int = 32,
         dtype: Optional[torch.dtype] = None,
         weight_quant: Optional[str] = None,    # None | "fp8" (decode path)
+        tokenizer=None,                        # default ByteTokenizer; or
+                                               # SpmTokenizer(model_path)
     ):
         self.spec = spec
         if device is None:
@@ -87,7 +89,11 @@ class LocalEngine:
             (self.is_gpu and not self.tp.is_distributed)
         self.seed = seed
         self.temperature = 0.0       # graph-captured; set before capture
-        self.tokenizer = ByteTokenizer()
+        self.tokenizer = tokenizer or ByteTokenizer()
+        if self.tokenizer.vocab_size > spec.vocab_size:
+            raise ValueError(
+                f"tokenizer vocab {self.tokenizer.vocab_size} exceeds model "
+                f"vocab {spec.vocab_size}")
 
         if self.is_gpu and not ops.kernels_available():
             ops.require_lib()        # fail loudly: no eager fallback on GPU

@@ -325,3 +325,48 @@ def test_generate_stream_matches_generate():
     assert chunks[-1]["done"] and chunks[-1]["text"] == plain["text"]
     assert "decode_tok_s" in chunks[-1]
     assert all(not c["done"] for c in chunks[:-1])
+
+
+def test_spm_tokenizer_roundtrip_and_engine(tmp_path):
+    """Train a tiny SentencePiece model offline and run the engine with it
+    (real-tokenizer support for loadable checkpoints)."""
+    import io
+    import sentencepiece as spm
+    from fei_amd.engine.engine import LocalEngine
+    from fei_amd.engine.tokenizer import SpmTokenizer
+    corpus = ["the quick brown fox jumps over the lazy dog",
+              "def add(a, b): return a + b",
+              "hello world example text for sentencepiece"] * 20
+    model = io.BytesIO()
+    spm.SentencePieceTrainer.train(
+        sentence_iterator=iter(corpus), model_writer=model,
+        vocab_size=80, model_type="bpe")
+    path = tmp_path / "tok.model"
+    path.write_bytes(model.getvalue())
+
+    tok = SpmTokenizer(str(path))
+    ids = tok.encode("the quick brown fox")
+    assert ids[0] == tok.bos_id
+    assert "quick" in tok.decode(ids)
+
+    eng = LocalEngine.create("llama3-tiny", tokenizer=tok)
+    out = eng.generate("hello world", max_new_tokens=6, stop_on_eos=False)
+    assert len(out["token_ids"]) == 6
+    assert isinstance(out["text"], str)
+
+
+def test_spm_vocab_overflow_rejected(tmp_path):
+    import pytest
+    from fei_amd.engine.engine import LocalEngine
+
+    class Fake:
+        vocab_size = 10 ** 9
+        bos_id = eos_id = pad_id = 0
+
+        def encode(self, t, **k):
+            return [0]
+
+        def decode(self, ids):
+            return ""
+    with pytest.raises(ValueError, match="exceeds model vocab"):
+        LocalEngine.create("llama3-tiny", tokenizer=Fake())
