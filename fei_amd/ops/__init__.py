@@ -370,7 +370,11 @@ def mfma_probe(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
 
 # -- decode GEMV dispatch ----------------------------------------------------
 
-_GEMV_NT_MIN_BYTES = 512 * 1024 * 1024   # NT pays only on >L3-size streams (lm_head)
+# NT pays only on >L3-size streams (lm_head) in MICRObenches, where the
+# repeated-run operand is L3-warm; FEI_GEMV_NT_MIN (bytes) overrides for
+# in-loop measurement where every weight read is cache-cold.
+_GEMV_NT_MIN_BYTES = int(os.environ.get("FEI_GEMV_NT_MIN",
+                                        512 * 1024 * 1024))
 
 
 def _gemv_ok(M: int, K: int) -> bool:
