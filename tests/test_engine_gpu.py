@@ -172,3 +172,16 @@ def test_speculative_gpu(engine):
     assert len(out["token_ids"]) == 24
     assert out["spec_blocks"] >= 1
     assert out["spec_tokens_per_block"] >= 1.0
+
+
+def test_generate_stream_gpu(engine):
+    """Streaming (graph-replay chunks) reassembles to the plain greedy
+    output on the hipGraph path."""
+    plain = engine.generate("gpu stream check", max_new_tokens=20,
+                            stop_on_eos=False)
+    chunks = list(engine.generate_stream("gpu stream check",
+                                         max_new_tokens=20,
+                                         stop_on_eos=False, chunk=6))
+    ids = [t for c in chunks for t in c["new_token_ids"]]
+    assert ids == plain["token_ids"]
+    assert chunks[-1]["done"]
