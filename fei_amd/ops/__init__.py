@@ -370,11 +370,13 @@ def mfma_probe(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
 
 # -- decode GEMV dispatch ----------------------------------------------------
 
-# NT pays only on >L3-size streams (lm_head) in MICRObenches, where the
-# repeated-run operand is L3-warm; FEI_GEMV_NT_MIN (bytes) overrides for
-# in-loop measurement where every weight read is cache-cold.
-_GEMV_NT_MIN_BYTES = int(os.environ.get("FEI_GEMV_NT_MIN",
-                                        512 * 1024 * 1024))
+# Nontemporal weight loads for ALL decode GEMVs: in the real decode loop
+# every weight byte is cache-cold (the 15 GB per-step cycle flushes the
+# 256 MB L3), so NT never hurts and avoids polluting L2/L3 for the hot
+# activations — measured 257.2 vs 255.6 tok/s on the 8B headline. The
+# microbench that suggested a 512 MB threshold was L3-warm across runs.
+# FEI_GEMV_NT_MIN (bytes) restores a threshold.
+_GEMV_NT_MIN_BYTES = int(os.environ.get("FEI_GEMV_NT_MIN", 0))
 
 
 def _gemv_ok(M: int, K: int) -> bool:
