@@ -25,6 +25,7 @@ run python bench.py --steps 256 --warmup 32 --batch 8             # 8 sessions
 run python bench.py --steps 256 --warmup 32 --weights-fp8         # fp8 mode
 run python bench.py --model llama3-70b --prompt-len 256 --steps 48 --warmup 8
 run python bench.py --model llama3-1b --steps 256 --warmup 32
+run python scripts/bench_spec.py llama3-8b 256
 run python scripts/bench_gemv.py
 run python scripts/bench_agent_turn.py
 run env CORPUS_N=1000000 python scripts/bench_memdir.py
