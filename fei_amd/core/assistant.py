@@ -116,6 +116,72 @@ class ConversationManager:
     def clear(self) -> None:
         self.messages.clear()
 
+    # -- context compaction ---------------------------------------------------
+
+    @staticmethod
+    def _msg_chars(msg: Dict[str, Any]) -> int:
+        content = msg.get("content")
+        if isinstance(content, str):
+            return len(content)
+        total = 0
+        for b in content or []:
+            if isinstance(b, dict):
+                total += len(str(b.get("text", ""))) + \
+                    len(str(b.get("content", ""))) + len(str(b.get("input", "")))
+        return total
+
+    def size_chars(self) -> int:
+        return sum(self._msg_chars(m) for m in self.messages)
+
+    def compact(self, keep_last: int = 4,
+                summarizer=None) -> Optional[str]:
+        """Replace everything but the last ``keep_last`` messages with one
+        summary message (the local engine has a HARD context window; the
+        reference leaned on the remote provider's). ``summarizer(text) ->
+        str`` produces the summary — pass the model itself, or leave None
+        for a head/tail excerpt. Keeps the tail boundary on a user message
+        so tool_use/tool_result pairs are never split. Returns the summary
+        or None if nothing was compacted."""
+        if len(self.messages) <= keep_last + 1:
+            return None
+        cut = len(self.messages) - keep_last
+        # never split an assistant(tool_use) from its user(tool_result)
+        while cut < len(self.messages) and \
+                isinstance(self.messages[cut].get("content"), list) and \
+                any(isinstance(b, dict) and b.get("type") == "tool_result"
+                    for b in self.messages[cut]["content"]):
+            cut += 1
+        if cut <= 0 or cut >= len(self.messages):
+            return None
+        old = self.messages[:cut]
+        flat = []
+        for m in old:
+            c = m.get("content")
+            if isinstance(c, str):
+                flat.append(f"{m['role']}: {c}")
+            else:
+                for b in c or []:
+                    if isinstance(b, dict) and b.get("type") == "text":
+                        flat.append(f"{m['role']}: {b['text']}")
+                    elif isinstance(b, dict) and b.get("type") == "tool_use":
+                        flat.append(f"[tool {b.get('name')}]")
+                    elif isinstance(b, dict) and b.get("type") == "tool_result":
+                        flat.append(f"[result {str(b.get('content'))[:200]}]")
+        text = "\n".join(flat)
+        if summarizer is not None:
+            summary = summarizer(text)
+        else:
+            summary = (text[:1500] + "\n...\n" + text[-500:]
+                       if len(text) > 2000 else text)
+        self.messages = (
+            [{"role": "user",
+              "content": f"[conversation summary of {cut} earlier messages]\n"
+                         f"{summary}"},
+             {"role": "assistant",
+              "content": "Understood — continuing from that summary."}]
+            + self.messages[cut:])
+        return summary
+
     def last_text(self) -> str:
         for msg in reversed(self.messages):
             if msg["role"] == "assistant":
@@ -169,6 +235,11 @@ class Assistant:
                 pass
         self.max_tokens = self.config.get_typed("llm.max_tokens", 4000)
         self.temperature = self.config.get_typed("llm.temperature", 0.0)
+        # auto-compaction: local engines have a HARD context window, so
+        # past this many conversation chars the older turns are folded
+        # into a summary message (0 = off; llm.auto_compact_chars)
+        self.auto_compact_chars = int(
+            self.config.get_typed("llm.auto_compact_chars", 0) or 0)
         # per-turn metrics (prefill/decode tok/s, tool latency) — SURVEY §5
         self.turn_metrics: List[Dict[str, Any]] = []
 
@@ -221,9 +292,18 @@ class Assistant:
 
     # -- public API ----------------------------------------------------------
 
+    def maybe_compact(self) -> Optional[str]:
+        """Fold older turns into a summary once the conversation exceeds
+        the configured budget (no-op when auto_compact_chars is 0)."""
+        if self.auto_compact_chars and \
+                self.conversation.size_chars() > self.auto_compact_chars:
+            return self.conversation.compact()
+        return None
+
     def chat(self, message: str, system_prompt: Optional[str] = None) -> str:
         """One turn with at most ONE tool round (reference: assistant.py:440-489).
         Tool calls in the continuation are recorded but not executed."""
+        self.maybe_compact()
         self.conversation.add_user_message(message)
         response = self._one_round(system_prompt)
         self.conversation.add_assistant_message(response.content, response.tool_calls)
@@ -236,6 +316,7 @@ class Assistant:
             max_tool_rounds: int = 8) -> str:
         """Iterating turn: keep executing tool rounds until the model
         answers without tool calls (or the round cap is hit)."""
+        self.maybe_compact()
         self.conversation.add_user_message(message)
         t0 = time.perf_counter()
         response = self._send(system_prompt)

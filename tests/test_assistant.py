@@ -159,3 +159,47 @@ def test_async_facades(cfg, tmp_path):
     r1, r2 = asyncio.run(run())
     assert "hello async" in r1
     assert "found z" in r2
+
+
+def test_conversation_compact_preserves_tail_and_pairs():
+    c = ConversationManager()
+    for i in range(6):
+        c.add_user_message(f"question {i} " + "x" * 50)
+        c.add_assistant_message(f"answer {i}")
+    c.add_assistant_message("", [{"id": "t1", "name": "Tool", "input": {}}])
+    c.add_tool_results([{"tool_use_id": "t1", "content": "tool says 42"}])
+    c.add_assistant_message("final answer")
+    n_before = len(c.messages)
+    summary = c.compact(keep_last=3)
+    assert summary is not None
+    assert len(c.messages) < n_before
+    # tail kept verbatim; tool_use/tool_result pair not split
+    roles = [m["role"] for m in c.messages]
+    flat = str(c.messages)
+    assert "final answer" in flat
+    assert ("tool_result" in flat) == ("tool_use" in flat)
+    assert c.messages[0]["role"] == "user"
+    assert "conversation summary" in c.messages[0]["content"]
+
+
+def test_assistant_auto_compact(cfg):
+    a = Assistant(config=cfg, provider="stub")
+    a.auto_compact_chars = 400
+    for i in range(8):
+        a.chat(f"msg {i} " + "y" * 120)
+    flat = str(a.conversation.messages)
+    assert "conversation summary" in flat
+    assert len(a.conversation.messages) < 16     # compaction actually fired
+    # the assistant still answers after compaction
+    out = a.chat("still alive?")
+    assert "still alive?" in out
+
+
+def test_compact_with_model_summarizer():
+    c = ConversationManager()
+    for i in range(8):
+        c.add_user_message(f"fact {i}: the sky is blue")
+        c.add_assistant_message("noted")
+    s = c.compact(keep_last=2, summarizer=lambda text: "SUMMARY(%d chars)" % len(text))
+    assert s.startswith("SUMMARY(")
+    assert "SUMMARY(" in c.messages[0]["content"]
