@@ -199,7 +199,18 @@ class MCPClient:
             raise MCPError(f"unknown MCP server: {name}")
         params = params or {}
         if self.servers[name]["type"] == "stdio":
-            return self._call_stdio_service(name, method, params)
+            try:
+                return self._call_stdio_service(name, method, params)
+            except MCPError as e:
+                # one restart on a dead subprocess (crash between calls,
+                # OOM-killed, etc.) — the reference re-raised and lost the
+                # turn; a fresh process usually recovers stateless servers
+                if "died" not in str(e) and "exited" not in str(e):
+                    raise
+                logger.warning("restarting stdio MCP server %s after: %s",
+                               name, e)
+                self.procs.stop(name)
+                return self._call_stdio_service(name, method, params)
         return self._call_http_service(name, method, params)
 
     def stop_server(self, name: str) -> bool:

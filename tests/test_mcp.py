@@ -99,3 +99,30 @@ def test_memory_service_facade(client):
     mgr.memory.server = "echo"
     out = mgr.memory.search_nodes("hello")
     assert out["params"]["query"] == "hello"
+
+
+def test_stdio_server_restart_after_crash(tmp_path):
+    """A stdio server that dies mid-conversation is restarted once and the
+    call retried (stateless-server recovery)."""
+    import sys
+    import textwrap
+    from fei_amd.core.mcp import MCPClient, ProcessManager
+    flag = tmp_path / "started_once"
+    server = tmp_path / "crashy.py"
+    server.write_text(textwrap.dedent(f"""
+        import json, os, sys
+        flag = {str(flag)!r}
+        if not os.path.exists(flag):
+            open(flag, "w").close()
+            sys.exit(1)            # crash on first launch
+        for line in sys.stdin:
+            req = json.loads(line)
+            sys.stdout.write(json.dumps({{"jsonrpc": "2.0", "id": req["id"],
+                                          "result": {{"ok": True}}}}) + "\\n")
+            sys.stdout.flush()
+    """))
+    client = MCPClient(process_manager=ProcessManager())
+    client.add_server("crashy", command=[sys.executable, str(server)])
+    out = client.call_service("crashy", "ping", {})
+    assert out == {"ok": True}
+    client.stop_server("crashy")
