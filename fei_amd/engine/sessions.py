@@ -53,6 +53,18 @@ class PagedSessionManager:
             device=m.device, dtype=m.dtype, mem_fraction=mem_fraction)
         self.sessions: Dict[int, Session] = {}
         self.eos = engine.tokenizer.eos_id
+        self._ws: Dict[int, tuple] = {}     # split-K workspace per batch size
+
+    def _workspace(self, B: int):
+        if B not in self._ws:
+            m = self.model
+            sp = self.engine.attn_splits
+            self._ws[B] = (
+                torch.zeros(B, m.hq_l, sp, m.D, dtype=torch.float32,
+                            device=m.device),
+                torch.zeros(B, m.hq_l, sp, 2, dtype=torch.float32,
+                            device=m.device))
+        return self._ws[B]
 
     # -- admission -----------------------------------------------------------
 
@@ -178,7 +190,8 @@ class PagedSessionManager:
                 vp[blk_of[i], :, off_of[i], :] = v[i]
             att = ops.attn_decode_paged(q_r, kp, vp, table, pos,
                                         splits=self.engine.attn_splits,
-                                        scale=scale)
+                                        scale=scale,
+                                        workspace=self._workspace(B))
             o = ops.linear_decode(att.reshape(B, -1), lw.wo)
             x, h = ops.fused_add_rmsnorm(o, h, lw.norm_mlp, s.norm_eps)
             act = ops.gemv_swiglu(x, lw.wgu)
