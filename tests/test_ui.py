@@ -116,3 +116,14 @@ def test_doctor(home, capsys):
     out = capsys.readouterr().out
     assert rc == 0
     assert "fei_amd" in out and "torch" in out and "kernel lib" in out
+
+
+def test_cli_local_engine_end_to_end(home, capsys):
+    """Full local path: CLI -> Assistant -> LocalBackend -> LocalEngine
+    (tiny model on CPU; output is random-weights text but the whole agent
+    stack runs for real)."""
+    from fei_amd.ui.cli import main
+    rc = main(["--provider", "local", "--model", "llama3-tiny",
+               "--no-memory", "-m", "hello"])
+    assert rc == 0
+    assert capsys.readouterr().out.strip()
