@@ -231,14 +231,28 @@ class LocalBackend(Backend):
         # one, so only the delta needs prefilling (the engine's KV caches
         # already hold the prefix + its generated tokens are NOT part of
         # the rendered prompt — cache only the prompt prefix).
+        # longest common prefix (not all-or-nothing: generated specials
+        # may not re-render byte-identically, but everything before them
+        # still reuses the cache)
         n_common = 0
-        if self._cached_ids and len(ids) > len(self._cached_ids) and                 ids[:len(self._cached_ids)] == self._cached_ids:
-            n_common = len(self._cached_ids)
+        limit = min(len(self._cached_ids), len(ids) - 1)
+        while n_common < limit and ids[n_common] == self._cached_ids[n_common]:
+            n_common += 1
         out = self.engine.generate(ids, max_new_tokens=max_tokens,
                                    temperature=temperature,
                                    stop_on_eos=self.stop_on_eos,
                                    from_pos=n_common)
-        self._cached_ids = ids
+        # extend the cache through the GENERATED tokens too: the response
+        # is already in the KV caches, and the next turn's rendered prompt
+        # repeats it byte-for-byte (byte tokenizer: decode->render->encode
+        # is identity for non-special tokens). If the rendering ever
+        # differs (specials got dropped), the next turn's prefix compare
+        # simply fails and we re-prefill — correctness never depends on it.
+        gen = list(out["token_ids"])
+        eos = self.engine.tokenizer.eos_id
+        if gen and gen[-1] == eos:
+            gen = gen[:-1]
+        self._cached_ids = ids + gen
         text = out["text"]
         calls = extract_tool_call_blocks(text)
         return BackendResponse(

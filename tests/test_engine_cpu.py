@@ -370,3 +370,22 @@ def test_spm_vocab_overflow_rejected(tmp_path):
             return ""
     with pytest.raises(ValueError, match="exceeds model vocab"):
         LocalEngine.create("llama3-tiny", tokenizer=Fake())
+
+
+def test_backend_prefix_cache_spans_generated_tokens(tmp_path):
+    """Turn 2 must reuse the KV of turn 1's RESPONSE, not just its prompt
+    (cached_prefix covers prompt + generated tokens)."""
+    from fei_amd.core.backends import LocalBackend
+    from fei_amd.engine.engine import LocalEngine
+    be = LocalBackend(engine=LocalEngine.create("llama3-tiny"))
+    msgs = [{"role": "user", "content": "first question"}]
+    r1 = be.complete(msgs, max_tokens=8)
+    n_cache_after_1 = len(be._cached_ids)
+    msgs = msgs + [{"role": "assistant", "content": r1.content},
+                   {"role": "user", "content": "second question"}]
+    be.complete(msgs, max_tokens=8)
+    cached = be.engine.last_metrics["cached_prefix"]
+    # the reused prefix must reach past the turn-1 prompt into its response
+    prompt1_len = n_cache_after_1 - 8  # at most; response added up to 8
+    assert cached >= prompt1_len
+    assert cached > 0
