@@ -54,6 +54,7 @@ class ChatRequest(BaseModel):
     temperature: float = 0.0
     top_k: int = 0
     top_p: float = 1.0
+    stream: bool = False
 
 
 def flatten_chat(messages: List[ChatMessage]) -> str:
@@ -147,6 +148,31 @@ def create_app(engine: Optional[LocalEngine] = None,
     @app.post("/v1/chat/completions")
     def chat(req: ChatRequest):
         t0 = time.time()
+        if req.stream:
+            import json
+
+            from fastapi.responses import StreamingResponse
+
+            prompt = flatten_chat(req.messages)
+
+            def sse():
+                with lock:
+                    prev = ""
+                    for c in eng.generate_stream(
+                            prompt, max_new_tokens=req.max_tokens,
+                            temperature=req.temperature):
+                        delta = c["text"][len(prev):]
+                        prev = c["text"]
+                        payload = {"object": "chat.completion.chunk",
+                                   "model": model_name,
+                                   "choices": [{"index": 0,
+                                                "delta": {"content": delta},
+                                                "finish_reason":
+                                                "stop" if c["done"] else None}]}
+                        yield f"data: {json.dumps(payload)}\n\n"
+                yield "data: [DONE]\n\n"
+
+            return StreamingResponse(sse(), media_type="text/event-stream")
         out = _generate(flatten_chat(req.messages), req)
         return {
             "id": f"chatcmpl-{int(t0 * 1000)}",

@@ -88,3 +88,12 @@ def test_completions_stream_sse(client):
     assert chunks and chunks[-1]["choices"][0]["finish_reason"] == "stop"
     text = "".join(c["choices"][0]["text"] for c in chunks)
     assert len(text) >= 1
+
+
+def test_chat_stream_sse(client):
+    with client.stream("POST", "/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": "hi"}],
+            "max_tokens": 10, "stream": True}) as r:
+        assert r.status_code == 200
+        lines = [l for l in r.iter_lines() if l.startswith("data: ")]
+    assert lines[-1] == "data: [DONE]"
