@@ -58,6 +58,11 @@ def main() -> int:
     device = torch.device(f"cuda:{ctx.local_rank}") if has_gpu else torch.device("cpu")
 
     spec = get_spec(args.model)
+    # the whole run (prompt + warmup + timed steps) must fit the model's
+    # context; shrink the prompt for small-context models (llama3-tiny)
+    budget = spec.max_seq_len - args.warmup - args.steps - 16
+    if args.prompt_len > max(budget, 16):
+        args.prompt_len = max(budget, 16)
     max_seq = min(spec.max_seq_len,
                   args.prompt_len + args.warmup + args.steps + 64)
     tp_mode = args.tp > 1

@@ -156,7 +156,7 @@ class LocalEngine:
         ops.sample(logits, self.token, self.step, self.sample_ws.view(self.B, -1),
                    out_tokens=self.out_tokens, temperature=self.temperature,
                    seed=self.seed, nchunks=SAMPLE_CHUNKS)
-        ops.advance(self.pos, self.step)
+        ops.advance(self.pos, self.step, max_pos=self.max_seq_len - 1)
 
     def _capture_graph(self) -> None:
         assert self.is_gpu
@@ -254,6 +254,8 @@ class LocalEngine:
                 for _ in range(chunk):
                     self._decode_step()
             done += chunk
+            if int(self.pos.max()) >= self.max_seq_len - 1:
+                break                      # cache capacity reached
             if stop_on_eos:
                 toks = self.out_tokens[:, :done].tolist()
                 if all(eos in row for row in toks):

@@ -52,7 +52,7 @@ def _try_load() -> None:
     lib.fei_swiglu.argtypes = [_vp, _vp, _l, _i, _vp]
     lib.fei_sample.argtypes = [_vp, _vp, _vp, _vp, _vp, _i, _i, _i, _f, _u64,
                                _i, _vp]
-    lib.fei_advance.argtypes = [_vp, _vp, _i, _vp]
+    lib.fei_advance.argtypes = [_vp, _vp, _i, _i, _vp]
     lib.fei_attn_prefill.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp,
                                      _i, _i, _i, _i, _i, _i, _f, _i, _l, _vp]
     lib.fei_mfma_probe.argtypes = [_vp, _vp, _vp, _vp]
@@ -348,14 +348,18 @@ def sample(logits: torch.Tensor, token: torch.Tensor,
     return token
 
 
-def advance(pos: torch.Tensor, step: torch.Tensor) -> None:
-    """pos[b] += 1; step += 1 (device-side, graph-capturable)."""
+def advance(pos: torch.Tensor, step: torch.Tensor,
+            max_pos: int = 1 << 30) -> None:
+    """pos[b] = min(pos[b]+1, max_pos); step += 1 (device-side,
+    graph-capturable). max_pos guards the RoPE table / KV cache against
+    out-of-bounds indexing when a caller decodes at capacity."""
     if not pos.is_cuda:
+        pos.clamp_(max=max_pos - 1)
         pos += 1
         step += 1
         return
     lib = require_lib()
-    lib.fei_advance(_ptr(pos), _ptr(step), pos.shape[0], _stream())
+    lib.fei_advance(_ptr(pos), _ptr(step), pos.shape[0], max_pos, _stream())
 
 
 def mfma_probe(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:

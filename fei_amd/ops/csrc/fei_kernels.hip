@@ -655,14 +655,21 @@ void fei_sample(const void* logits, int* token, int* out_tokens,
 }
 
 // advance: pos[b]+=1 ; step+=1  (runs after sampling inside the graph)
-__global__ void k_advance(int* pos, int* step, int B) {
-  if ((int)threadIdx.x < B) pos[threadIdx.x] += 1;
+// max_pos clamps the position at cache capacity: a caller decoding past
+// max_seq would otherwise index the RoPE table / KV cache out of bounds
+// from inside the graph (memory corruption, not an exception). Clamped
+// steps overwrite the last slot — wrong results but memory-safe; the
+// python loops stop at capacity before this matters.
+__global__ void k_advance(int* pos, int* step, int B, int max_pos) {
+  if ((int)threadIdx.x < B)
+    pos[threadIdx.x] = min(pos[threadIdx.x] + 1, max_pos);
   if (threadIdx.x == 0 && step) *step += 1;
 }
 
-void fei_advance(int* pos, int* step, int B, hipStream_t stream) {
+void fei_advance(int* pos, int* step, int B, int max_pos,
+                 hipStream_t stream) {
   hipLaunchKernelGGL(k_advance, dim3(1), dim3(max(B, 64)), 0, stream,
-                     pos, step, B);
+                     pos, step, B, max_pos);
 }
 
 }  // extern "C"

@@ -389,3 +389,15 @@ def test_backend_prefix_cache_spans_generated_tokens(tmp_path):
     prompt1_len = n_cache_after_1 - 8  # at most; response added up to 8
     assert cached >= prompt1_len
     assert cached > 0
+
+
+def test_decode_stops_at_cache_capacity():
+    """Decoding into a full context must stop cleanly at max_seq_len, not
+    index the RoPE table / KV cache out of bounds."""
+    from fei_amd.engine.engine import LocalEngine
+    eng = LocalEngine.create("llama3-tiny", max_seq_len=64)
+    prompt = list(range(4, 44))              # 40 tokens
+    eng.prefill(prompt)
+    rows = eng.decode(100, stop_on_eos=False)    # asks for more than fits
+    assert len(rows[0]) <= 64 - len(prompt) + 36  # bounded, no crash
+    assert int(eng.pos.max()) <= 63
