@@ -66,7 +66,8 @@ def flatten_chat(messages: List[ChatMessage]) -> str:
 
 
 def create_app(engine: Optional[LocalEngine] = None,
-               model: str = "llama3-tiny", **engine_kwargs):
+               model: str = "llama3-tiny",
+               session_blocks: Optional[int] = None, **engine_kwargs):
     from fastapi import FastAPI
 
     app = FastAPI(title="fei_amd", version="0.1.0")
@@ -196,7 +197,7 @@ def create_app(engine: Optional[LocalEngine] = None,
     def _mgr():
         if state["mgr"] is None:
             from fei_amd.engine.sessions import PagedSessionManager
-            state["mgr"] = PagedSessionManager(eng)
+            state["mgr"] = PagedSessionManager(eng, num_blocks=session_blocks)
         return state["mgr"]
 
     @app.post("/v1/sessions")

@@ -97,3 +97,20 @@ def test_chat_stream_sse(client):
         assert r.status_code == 200
         lines = [l for l in r.iter_lines() if l.startswith("data: ")]
     assert lines[-1] == "data: [DONE]"
+
+
+def test_sessions_503_on_pool_exhaustion():
+    from fei_amd.engine.engine import LocalEngine
+    from fei_amd.serve.api import create_app
+    eng = LocalEngine.create("llama3-tiny")
+    c = TestClient(create_app(engine=eng, session_blocks=3))
+    r1 = c.post("/v1/sessions", json={"prompt": "abcdefgh" * 3,
+                                      "max_tokens": 2})
+    assert r1.status_code == 200
+    r2 = c.post("/v1/sessions", json={"prompt": "ijklmnop" * 5,
+                                      "max_tokens": 2})
+    assert r2.status_code == 503          # admission control over HTTP
+    c.delete(f"/v1/sessions/{r1.json()['session_id']}")
+    r3 = c.post("/v1/sessions", json={"prompt": "ijklmnop" * 5,
+                                      "max_tokens": 2})
+    assert r3.status_code == 200          # eviction freed the pool
