@@ -201,7 +201,7 @@ REPO_SUMMARY_TOOL: Dict[str, Any] = {
 }
 
 REPO_DEPS_TOOL: Dict[str, Any] = {
-    "name": "RepoDeps",
+    "name": "RepoDependencies",   # reference name (definitions.py RepoDependencies)
     "description": "Extract module-level dependencies (imports) of a repository, aggregated per file.",
     "input_schema": {
         "type": "object",

@@ -76,5 +76,5 @@ def test_code_tools_registered():
     assert len(reg.list_tools()) == 14
     for name in ["GlobTool", "GrepTool", "View", "Edit", "Replace", "LS",
                  "RegexEdit", "BatchGlob", "FindInFiles", "SmartSearch",
-                 "RepoMap", "RepoSummary", "RepoDeps", "Shell"]:
+                 "RepoMap", "RepoSummary", "RepoDependencies", "Shell"]:
         assert name in reg.list_tools()
