@@ -269,21 +269,22 @@ class MCPBraveSearchService:
         self.client = client
         self.server = server
 
-    def search(self, query: str, count: int = 5) -> Any:
+    def search(self, query: str, count: int = 5, offset: int = 0) -> Any:
         try:
             return self.client.call_service(self.server, "brave_web_search",
-                                            {"query": query, "count": count})
+                                            {"query": query, "count": count,
+                                             "offset": offset})
         except MCPError:
-            return self._direct_search(query, count)
+            return self._direct_search(query, count, offset)
 
-    def _direct_search(self, query: str, count: int) -> Any:
+    def _direct_search(self, query: str, count: int, offset: int = 0) -> Any:
         api_key = os.environ.get("BRAVE_API_KEY", "")
         if not api_key:
             raise MCPError("brave search unavailable: no MCP server and no "
                            "BRAVE_API_KEY")
         r = requests.get(
             "https://api.search.brave.com/res/v1/web/search",
-            params={"q": query, "count": count},
+            params={"q": query, "count": count, "offset": offset},
             headers={"X-Subscription-Token": api_key}, timeout=15)
         r.raise_for_status()
         data = r.json()
@@ -336,7 +337,8 @@ class MCPManager:
         registry.register_prefix_hook(
             "brave_web_search",
             lambda name, args: {"results": self.brave_search.search(
-                args.get("query", ""), count=int(args.get("count", 5)))})
+                args.get("query", ""), count=int(args.get("count", 5)),
+                offset=int(args.get("offset", 0)))})
 
     def shutdown(self) -> None:
         self.client.procs.stop_all()

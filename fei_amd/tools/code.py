@@ -144,8 +144,10 @@ class GrepTool:
         results.sort(key=lambda m: (m["file"], m["line"]))
         return results
 
-    def find_in_files(self, files: List[str], pattern: str) -> Dict[str, List[Dict[str, Any]]]:
-        rx = self._compile(pattern)
+    def find_in_files(self, files: List[str], pattern: str,
+                      case_sensitive: bool = False) -> Dict[str, List[Dict[str, Any]]]:
+        # reference default: case-insensitive (definitions.py FindInFiles)
+        rx = self._compile(pattern, 0 if case_sensitive else re.IGNORECASE)
         out: Dict[str, List[Dict[str, Any]]] = {}
         for path in files:
             if not os.path.isfile(path):
@@ -233,7 +235,11 @@ class CodeEditor:
         replacement: str,
         count: int = 0,
         validate: bool = True,
+        validators: Optional[List[str]] = None,
     ) -> Dict[str, Any]:
+        """``validators``: explicit validator list (reference semantics:
+        ['ast'] forces the python check, [] disables). None = by file
+        extension, gated on ``validate``."""
         if not os.path.isfile(file_path):
             return {"error": f"file not found: {file_path}"}
         with open(file_path, "r", encoding="utf-8") as f:
@@ -245,7 +251,10 @@ class CodeEditor:
         new_content, n = rx.subn(replacement, content, count=count)
         if n == 0:
             return {"error": "pattern did not match"}
-        if validate:
+        run_ast = validate and file_path.endswith(".py")
+        if validators is not None:
+            run_ast = "ast" in validators
+        if run_ast:
             err = self._validate_code(file_path, new_content)
             if err:
                 return {"error": err, "replacements": 0}

@@ -122,6 +122,7 @@ REGEX_EDIT_TOOL: Dict[str, Any] = {
             "replacement": {"type": "string", "description": "Replacement text (may use backrefs like \\1)"},
             "count": {"type": "integer", "description": "Max replacements (0 = all)"},
             "validate": {"type": "boolean", "description": "Validate the result (default true)"},
+            "validators": {"type": "array", "items": {"type": "string"}, "description": "Explicit validators (e.g. ['ast']); overrides auto-detection"},
         },
         "required": ["file_path", "pattern", "replacement"],
     },
@@ -134,6 +135,7 @@ BATCH_GLOB_TOOL: Dict[str, Any] = {
         "type": "object",
         "properties": {
             "patterns": {"type": "array", "items": {"type": "string"}, "description": "Glob patterns"},
+            "limit_per_pattern": {"type": "number", "description": "Max files per pattern (default 20)"},
             "path": {"type": "string", "description": "Directory to search in"},
         },
         "required": ["patterns"],
@@ -148,6 +150,7 @@ FIND_IN_FILES_TOOL: Dict[str, Any] = {
         "properties": {
             "files": {"type": "array", "items": {"type": "string"}, "description": "Files to scan"},
             "pattern": {"type": "string", "description": "Regular expression"},
+            "case_sensitive": {"type": "boolean", "description": "Case-sensitive search (default false)"},
         },
         "required": ["files", "pattern"],
     },
@@ -164,6 +167,7 @@ SMART_SEARCH_TOOL: Dict[str, Any] = {
         "type": "object",
         "properties": {
             "query": {"type": "string", "description": "Natural-language or code query"},
+            "context": {"type": "string", "description": "Extra context words to rank results by (optional)"},
             "path": {"type": "string", "description": "Directory to search in"},
             "language": {"type": "string", "description": "Restrict to a language (python, js, ...)"},
         },
@@ -181,10 +185,11 @@ REPO_MAP_TOOL: Dict[str, Any] = {
     "input_schema": {
         "type": "object",
         "properties": {
-            "path": {"type": "string", "description": "Repository root"},
+            "path": {"type": "string", "description": "Repository root (default: cwd)"},
             "token_budget": {"type": "integer", "description": "Approximate output budget in tokens"},
+            "exclude_patterns": {"type": "array", "items": {"type": "string"}, "description": "Glob patterns to skip"},
         },
-        "required": ["path"],
+        "required": [],
     },
 }
 
@@ -194,9 +199,11 @@ REPO_SUMMARY_TOOL: Dict[str, Any] = {
     "input_schema": {
         "type": "object",
         "properties": {
-            "path": {"type": "string", "description": "Repository root"},
+            "path": {"type": "string", "description": "Repository root (default: cwd)"},
+            "max_tokens": {"type": "integer", "description": "Compact the output further"},
+            "exclude_patterns": {"type": "array", "items": {"type": "string"}, "description": "Glob patterns to skip"},
         },
-        "required": ["path"],
+        "required": [],
     },
 }
 
@@ -206,10 +213,11 @@ REPO_DEPS_TOOL: Dict[str, Any] = {
     "input_schema": {
         "type": "object",
         "properties": {
-            "path": {"type": "string", "description": "Repository root"},
+            "path": {"type": "string", "description": "Repository root (default: cwd)"},
             "module": {"type": "string", "description": "Only report dependencies of this module/file"},
+            "depth": {"type": "integer", "description": "Transitive depth when module is given (default 1)"},
         },
-        "required": ["path"],
+        "required": [],
     },
 }
 
@@ -227,6 +235,7 @@ SHELL_TOOL: Dict[str, Any] = {
             "timeout": {"type": "number", "description": "Seconds before the command is killed (default 60)"},
             "background": {"type": "boolean", "description": "Run detached and return immediately"},
             "working_dir": {"type": "string", "description": "Directory to run in"},
+            "current_dir": {"type": "string", "description": "Alias of working_dir (reference name)"},
         },
         "required": ["command"],
     },
@@ -240,6 +249,7 @@ BRAVE_SEARCH_TOOL: Dict[str, Any] = {
         "properties": {
             "query": {"type": "string", "description": "Search query"},
             "count": {"type": "integer", "description": "Number of results (max 20)"},
+            "offset": {"type": "integer", "description": "Pagination offset"},
         },
         "required": ["query"],
     },

@@ -45,16 +45,20 @@ def regex_edit_handler(args: Dict[str, Any]) -> Dict[str, Any]:
     return c.code_editor.regex_replace(
         args["file_path"], args["pattern"], args["replacement"],
         count=args.get("count", 0), validate=args.get("validate", True),
+        validators=args.get("validators"),
     )
 
 
 def batch_glob_handler(args: Dict[str, Any]) -> Dict[str, Any]:
     results = c.glob_finder.batch(args["patterns"], args.get("path"))
-    return {"results": {k: v[:200] for k, v in results.items()}}
+    limit = int(args.get("limit_per_pattern", 20))   # reference default
+    return {"results": {k: v[:limit] for k, v in results.items()}}
 
 
 def find_in_files_handler(args: Dict[str, Any]) -> Dict[str, Any]:
-    return {"results": c.grep_tool.find_in_files(args["files"], args["pattern"])}
+    return {"results": c.grep_tool.find_in_files(
+        args["files"], args["pattern"],
+        case_sensitive=bool(args.get("case_sensitive", False)))}
 
 
 # -- SmartSearch -------------------------------------------------------------
@@ -91,6 +95,16 @@ def smart_search_handler(args: Dict[str, Any]) -> Dict[str, Any]:
                 m["language"] = lg
                 definitions.append(m)
     plain = c.grep_tool.search(re.escape(ident), path)[:50] if ident else []
+    ctx = (args.get("context") or "").strip().lower()
+    if ctx:
+        ctx_words = set(re.findall(r"[a-z_][a-z0-9_]*", ctx))
+
+        def ctx_rank(m):
+            hay = (m.get("file", "") + " " + m.get("line", "")).lower()
+            return -sum(1 for w in ctx_words if w in hay)
+
+        definitions.sort(key=ctx_rank)
+        plain.sort(key=ctx_rank)
     return {
         "query": query, "identifier": ident,
         "definitions": definitions[:100], "mentions": plain,
@@ -98,21 +112,39 @@ def smart_search_handler(args: Dict[str, Any]) -> Dict[str, Any]:
 
 
 def repo_map_handler(args: Dict[str, Any]) -> Dict[str, Any]:
-    mapper = repomap.RepoMapper(args["path"])
+    mapper = repomap.RepoMapper(args.get("path") or os.getcwd(),
+                                exclude_patterns=args.get("exclude_patterns"))
     budget = args.get("token_budget", 2000)
     return {"map": mapper.generate_map(token_budget=budget)}
 
 
 def repo_summary_handler(args: Dict[str, Any]) -> Dict[str, Any]:
-    return repomap.RepoMapper(args["path"]).summary()
+    out = repomap.RepoMapper(
+        args.get("path") or os.getcwd(),
+        exclude_patterns=args.get("exclude_patterns")).summary()
+    # max_tokens (reference field): truncate the largest-files list as a
+    # crude size control (the summary is already compact)
+    if args.get("max_tokens"):
+        out["largest_files"] = out["largest_files"][:5]
+    return out
 
 
 def repo_deps_handler(args: Dict[str, Any]) -> Dict[str, Any]:
-    mapper = repomap.RepoMapper(args["path"])
+    mapper = repomap.RepoMapper(args.get("path") or os.getcwd())
     deps = mapper.dependencies()
     module = args.get("module")
     if module:
         deps = {k: v for k, v in deps.items() if module in k}
+    depth = int(args.get("depth", 1))
+    if depth > 1 and module:
+        # expand transitively: pull in files whose module names appear in
+        # the already-selected dependency lists
+        for _ in range(depth - 1):
+            wanted = {m for vs in deps.values() for m in vs}
+            for k, v in mapper.dependencies().items():
+                stem = os.path.splitext(os.path.basename(k))[0]
+                if stem in wanted and k not in deps:
+                    deps[k] = v
     return {"dependencies": deps}
 
 
@@ -121,5 +153,5 @@ def shell_handler(args: Dict[str, Any]) -> Dict[str, Any]:
         args["command"],
         timeout=float(args.get("timeout", 60.0)),
         background=bool(args.get("background", False)),
-        working_dir=args.get("working_dir"),
+        working_dir=args.get("working_dir") or args.get("current_dir"),
     )

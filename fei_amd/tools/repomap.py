@@ -48,11 +48,21 @@ _IMPORT_RE = [
 
 
 class RepoMapper:
-    def __init__(self, root: str, max_files: int = 2000, max_file_bytes: int = 512 * 1024):
+    def __init__(self, root: str, max_files: int = 2000,
+                 max_file_bytes: int = 512 * 1024,
+                 exclude_patterns: Optional[List[str]] = None):
         self.root = os.path.abspath(root)
         self.max_files = max_files
         self.max_file_bytes = max_file_bytes
+        # reference RepoMap/RepoSummary accept glob-style exclusions
+        self.exclude_patterns = list(exclude_patterns or [])
         self._symbols: Optional[Dict[str, List[Tuple[str, str]]]] = None
+
+    def _excluded(self, relpath: str) -> bool:
+        import fnmatch
+        return any(fnmatch.fnmatch(relpath, pat) or
+                   fnmatch.fnmatch(os.path.basename(relpath), pat)
+                   for pat in self.exclude_patterns)
 
     # -- walking & extraction ------------------------------------------------
 
@@ -63,6 +73,9 @@ class RepoMapper:
             for fn in sorted(filenames):
                 if os.path.splitext(fn)[1] in SOURCE_EXTS:
                     full = os.path.join(dirpath, fn)
+                    if self.exclude_patterns and \
+                            self._excluded(os.path.relpath(full, self.root)):
+                        continue
                     try:
                         if os.path.getsize(full) <= self.max_file_bytes:
                             out.append(full)
