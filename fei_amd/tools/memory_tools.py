@@ -34,6 +34,8 @@ MEMORY_SEARCH_TOOL = {
         "properties": {
             "query": {"type": "string", "description": "Query string"},
             "folder": {"type": "string", "description": "Restrict to one folder"},
+            "status": {"type": "string", "enum": ["cur", "new", "tmp"]},
+            "limit": {"type": "integer", "description": "Max results (default 50)"},
             "with_content": {"type": "boolean"},
         },
         "required": ["query"],
@@ -69,6 +71,7 @@ MEMORY_CREATE_TOOL = {
         "properties": {
             "subject": {"type": "string"},
             "body": {"type": "string"},
+            "content": {"type": "string", "description": "Alias of body (reference field name)"},
             "tags": {"type": "string", "description": "comma-separated"},
             "folder": {"type": "string"},
             "priority": {"type": "string", "enum": ["low", "normal", "high"]},
@@ -96,6 +99,7 @@ MEMORY_LIST_TOOL = {
         "properties": {
             "folder": {"type": "string"},
             "status": {"type": "string", "enum": ["cur", "new", "tmp"]},
+            "limit": {"type": "integer", "description": "Max results (default 50)"},
         },
     },
 }
@@ -151,10 +155,13 @@ class MemoryTools:
         q = msearch.parse_search_args(args["query"])
         if args.get("folder"):
             q.folders = [args["folder"]]
+        if args.get("status"):
+            q.statuses = [args["status"]]
         if args.get("with_content"):
             q.with_content = True
         results = msearch.search_memories(q, base=self.base)
-        return {"count": len(results), "results": results[:50]}
+        limit = int(args.get("limit", 50))
+        return {"count": len(results), "results": results[:limit]}
 
     def semantic_search(self, args: Dict[str, Any]) -> Dict[str, Any]:
         results = self.index().search_memories(args["query"],
@@ -174,7 +181,7 @@ class MemoryTools:
         if args.get("priority"):
             headers["Priority"] = args["priority"]
         filename = mu.create_memory(args.get("folder", ""), headers,
-                                    args.get("body", ""),
+                                    args.get("body") or args.get("content", ""),
                                     flags=args.get("flags", ""),
                                     base=self.base)
         meta = mu.parse_memory_filename(filename)
@@ -195,6 +202,7 @@ class MemoryTools:
         for st in statuses:
             out.extend(mu.list_memories(folder, st, include_content=True,
                                         base=self.base))
+        out = out[: int(args.get("limit", 50))]
         brief = [{"memory_id": m["metadata"]["unique"],
                   "subject": m.get("headers", {}).get("Subject", ""),
                   "tags": m.get("headers", {}).get("Tags", ""),

@@ -118,3 +118,17 @@ def test_embedding_index_add_texts(memdir_base):
     res = idx.search("alpha doc about gpus", topk=2)
     assert res[0][0] == "id1"
     assert res[0][1] > 0.999
+
+
+def test_reference_field_names(tools):
+    """Reference argument names: content (=body), limit, status."""
+    out = tools.create({"subject": "ref style", "content": "via content field"})
+    assert out["success"]
+    got = tools.view({"memory_id": out["memory_id"]})
+    assert got["content"] == "via content field"
+    for i in range(5):
+        tools.create({"subject": f"bulk {i}", "tags": "bulk"})
+    assert tools.list({"limit": 2})["count"] >= 2
+    assert len(tools.list({"limit": 2})["memories"]) == 2
+    res = tools.search({"query": "bulk", "limit": 3})
+    assert len(res["results"]) == 3
