@@ -164,6 +164,9 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--provider", choices=["local", "stub", "scripted"],
                    default=None)
     p.add_argument("--model", default=None)
+    p.add_argument("--api-key", default=None,
+                   help="API key (reference-compatible; unused by local)")
+    p.add_argument("--debug", action="store_true", help="debug logging")
     p.add_argument("--message", "-m", help="single message, print reply, exit")
     p.add_argument("--task", help="run a continuous task to completion")
     p.add_argument("--max-iterations", type=int, default=10)
@@ -176,7 +179,11 @@ def build_parser() -> argparse.ArgumentParser:
     a = sub.add_parser("ask", help="search-augmented one-shot question")
     a.add_argument("question", nargs="+")
 
-    sub.add_parser("history", help="show recent chat history")
+    h = sub.add_parser("history", help="show recent chat history")
+    h.add_argument("--limit", type=int, default=20)
+    h.add_argument("--clear", action="store_true", help="clear the history")
+    h.add_argument("--load", type=int, default=None,
+                   help="print one entry (prompt + reply) by index")
 
     m = sub.add_parser("mcp", help="list configured MCP servers")
 
@@ -245,15 +252,31 @@ def main(argv: Optional[List[str]] = None) -> int:
     setup_logging()
     parser = build_parser()
     args = parser.parse_args(argv)
+    if getattr(args, "debug", False):
+        import logging
+        logging.getLogger("fei_amd").setLevel(logging.DEBUG)
 
     if args.cmd == "ask":
         print(handle_ask_command(" ".join(args.question), args.provider,
                                  args.model))
         return 0
     if args.cmd == "history":
-        for e in ChatHistory().entries[-20:]:
+        hist = ChatHistory()
+        if args.clear:
+            hist.clear()
+            print("history cleared")
+            return 0
+        if args.load is not None:
+            try:
+                e = hist.entries[args.load]
+            except IndexError:
+                print(f"no history entry {args.load}")
+                return 1
+            print(f"> {e['prompt']}\n{e['response']}")
+            return 0
+        for i, e in enumerate(hist.entries[-args.limit:]):
             stamp = time.strftime("%m-%d %H:%M", time.localtime(e["ts"]))
-            print(f"[{stamp}] {e['prompt'][:70]}")
+            print(f"[{i}] [{stamp}] {e['prompt'][:70]}")
         return 0
     if args.cmd == "mcp":
         from fei_amd.core.mcp import MCPClient

@@ -127,3 +127,20 @@ def test_cli_local_engine_end_to_end(home, capsys):
                "--no-memory", "-m", "hello"])
     assert rc == 0
     assert capsys.readouterr().out.strip()
+
+
+def test_history_flags(home, capsys):
+    from fei_amd.ui.cli import ChatHistory, main
+    h = ChatHistory()
+    h.add("first prompt", "first reply")
+    h.add("second prompt", "second reply")
+    assert main(["history", "--limit", "1"]) == 0
+    out = capsys.readouterr().out
+    assert "second prompt" in out and "first prompt" not in out
+    assert main(["history", "--load", "0"]) == 0
+    assert "first reply" in capsys.readouterr().out
+    assert main(["history", "--clear"]) == 0
+    capsys.readouterr()
+    main(["history"])
+    assert "prompt" not in capsys.readouterr().out
+    assert main(["history", "--load", "5"]) == 1
