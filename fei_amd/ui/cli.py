@@ -67,7 +67,8 @@ class CLI:
     def __init__(self, provider: Optional[str] = None,
                  model: Optional[str] = None,
                  with_memory: bool = True,
-                 script: Optional[List[Dict[str, Any]]] = None):
+                 script: Optional[List[Dict[str, Any]]] = None,
+                 api_key: Optional[str] = None):
         from fei_amd.core.assistant import Assistant
         from fei_amd.tools.code import create_code_tools
         from fei_amd.tools.registry import ToolRegistry
@@ -83,8 +84,8 @@ class CLI:
             kwargs["script"] = script
             provider = "scripted"
         self.assistant = Assistant(config=self.config, provider=provider,
-                                   model=model, tool_registry=registry,
-                                   **kwargs)
+                                   model=model, api_key=api_key,
+                                   tool_registry=registry, **kwargs)
         self.history = ChatHistory()
 
     # -- chat loop -----------------------------------------------------------
@@ -304,7 +305,7 @@ def main(argv: Optional[List[str]] = None) -> int:
         return 0
 
     cli = CLI(provider=args.provider, model=args.model,
-              with_memory=not args.no_memory)
+              with_memory=not args.no_memory, api_key=args.api_key)
     if args.task:
         print(cli.single_message(args.task, task=True,
                                  max_iterations=args.max_iterations))
