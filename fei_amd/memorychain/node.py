@@ -4,6 +4,7 @@ Route parity with the reference Flask app (memdir_tools/memorychain.py:
 1263-1685) under ``/memorychain/*``:
   vote, update, propose, propose_task, claim_task, submit_solution,
   vote_solution, vote_difficulty, wallet/balance, wallet/transfer,
+  wallet/transactions,
   register, sync_nodes, chain, tasks, tasks/<id>, network_status,
   responsible_memories, health, node_status, update_status
 
@@ -202,6 +203,17 @@ class MemorychainNode:
         def wallet_balance():
             who = request.args.get("node", node.node_id)
             return jsonify({"node": who, "balance": chain.wallet.balance(who)})
+
+        @app.get("/memorychain/wallet/transactions")
+        def wallet_transactions():
+            who = request.args.get("node")
+            txs = chain.wallet.transactions
+            if who:
+                txs = [t for t in txs
+                       if t.get("from") == who or t.get("to") == who or
+                       t.get("node") == who]
+            limit = int(request.args.get("limit", 50))
+            return jsonify({"transactions": txs[-limit:]})
 
         @app.post("/memorychain/wallet/transfer")
         def wallet_transfer():

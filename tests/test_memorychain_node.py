@@ -164,3 +164,14 @@ def test_real_socket_two_node_network(tmp_path):
     assert len(nodes[0].chain.blocks) == 3        # update propagated
     assert nodes[0].chain.blocks[2].memory_id == "net-mem"
     assert nodes[0].chain.validate_chain()
+
+
+def test_wallet_transactions_route(node):
+    c = node.app.test_client()
+    node.chain.wallet.credit(node.node_id, 5.0, reason="test reward")
+    r = c.get("/memorychain/wallet/transactions")
+    assert r.status_code == 200
+    txs = r.get_json()["transactions"]
+    assert any(t.get("reason") == "test reward" for t in txs)
+    r = c.get(f"/memorychain/wallet/transactions?node={node.node_id}&limit=1")
+    assert len(r.get_json()["transactions"]) == 1
