@@ -474,6 +474,39 @@ class LocalEngine:
                 out.update(self.last_metrics)
             yield out
 
+    @torch.no_grad()
+    def loglikelihood(self, context: Union[str, List[int]],
+                      continuation: Union[str, List[int]]
+                      ) -> Dict[str, object]:
+        """Teacher-forced log P(continuation | context) — the evaluation
+        primitive (lm-eval style): ONE prefill forward over
+        context+continuation with all-positions logits; returns the summed
+        logprob, per-token logprobs, and whether the continuation is the
+        greedy decode."""
+        ctx = (self.tokenizer.encode(context) if isinstance(context, str)
+               else list(context))
+        cont = (self.tokenizer.encode(continuation, add_bos=False)
+                if isinstance(continuation, str) else list(continuation))
+        if not cont:
+            return {"logprob": 0.0, "token_logprobs": [], "is_greedy": True}
+        ids = (ctx + cont)[-(self.max_seq_len - 1):]
+        n_cont = min(len(cont), len(ids) - 1)
+        tokens = torch.tensor([ids], dtype=torch.int64, device=self.device)
+        pos0 = torch.zeros(1, dtype=torch.int32, device=self.device)
+        logits = self.model.forward_prefill(tokens, pos0, self.k_caches,
+                                            self.v_caches,
+                                            all_positions=True)[0].float()
+        # logits[i] predicts ids[i+1]
+        lp = torch.log_softmax(logits[:-1], dim=-1)
+        start = len(ids) - 1 - n_cont
+        tgt = torch.tensor(ids[start + 1:], device=self.device)
+        rows = lp[start:]
+        tok_lp = rows.gather(1, tgt.unsqueeze(1)).squeeze(1)
+        greedy = bool((rows.argmax(dim=-1) == tgt).all())
+        return {"logprob": float(tok_lp.sum()),
+                "token_logprobs": [float(x) for x in tok_lp],
+                "is_greedy": greedy}
+
     def shutdown(self) -> None:
         self._graph = None
         self.k_caches = self.v_caches = None  # release HBM

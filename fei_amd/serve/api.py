@@ -35,6 +35,8 @@ class CompletionRequest(BaseModel):
     stop_on_eos: bool = True
     speculative: Optional[bool] = None
     stream: bool = False
+    echo: bool = False
+    logprobs: Optional[int] = None
 
 
 class ChatMessage(BaseModel):
@@ -107,6 +109,26 @@ def create_app(engine: Optional[LocalEngine] = None,
     @app.post("/v1/completions")
     def completions(req: CompletionRequest):
         t0 = time.time()
+        if req.echo and req.max_tokens == 0:
+            # evaluation mode (lm-eval pattern): teacher-forced logprobs of
+            # the prompt itself, no generation
+            ids = eng.tokenizer.encode(req.prompt)
+            with lock:
+                ll = eng.loglikelihood(ids[:1], ids[1:])
+            return {
+                "id": f"cmpl-{int(t0 * 1000)}",
+                "object": "text_completion",
+                "model": model_name,
+                "choices": [{"index": 0, "text": req.prompt,
+                             "finish_reason": "length",
+                             "logprobs": {
+                                 "token_logprobs": [None] + ll["token_logprobs"],
+                                 "is_greedy": ll["is_greedy"],
+                             }}],
+                "usage": {"prompt_tokens": len(ids),
+                          "completion_tokens": 0,
+                          "total_tokens": len(ids)},
+            }
         if req.stream:
             import json
 

@@ -401,3 +401,30 @@ def test_decode_stops_at_cache_capacity():
     rows = eng.decode(100, stop_on_eos=False)    # asks for more than fits
     assert len(rows[0]) <= 64 - len(prompt) + 36  # bounded, no crash
     assert int(eng.pos.max()) <= 63
+
+
+def test_loglikelihood_greedy_and_chain_rule():
+    from fei_amd.engine.engine import LocalEngine
+    eng = LocalEngine.create("llama3-tiny")
+    prompt = "score this continuation"
+    out = eng.generate(prompt, max_new_tokens=8, stop_on_eos=False)
+    gen = out["token_ids"]
+    ctx = eng.tokenizer.encode(prompt)
+
+    ll = eng.loglikelihood(ctx, gen)
+    assert ll["is_greedy"] is True
+    assert len(ll["token_logprobs"]) == len(gen)
+    assert ll["logprob"] < 0.0
+
+    # a perturbed continuation scores lower and is not greedy
+    bad = list(gen)
+    bad[0] = (bad[0] + 7) % 200 + 4
+    llb = eng.loglikelihood(ctx, bad)
+    assert llb["logprob"] < ll["logprob"]
+    assert llb["is_greedy"] is False
+
+    # chain rule: ll(ctx, a+b) == ll(ctx, a) + ll(ctx+a, b)
+    a, b = gen[:3], gen[3:]
+    lab = eng.loglikelihood(ctx, a)["logprob"] + \
+        eng.loglikelihood(ctx + a, b)["logprob"]
+    assert abs(lab - ll["logprob"]) < 1e-3

@@ -114,3 +114,13 @@ def test_sessions_503_on_pool_exhaustion():
     r3 = c.post("/v1/sessions", json={"prompt": "ijklmnop" * 5,
                                       "max_tokens": 2})
     assert r3.status_code == 200          # eviction freed the pool
+
+
+def test_completions_eval_mode_logprobs(client):
+    r = client.post("/v1/completions", json={
+        "prompt": "evaluate this text", "max_tokens": 0, "echo": True,
+        "logprobs": 0})
+    assert r.status_code == 200
+    lp = r.json()["choices"][0]["logprobs"]
+    assert lp["token_logprobs"][0] is None
+    assert all(isinstance(x, float) and x <= 0 for x in lp["token_logprobs"][1:])
