@@ -180,11 +180,19 @@ class MemoryTools:
             headers["Tags"] = args["tags"]
         if args.get("priority"):
             headers["Priority"] = args["priority"]
+        body = args.get("body") or args.get("content", "")
         filename = mu.create_memory(args.get("folder", ""), headers,
-                                    args.get("body") or args.get("content", ""),
-                                    flags=args.get("flags", ""),
+                                    body, flags=args.get("flags", ""),
                                     base=self.base)
         meta = mu.parse_memory_filename(filename)
+        # keep a live semantic index fresh (full rebuilds stay explicit via
+        # memory_index_build; deletes leave a stale entry until rebuild)
+        if self._index is not None and self._index.embeddings is not None:
+            text = (headers.get("Subject", "") + "\n" +
+                    headers.get("Tags", "") + "\n" + body)[:2000]
+            folder = args.get("folder", "")
+            self._index.add_texts(
+                [text], [f"{folder}\x00new\x00{filename}"])
         return {"success": True, "memory_id": meta["unique"],
                 "filename": filename}
 

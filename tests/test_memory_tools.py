@@ -132,3 +132,16 @@ def test_reference_field_names(tools):
     assert len(tools.list({"limit": 2})["memories"]) == 2
     res = tools.search({"query": "bulk", "limit": 3})
     assert len(res["results"]) == 3
+
+
+def test_live_index_updated_on_create(tools):
+    tools.create({"subject": "seed memory", "body": "initial content"})
+    tools.index_build({})
+    n0 = tools.index().embeddings.shape[0]
+    tools.create({"subject": "fresh gpu insight",
+                  "body": "the new memory must be searchable immediately"})
+    assert tools.index().embeddings.shape[0] == n0 + 1
+    res = tools.semantic_search(
+        {"query": "fresh gpu insight\n\nthe new memory must be searchable immediately",
+         "topk": 1})
+    assert res["results"][0]["headers"]["Subject"] == "fresh gpu insight"
