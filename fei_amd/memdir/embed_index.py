@@ -111,6 +111,18 @@ class EmbeddingIndex:
         logger.info("indexed %d memories in %.2fs", len(ids), time.time() - t0)
         return len(ids)
 
+    def remove(self, id_substr: str) -> int:
+        """Drop rows whose id contains ``id_substr`` (e.g. a filename when
+        the memory was moved/deleted). Returns how many were removed."""
+        if self.embeddings is None:
+            return 0
+        keep = [i for i, k in enumerate(self.ids) if id_substr not in k]
+        removed = len(self.ids) - len(keep)
+        if removed:
+            self.embeddings = self.embeddings[keep]
+            self.ids = [self.ids[i] for i in keep]
+        return removed
+
     def add_texts(self, texts: List[str], ids: List[str]) -> None:
         """Incremental append (used by tests and live updates)."""
         emb = self.encoder.encode_texts(texts).float()

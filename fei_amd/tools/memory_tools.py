@@ -185,8 +185,8 @@ class MemoryTools:
                                     body, flags=args.get("flags", ""),
                                     base=self.base)
         meta = mu.parse_memory_filename(filename)
-        # keep a live semantic index fresh (full rebuilds stay explicit via
-        # memory_index_build; deletes leave a stale entry until rebuild)
+        # keep a live semantic index fresh (full rebuilds stay explicit
+        # via memory_index_build; delete() drops rows symmetrically)
         if self._index is not None and self._index.embeddings is not None:
             text = (headers.get("Subject", "") + "\n" +
                     headers.get("Tags", "") + "\n" + body)[:2000]
@@ -224,6 +224,9 @@ class MemoryTools:
         if loc is None:
             return {"error": f"memory not found: {args['memory_id']}"}
         folder, status, filename = loc
+        if self._index is not None and self._index.embeddings is not None:
+            # maildir moves keep the unique part of the filename stable
+            self._index.remove(filename.split(":", 1)[0])
         if args.get("permanent"):
             import os
             root = mu.get_memdir_base(self.base)

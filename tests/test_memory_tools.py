@@ -145,3 +145,15 @@ def test_live_index_updated_on_create(tools):
         {"query": "fresh gpu insight\n\nthe new memory must be searchable immediately",
          "topk": 1})
     assert res["results"][0]["headers"]["Subject"] == "fresh gpu insight"
+
+
+def test_live_index_removal_on_delete(tools):
+    a = tools.create({"subject": "kept entry", "body": "stay"})
+    b = tools.create({"subject": "doomed entry", "body": "go away"})
+    tools.index_build({})
+    n0 = tools.index().embeddings.shape[0]
+    tools.delete({"memory_id": b["memory_id"]})
+    assert tools.index().embeddings.shape[0] == n0 - 1
+    res = tools.semantic_search({"query": "anything", "topk": 10})
+    subs = [m["headers"]["Subject"] for m in res["results"]]
+    assert "doomed entry" not in subs and "kept entry" in subs
