@@ -144,3 +144,15 @@ def test_history_flags(home, capsys):
     main(["history"])
     assert "prompt" not in capsys.readouterr().out
     assert main(["history", "--load", "5"]) == 1
+
+
+def test_cli_task_mode_local_engine(home, capsys):
+    """--task through the REAL local engine (tiny): random weights emit no
+    valid tool calls, so the loop must hit the iteration cap and exit
+    cleanly rather than crash on garbage output."""
+    from fei_amd.ui.cli import main
+    rc = main(["--provider", "local", "--model", "llama3-tiny",
+               "--no-memory", "--task", "do something",
+               "--max-iterations", "2"])
+    assert rc == 0
+    assert capsys.readouterr().out is not None
