@@ -189,11 +189,13 @@ def list_folders(base: Optional[str] = None) -> List[str]:
 
 
 def find_memory(
-    memory_id: str, base: Optional[str] = None
+    memory_id: str, base: Optional[str] = None,
+    folder: Optional[str] = None,
 ) -> Optional[Tuple[str, str, str]]:
-    """Locate a memory by full filename or unique-id prefix.
-    Returns (folder, status, filename) or None."""
-    for folder in list_folders(base):
+    """Locate a memory by full filename or unique-id prefix, optionally
+    within one folder. Returns (folder, status, filename) or None."""
+    folders = [folder] if folder is not None else list_folders(base)
+    for folder in folders:
         root = get_memdir_base(base)
         folder_path = os.path.join(root, folder) if folder else root
         for status in STATUS_DIRS:

@@ -86,7 +86,10 @@ MEMORY_VIEW_TOOL = {
     "description": "View one memory by its id (8-hex unique or full filename).",
     "input_schema": {
         "type": "object",
-        "properties": {"memory_id": {"type": "string"}},
+        "properties": {
+            "memory_id": {"type": "string"},
+            "folder": {"type": "string", "description": "Restrict the lookup to one folder"},
+        },
         "required": ["memory_id"],
     },
 }
@@ -111,6 +114,7 @@ MEMORY_DELETE_TOOL = {
         "type": "object",
         "properties": {
             "memory_id": {"type": "string"},
+            "folder": {"type": "string", "description": "Restrict the lookup to one folder"},
             "permanent": {"type": "boolean"},
         },
         "required": ["memory_id"],
@@ -197,7 +201,8 @@ class MemoryTools:
                 "filename": filename}
 
     def view(self, args: Dict[str, Any]) -> Dict[str, Any]:
-        loc = mu.find_memory(args["memory_id"], base=self.base)
+        loc = mu.find_memory(args["memory_id"], base=self.base,
+                             folder=args.get("folder"))
         if loc is None:
             return {"error": f"memory not found: {args['memory_id']}"}
         return mu.read_memory(*loc, base=self.base)
@@ -220,7 +225,8 @@ class MemoryTools:
                 "memories": brief}
 
     def delete(self, args: Dict[str, Any]) -> Dict[str, Any]:
-        loc = mu.find_memory(args["memory_id"], base=self.base)
+        loc = mu.find_memory(args["memory_id"], base=self.base,
+                             folder=args.get("folder"))
         if loc is None:
             return {"error": f"memory not found: {args['memory_id']}"}
         folder, status, filename = loc
