@@ -17,7 +17,7 @@ script = [
     {"tool_calls": [{"name": "GrepTool",
                      "input": {"pattern": "def attn_decode", "path": repo,
                                "include": "*.py"}}]},
-    {"content": "attn_decode lives in fei_amd/ops/__init__.py [TASK_COMPLETE]"},
+    {"content": "attn_decode lives in fei_amd/ops/__init__.py"},
 ]
 a = Assistant(backend=ScriptedBackend(script), tool_registry=reg)
 print(a.ask("where is decode attention dispatched?"))
