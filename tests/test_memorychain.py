@@ -184,3 +184,55 @@ def test_federation_vote_transport_exception_safe(tmp_path):
     assert not out["accepted"]           # raising transport = "no" votes
     assert out["votes"] == 1
     assert h.chains[0].validate_chain()
+
+
+def test_randomized_consensus_soak(tmp_path):
+    """Randomized soak over the federation harness: proposals from random
+    nodes with randomly failing vote transports. Invariants after every
+    accepted round: (a) quorum really held, (b) all reachable nodes carry
+    identical chains after propagation, (c) the chain validates, (d) no
+    memory id appears twice."""
+    import random
+    rng = random.Random(1234)
+    h = Harness(tmp_path, 5)
+    by_addr = {f"addr{i}": c for i, c in enumerate(h.chains)}
+
+    fail_prob = {"addr%d" % i: 0.0 for i in range(5)}
+
+    def flaky_vote(peer, prop):
+        if rng.random() < fail_prob[peer]:
+            raise ConnectionError("injected partition")
+        return by_addr[peer].vote_on_proposal(prop)
+
+    for c in h.chains:
+        c.vote_transport = flaky_vote
+
+    accepted = 0
+    for round_no in range(40):
+        # randomly degrade 0-2 peers for this round
+        for k in fail_prob:
+            fail_prob[k] = rng.choice([0.0, 0.0, 0.0, 1.0]) \
+                if rng.random() < 0.3 else 0.0
+        proposer = h.chains[rng.randrange(5)]
+        out = proposer.propose_memory(
+            f"mem-{round_no}", {"Subject": f"subject {round_no}"})
+        if out.get("accepted"):
+            accepted += 1
+            assert out["votes"] * 2 > out["total"], "accepted without quorum"
+    assert accepted >= 10, "soak produced too few accepted rounds"
+
+    # heal the network and converge via anti-entropy sync
+    for k in fail_prob:
+        fail_prob[k] = 0.0
+    longest = max(h.chains, key=lambda c: len(c.blocks))
+    for c in h.chains:
+        if c is not longest:
+            c.receive_chain_update(longest.serialize())
+    lengths = {len(c.blocks) for c in h.chains}
+    assert len(lengths) == 1, f"nodes diverged: {lengths}"
+    tips = {c.blocks[-1].hash for c in h.chains}
+    assert len(tips) == 1
+    for c in h.chains:
+        assert c.validate_chain()
+        ids = [b.memory_id for b in c.blocks[1:]]
+        assert len(ids) == len(set(ids)), "duplicate memory committed"
