@@ -124,3 +124,20 @@ def test_completions_eval_mode_logprobs(client):
     lp = r.json()["choices"][0]["logprobs"]
     assert lp["token_logprobs"][0] is None
     assert all(isinstance(x, float) and x <= 0 for x in lp["token_logprobs"][1:])
+
+
+def test_concurrent_completions_serialized(client):
+    """Parallel HTTP requests must serialize cleanly on the engine lock
+    (no interleaved decode state, every response well-formed)."""
+    import concurrent.futures as cf
+
+    def hit(i):
+        r = client.post("/v1/completions", json={
+            "prompt": f"concurrent request {i}", "max_tokens": 6})
+        assert r.status_code == 200
+        return r.json()["choices"][0]["text"]
+
+    with cf.ThreadPoolExecutor(max_workers=4) as ex:
+        texts = list(ex.map(hit, range(8)))
+    assert len(texts) == 8
+    assert all(isinstance(t, str) for t in texts)
