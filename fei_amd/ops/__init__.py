@@ -326,8 +326,15 @@ def sample(logits: torch.Tensor, token: torch.Tensor,
     out_tokens[b, step] when given). All state on device (hipGraph-safe)."""
     B, V = logits.shape
     if not logits.is_cuda:
-        t = (ref.argmax_sample(logits) if temperature <= 0
-             else ref.gumbel_sample(logits, temperature))
+        if temperature <= 0:
+            t = ref.argmax_sample(logits)
+        else:
+            # deterministic by (seed, step) like the GPU kernel's splitmix
+            # hash (not bit-identical, but the same reproducibility
+            # contract on both paths)
+            g = torch.Generator().manual_seed(
+                (int(seed) * 1000003 + int(step)) & 0x7FFFFFFF)
+            t = ref.gumbel_sample(logits, temperature, g)
         token.copy_(t)
         if out_tokens is not None:
             st = int(step)
