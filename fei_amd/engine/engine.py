@@ -98,6 +98,12 @@ class LocalEngine:
         if self.is_gpu and not ops.kernels_available():
             ops.require_lib()        # fail loudly: no eager fallback on GPU
 
+        if not self.is_gpu and spec.hidden_size >= 2048:
+            logger.warning(
+                "building %s on CPU: ~%d GB of fp32 host RAM and minutes of "
+                "init — use llama3-tiny for CPU work", spec.name,
+                2 * spec.num_layers * spec.hidden_size *
+                spec.intermediate_size * 12 // 10 ** 9)
         t0 = time.perf_counter()
         self.model = LlamaModel(spec, device, dtype, tp=self.tp, seed=seed,
                                 max_seq_len=self.max_seq_len)
