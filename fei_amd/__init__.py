@@ -23,4 +23,17 @@ __version__ = "0.1.0"
 from fei_amd.core.assistant import Assistant  # noqa: F401
 from fei_amd.core.task_executor import TaskExecutor  # noqa: F401
 
-__all__ = ["Assistant", "TaskExecutor", "__version__"]
+__all__ = ["Assistant", "TaskExecutor", "LocalEngine",
+           "PagedSessionManager", "__version__"]
+
+
+def __getattr__(name):
+    # lazy: keep `import fei_amd` torch-free (torch costs ~1.5 s and the
+    # agent/memdir layers don't need it)
+    if name == "LocalEngine":
+        from fei_amd.engine.engine import LocalEngine
+        return LocalEngine
+    if name == "PagedSessionManager":
+        from fei_amd.engine.sessions import PagedSessionManager
+        return PagedSessionManager
+    raise AttributeError(name)
