@@ -122,3 +122,16 @@ def test_fei_history_and_mcp_subcommands(tmp_path, monkeypatch, capsys):
     monkeypatch.setenv("FEI_MCP_SERVER_DEMO", "http://localhost:9/rpc")
     assert cli_mod.main(["mcp"]) == 0
     assert "demo" in capsys.readouterr().out
+
+
+def test_package_import_is_torch_free():
+    """`import fei_amd` must stay light (agent/memdir layers don't need
+    torch; it costs ~1.5 s) — engine exports are lazy."""
+    import subprocess
+    import sys
+    r = subprocess.run(
+        [sys.executable, "-c",
+         "import sys, fei_amd; assert 'torch' not in sys.modules; "
+         "fei_amd.LocalEngine; assert 'torch' in sys.modules"],
+        capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
