@@ -234,6 +234,17 @@ def run_doctor() -> int:
     else:
         rows.append(("hipcc", "not found (kernel rebuilds unavailable)"))
     rows.append(("memdir base", mu.get_memdir_base()))
+    try:
+        import fastapi  # noqa: F401
+        import uvicorn  # noqa: F401
+        rows.append(("serving api", "available (fei api)"))
+    except ImportError as e:
+        rows.append(("serving api", f"unavailable ({e.name} missing)"))
+    try:
+        import sentencepiece  # noqa: F401
+        rows.append(("tokenizers", "byte + sentencepiece"))
+    except ImportError:
+        rows.append(("tokenizers", "byte only"))
     cfg = get_config()
     rows.append(("provider", cfg.get("llm.provider")))
     rows.append(("model", cfg.get("llm.model")))
