@@ -89,6 +89,11 @@ class LocalEngine:
             (self.is_gpu and not self.tp.is_distributed)
         self.seed = seed
         self.temperature = 0.0       # graph-captured; set before capture
+        if tokenizer is None:
+            tok_path = _os.environ.get("FEI_TOKENIZER")
+            if tok_path:
+                from fei_amd.engine.tokenizer import SpmTokenizer
+                tokenizer = SpmTokenizer(tok_path)
         self.tokenizer = tokenizer or ByteTokenizer()
         if self.tokenizer.vocab_size > spec.vocab_size:
             raise ValueError(

@@ -453,3 +453,19 @@ def test_generate_sampled_seeded():
     o2 = eng.generate_sampled("nucleus", max_new_tokens=10, temperature=0.8,
                               top_p=0.9, seed=42)
     assert o1["token_ids"] == o2["token_ids"]
+
+
+def test_fei_tokenizer_env(tmp_path, monkeypatch):
+    import io
+    import sentencepiece as spm
+    corpus = ["env tokenizer test sentence"] * 30
+    model = io.BytesIO()
+    spm.SentencePieceTrainer.train(sentence_iterator=iter(corpus),
+                                   model_writer=model, vocab_size=40,
+                                   model_type="bpe")
+    path = tmp_path / "t.model"
+    path.write_bytes(model.getvalue())
+    monkeypatch.setenv("FEI_TOKENIZER", str(path))
+    from fei_amd.engine.engine import LocalEngine
+    eng = LocalEngine.create("llama3-tiny")
+    assert type(eng.tokenizer).__name__ == "SpmTokenizer"
