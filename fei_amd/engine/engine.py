@@ -112,6 +112,12 @@ class LocalEngine:
         t0 = time.perf_counter()
         self.model = LlamaModel(spec, device, dtype, tp=self.tp, seed=seed,
                                 max_seq_len=self.max_seq_len)
+        w_path = _os.environ.get("FEI_WEIGHTS")
+        if w_path:
+            t1 = time.perf_counter()
+            self.model.load_weights(w_path)
+            logger.info("loaded weights from %s in %.1fs", w_path,
+                        time.perf_counter() - t1)
         self.k_caches, self.v_caches = self.model.new_kv_cache(self.B, self.max_seq_len)
         logger.info("model %s init in %.1fs (%.2f GB params)", spec.name,
                     time.perf_counter() - t0, self.model.param_bytes() / 2**30)

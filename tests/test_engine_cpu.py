@@ -469,3 +469,20 @@ def test_fei_tokenizer_env(tmp_path, monkeypatch):
     from fei_amd.engine.engine import LocalEngine
     eng = LocalEngine.create("llama3-tiny")
     assert type(eng.tokenizer).__name__ == "SpmTokenizer"
+
+
+def test_fei_weights_env(tmp_path, monkeypatch):
+    """FEI_WEIGHTS loads a safetensors checkpoint at engine init: two
+    engines with different seeds converge once one loads the other's
+    weights (full real-checkpoint path: FEI_WEIGHTS + FEI_TOKENIZER)."""
+    from fei_amd.engine.engine import LocalEngine
+    src = LocalEngine.create("llama3-tiny", seed=11)
+    path = str(tmp_path / "ck")
+    src.model.save_weights(path)
+    want = src.generate("checkpoint check", max_new_tokens=8,
+                        stop_on_eos=False)["token_ids"]
+    monkeypatch.setenv("FEI_WEIGHTS", path)
+    dst = LocalEngine.create("llama3-tiny", seed=999)   # different init
+    got = dst.generate("checkpoint check", max_new_tokens=8,
+                       stop_on_eos=False)["token_ids"]
+    assert got == want
