@@ -37,6 +37,7 @@ class CompletionRequest(BaseModel):
     stream: bool = False
     echo: bool = False
     logprobs: Optional[int] = None
+    cache_prefix: bool = True
 
 
 class ChatMessage(BaseModel):
@@ -77,6 +78,10 @@ def create_app(engine: Optional[LocalEngine] = None,
     lock = threading.Lock()
     model_name = eng.spec.name
 
+    cache = {"ids": []}     # cross-request prefix cache (LCP, like the
+                            # agent backend): repeated/extended prompts
+                            # re-prefill only the delta
+
     def _generate(prompt: str, req) -> dict:
         with lock:
             if req.temperature > 0 and (req.top_k or req.top_p < 1.0):
@@ -84,12 +89,24 @@ def create_app(engine: Optional[LocalEngine] = None,
                     prompt, max_new_tokens=req.max_tokens,
                     temperature=req.temperature, top_k=req.top_k,
                     top_p=req.top_p)
+                cache["ids"] = []
             else:
+                ids = eng.tokenizer.encode(prompt)
+                n = 0
+                if getattr(req, "cache_prefix", True) and eng.B == 1:
+                    limit = min(len(cache["ids"]), len(ids) - 1)
+                    while n < limit and ids[n] == cache["ids"][n]:
+                        n += 1
                 out = eng.generate(
-                    prompt, max_new_tokens=req.max_tokens,
+                    ids, max_new_tokens=req.max_tokens,
                     temperature=req.temperature,
                     stop_on_eos=getattr(req, "stop_on_eos", True),
-                    speculative=getattr(req, "speculative", None))
+                    speculative=getattr(req, "speculative", None),
+                    from_pos=n)
+                gen = list(out["token_ids"])
+                if gen and gen[-1] == eng.tokenizer.eos_id:
+                    gen = gen[:-1]
+                cache["ids"] = ids + gen
         return out
 
     @app.get("/health")

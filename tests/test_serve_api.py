@@ -141,3 +141,17 @@ def test_concurrent_completions_serialized(client):
         texts = list(ex.map(hit, range(8)))
     assert len(texts) == 8
     assert all(isinstance(t, str) for t in texts)
+
+
+def test_cross_request_prefix_cache(client):
+    p1 = "shared long preamble " * 20 + "question one"
+    client.post("/v1/completions", json={"prompt": p1, "max_tokens": 4})
+    r = client.post("/v1/completions", json={
+        "prompt": p1 + " and question two", "max_tokens": 4})
+    eng = client.app.state.engine
+    assert eng.last_metrics["cached_prefix"] > 50     # reused the preamble
+    # opt-out works
+    r = client.post("/v1/completions", json={
+        "prompt": p1, "max_tokens": 4, "cache_prefix": False})
+    assert r.status_code == 200
+    assert eng.last_metrics["cached_prefix"] == 0
