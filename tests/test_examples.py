@@ -18,6 +18,7 @@ EXAMPLES = [
     "http_api.py",
     "paged_serving.py",
     "speculative_decode.py",
+    "status_reporting.py",
 ]
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
