@@ -9,9 +9,11 @@
 //   k_gemv_norm_fp8        (QKV, lm_head: norm prologue)
 //   k_gemv_res_fp8         (O, down: residual epilogue)
 //   k_gemv_swiglu_norm_fp8 (gate/up + SwiGLU + norm prologue)
-// Dequant via a 256-entry LDS LUT: 1 broadcast-class ds_read per element
-// (~33 % of LDS issue at the 2x-bandwidth target) — an arithmetic decode
-// would cost ~8 VALU/element and become the bottleneck.
+// Dequant via the HARDWARE converter: v_cvt_pk_f32_fp8 (2 elems/instr)
+// into packed bf16 + v_dot2 f32 accumulation — measured 277.9 tok/s vs
+// 272.8 for a 256-entry LDS lookup table (docs/BENCHMARKS.md; the LUT
+// attempt lives in git history). These kernels are VALU-issue-bound, not
+// HBM-bound like their bf16 siblings.
 //
 // Quantization itself: k_quant_fp8_rows (one block per row: absmax ->
 // scale -> round-to-nearest-even encode), done once at load time.
