@@ -257,9 +257,12 @@ class LocalEngine:
         self.step.fill_(1)
 
     def decode(self, n_tokens: int, eos_check_every: int = 32,
-               stop_on_eos: bool = True) -> List[List[int]]:
+               stop_on_eos: bool = True,
+               stop: Optional[List[str]] = None) -> List[List[int]]:
         """Generate up to n_tokens per sequence; returns new token ids
-        (including the one sampled by prefill)."""
+        (including the one sampled by prefill). ``stop``: stop strings —
+        generation halts once any appears in the decoded text (checked at
+        chunk boundaries; the match itself is trimmed by the caller)."""
         done = 1                      # prefill already produced token 0
         eos = self.tokenizer.eos_id
         while done < n_tokens:
@@ -273,9 +276,13 @@ class LocalEngine:
             done += chunk
             if int(self.pos.max()) >= self.max_seq_len - 1:
                 break                      # cache capacity reached
-            if stop_on_eos:
+            if stop_on_eos or stop:
                 toks = self.out_tokens[:, :done].tolist()
-                if all(eos in row for row in toks):
+                if stop_on_eos and all(eos in row for row in toks):
+                    break
+                if stop and all(
+                        any(ss in self.tokenizer.decode(row) for ss in stop)
+                        for row in toks):
                     break
         if self.is_gpu:
             torch.cuda.synchronize(self.device)
@@ -358,7 +365,8 @@ class LocalEngine:
     def generate(self, prompt: Union[str, List[int]], max_new_tokens: int = 256,
                  temperature: float = 0.0, stop_on_eos: bool = True,
                  from_pos: int = 0,
-                 speculative: Optional[bool] = None) -> Dict[str, object]:
+                 speculative: Optional[bool] = None,
+                 stop: Optional[List[str]] = None) -> Dict[str, object]:
         """Prefill + decode; returns text and timing metrics. ``from_pos``
         enables prefix caching: the prompt's first ``from_pos`` tokens are
         already in the KV caches and only the remainder is prefilled.
@@ -384,7 +392,8 @@ class LocalEngine:
                                            stop_on_eos=stop_on_eos)
             spec_metrics = dict(self.last_metrics)
         else:
-            rows = self.decode(max_new_tokens, stop_on_eos=stop_on_eos)
+            rows = self.decode(max_new_tokens, stop_on_eos=stop_on_eos,
+                               stop=stop)
         t2 = time.perf_counter()
         new_tokens = len(rows[0])
         decode_s = t2 - t1
@@ -399,9 +408,18 @@ class LocalEngine:
             "decode_tok_s": (max(new_tokens - 1, 0) * self.B) / max(decode_s, 1e-9),
         }
         text = self.tokenizer.decode(rows[0])
+        finish = "stop" if (stop_on_eos and rows[0] and
+                            rows[0][-1] == self.tokenizer.eos_id) else "length"
+        if stop:
+            cut = min((text.find(ss) for ss in stop if ss in text),
+                      default=-1)
+            if cut >= 0:
+                text = text[:cut]
+                finish = "stop"
         return {
             "text": text,
             "token_ids": rows[0],
+            "finish_reason": finish,
             **self.last_metrics,
         }
 

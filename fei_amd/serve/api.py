@@ -32,6 +32,8 @@ class CompletionRequest(BaseModel):
     temperature: float = 0.0
     top_k: int = 0
     top_p: float = 1.0
+    stop: Optional[List[str]] = None
+    n: int = 1
     stop_on_eos: bool = True
     speculative: Optional[bool] = None
     stream: bool = False
@@ -102,6 +104,7 @@ def create_app(engine: Optional[LocalEngine] = None,
                     temperature=req.temperature,
                     stop_on_eos=getattr(req, "stop_on_eos", True),
                     speculative=getattr(req, "speculative", None),
+                    stop=getattr(req, "stop", None),
                     from_pos=n)
                 gen = list(out["token_ids"])
                 if gen and gen[-1] == eng.tokenizer.eos_id:
@@ -169,15 +172,23 @@ def create_app(engine: Optional[LocalEngine] = None,
                 yield "data: [DONE]\n\n"
 
             return StreamingResponse(sse(), media_type="text/event-stream")
-        out = _generate(req.prompt, req)
+        choices = []
+        n_req = max(1, min(int(getattr(req, "n", 1)), 8))
+        for i in range(n_req):
+            if n_req > 1 and req.temperature > 0:
+                # distinct samples per choice: nudge the sampling stream
+                eng.seed = eng.seed + 1
+            out = _generate(req.prompt, req)
+            choices.append({"index": i, "text": out["text"],
+                            "finish_reason": out.get("finish_reason",
+                                                     "stop")})
         n_new = len(out["token_ids"])
         return {
             "id": f"cmpl-{int(t0 * 1000)}",
             "object": "text_completion",
             "created": int(t0),
             "model": model_name,
-            "choices": [{"index": 0, "text": out["text"],
-                         "finish_reason": "stop"}],
+            "choices": choices,
             "usage": {"prompt_tokens": out.get("prompt_tokens", 0),
                       "completion_tokens": n_new,
                       "total_tokens": out.get("prompt_tokens", 0) + n_new},

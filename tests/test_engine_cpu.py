@@ -486,3 +486,20 @@ def test_fei_weights_env(tmp_path, monkeypatch):
     got = dst.generate("checkpoint check", max_new_tokens=8,
                        stop_on_eos=False)["token_ids"]
     assert got == want
+
+
+def test_stop_sequences():
+    from fei_amd.engine.engine import LocalEngine
+    eng = LocalEngine.create("llama3-tiny")
+    base = eng.generate("stop test", max_new_tokens=40, stop_on_eos=False)
+    # choose a stop string that actually occurs mid-text
+    text = base["text"]
+    assert len(text) > 4
+    ss = text[3:5]
+    out = eng.generate("stop test", max_new_tokens=40, stop_on_eos=False,
+                       stop=[ss])
+    assert ss not in out["text"]
+    assert out["text"] == text[: text.find(ss)]
+    assert out["finish_reason"] == "stop"
+    out2 = eng.generate("stop test", max_new_tokens=8, stop_on_eos=False)
+    assert out2["finish_reason"] in ("length", "stop")

@@ -155,3 +155,16 @@ def test_cross_request_prefix_cache(client):
         "prompt": p1, "max_tokens": 4, "cache_prefix": False})
     assert r.status_code == 200
     assert eng.last_metrics["cached_prefix"] == 0
+
+
+def test_stop_and_n(client):
+    r = client.post("/v1/completions", json={
+        "prompt": "abc abc abc", "max_tokens": 16, "n": 2,
+        "stop_on_eos": False})
+    body = r.json()
+    assert len(body["choices"]) == 2
+    assert body["choices"][1]["index"] == 1
+    r = client.post("/v1/completions", json={
+        "prompt": "abc abc abc", "max_tokens": 16, "stop": ["zzqq"],
+        "stop_on_eos": False})
+    assert r.status_code == 200
