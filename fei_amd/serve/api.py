@@ -174,14 +174,17 @@ def create_app(engine: Optional[LocalEngine] = None,
             return StreamingResponse(sse(), media_type="text/event-stream")
         choices = []
         n_req = max(1, min(int(getattr(req, "n", 1)), 8))
-        for i in range(n_req):
-            if n_req > 1 and req.temperature > 0:
-                # distinct samples per choice: nudge the sampling stream
-                eng.seed = eng.seed + 1
-            out = _generate(req.prompt, req)
-            choices.append({"index": i, "text": out["text"],
-                            "finish_reason": out.get("finish_reason",
-                                                     "stop")})
+        seed0 = eng.seed
+        try:
+            for i in range(n_req):
+                if n_req > 1 and req.temperature > 0:
+                    eng.seed = seed0 + i   # distinct samples per choice
+                out = _generate(req.prompt, req)
+                choices.append({"index": i, "text": out["text"],
+                                "finish_reason": out.get("finish_reason",
+                                                         "stop")})
+        finally:
+            eng.seed = seed0
         n_new = len(out["token_ids"])
         return {
             "id": f"cmpl-{int(t0 * 1000)}",
