@@ -191,10 +191,21 @@ class LocalEngine:
             self._decode_step()
         torch.cuda.synchronize()
         self._graph = g
+        self._graph_params = (self.temperature, self.seed)
         logger.info("decode step captured into hipGraph")
 
     def ensure_graph(self) -> None:
-        if self.use_graph and self._graph is None:
+        if not self.use_graph:
+            return
+        # temperature and seed are kernel ARGUMENTS baked into the capture:
+        # a later generate() with different sampling params must recapture
+        # or the graph would silently sample with the old ones
+        if self._graph is not None and \
+                getattr(self, "_graph_params", None) != (self.temperature,
+                                                         self.seed):
+            logger.info("sampling params changed; recapturing decode graph")
+            self._graph = None
+        if self._graph is None:
             t0 = time.perf_counter()
             self._capture_graph()
             logger.info("graph capture took %.2fs", time.perf_counter() - t0)

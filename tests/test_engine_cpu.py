@@ -503,3 +503,24 @@ def test_stop_sequences():
     assert out["finish_reason"] == "stop"
     out2 = eng.generate("stop test", max_new_tokens=8, stop_on_eos=False)
     assert out2["finish_reason"] in ("length", "stop")
+
+
+def test_graph_params_tracked(monkeypatch):
+    """ensure_graph must recapture when temperature/seed change (they are
+    baked kernel arguments). CPU has no graphs, so simulate the gate."""
+    from fei_amd.engine.engine import LocalEngine
+    eng = LocalEngine.create("llama3-tiny")
+    eng.use_graph = True
+    captured = []
+    monkeypatch.setattr(eng, "_capture_graph",
+                        lambda: (captured.append(eng.temperature),
+                                 setattr(eng, "_graph", object()),
+                                 setattr(eng, "_graph_params",
+                                         (eng.temperature, eng.seed))))
+    eng.temperature = 0.0
+    eng.ensure_graph()
+    eng.ensure_graph()                     # no change -> no recapture
+    assert captured == [0.0]
+    eng.temperature = 0.8
+    eng.ensure_graph()                     # param change -> recapture
+    assert captured == [0.0, 0.8]
