@@ -464,7 +464,8 @@ class LocalEngine:
     def generate_stream(self, prompt: Union[str, List[int]],
                         max_new_tokens: int = 256,
                         temperature: float = 0.0, stop_on_eos: bool = True,
-                        from_pos: int = 0, chunk: int = 16):
+                        from_pos: int = 0, chunk: int = 16,
+                        stop: Optional[List[str]] = None):
         """Streaming generate (batch 1): yields a dict per decoded chunk —
         {"new_token_ids", "text", "done"} where ``text`` is the cumulative
         decode (byte tokens can split multi-byte characters, so deltas are
@@ -500,10 +501,17 @@ class LocalEngine:
                 finished = True
             if done >= max_new_tokens:
                 finished = True
+            text = self.tokenizer.decode(row)
+            if stop:
+                cut = min((text.find(ss) for ss in stop if ss in text),
+                          default=-1)
+                if cut >= 0:
+                    text = text[:cut]
+                    finished = True
             new = row[emitted:]
             emitted = len(row)
             out = {"new_token_ids": new,
-                   "text": self.tokenizer.decode(row),
+                   "text": text,
                    "done": finished}
             if finished:
                 decode_s = time.perf_counter() - t1

@@ -160,7 +160,8 @@ def create_app(engine: Optional[LocalEngine] = None,
                     for c in eng.generate_stream(
                             req.prompt, max_new_tokens=req.max_tokens,
                             temperature=req.temperature,
-                            stop_on_eos=req.stop_on_eos):
+                            stop_on_eos=req.stop_on_eos,
+                            stop=req.stop):
                         delta = c["text"][len(prev):]
                         prev = c["text"]
                         payload = {"object": "text_completion.chunk",

@@ -524,3 +524,14 @@ def test_graph_params_tracked(monkeypatch):
     eng.temperature = 0.8
     eng.ensure_graph()                     # param change -> recapture
     assert captured == [0.0, 0.8]
+
+
+def test_stream_stop_sequences():
+    from fei_amd.engine.engine import LocalEngine
+    eng = LocalEngine.create("llama3-tiny")
+    base = eng.generate("stream stop", max_new_tokens=40, stop_on_eos=False)
+    ss = base["text"][5:7]
+    chunks = list(eng.generate_stream("stream stop", max_new_tokens=40,
+                                      stop_on_eos=False, stop=[ss], chunk=5))
+    assert chunks[-1]["done"]
+    assert ss not in chunks[-1]["text"]
