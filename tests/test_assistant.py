@@ -203,3 +203,13 @@ def test_compact_with_model_summarizer():
     s = c.compact(keep_last=2, summarizer=lambda text: "SUMMARY(%d chars)" % len(text))
     assert s.startswith("SUMMARY(")
     assert "SUMMARY(" in c.messages[0]["content"]
+
+
+def test_auto_compact_config_key(cfg, monkeypatch, tmp_path):
+    """llm.auto_compact_chars reaches the Assistant via the standard
+    config precedence (env FEI_LLM_AUTO_COMPACT_CHARS)."""
+    monkeypatch.setenv("FEI_LLM_AUTO_COMPACT_CHARS", "123")
+    from fei_amd.utils.config import Config
+    c = Config(ini_path=str(tmp_path / "x.ini"), load_dotenv=False)
+    a = Assistant(config=c, provider="stub")
+    assert a.auto_compact_chars == 123
