@@ -1,8 +1,12 @@
 """OpenAI-style HTTP serving front-end for the local engine.
 
-POST /v1/completions        {model?, prompt, max_tokens, temperature, top_k,
-                             top_p, stop_on_eos}
-POST /v1/chat/completions   {model?, messages=[{role, content}], ...}
+POST /v1/completions        prompt, max_tokens, temperature, top_k, top_p,
+                            stop, n, stream (SSE), echo+logprobs (eval),
+                            speculative, cache_prefix
+POST /v1/chat/completions   messages=[{role, content}], ... (+ stream SSE)
+POST /v1/sessions           open a paged continuous-batching session
+POST /v1/sessions/step      advance all sessions (?n=K)
+GET  /v1/sessions/{id}      poll result     DELETE /v1/sessions/{id}  close
 GET  /v1/models             available ModelSpecs
 GET  /health                engine + device status
 

@@ -234,6 +234,12 @@ def run_doctor() -> int:
     else:
         rows.append(("hipcc", "not found (kernel rebuilds unavailable)"))
     rows.append(("memdir base", mu.get_memdir_base()))
+    for env in ("FEI_WEIGHTS", "FEI_TOKENIZER"):
+        val = os.environ.get(env)
+        if val:
+            exists = os.path.exists(val)
+            rows.append((env.lower(), val if exists
+                         else f"MISSING ({val})"))
     try:
         import fastapi  # noqa: F401
         import uvicorn  # noqa: F401
