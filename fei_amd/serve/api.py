@@ -309,8 +309,15 @@ def main(argv=None) -> int:
     p.add_argument("--model", default="llama3-8b")
     p.add_argument("--host", default="127.0.0.1")
     p.add_argument("--port", type=int, default=8123)
+    p.add_argument("--check", action="store_true",
+                   help="build the app (loads the model) and exit")
     args = p.parse_args(argv)
-    uvicorn.run(create_app(model=args.model), host=args.host, port=args.port)
+    app = create_app(model=args.model)
+    if args.check:
+        print(f"[fei-api] ok: {app.state.engine.spec.name} on "
+              f"{app.state.engine.device}")
+        return 0
+    uvicorn.run(app, host=args.host, port=args.port)
     return 0
 
 

@@ -168,3 +168,9 @@ def test_stop_and_n(client):
         "prompt": "abc abc abc", "max_tokens": 16, "stop": ["zzqq"],
         "stop_on_eos": False})
     assert r.status_code == 200
+
+
+def test_api_main_check_flag(capsys):
+    from fei_amd.serve.api import main
+    assert main(["--model", "llama3-tiny", "--check"]) == 0
+    assert "ok: llama3-tiny" in capsys.readouterr().out
