@@ -259,7 +259,10 @@ class LocalBackend(Backend):
             content=strip_tool_call_blocks(text),
             tool_calls=calls,
             usage={"input_tokens": out.get("prompt_tokens", 0),
-                   "output_tokens": out.get("new_tokens", 0)},
+                   "output_tokens": out.get("new_tokens", 0),
+                   "cached_prefix": out.get("cached_prefix", 0),
+                   "decode_tok_s": round(out.get("decode_tok_s", 0.0), 1),
+                   "prefill_tok_s": round(out.get("prefill_tok_s", 0.0), 1)},
             raw=out,
         )
 
