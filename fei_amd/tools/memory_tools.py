@@ -59,8 +59,25 @@ MEMORY_SEMANTIC_SEARCH_TOOL = {
 
 MEMORY_INDEX_BUILD_TOOL = {
     "name": "memory_index_build",
-    "description": "(Re)build the semantic embedding index over all memories.",
+    "description": ("(Re)build the search indexes over all memories: the "
+                    "semantic embedding index and the lexical FTS index."),
     "input_schema": {"type": "object", "properties": {}},
+}
+
+MEMORY_KEYWORD_SEARCH_TOOL = {
+    "name": "memory_keyword_search",
+    "description": ("Ranked keyword search (bm25 over subject/tags/content "
+                    "via the FTS index; one query instead of a corpus "
+                    "scan). Build with memory_index_build."),
+    "input_schema": {
+        "type": "object",
+        "properties": {
+            "query": {"type": "string"},
+            "limit": {"type": "integer"},
+            "with_content": {"type": "boolean"},
+        },
+        "required": ["query"],
+    },
 }
 
 MEMORY_CREATE_TOOL = {
@@ -147,6 +164,7 @@ class MemoryTools:
     def __init__(self, base: Optional[str] = None):
         self.base = base
         self._index = None
+        self._fts = None
 
     # lazily built embedding index
     def index(self):
@@ -174,9 +192,22 @@ class MemoryTools:
             return {"error": "no semantic index; run memory_index_build first"}
         return {"count": len(results), "results": results}
 
+    def fts(self):
+        if self._fts is None:
+            from fei_amd.memdir.fts_index import FtsIndex
+            self._fts = FtsIndex(base=self.base)
+        return self._fts
+
     def index_build(self, args: Dict[str, Any]) -> Dict[str, Any]:
         n = self.index().build()
-        return {"success": True, "indexed": n}
+        n_fts = self.fts().build()
+        return {"success": True, "indexed": n, "fts_indexed": n_fts}
+
+    def keyword_search(self, args: Dict[str, Any]) -> Dict[str, Any]:
+        res = self.fts().search_memories(
+            args["query"], limit=int(args.get("limit", 20)),
+            with_content=bool(args.get("with_content", True)))
+        return {"count": len(res), "results": res}
 
     def create(self, args: Dict[str, Any]) -> Dict[str, Any]:
         headers = {"Subject": args["subject"]}
@@ -197,6 +228,10 @@ class MemoryTools:
             folder = args.get("folder", "")
             self._index.add_texts(
                 [text], [f"{folder}\x00new\x00{filename}"])
+        if self._fts is not None and self._fts.count() > 0:
+            self._fts.add(args.get("folder", ""), "new", filename,
+                          headers.get("Subject", ""),
+                          headers.get("Tags", ""), body)
         return {"success": True, "memory_id": meta["unique"],
                 "filename": filename}
 
@@ -233,6 +268,8 @@ class MemoryTools:
         if self._index is not None and self._index.embeddings is not None:
             # maildir moves keep the unique part of the filename stable
             self._index.remove(filename.split(":", 1)[0])
+        if self._fts is not None and self._fts.count() > 0:
+            self._fts.remove(filename.split(":", 1)[0])
         if args.get("permanent"):
             import os
             root = mu.get_memdir_base(self.base)
@@ -255,6 +292,7 @@ def create_memory_tools(registry, base: Optional[str] = None,
     pairs = [
         (MEMORY_SEARCH_TOOL, tools.search),
         (MEMORY_SEMANTIC_SEARCH_TOOL, tools.semantic_search),
+        (MEMORY_KEYWORD_SEARCH_TOOL, tools.keyword_search),
         (MEMORY_INDEX_BUILD_TOOL, tools.index_build),
         (MEMORY_CREATE_TOOL, tools.create),
         (MEMORY_VIEW_TOOL, tools.view),

@@ -157,3 +157,27 @@ def test_live_index_removal_on_delete(tools):
     res = tools.semantic_search({"query": "anything", "topk": 10})
     subs = [m["headers"]["Subject"] for m in res["results"]]
     assert "doomed entry" not in subs and "kept entry" in subs
+
+
+def test_fts_keyword_search(tools):
+    tools.create({"subject": "rocprof profiling guide", "tags": "gpu",
+                  "body": "collect kernel traces with rocprofv3 on MI355X"})
+    tools.create({"subject": "pasta recipe", "tags": "food",
+                  "body": "boil water and add the rocprof... no, salt"})
+    tools.create({"subject": "unrelated", "body": "nothing to see"})
+    out = tools.index_build({})
+    assert out["fts_indexed"] == 3
+    res = tools.keyword_search({"query": "rocprof kernel traces"})
+    assert res["count"] >= 1
+    assert res["results"][0]["headers"]["Subject"] == "rocprof profiling guide"
+
+    # live updates: create is searchable, delete disappears
+    made = tools.create({"subject": "fresh fts entry",
+                         "body": "immediately findable zebra"})
+    res = tools.keyword_search({"query": "zebra"})
+    assert res["count"] == 1
+    tools.delete({"memory_id": made["memory_id"]})
+    assert tools.keyword_search({"query": "zebra"})["count"] == 0
+
+    # operator/punctuation injection cannot break the query
+    assert tools.keyword_search({"query": 'NEAR( "unbalanced OR *'})["count"] >= 0
