@@ -19,7 +19,7 @@ from textual.widgets import Footer, Header, Input, Static
 MEM_COMMANDS = [
     "/mem help", "/mem list", "/mem search ", "/mem view ", "/mem save ",
     "/mem tag ", "/mem server start", "/mem server stop", "/mem server status",
-    "/mem index", "/mem semantic ",
+    "/mem index", "/mem semantic ", "/mem keyword ",
 ]
 
 MEM_HELP = """\
@@ -27,6 +27,7 @@ MEM_HELP = """\
 /mem list [folder]        list memories
 /mem search <query>       query-language search
 /mem semantic <query>     embedding-index search (GPU)
+/mem keyword <query>      bm25 FTS search
 /mem index                rebuild the embedding index
 /mem view <id>            show one memory
 /mem save <subject>       save the conversation as a memory
@@ -134,6 +135,12 @@ class FeiChatApp(App):
             if cmd == "search":
                 out = tools.search({"query": " ".join(rest),
                                     "with_content": False})
+                return "\n".join(
+                    f"{m['metadata']['unique']} {m.get('headers', {}).get('Subject', '')}"
+                    for m in out.get("results", [])) or "(no matches)"
+            if cmd == "keyword":
+                out = tools.keyword_search({"query": " ".join(rest),
+                                            "with_content": False})
                 return "\n".join(
                     f"{m['metadata']['unique']} {m.get('headers', {}).get('Subject', '')}"
                     for m in out.get("results", [])) or "(no matches)"

@@ -156,3 +156,14 @@ def test_cli_task_mode_local_engine(home, capsys):
                "--max-iterations", "2"])
     assert rc == 0
     assert capsys.readouterr().out is not None
+
+
+def test_tui_mem_keyword(home):
+    from fei_amd.ui.tui import FeiChatApp
+    from fei_amd.tools.memory_tools import MemoryTools
+    tools = MemoryTools(base=str(home / "Memdir"))
+    tools.create({"subject": "keyword target", "body": "findable giraffe"})
+    tools.index_build({})
+    app = FeiChatApp(assistant=None, memory_tools=tools)
+    out = app.handle_memory_command("/mem keyword giraffe")
+    assert "keyword target" in out
