@@ -38,6 +38,11 @@ def main(argv: Optional[List[str]] = None) -> int:
     p = sub.add_parser("search", help="search with the query language")
     p.add_argument("query", nargs="+")
     p.add_argument("--format", default="text", choices=["text", "json", "csv", "compact"])
+    p.add_argument("--fts", action="store_true",
+                   help="bm25 FTS index instead of the grammar scan "
+                        "(build: memdir index)")
+
+    sub.add_parser("index", help="(re)build the lexical FTS index")
 
     p = sub.add_parser("flag", help="set flags on a memory")
     p.add_argument("memory_id")
@@ -86,8 +91,17 @@ def main(argv: Optional[List[str]] = None) -> int:
         print("moved" if ok else "failed")
         return 0 if ok else 1
     if args.cmd == "search":
-        results = msearch.search(" ".join(args.query), base=base)
+        if args.fts:
+            from fei_amd.memdir.fts_index import FtsIndex
+            results = FtsIndex(base=base).search_memories(" ".join(args.query))
+        else:
+            results = msearch.search(" ".join(args.query), base=base)
         print(msearch.format_results(results, args.format))
+        return 0
+    if args.cmd == "index":
+        from fei_amd.memdir.fts_index import FtsIndex
+        n = FtsIndex(base=base).build()
+        print(f"indexed {n} memories")
         return 0
     if args.cmd == "flag":
         loc = mu.find_memory(args.memory_id, base=base)

@@ -135,3 +135,14 @@ def test_package_import_is_torch_free():
          "fei_amd.LocalEngine; assert 'torch' in sys.modules"],
         capture_output=True, text=True, timeout=120)
     assert r.returncode == 0, r.stderr
+
+
+def test_memdir_cli_fts(memdir_base, capsys):
+    from fei_amd.memdir import utils as mu
+    from fei_amd.memdir.cli import main
+    mu.create_memory("", {"Subject": "fts cli target"},
+                     "zebra content here", status="cur", base=memdir_base)
+    assert main(["--base", memdir_base, "index"]) == 0
+    assert "indexed 1" in capsys.readouterr().out
+    assert main(["--base", memdir_base, "search", "--fts", "zebra"]) == 0
+    assert "fts cli target" in capsys.readouterr().out
