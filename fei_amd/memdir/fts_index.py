@@ -54,9 +54,13 @@ class FtsIndex:
         logger.info("fts indexed %d memories", n)
         return n
 
+    # key separator: 0x1f (unit separator), NOT NUL — SQLite's LIKE and
+    # other text functions stop at an embedded NUL, which silently broke
+    # remove() on root-folder keys
+
     def _insert(self, folder: str, status: str, mem: Dict[str, Any]) -> None:
         headers = mem.get("headers", {})
-        key = f"{folder}\x00{status}\x00{mem['filename']}"
+        key = f"{folder}\x1f{status}\x1f{mem['filename']}"
         self._db.execute(
             "INSERT INTO mem_fts (key, subject, tags, content) VALUES (?,?,?,?)",
             (key, headers.get("Subject", ""), headers.get("Tags", ""),
@@ -64,17 +68,19 @@ class FtsIndex:
 
     def add(self, folder: str, status: str, filename: str, subject: str,
             tags: str, content: str) -> None:
-        key = f"{folder}\x00{status}\x00{filename}"
+        key = f"{folder}\x1f{status}\x1f{filename}"
         self._db.execute(
             "INSERT INTO mem_fts (key, subject, tags, content) VALUES (?,?,?,?)",
             (key, subject, tags, content[:20000]))
         self._db.commit()
 
     def remove(self, key_substr: str) -> int:
-        cur = self._db.execute(
+        # virtual tables don't report rowcount; count explicitly
+        before = self.count()
+        self._db.execute(
             "DELETE FROM mem_fts WHERE key LIKE ?", (f"%{key_substr}%",))
         self._db.commit()
-        return cur.rowcount
+        return before - self.count()
 
     def count(self) -> int:
         return self._db.execute("SELECT COUNT(*) FROM mem_fts").fetchone()[0]
@@ -103,7 +109,7 @@ class FtsIndex:
                         with_content: bool = True) -> List[Dict[str, Any]]:
         out = []
         for key, score in self.search(query, limit):
-            folder, status, filename = key.split("\x00")
+            folder, status, filename = key.split("\x1f")
             mem = mu.read_memory(folder, status, filename, base=self.base)
             if mem is None:
                 continue

@@ -181,3 +181,22 @@ def test_memory_content_without_separator():
 def test_move_memory_missing_file(memdir_base):
     mu.ensure_folder("", memdir_base)
     assert mu.move_memory("nope", "", ".Trash", base=memdir_base) is False
+
+
+def test_fts_index_edges(memdir_base):
+    from fei_amd.memdir.fts_index import FtsIndex
+    idx = FtsIndex(base=memdir_base)
+    assert idx.build() == 0
+    assert idx.search("anything") == []          # empty index
+    assert idx.search("") == []                  # empty query
+    idx.add("", "cur", "f1:2,", "subj one", "t", "alpha beta")
+    idx.add("", "cur", "f2:2,", "subj two", "t", "beta gamma")
+    assert idx.count() == 2
+    keys = [k for k, _ in idx.search("beta")]
+    assert len(keys) == 2
+    assert idx.remove("f1") == 1
+    assert idx.count() == 1
+    # persistence across instances (same sqlite file)
+    idx.close()
+    idx2 = FtsIndex(base=memdir_base)
+    assert idx2.count() == 1
