@@ -180,9 +180,11 @@ int fei_stream_gemv_proto(float* y, const void* x, const void* w,
                           hipStream_t stream) {
   if (K != 4096) return -2;
   int max_blocks_per_cu = 0;
+  // dynamic-LDS arg is 0: the ring + flags are STATIC shared memory and
+  // already in the kernel's group_segment_fixed_size (passing them again
+  // double-counts and the guard refuses everything)
   hipError_t e = hipOccupancyMaxActiveBlocksPerMultiprocessor(
-      &max_blocks_per_cu, (const void*)k_stream_gemv_proto, 256,
-      SLOTS * SLOT_BYTES + SLOTS * 4);
+      &max_blocks_per_cu, (const void*)k_stream_gemv_proto, 256, 0);
   if (e != hipSuccess || max_blocks_per_cu < 1) return -1;
   hipDeviceProp_t prop;
   if (hipGetDeviceProperties(&prop, 0) != hipSuccess) return -1;

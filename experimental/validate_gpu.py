@@ -1,0 +1,62 @@
+"""GPU validation of the round-2 prototype (run via gpurun; not a test).
+Builds the prototype as a shared lib, runs ONE bounded GEMV, compares to
+torch. Spins give up after ~30 ms; any fault kills only this process."""
+import ctypes
+import os
+import subprocess
+import sys
+import time
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+SO = os.path.join(HERE, "libstream_proto.so")
+
+subprocess.run(["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17",
+                "-shared", "-fPIC",
+                os.path.join(HERE, "stream_engine_proto.hip"), "-o", SO],
+               check=True)
+
+import torch
+assert torch.cuda.is_available()
+lib = ctypes.CDLL(SO)
+lib.fei_stream_gemv_proto.restype = ctypes.c_int
+lib.fei_stream_gemv_proto.argtypes = [ctypes.c_void_p] * 3 + \
+    [ctypes.c_int] * 2 + [ctypes.c_void_p] * 2
+
+N, K = 4096, 4096
+torch.manual_seed(0)
+x = (torch.randn(K, device="cuda") * 0.5).to(torch.bfloat16)
+w = (torch.randn(N, K, device="cuda") * 0.02).to(torch.bfloat16)
+y = torch.full((N,), float("nan"), device="cuda", dtype=torch.float32)  # poison
+fail = torch.zeros(1, device="cuda", dtype=torch.int32)
+
+rc = lib.fei_stream_gemv_proto(y.data_ptr(), x.data_ptr(), w.data_ptr(),
+                               N, K, fail.data_ptr(),
+                               torch.cuda.current_stream().cuda_stream)
+torch.cuda.synchronize()
+print("launch rc:", rc, "fail flag:", int(fail))
+assert rc == 0 and int(fail) == 0
+
+ref = (w.float() @ x.float())
+err = (y - ref).abs().max().item()
+rel = (y - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
+print(f"max abs err {err:.5f}  rel {rel:.6f}  (bf16 dot vs fp32 ref)")
+assert rel < 2e-2, "numerics mismatch"
+
+# crude slot-cadence timing: repeat and report us per call
+for _ in range(3):
+    lib.fei_stream_gemv_proto(y.data_ptr(), x.data_ptr(), w.data_ptr(),
+                              N, K, fail.data_ptr(),
+                              torch.cuda.current_stream().cuda_stream)
+torch.cuda.synchronize()
+t0 = time.perf_counter()
+iters = 50
+for _ in range(iters):
+    lib.fei_stream_gemv_proto(y.data_ptr(), x.data_ptr(), w.data_ptr(),
+                              N, K, fail.data_ptr(),
+                              torch.cuda.current_stream().cuda_stream)
+torch.cuda.synchronize()
+dt = (time.perf_counter() - t0) / iters
+gb = N * K * 2 / 1e9
+print(f"{dt * 1e6:.1f} us/call  {gb / dt / 1e3:.2f} TB/s effective "
+      f"(34 MB GEMV; shipped k_gemv does ~6.9 us)")
+print("VALIDATED: streaming-engine core is correct on silicon")
