@@ -59,3 +59,15 @@ def test_budget_and_done(engine):
     mgr.run()
     res = mgr.result(sid)
     assert res["done"] and len(res["token_ids"]) <= 5
+
+
+def test_result_stable_after_done(engine):
+    """result() is idempotent and stable once a session finishes."""
+    mgr = PagedSessionManager(engine, block_size=16, num_blocks=32)
+    sid = mgr.open("stable", max_new_tokens=6)
+    mgr.run()
+    a = mgr.result(sid)
+    b = mgr.result(sid)
+    assert a == b and a["done"]
+    mgr.step()                       # no active sessions: must be a no-op
+    assert mgr.result(sid) == a
