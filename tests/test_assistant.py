@@ -213,3 +213,16 @@ def test_auto_compact_config_key(cfg, monkeypatch, tmp_path):
     c = Config(ini_path=str(tmp_path / "x.ini"), load_dotenv=False)
     a = Assistant(config=c, provider="stub")
     assert a.auto_compact_chars == 123
+
+
+def test_auto_compact_during_task_loop(cfg):
+    """A long TaskExecutor run with auto-compaction enabled keeps the
+    conversation bounded and still completes."""
+    script = [{"content": "working " + "z" * 200}] * 20 + \
+             [{"content": "done [TASK_COMPLETE]"}]
+    a = Assistant(config=cfg, backend=ScriptedBackend(script))
+    a.auto_compact_chars = 1500
+    result = TaskExecutor(a).execute_task("long job", max_iterations=30)
+    assert result["complete"]
+    assert "conversation summary" in str(a.conversation.messages)
+    assert a.conversation.size_chars() < 6000
