@@ -55,7 +55,15 @@ class FederationAgent:
             device = (torch.device(f"cuda:{self.ctx.local_rank}")
                       if torch.cuda.is_available() else torch.device("cpu"))
             from fei_amd.core.backends import LocalBackend
+            from fei_amd.engine.config import get_spec
             from fei_amd.engine.engine import LocalEngine
+            if device.type == "cpu" and get_spec(model).hidden_size >= 2048:
+                # building 8B+ at fp32 in host RAM takes minutes and tens
+                # of GB per rank — same guard as LocalBackend, demo-grade
+                # fallback instead of a hang-like init
+                logger.warning("no GPU: federation falls back to "
+                               "llama3-tiny (asked for %s)", model)
+                model = "llama3-tiny"
             engine = LocalEngine.create(model, device=device,
                                         **(engine_kwargs or {}))
             kwargs["backend"] = LocalBackend(engine=engine)
