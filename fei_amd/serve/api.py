@@ -306,12 +306,17 @@ def main(argv=None) -> int:
 
     p = argparse.ArgumentParser("fei-api",
                                 description="OpenAI-style local serving API")
-    p.add_argument("--model", default="llama3-8b")
+    p.add_argument("--model", default=None,
+                   help="default: llama3-8b on GPU, llama3-tiny on CPU")
     p.add_argument("--host", default="127.0.0.1")
     p.add_argument("--port", type=int, default=8123)
     p.add_argument("--check", action="store_true",
                    help="build the app (loads the model) and exit")
     args = p.parse_args(argv)
+    if args.model is None:
+        import torch
+        args.model = ("llama3-8b" if torch.cuda.is_available()
+                      else "llama3-tiny")
     app = create_app(model=args.model)
     if args.check:
         print(f"[fei-api] ok: {app.state.engine.spec.name} on "

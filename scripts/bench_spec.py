@@ -21,7 +21,9 @@ def run(engine, prompt, n, speculative):
 
 
 def main():
-    model = sys.argv[1] if len(sys.argv) > 1 else "llama3-8b"
+    import torch
+    model = sys.argv[1] if len(sys.argv) > 1 else (
+        "llama3-8b" if torch.cuda.is_available() else "llama3-tiny")
     n = int(sys.argv[2]) if len(sys.argv) > 2 else 256
     engine = LocalEngine.create(model)
     prompts = {

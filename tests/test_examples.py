@@ -10,6 +10,7 @@ import pytest
 
 EXAMPLES = [
     "basic_chat.py",
+    "engine_generate.py",
     "agent_task.py",
     "memory_workflow.py",
     "federation_demo.py",
