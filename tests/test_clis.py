@@ -146,3 +146,17 @@ def test_memdir_cli_fts(memdir_base, capsys):
     assert "indexed 1" in capsys.readouterr().out
     assert main(["--base", memdir_base, "search", "--fts", "zebra"]) == 0
     assert "fts cli target" in capsys.readouterr().out
+
+
+@pytest.mark.parametrize("mod", [
+    "fei_amd", "fei_amd.memdir.cli", "fei_amd.memorychain.cli",
+    "fei_amd.serve.api", "fei_amd.memdir.run_server",
+])
+def test_cli_help_surfaces(mod):
+    """Every console entry point parses --help without importing torch-heavy
+    or crashing (first thing a new user runs)."""
+    import subprocess
+    import sys
+    r = subprocess.run([sys.executable, "-m", mod, "--help"],
+                       capture_output=True, text=True, timeout=90)
+    assert r.returncode == 0, r.stderr[-500:]
