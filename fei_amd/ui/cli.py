@@ -162,6 +162,9 @@ def handle_ask_command(question: str, provider: Optional[str],
 def build_parser() -> argparse.ArgumentParser:
     p = argparse.ArgumentParser(prog="fei",
                                 description="MI355X-local coding assistant")
+    from fei_amd import __version__
+    p.add_argument("--version", action="version",
+                   version=f"fei_amd {__version__}")
     p.add_argument("--provider", choices=["local", "stub", "scripted"],
                    default=None)
     p.add_argument("--model", default=None)
