@@ -22,7 +22,8 @@ lib.fei_stream_gemv_proto.restype = ctypes.c_int
 lib.fei_stream_gemv_proto.argtypes = [ctypes.c_void_p] * 3 + \
     [ctypes.c_int] * 2 + [ctypes.c_void_p] * 2
 
-N, K = 4096, 4096
+N = int(sys.argv[1]) if len(sys.argv) > 1 else 4096
+K = 4096
 torch.manual_seed(0)
 x = (torch.randn(K, device="cuda") * 0.5).to(torch.bfloat16)
 w = (torch.randn(N, K, device="cuda") * 0.02).to(torch.bfloat16)
