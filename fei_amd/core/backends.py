@@ -248,10 +248,13 @@ class LocalBackend(Backend):
         # is identity for non-special tokens). If the rendering ever
         # differs (specials got dropped), the next turn's prefix compare
         # simply fails and we re-prefill — correctness never depends on it.
-        gen = list(out["token_ids"])
-        eos = self.engine.tokenizer.eos_id
-        if gen and gen[-1] == eos:
-            gen = gen[:-1]
+        # The FINAL generated token was only sampled — its KV row is written
+        # when it becomes the next decode step's input, which never happens
+        # for the last token of a turn. Drop it unconditionally (covers both
+        # the eos and the length-finish case): letting the next turn's LCP
+        # reach a position with no KV silently corrupts attention
+        # (ADVICE r01, medium).
+        gen = list(out["token_ids"])[:-1]
         self._cached_ids = ids + gen
         text = out["text"]
         calls = extract_tool_call_blocks(text)

@@ -177,6 +177,23 @@ class LocalEngine:
 
     def _capture_graph(self) -> None:
         assert self.is_gpu
+        # The two warmup decode steps and the capture itself mutate real
+        # engine state: KV rows at the warmup positions, token/pos/step and
+        # out_tokens[:, :2]. ensure_graph() can run AFTER a prefill (e.g. a
+        # recapture when sampling params change between generate() calls, or
+        # first capture on a chunked >PREFILL_CHUNK prompt), so snapshot and
+        # restore everything the warmup touches — otherwise the cached
+        # prefix is silently corrupted.
+        saved = {
+            "token": self.token.clone(),
+            "pos": self.pos.clone(),
+            "step": self.step.clone(),
+            "out": self.out_tokens[:, :4].clone(),
+        }
+        # warmup at positions 1..2: snapshot those KV rows in every layer
+        kv_rows = slice(1, 4)
+        saved_k = [k[:, :, kv_rows].clone() for k in self.k_caches]
+        saved_v = [v[:, :, kv_rows].clone() for v in self.v_caches]
         # state must be valid during warmup/capture: pretend one token exists
         self.pos.fill_(1)
         self.step.zero_()
@@ -189,6 +206,16 @@ class LocalEngine:
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
             self._decode_step()
+        torch.cuda.synchronize()
+        # restore the clobbered state
+        self.token.copy_(saved["token"])
+        self.pos.copy_(saved["pos"])
+        self.step.copy_(saved["step"])
+        self.out_tokens[:, :4].copy_(saved["out"])
+        for k, sk in zip(self.k_caches, saved_k):
+            k[:, :, kv_rows].copy_(sk)
+        for v, sv in zip(self.v_caches, saved_v):
+            v[:, :, kv_rows].copy_(sv)
         torch.cuda.synchronize()
         self._graph = g
         self._graph_params = (self.temperature, self.seed)
@@ -221,11 +248,20 @@ class LocalEngine:
         first new token. Long prompts are processed in PREFILL_CHUNK
         slices (each slice extends the KV caches; only the last slice's
         logits are sampled)."""
+        # Cap to the context budget BEFORE chunking (mirrors _prep_prompt's
+        # cap for the from_pos>0 prefix-cached path, which bypasses it):
+        # keep the TAIL of the new tokens — a mid-loop break would silently
+        # drop a chunk from the MIDDLE and prefill the remainder at the
+        # dropped chunk's positions (ADVICE r01, low).
+        budget = self.max_seq_len - 1 - from_pos
+        if budget <= 0:
+            from_pos = 0
+            budget = self.max_seq_len - 1
+        if len(token_ids) > budget:
+            token_ids = token_ids[-budget:]
         while len(token_ids) > self.PREFILL_CHUNK:
             head, token_ids = (token_ids[:self.PREFILL_CHUNK],
                                token_ids[self.PREFILL_CHUNK:])
-            if from_pos + len(head) >= self.max_seq_len:
-                break                      # the tail-truncation below handles it
             tokens = torch.tensor([head] * self.B, dtype=torch.int64,
                                   device=self.device)
             pos0 = torch.full((self.B,), from_pos, dtype=torch.int32,
@@ -234,16 +270,6 @@ class LocalEngine:
                                        self.v_caches)
             from_pos += len(head)
         S = len(token_ids)
-        if from_pos + S >= self.max_seq_len:
-            # drop the oldest suffix tokens; a full-context agent should
-            # summarise instead (UI concern)
-            keep = self.max_seq_len - 1 - from_pos
-            if keep <= 0:
-                from_pos = 0
-                token_ids = token_ids[-(self.max_seq_len - 1):]
-            else:
-                token_ids = token_ids[:keep]
-            S = len(token_ids)
         self.ensure_graph()
         tokens = torch.tensor([token_ids] * self.B, dtype=torch.int64,
                               device=self.device)

@@ -22,7 +22,8 @@ class MemdirFolderManager:
         mu.ensure_folder("", self.base)
 
     def _folder_path(self, folder: str) -> str:
-        return os.path.join(self.base, folder) if folder else self.base
+        # validated join: rejects names escaping the memdir base
+        return mu.folder_path(folder, self.base)
 
     # -- CRUD ---------------------------------------------------------------
 

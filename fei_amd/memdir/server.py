@@ -44,6 +44,12 @@ def create_app(base: Optional[str] = None, api_key: Optional[str] = None) -> Fla
         provided = request.headers.get("X-API-Key", "")
         return hmac.compare_digest(provided, key)
 
+    @app.errorhandler(ValueError)
+    def _bad_value(e):
+        # traversal-rejecting folder/status/filename validation
+        # (fei_amd.memdir.utils.folder_path) surfaces as a 400, not a 500
+        return jsonify({"error": str(e)}), 400
+
     @app.before_request
     def _check_key():
         if request.path == "/health":
@@ -109,9 +115,8 @@ def create_app(base: Optional[str] = None, api_key: Optional[str] = None) -> Fla
             return jsonify({"error": "not found"}), 404
         folder, status, filename = loc
         if request.args.get("permanent", "0") in ("1", "true"):
-            root = mu.get_memdir_base(base)
-            path = os.path.join(root, folder, status, filename) if folder else \
-                os.path.join(root, status, filename)
+            path = os.path.join(mu.folder_path(folder, base), status,
+                                filename)
             os.unlink(path)
             return jsonify({"success": True, "permanent": True})
         ok = mu.move_memory(filename, folder, ".Trash", src_status=status,

@@ -49,15 +49,45 @@ def get_memdir_base(base: Optional[str] = None) -> str:
     return os.path.abspath("Memdir")
 
 
+def folder_path(folder: str = "", base: Optional[str] = None) -> str:
+    """Resolve a folder name to its path under the memdir base, REJECTING
+    names whose normalized path escapes the base. Folder names arrive from
+    the HTTP server (unauthenticated by default), so '../../x' must never
+    reach a filesystem operation. normpath (not realpath)
+    so the folder manager's intentional in-tree symlinks keep working."""
+    root = get_memdir_base(base)
+    if not folder:
+        return root
+    p = os.path.normpath(os.path.join(root, folder))
+    if p != root and not p.startswith(root + os.sep):
+        raise ValueError(f"folder name escapes the memdir base: {folder!r}")
+    return p
+
+
+def check_status(status: str) -> str:
+    """Statuses are exactly the maildir trio; anything else (e.g. a
+    traversal attempt through the status field) is rejected."""
+    if status not in STATUS_DIRS:
+        raise ValueError(f"invalid status {status!r} (expected cur/new/tmp)")
+    return status
+
+
+def check_filename(filename: str) -> str:
+    """Reject filenames containing path separators or dot-dot."""
+    if (not filename or filename in (".", "..") or "/" in filename
+            or "\\" in filename or os.sep in filename):
+        raise ValueError(f"invalid memory filename: {filename!r}")
+    return filename
+
+
 def ensure_folder(folder: str = "", base: Optional[str] = None) -> str:
     """Create the folder (and its cur/new/tmp) if needed; return its path.
     ``folder`` is '' for the root folder, or a dotted name like '.Projects'
     or a path like '.Projects/python'."""
-    root = get_memdir_base(base)
-    folder_path = os.path.join(root, folder) if folder else root
+    fpath = folder_path(folder, base)
     for status in STATUS_DIRS:
-        os.makedirs(os.path.join(folder_path, status), exist_ok=True)
-    return folder_path
+        os.makedirs(os.path.join(fpath, status), exist_ok=True)
+    return fpath
 
 
 def generate_filename(flags: str = "", timestamp: Optional[float] = None) -> str:
@@ -122,10 +152,11 @@ def create_memory(
     (default ``new``; reference: utils.py:153-200). Returns the filename."""
     headers = dict(headers or {})
     headers.setdefault("Date", time.strftime("%a, %d %b %Y %H:%M:%S +0000", time.gmtime()))
-    folder_path = ensure_folder(folder, base)
+    fpath = ensure_folder(folder, base)
+    check_status(status)
     filename = generate_filename(flags)
-    tmp_path = os.path.join(folder_path, "tmp", filename)
-    final_path = os.path.join(folder_path, status, filename)
+    tmp_path = os.path.join(fpath, "tmp", filename)
+    final_path = os.path.join(fpath, status, filename)
     with open(tmp_path, "w", encoding="utf-8") as f:
         f.write(format_memory_content(headers, body))
         f.flush()
@@ -148,10 +179,10 @@ def list_memories(
     base: Optional[str] = None,
 ) -> List[Dict[str, Any]]:
     """List memories in one folder+status (reference: utils.py:202-253)."""
-    root = get_memdir_base(base)
-    folder_path = os.path.join(root, folder) if folder else root
+    fpath = folder_path(folder, base)
+    check_status(status)
     out: List[Dict[str, Any]] = []
-    for filename in _iter_status_files(folder_path, status):
+    for filename in _iter_status_files(fpath, status):
         meta = parse_memory_filename(filename)
         if meta is None:
             continue
@@ -161,7 +192,7 @@ def list_memories(
             "status": status,
             "metadata": meta,
         }
-        path = os.path.join(folder_path, status, filename)
+        path = os.path.join(fpath, status, filename)
         if include_content:
             try:
                 with open(path, "r", encoding="utf-8", errors="replace") as f:
@@ -196,10 +227,9 @@ def find_memory(
     within one folder. Returns (folder, status, filename) or None."""
     folders = [folder] if folder is not None else list_folders(base)
     for folder in folders:
-        root = get_memdir_base(base)
-        folder_path = os.path.join(root, folder) if folder else root
+        fpath = folder_path(folder, base)
         for status in STATUS_DIRS:
-            for filename in _iter_status_files(folder_path, status):
+            for filename in _iter_status_files(fpath, status):
                 meta = parse_memory_filename(filename)
                 if meta is None:
                     continue
@@ -212,9 +242,8 @@ def find_memory(
 def read_memory(
     folder: str, status: str, filename: str, base: Optional[str] = None
 ) -> Optional[Dict[str, Any]]:
-    root = get_memdir_base(base)
-    path = os.path.join(root, folder, status, filename) if folder else \
-        os.path.join(root, status, filename)
+    path = os.path.join(folder_path(folder, base), check_status(status),
+                        check_filename(filename))
     try:
         with open(path, "r", encoding="utf-8", errors="replace") as f:
             content = f.read()
@@ -242,9 +271,8 @@ def move_memory(
 ) -> bool:
     """Move by rename, optionally rewriting the flags part of the filename
     (reference: utils.py:255-297)."""
-    root = get_memdir_base(base)
-    src = os.path.join(root, src_folder, src_status, filename) if src_folder else \
-        os.path.join(root, src_status, filename)
+    src = os.path.join(folder_path(src_folder, base),
+                       check_status(src_status), check_filename(filename))
     if not os.path.exists(src):
         return False
     new_name = filename
@@ -253,8 +281,8 @@ def move_memory(
         flags = "".join(sorted(set(c for c in new_flags.upper() if c in FLAGS)))
         new_name = f"{base_part}:2,{flags}"
     ensure_folder(dst_folder, base)
-    dst = os.path.join(root, dst_folder, dst_status, new_name) if dst_folder else \
-        os.path.join(root, dst_status, new_name)
+    dst = os.path.join(folder_path(dst_folder, base),
+                       check_status(dst_status), new_name)
     os.rename(src, dst)
     return True
 
@@ -268,15 +296,14 @@ def update_memory_flags(
 ) -> Optional[str]:
     """Rewrite flags via rename in place (reference: utils.py:354-388).
     Returns the new filename or None."""
-    root = get_memdir_base(base)
-    folder_path = os.path.join(root, folder) if folder else root
-    src = os.path.join(folder_path, status, filename)
+    fpath = folder_path(folder, base)
+    src = os.path.join(fpath, check_status(status), check_filename(filename))
     if not os.path.exists(src):
         return None
     base_part = filename.split(":2,")[0]
     norm = "".join(sorted(set(c for c in flags.upper() if c in FLAGS)))
     new_name = f"{base_part}:2,{norm}"
-    os.rename(src, os.path.join(folder_path, status, new_name))
+    os.rename(src, os.path.join(fpath, status, new_name))
     return new_name
 
 

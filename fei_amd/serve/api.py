@@ -110,10 +110,9 @@ def create_app(engine: Optional[LocalEngine] = None,
                     speculative=getattr(req, "speculative", None),
                     stop=getattr(req, "stop", None),
                     from_pos=n)
-                gen = list(out["token_ids"])
-                if gen and gen[-1] == eng.tokenizer.eos_id:
-                    gen = gen[:-1]
-                cache["ids"] = ids + gen
+                # drop the final generated token: only sampled, its KV row
+                # is never written (same rule as the agent backend)
+                cache["ids"] = ids + list(out["token_ids"])[:-1]
         return out
 
     @app.get("/health")
@@ -138,6 +137,9 @@ def create_app(engine: Optional[LocalEngine] = None,
             # the prompt itself, no generation
             ids = eng.tokenizer.encode(req.prompt)
             with lock:
+                # loglikelihood overwrites the KV caches from position 0:
+                # the cross-request prefix cache no longer describes them
+                cache["ids"] = []
                 ll = eng.loglikelihood(ids[:1], ids[1:])
             return {
                 "id": f"cmpl-{int(t0 * 1000)}",
@@ -160,6 +162,9 @@ def create_app(engine: Optional[LocalEngine] = None,
 
             def sse():
                 with lock:
+                    # streaming prefills from position 0: invalidate the
+                    # cross-request prefix cache
+                    cache["ids"] = []
                     prev = ""
                     for c in eng.generate_stream(
                             req.prompt, max_new_tokens=req.max_tokens,
@@ -216,6 +221,7 @@ def create_app(engine: Optional[LocalEngine] = None,
 
             def sse():
                 with lock:
+                    cache["ids"] = []       # stream prefills from position 0
                     prev = ""
                     for c in eng.generate_stream(
                             prompt, max_new_tokens=req.max_tokens,

@@ -185,3 +185,28 @@ def test_generate_stream_gpu(engine):
     ids = [t for c in chunks for t in c["new_token_ids"]]
     assert ids == plain["token_ids"]
     assert chunks[-1]["done"]
+
+
+def test_recapture_preserves_kv_state():
+    """a graph RECAPTURE between generate() calls (here
+    forced by a temperature change) runs two warmup decode steps that used
+    to clobber KV rows 1-2 and the token/pos/step state. With the
+    save/restore fix, a prefix-cached continuation across a recapture must
+    match the same continuation computed with no prefix cache."""
+    from fei_amd.engine.engine import LocalEngine
+
+    def run(use_cache: bool):
+        eng = LocalEngine.create("llama3-tiny", max_seq_len=256, seed=7,
+                                 use_hip_graph=True)
+        p1 = eng.tokenizer.encode("first turn prompt body")
+        o1 = eng.generate(p1, max_new_tokens=8, temperature=0.0,
+                          stop_on_eos=False)
+        ctx = p1 + o1["token_ids"][:-1]   # prefix with KV present
+        p2 = ctx + eng.tokenizer.encode(" and more", add_bos=False)
+        # temperature change forces a recapture inside the next prefill
+        o2 = eng.generate(p2, max_new_tokens=8, temperature=0.5,
+                          stop_on_eos=False,
+                          from_pos=len(ctx) if use_cache else 0)
+        return o2["token_ids"]
+
+    assert run(True) == run(False)
