@@ -63,8 +63,11 @@ def main() -> int:
     budget = spec.max_seq_len - args.warmup - args.steps - 16
     if args.prompt_len > max(budget, 16):
         args.prompt_len = max(budget, 16)
+    # floor of 2048 on GPU so the post-timing agent tool-turn measurement
+    # (tool_turn_p50_s below) has context for real rendered prompts
     max_seq = min(spec.max_seq_len,
-                  args.prompt_len + args.warmup + args.steps + 64)
+                  max(args.prompt_len + args.warmup + args.steps + 64,
+                      2048 if has_gpu else 0))
     tp_mode = args.tp > 1
     if tp_mode:
         assert world == args.tp, "--tp requires launching exactly tp ranks"
@@ -128,6 +131,17 @@ def main() -> int:
     value = total_tokens / t_max
     ms_per_step = t_max / args.steps * 1000.0
 
+    # Second half of the BASELINE metric: p50 agent tool-turn latency,
+    # measured AFTER the timed region through the real Assistant stack
+    # (rank 0, single-agent configs only). Also gives the driver's SMI
+    # sampler several seconds of real GPU work to observe (r01's
+    # gpu_busy record was empty because the timed region was 77 ms).
+    turn_stats = {}
+    if rank == 0 and not tp_mode and world <= 1 and args.batch == 1 \
+            and has_gpu:
+        from fei_amd.core.turn_bench import measure_tool_turns
+        turn_stats = measure_tool_turns(engine, n_turns=9, max_new=96)
+
     if rank == 0:
         result = {
             "metric": (f"agent tok/s ({args.model} local decode, "
@@ -153,6 +167,7 @@ def main() -> int:
                                 f"dp{n_gpus} (1 agent per GPU, weak scaling)"),
                 "hip_graph": engine._graph is not None,
                 "prefill_tok_s": round(prefill_tok_s, 1),
+                **turn_stats,
             },
         }
         print(json.dumps(result))

@@ -16,7 +16,7 @@ import torch
 
 def main():
     from fei_amd.core.assistant import Assistant
-    from fei_amd.core.backends import LocalBackend
+    from fei_amd.core.turn_bench import TurnBackend
     from fei_amd.engine.engine import LocalEngine
     from fei_amd.tools.code import create_code_tools
     from fei_amd.tools.registry import ToolRegistry
@@ -26,28 +26,6 @@ def main():
                            "llama3-8b" if has_gpu else "llama3-tiny")
     max_new = 128 if has_gpu else 16
     engine = LocalEngine.create(model, max_seq_len=8192 if has_gpu else 512)
-
-    class TurnBackend(LocalBackend):
-        """Real engine inference; deterministic tool-call injection on the
-        first round of each turn (random-init weights can't emit JSON)."""
-        def __init__(self, engine, max_new):
-            super().__init__(engine=engine)
-            self.max_new = max_new
-            self._round = 0
-
-        def complete(self, messages, tools=None, system=None, max_tokens=4000,
-                     temperature=0.0):
-            out = super().complete(messages, tools, system,
-                                   max_tokens=self.max_new,
-                                   temperature=temperature)
-            self._round += 1
-            if self._round % 2 == 1:       # first round of a turn: call a tool
-                out.tool_calls = [{"id": f"c{self._round}", "name": "GlobTool",
-                                   "input": {"pattern": "**/*.py",
-                                             "path": os.getcwd()}}]
-            else:
-                out.tool_calls = []
-            return out
 
     registry = ToolRegistry()
     create_code_tools(registry)
