@@ -746,6 +746,77 @@ void fei_sample_onepass(const void* logits, int* token, int* out_tokens,
                      temperature, seed, max_new);
 }
 
+// Shard sampler for tensor-parallel decode: each rank samples over its OWN
+// lm_head shard [B, vocab_local] and emits (best value, best GLOBAL index)
+// into out[b] = {f32 val, i32 idx}; the ranks then all-gather 8 bytes per
+// sequence instead of the full vocab row (replaces the 513 KB logits
+// all-gather per step at Llama-3 vocab). Gumbel noise is keyed by the
+// GLOBAL index, so the sharded sample is bit-identical to the full-vocab
+// k_sample_onepass on the gathered logits.
+__global__ void __launch_bounds__(1024)
+k_sample_shard(const u16* __restrict__ logits, float* __restrict__ out,
+               const int* __restrict__ step, int B, int vocab_local,
+               int v_offset, float temperature, u64 seed) {
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  const u64 st = (u64)(*step);
+  const float invT = temperature > 0.f ? 1.f / temperature : 1.f;
+  float best = -1.0f / 0.0f;
+  int besti = 0x7fffffff;
+  const s16x8* row = (const s16x8*)(logits + (long)b * vocab_local);
+  const int nv = vocab_local >> 3;
+  for (int i = tid; i < nv; i += blockDim.x) {
+    s16x8 v8 = row[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int idx = v_offset + i * 8 + j;
+      float v = bf2f((u16)v8[j]) * invT;
+      if (temperature > 0.f) {
+        float u = hash_uniform(seed ^ (st * 0x51ed27f1ull) ^ ((u64)b << 40) ^ (u64)idx);
+        v += -__logf(-__logf(u));
+      }
+      if (v > best || (v == best && idx < besti)) { best = v; besti = idx; }
+    }
+  }
+  for (int k = nv * 8 + tid; k < vocab_local; k += blockDim.x) {
+    const int idx = v_offset + k;
+    float v = bf2f(logits[(long)b * vocab_local + k]) * invT;
+    if (temperature > 0.f) {
+      float u = hash_uniform(seed ^ (st * 0x51ed27f1ull) ^ ((u64)b << 40) ^ (u64)idx);
+      v += -__logf(-__logf(u));
+    }
+    if (v > best || (v == best && idx < besti)) { best = v; besti = idx; }
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) {
+    const float ov = __shfl_xor(best, off);
+    const int oi = __shfl_xor(besti, off);
+    if (ov > best || (ov == best && oi < besti)) { best = ov; besti = oi; }
+  }
+  __shared__ float rv[16];
+  __shared__ int ri[16];
+  const int wid = tid >> 6;
+  if ((tid & 63) == 0) { rv[wid] = best; ri[wid] = besti; }
+  __syncthreads();
+  if (tid == 0) {
+    const int nw = blockDim.x >> 6;
+    for (int i = 1; i < nw; ++i)
+      if (rv[i] > best || (rv[i] == best && ri[i] < besti)) {
+        best = rv[i]; besti = ri[i];
+      }
+    out[(long)b * 2] = best;
+    ((int*)out)[(long)b * 2 + 1] = besti;
+  }
+}
+
+void fei_sample_shard(const void* logits, float* out, const int* step,
+                      int B, int vocab_local, int v_offset, float temperature,
+                      u64 seed, hipStream_t stream) {
+  hipLaunchKernelGGL(k_sample_shard, dim3(B), dim3(1024), 0, stream,
+                     (const u16*)logits, out, step, B, vocab_local, v_offset,
+                     temperature, seed);
+}
+
 }  // extern "C"
 
 // ---------------------------------------------------------------------------
