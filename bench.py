@@ -73,9 +73,8 @@ def main() -> int:
         assert world == args.tp, "--tp requires launching exactly tp ranks"
         engine = LocalEngine(
             spec, device=device, batch_size=args.batch, max_seq_len=max_seq,
-            use_hip_graph=False,            # collectives not graph-captured
-            seed=1234, tp=ctx,
-        )
+            seed=1234, tp=ctx,              # hipGraph on: RCCL collectives
+        )                                   # captured (eager fallback inside)
     else:
         engine = LocalEngine(
             spec, device=device, batch_size=args.batch, max_seq_len=max_seq,

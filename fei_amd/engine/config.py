@@ -66,6 +66,13 @@ MODEL_SPECS: Dict[str, ModelSpec] = {
         num_heads=4, num_kv_heads=2, head_dim=64, intermediate_size=512,
         max_seq_len=512,
     ),
+    # tiny8: every sharded axis divisible by 8 — the gloo-8 TP harness
+    # (70B's tp=8 shard shape ratios at toy scale)
+    "llama3-tiny8": ModelSpec(
+        name="llama3-tiny8", vocab_size=512, hidden_size=256, num_layers=2,
+        num_heads=8, num_kv_heads=8, head_dim=32, intermediate_size=512,
+        max_seq_len=512,
+    ),
     # embedding encoder shape (bge-base class): 12L/768h/12heads, D=64
     "bge-base": ModelSpec(
         name="bge-base", vocab_size=30522, hidden_size=768, num_layers=12,

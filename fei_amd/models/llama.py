@@ -128,6 +128,8 @@ class LlamaModel:
         self.norm_f = torch.ones(C, device=self.device, dtype=self.dtype)
         lm = self._rand(gen, s.vocab_size, C, scale=sc)
         self.lm_head = lm[r * self.vocab_l:(r + 1) * self.vocab_l].contiguous()
+        self.vocab_offset = r * self.vocab_l   # first GLOBAL vocab id of
+                                               # this rank's lm_head shard
 
     def quantize_fp8(self) -> None:
         """Quantize the DECODE-path weights to e4m3fn (per-row scales).
@@ -194,8 +196,12 @@ class LlamaModel:
         fused_attn: bool = False,
         attn_out: Optional[torch.Tensor] = None,
         fused_norm: bool = False,
+        gather_logits: bool = True,
     ) -> torch.Tensor:
-        """One decode step -> logits [B, vocab] (gathered across TP).
+        """One decode step -> logits [B, vocab] (gathered across TP;
+        ``gather_logits=False`` returns THIS rank's [B, vocab/W] shard —
+        the TP hot loop samples on shards and gathers 8 bytes/seq instead,
+        engine._decode_step_tp).
         ``fused_norm`` (tp=1 only): the residual stream stays inside the
         GEMV kernels — norm-prologue QKV/gate-up, residual-epilogue O/down;
         5 kernels per layer instead of 9."""
@@ -236,7 +242,8 @@ class LlamaModel:
                          else self.norm_f)
             x, h = ops.fused_add_rmsnorm(d, h, next_norm, s.norm_eps)
         logits = ops.linear_decode(x, self.lm_head)
-        logits = all_gather_cat(logits, dim=-1, ctx=self.tp)
+        if gather_logits:
+            logits = all_gather_cat(logits, dim=-1, ctx=self.tp)
         return logits
 
     def _ssq_slots(self, B: int, device) -> torch.Tensor:

@@ -62,6 +62,7 @@ def _try_load() -> None:
                                           _l, _l, _vp]
     lib.fei_sample_onepass.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _f, _u64,
                                        _i, _vp]
+    lib.fei_sample_shard.argtypes = [_vp, _vp, _vp, _i, _i, _i, _f, _u64, _vp]
     lib.fei_attn_decode_paged.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp,
                                           _i, _i, _i, _i, _i, _i, _i, _f,
                                           _l, _vp]
@@ -353,6 +354,34 @@ def sample(logits: torch.Tensor, token: torch.Tensor,
                        _ptr(workspace), B, V, nchunks, temperature, seed,
                        max_new, _stream())
     return token
+
+
+def sample_shard(logits: torch.Tensor, step: torch.Tensor,
+                 v_offset: int, out: torch.Tensor,
+                 temperature: float = 0.0, seed: int = 0) -> torch.Tensor:
+    """Tensor-parallel shard sampler: reduce THIS rank's logits shard
+    [B, V_local] (bf16 on GPU) to (best value, best GLOBAL index) per
+    sequence, written into ``out`` [B, 2] f32 (index stored as int bits).
+    The ranks then all-gather 8 bytes/seq instead of the vocab row.
+    Gumbel noise is keyed by the global index (same splitmix hash as the
+    full sampler), so shard+combine is bit-identical to sampling the
+    gathered logits. CPU path mirrors the hash exactly."""
+    B, Vl = logits.shape
+    if not logits.is_cuda:
+        vals = logits.float()
+        if temperature > 0:
+            vals = vals / temperature + ref.hash_gumbel(
+                B, Vl, v_offset, int(seed), int(step))
+        best = vals.max(dim=-1)
+        out[:, 0] = best.values
+        out[:, 1].view(torch.int32).copy_(
+            (best.indices + v_offset).to(torch.int32))
+        return out
+    lib = require_lib()
+    lib.fei_sample_shard(_ptr(logits), _ptr(out), _ptr(step), B, Vl,
+                         int(v_offset), float(temperature), int(seed),
+                         _stream())
+    return out
 
 
 def advance(pos: torch.Tensor, step: torch.Tensor,

@@ -247,3 +247,30 @@ def quant_fp8(w: torch.Tensor):
 
 def dequant_fp8(w8: torch.Tensor, scales: torch.Tensor) -> torch.Tensor:
     return w8.view(torch.float8_e4m3fn).float() * scales.unsqueeze(-1).float()
+
+
+def hash_gumbel(B: int, v_local: int, v_offset: int, seed: int,
+                step: int) -> torch.Tensor:
+    """CPU replica of the GPU sampler's per-element Gumbel noise
+    (fei_common.h hash_uniform: splitmix64 -> 24-bit uniform in (0,1]),
+    keyed by (seed, step, batch row, GLOBAL vocab index) — so the
+    tensor-parallel shard sampler matches the full sampler bit-for-bit
+    in its noise stream."""
+    import numpy as np
+
+    idx = np.arange(v_offset, v_offset + v_local, dtype=np.uint64)
+    out = np.empty((B, v_local), dtype=np.float32)
+    with np.errstate(over="ignore"):
+        for b in range(B):
+            x = (np.uint64(seed)
+                 ^ (np.uint64(step) * np.uint64(0x51ED27F1))
+                 ^ (np.uint64(b) << np.uint64(40))
+                 ^ idx)
+            x = x + np.uint64(0x9E3779B97F4A7C15)
+            x = (x ^ (x >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+            x = (x ^ (x >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+            x = x ^ (x >> np.uint64(31))
+            m = (x >> np.uint64(40)).astype(np.float32)
+            out[b] = (m + 1.0) * (1.0 / 16777216.0)
+    u = torch.from_numpy(out)
+    return -torch.log(-torch.log(u))
