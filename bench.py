@@ -39,12 +39,17 @@ def main() -> int:
                    help="e4m3fn weight quantization on the decode path "
                         "(REDUCED weight precision: reported as a separate "
                         "labeled number, never the bf16 headline)")
+    p.add_argument("--stream", action="store_true",
+                   help="persistent weight-streaming decode engine "
+                        "(FEI_STREAM_DECODE=1; launch path is the default)")
     p.add_argument("--tp", type=int, default=1,
                    help="tensor-parallel degree: the W ranks run ONE agent "
                         "sharded over RCCL/xGMI instead of W independent "
                         "agents (requires world_size == tp)")
     args = p.parse_args()
 
+    if args.stream:
+        os.environ["FEI_STREAM_DECODE"] = "1"
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     n_gpus = max(args.gpus, world)
@@ -165,6 +170,8 @@ def main() -> int:
                                 if tp_mode else
                                 f"dp{n_gpus} (1 agent per GPU, weak scaling)"),
                 "hip_graph": engine._graph is not None,
+                "stream_engine": bool(getattr(engine, "stream_decode",
+                                              False)),
                 "prefill_tok_s": round(prefill_tok_s, 1),
                 **turn_stats,
             },

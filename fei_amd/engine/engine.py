@@ -162,6 +162,32 @@ class LocalEngine:
         )
         self.attn_out = torch.zeros(self.B, Hq_l, D, dtype=self.dtype,
                                     device=device)
+        # Persistent weight-streaming decode engine (FEI_STREAM_DECODE=1):
+        # one launch per layer on the LDS-DMA loader/consumer structure
+        # (ops/csrc/stream_layer.hip) instead of six. Guarded by the
+        # residency/shape check; batch-1 bf16 tp=1 only. The launch path
+        # stays the default and the correctness oracle.
+        self.stream_decode = False
+        self._stream_ws = None
+        if (self.is_gpu and _os.environ.get("FEI_STREAM_DECODE", "0") == "1"
+                and self.B == 1 and not self.tp.is_distributed
+                and self.weight_quant is None and spec.arch == "llama"):
+            rc = ops.stream_layer_check(
+                spec.hidden_size, spec.num_heads, spec.num_kv_heads,
+                spec.head_dim, spec.intermediate_size)
+            if rc == 0:
+                self._stream_ws = ops.stream_workspace(spec, device)
+                self._stream_bufs = (
+                    torch.zeros(1, spec.hidden_size, dtype=self.dtype,
+                                device=device),
+                    torch.zeros(1, spec.hidden_size, dtype=self.dtype,
+                                device=device),
+                )
+                self.stream_decode = True
+                logger.info("persistent stream-decode engine enabled")
+            else:
+                logger.warning("stream-decode refused (rc=%d); launch path",
+                               rc)
         self._graph: Optional[torch.cuda.CUDAGraph] = None
         if self.tp.is_distributed:
             # TP decode samples on each rank's logits SHARD and gathers
@@ -193,6 +219,17 @@ class LocalEngine:
     def _decode_step(self) -> None:
         if self.tp.is_distributed:
             return self._decode_step_tp()
+        if self.stream_decode:
+            logits = self.model.forward_decode_stream(
+                self.token, self.pos, self.k_caches, self.v_caches,
+                self._stream_ws, self._stream_bufs)
+            ops.sample(logits, self.token, self.step,
+                       self.sample_ws.view(self.B, -1),
+                       out_tokens=self.out_tokens,
+                       temperature=self.temperature, seed=self.seed,
+                       nchunks=SAMPLE_CHUNKS)
+            ops.advance(self.pos, self.step, max_pos=self.max_seq_len - 1)
+            return
         logits = self.model.forward_decode(
             self.token, self.pos, self.k_caches, self.v_caches,
             attn_splits=self.attn_splits, workspace=self.attn_ws,
@@ -318,6 +355,11 @@ class LocalEngine:
         first new token. Long prompts are processed in PREFILL_CHUNK
         slices (each slice extends the KV caches; only the last slice's
         logits are sampled)."""
+        if self.stream_decode:
+            # pos can move backwards across prefills; the stream engine's
+            # granule tags are pos-keyed, so stale tags must be cleared
+            for t in self._stream_ws.values():
+                t.zero_()
         # Cap to the context budget BEFORE chunking (mirrors _prep_prompt's
         # cap for the from_pos>0 prefix-cached path, which bypasses it):
         # keep the TAIL of the new tokens — a mid-loop break would silently
@@ -393,11 +435,22 @@ class LocalEngine:
                     break
         if self.is_gpu:
             torch.cuda.synchronize(self.device)
+        self._check_stream_fail()
         rows = self.out_tokens[:, :done].tolist()
         if stop_on_eos:
             rows = [row[: row.index(eos) + 1] if eos in row else row
                     for row in rows]
         return rows
+
+    def _check_stream_fail(self) -> None:
+        """Stream-engine give-up detection: a bounded spin that timed out
+        stamps the fail word instead of hanging the GPU; surface it."""
+        if self.stream_decode and int(self._stream_ws["fail"][0]):
+            code = int(self._stream_ws["fail"][0])
+            self._stream_ws["fail"].zero_()
+            raise RuntimeError(
+                f"stream-decode engine gave up (code {code}); outputs are "
+                "poisoned — rerun with FEI_STREAM_DECODE=0")
 
     def decode_speculative(self, n_tokens: int, ctx_tokens: List[int],
                            spec_k: int = 8, stop_on_eos: bool = True,
@@ -591,6 +644,7 @@ class LocalEngine:
             done += n_chunk
             if self.is_gpu:
                 torch.cuda.synchronize(self.device)
+            self._check_stream_fail()
             row = self.out_tokens[0, :done].tolist()
             if stop_on_eos and eos in row:
                 row = row[: row.index(eos) + 1]

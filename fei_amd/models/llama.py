@@ -246,6 +246,25 @@ class LlamaModel:
             logits = all_gather_cat(logits, dim=-1, ctx=self.tp)
         return logits
 
+    def forward_decode_stream(self, token, pos, k_caches, v_caches,
+                              ws: dict, bufs) -> torch.Tensor:
+        """One decode step through the persistent weight-streaming layer
+        engine (ops.stream_layer: ONE launch per layer instead of six;
+        csrc/stream_layer.hip). Batch 1, tp=1, bf16 only — the engine
+        guards with ops.stream_layer_check before enabling. ``bufs`` is a
+        persistent pair of [1, C] bf16 residual buffers (stable addresses
+        for hipGraph capture)."""
+        s = self.spec
+        h = F.embedding(token.long(), self.emb)
+        bufs[0].copy_(h.to(bufs[0].dtype))
+        scale = 1.0 / math.sqrt(self.D)
+        cur, nxt = bufs
+        for li, lw in enumerate(self.layers):
+            ops.stream_layer(cur, nxt, lw, s, k_caches[li], v_caches[li],
+                             self.rope, pos, ws, li, scale)
+            cur, nxt = nxt, cur
+        return ops.gemv_norm(cur, self.norm_f, self.lm_head, s.norm_eps)
+
     def _ssq_slots(self, B: int, device) -> torch.Tensor:
         """[2L, B] f32 sum-of-squares slots for the norm-precompute chain:
         slot 2i  = residual after layer i's O projection (feeds the MLP norm),

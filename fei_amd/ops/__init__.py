@@ -63,6 +63,12 @@ def _try_load() -> None:
     lib.fei_sample_onepass.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _f, _u64,
                                        _i, _vp]
     lib.fei_sample_shard.argtypes = [_vp, _vp, _vp, _i, _i, _i, _f, _u64, _vp]
+    lib.fei_stream_layer_check.argtypes = [_i, _i, _i, _i, _i]
+    lib.fei_stream_layer_check.restype = _i
+    lib.fei_stream_layer.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp, _vp,
+                                     _vp, _vp, _vp, _vp, _vp, _vp, _vp, _vp,
+                                     _vp, _vp, _i, _i, _i, _i, _i, _i, _i,
+                                     _f, _f, _vp]
     lib.fei_attn_decode_paged.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp,
                                           _i, _i, _i, _i, _i, _i, _i, _f,
                                           _l, _vp]
@@ -382,6 +388,55 @@ def sample_shard(logits: torch.Tensor, step: torch.Tensor,
                          int(v_offset), float(temperature), int(seed),
                          _stream())
     return out
+
+
+# -- persistent weight-streaming decode layer (csrc/stream_layer.hip) -------
+
+STREAM_NSPLIT = 32
+
+
+def stream_layer_check(C: int, Hq: int, Hkv: int, D: int, I: int) -> int:
+    """Residency + shape guard for the persistent layer engine. 0 = OK;
+    negative = unsupported (caller uses the launch path). GPU only."""
+    lib = require_lib()
+    return int(lib.fei_stream_layer_check(C, Hq, Hkv, D, I))
+
+
+def stream_workspace(spec, device) -> dict:
+    """Granule buffers (u64 {tag, payload}) + fail word for the stream
+    engine. ~1.3 MB for llama3-8b; zeroed by the engine whenever pos can
+    move backwards (prefill) — tags are pos-keyed otherwise."""
+    C, D = spec.hidden_size, spec.head_dim
+    Hq, Hkv, I = spec.num_heads, spec.num_kv_heads, spec.intermediate_size
+    mk = lambda n: torch.zeros(n, dtype=torch.int64, device=device)
+    return {
+        "g_qkv": mk((Hq + 2 * Hkv) * D),
+        "g_part": mk(Hq * STREAM_NSPLIT * (D + 2)),
+        "g_att": mk(Hq * D // 2),
+        "g_h2": mk(C // 2),
+        "g_act": mk(I),
+        "fail": torch.zeros(1, dtype=torch.int32, device=device),
+    }
+
+
+def stream_layer(x_in: torch.Tensor, h_out: torch.Tensor, lw, spec,
+                 k_cache: torch.Tensor, v_cache: torch.Tensor,
+                 cos_sin: torch.Tensor, pos: torch.Tensor, ws: dict,
+                 layer: int, scale: float) -> None:
+    """ONE persistent launch = one decode layer at batch 1 (S1 qkv GEMV ->
+    attention -> combine -> O GEMV -> SwiGLU GEMV -> down GEMV with the
+    LDS-DMA loader streaming weights ahead across every stage edge).
+    Replaces six launch-path kernels; see csrc/stream_layer.hip."""
+    lib = require_lib()
+    lib.fei_stream_layer(
+        _ptr(x_in), _ptr(h_out), _ptr(lw.wqkv), _ptr(lw.wo), _ptr(lw.wgu),
+        _ptr(lw.wdown), _ptr(lw.norm_attn), _ptr(lw.norm_mlp),
+        _ptr(k_cache), _ptr(v_cache), _ptr(cos_sin), _ptr(pos),
+        _ptr(ws["g_qkv"]), _ptr(ws["g_part"]), _ptr(ws["g_att"]),
+        _ptr(ws["g_h2"]), _ptr(ws["g_act"]), _ptr(ws["fail"]),
+        spec.hidden_size, spec.num_heads, spec.num_kv_heads, spec.head_dim,
+        spec.intermediate_size, k_cache.shape[-2], layer, spec.norm_eps,
+        scale, _stream())
 
 
 def advance(pos: torch.Tensor, step: torch.Tensor,

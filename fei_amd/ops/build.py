@@ -13,7 +13,7 @@ import sys
 
 CSRC = os.path.join(os.path.dirname(os.path.abspath(__file__)), "csrc")
 OUT = os.path.join(os.path.dirname(os.path.abspath(__file__)), "libfei_kernels.so")
-SOURCES = ["fei_kernels.hip", "attn_prefill.hip", "gemv.hip", "attn_decode_fused.hip", "gemv_fp8.hip"]
+SOURCES = ["fei_kernels.hip", "attn_prefill.hip", "gemv.hip", "attn_decode_fused.hip", "gemv_fp8.hip", "stream_layer.hip"]
 HIPCC = os.environ.get("HIPCC", "hipcc")
 ARCH = os.environ.get("FEI_AMD_ARCH", "gfx950")
 

@@ -1,0 +1,795 @@
+// Persistent weight-streaming decode layer (gfx950) — the §3.4 LDS-DMA
+// engine structure (MI355X_MICROARCH.md rows: engine-vs-launches 0.87-0.89x,
+// ldsdma-fill, prefetch-credit, gather-pass, nt-weights).
+//
+// ONE launch computes a whole Llama decode layer at batch 1:
+//   S1 qkv = rmsnorm(h)*wa @ Wqkv^T          (norm prologue fused)
+//   S2 split-K decode attention (in-kernel RoPE + KV append)
+//   S2b per-head split combine
+//   S3 h2 = h + att @ Wo^T                   (residual epilogue)
+//   S4 act = swiglu(rmsnorm(h2)*wm @ Wgu^T)
+//   S5 out = h2 + act @ Wdown^T
+// replacing six kernel launches; the per-CU loader wave streams the next
+// stage's weights (global_load_lds ... nt) ahead across every stage edge
+// while consumers gather hand-offs, which is where the engine's win lives.
+//
+// Geometry: grid = 256 workgroups (one per CU; >80 KiB dynamic LDS pins
+// residency), 256 threads = 4 waves: wave 3 = loader, waves 0-2 consumers.
+// Wave 0 is the lead consumer: it gathers each stage's input from the
+// granule buffers into LDS, preps the norm, runs this WG's attention
+// split, then joins slot processing.
+//
+// Cross-WG hand-offs use the Guideline-16 R2 granule protocol: one
+// naturally-aligned 8-byte {tag, payload} written by ONE relaxed
+// agent-scope (sc1) store, swept by relaxed agent loads — no fences, no
+// flags, placement-independent. tag = pos*1024 + layer*8 + stage, so tags
+// are unique across the step's 32 layer launches and across decode steps
+// (pos is device state, monotone within a decode run); the engine zeroes
+// the workspace whenever pos can move backwards (prefill).
+//
+// Every spin is bounded: on timeout the WG stamps the fail word and
+// RETURNS (no hang; the host checks the fail word at its next sync).
+//
+// Numerics mirror the launch-path kernels bit-for-bit where the operand
+// order matches (same bf16 rounding points: gemv.hip k_gemv_norm /
+// k_gemv_res, fei_kernels.hip k_attn_decode/_combine); reduction orders
+// differ, so tests compare against the fp32 torch reference with the same
+// tolerances as the launch path (tests/test_stream_gpu.py).
+#include "fei_common.h"
+
+namespace {
+
+typedef unsigned long long u64g;
+typedef __attribute__((address_space(1))) unsigned long long gu64;
+
+constexpr int NWG = 256;          // one workgroup per CU
+constexpr int SLOT_BYTES = 16384; // ring slot capacity (16 KiB)
+constexpr int RING_SLOTS = 7;     // 7 x 16 KiB = 112 KiB of the 160
+constexpr int SPIN_LIMIT = 1 << 21;
+constexpr int FLAG_FREE = 0;
+constexpr int FLAG_READY = 1;
+
+// dynamic-LDS carve offsets (bytes; all multiples of 16 — Guideline 17)
+constexpr int LDS_RING = 0;
+constexpr int LDS_X = RING_SLOTS * SLOT_BYTES;            // 8 KiB x-region
+constexpr int LDS_ACT = LDS_X + 8192;                     // 28 KiB act
+constexpr int LDS_CTRL = LDS_ACT + 28672;                 // control words
+constexpr int LDS_TOTAL = LDS_CTRL + 256;
+
+// control-word indices within the int32 control block
+constexpr int C_RINGF = 0;                 // [RING_SLOTS] ring flags
+constexpr int C_STAGE = C_RINGF + RING_SLOTS;  // stage-input-ready counter
+constexpr int C_DONE1 = C_STAGE + 1;       // consumers done with S1 slots
+constexpr int C_DONE3 = C_DONE1 + 1;
+constexpr int C_DONE4 = C_DONE3 + 1;
+
+#define dot8 dot8_bf16
+#define wave_sum wave_reduce_sum
+#define wave_max wave_reduce_max
+
+__device__ __forceinline__ u32 f2u(float f) {
+  union { float f; u32 i; } cv; cv.f = f; return cv.i;
+}
+__device__ __forceinline__ float u2f(u32 u) {
+  union { float f; u32 i; } cv; cv.i = u; return cv.f;
+}
+
+// one 1-KiB LDS-DMA piece: 64 lanes x 16 B, nt (streamed-once weights;
+// guide §5.7 recipe — M0 saved/restored in-statement, ring address passed
+// through the asm so the allocation cannot be eliminated)
+__device__ __forceinline__ void glds16_nt(const void* gsrc, unsigned lds_dst) {
+  unsigned keep;
+  asm volatile(
+      "s_mov_b32 %0, m0\n\t"
+      "s_mov_b32 m0, %2\n\t"
+      "s_nop 0\n\t"
+      "global_load_lds_dwordx4 %1, off nt\n\t"
+      "s_mov_b32 m0, %0"
+      : "=&s"(keep)
+      : "v"(gsrc), "s"(lds_dst)
+      : "memory");
+}
+
+// R2 granule ops (relaxed agent scope = sc1, untorn 8-byte)
+__device__ __forceinline__ void put_granule(u64g* g, unsigned tag,
+                                            unsigned payload) {
+  __hip_atomic_store((gu64*)g, ((u64g)tag << 32) | payload,
+                     __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+__device__ __forceinline__ u64g get_granule(const u64g* g) {
+  return __hip_atomic_load((const gu64*)g, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+}
+
+__device__ __forceinline__ unsigned pack_bf16(u16 lo, u16 hi) {
+  return (unsigned)lo | ((unsigned)hi << 16);
+}
+
+struct StreamArgs {
+  // activations / weights (bf16 unless noted)
+  const u16* x_in;        // [C] layer input residual h
+  u16* h_out;             // [C] layer output residual
+  const u16* wqkv;        // [Nqkv, C]
+  const u16* wo;          // [C, HqD]
+  const u16* wgu;         // [2I, C]
+  const u16* wdown;       // [C, I]
+  const u16* norm_attn;   // [C]
+  const u16* norm_mlp;    // [C]
+  u16* k_cache;           // [Hkv, max_seq, D]
+  u16* v_cache;
+  const float* cos_sin;   // [max_seq, D/2, 2]
+  const int* pos;         // [1]
+  // granule workspace (u64 each)
+  u64g* g_qkv;            // [Nqkv] f32 payload
+  u64g* g_part;           // [Hq, NSPLIT, D+2] f32 payload (o..., m, l)
+  u64g* g_att;            // [HqD] f32 payload
+  u64g* g_h2;             // [C/2] 2xbf16 payload
+  u64g* g_act;            // [I] f32 payload
+  int* fail;
+  // shape
+  int C, Hq, Hkv, D, I, max_seq, layer;
+  float eps, scale;
+};
+
+constexpr int NSPLIT = 32;
+
+// bounded LDS-flag spin; returns false on timeout (fail stamped)
+__device__ __forceinline__ bool lds_wait_eq(volatile int* w, int want,
+                                            int* fail, int code) {
+  int spins = 0;
+  while (__builtin_amdgcn_readfirstlane(*w) != want) {
+    __builtin_amdgcn_s_sleep(1);
+    if (++spins > SPIN_LIMIT) {
+      if ((threadIdx.x & 63) == 0) atomicExch(fail, code);
+      return false;
+    }
+  }
+  return true;
+}
+
+__device__ __forceinline__ bool lds_wait_ge(volatile int* w, int want,
+                                            int* fail, int code) {
+  int spins = 0;
+  while (__builtin_amdgcn_readfirstlane(*w) < want) {
+    __builtin_amdgcn_s_sleep(1);
+    if (++spins > SPIN_LIMIT) {
+      if ((threadIdx.x & 63) == 0) atomicExch(fail, code);
+      return false;
+    }
+  }
+  return true;
+}
+
+// ---------------------------------------------------------------------------
+// LOADER (wave 3): streams this WG's weight slots in stage order through the
+// LDS ring with nt LDS-DMA; one slot of run-ahead beyond the issuing slot
+// (counted s_waitcnt vmcnt gate — the 'thinned' loader of the gather-pass
+// row), ring-credit run-ahead across stage edges (prefetch-credit).
+// Slot map for one WG (8B shapes; rows are the WG's contiguous blocks):
+//   S1: 12 slots x 2 qkv rows (8 KiB row)
+//   S3:  8 slots x 2 o rows
+//   S4: 56 slots x {gate row i, up row i}
+//   S5: 2*DR slots = half down-rows (K*2/2 bytes each)
+// ---------------------------------------------------------------------------
+__device__ void loader_wave(const StreamArgs& a, char* lds, volatile int* ctrl,
+                            int wg, int lane) {
+  const int C = a.C, I = a.I, D = a.D;
+  const int Nqkv = (a.Hq + 2 * a.Hkv) * D;
+  const int r1 = Nqkv / NWG;                  // qkv rows per WG (24)
+  const int r3 = C / NWG;                     // o rows per WG (16)
+  const int r4 = I / NWG;                     // act rows per WG (56)
+  const int r5 = C / NWG;                     // down rows per WG (16)
+  const int s1 = r1 / 2, s3 = r3 / 2, s4 = r4, s5 = 2 * r5;
+  const int n_slots = s1 + s3 + s4 + s5;
+  const long rowb = (long)C * 2;              // 8 KiB row bytes
+  const long halfb = (long)I;                 // half down-row bytes (I*2/2)
+
+  int outstanding = 0;                        // slots issued, not yet gated
+  int pending_slot = -1;                      // slot awaiting READY publish
+  for (int s = 0; s < n_slots; ++s) {
+    const int slot = s % RING_SLOTS;
+    if (s >= RING_SLOTS) {
+      if (!lds_wait_eq(&ctrl[C_RINGF + slot], FLAG_FREE, a.fail, 10)) return;
+    }
+    const unsigned ring_base = __builtin_amdgcn_readfirstlane(
+        (unsigned)(unsigned long)(lds + LDS_RING + slot * SLOT_BYTES));
+    int pieces;
+    if (s < s1) {                             // S1: two qkv rows
+      const long row = (long)wg * r1 + (long)(s) * 2;
+      const char* src = (const char*)a.wqkv + row * rowb;
+      pieces = (int)(2 * rowb) / 1024;
+      for (int j = 0; j < pieces; ++j)
+        glds16_nt(src + (long)j * 1024 + lane * 16, ring_base + j * 1024);
+    } else if (s < s1 + s3) {                 // S3: two o rows
+      const long row = (long)wg * r3 + (long)(s - s1) * 2;
+      const char* src = (const char*)a.wo + row * rowb;
+      pieces = (int)(2 * rowb) / 1024;
+      for (int j = 0; j < pieces; ++j)
+        glds16_nt(src + (long)j * 1024 + lane * 16, ring_base + j * 1024);
+    } else if (s < s1 + s3 + s4) {            // S4: gate row + up row
+      const long n = (long)wg * r4 + (s - s1 - s3);
+      const char* gsrc = (const char*)a.wgu + n * rowb;
+      const char* usrc = (const char*)a.wgu + ((long)I + n) * rowb;
+      const int half = (int)rowb / 1024;      // 8 pieces each
+      for (int j = 0; j < half; ++j)
+        glds16_nt(gsrc + (long)j * 1024 + lane * 16, ring_base + j * 1024);
+      for (int j = 0; j < half; ++j)
+        glds16_nt(usrc + (long)j * 1024 + lane * 16,
+                  ring_base + (half + j) * 1024);
+      pieces = 2 * half;
+    } else {                                  // S5: half a down row
+      const int hs = s - s1 - s3 - s4;
+      const long row = (long)wg * r5 + hs / 2;
+      const char* src = (const char*)a.wdown + row * (halfb * 2)
+                        + (hs & 1) * halfb;
+      pieces = (int)halfb / 1024;             // 14 for I=14336
+      for (int j = 0; j < pieces; ++j)
+        glds16_nt(src + (long)j * 1024 + lane * 16, ring_base + j * 1024);
+    }
+    // gate: leave only the just-issued slot's pieces in flight (counted
+    // wait = its piece count: vmcnt retires in issue order, so everything
+    // older — the previous slot — has landed), then publish the PREVIOUS
+    // slot. hipcc does not track asm loads; the counts are ours (§5.7).
+    (void)outstanding;
+    if (pending_slot >= 0) {
+      if (pieces >= 16)
+        asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
+      else if (pieces >= 14)
+        asm volatile("s_waitcnt vmcnt(14)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      if (lane == 0) ctrl[C_RINGF + pending_slot] = FLAG_READY;
+    }
+    pending_slot = slot;
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  if (pending_slot >= 0 && lane == 0)
+    ctrl[C_RINGF + pending_slot] = FLAG_READY;
+}
+
+// ---------------------------------------------------------------------------
+// consumer slot processing helpers
+// ---------------------------------------------------------------------------
+
+// dot one weight row (LDS, base byte offset) against x (LDS bf16[C])
+__device__ __forceinline__ float lds_row_dot(const char* lds, int row_off,
+                                             const char* xoff, int C,
+                                             int lane) {
+  const s16x8* w = (const s16x8*)(lds + row_off);
+  const s16x8* x = (const s16x8*)xoff;
+  float acc = 0.f;
+  const int nv = C >> 3;
+  for (int i = lane; i < nv; i += 64) acc += dot8(w[i], x[i]);
+  return wave_sum(acc);
+}
+
+// ---------------------------------------------------------------------------
+// the kernel
+// ---------------------------------------------------------------------------
+template <int G>
+__global__ void __launch_bounds__(256, 1)
+k_stream_layer(StreamArgs a) {
+  extern __shared__ __attribute__((aligned(16))) char lds[];
+  volatile int* ctrl = (volatile int*)(lds + LDS_CTRL);
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wg = blockIdx.x;
+  const int C = a.C, D = a.D, I = a.I, Hq = a.Hq, Hkv = a.Hkv;
+  const int Nqkv = (Hq + 2 * Hkv) * D;
+  const int r1 = Nqkv / NWG, r3 = C / NWG, r4 = I / NWG, r5 = C / NWG;
+  const int s1 = r1 / 2, s3 = r3 / 2, s4 = r4, s5 = 2 * r5;
+  const int pos_now = a.pos[0];
+  const unsigned tagbase = (unsigned)pos_now * 1024u + (unsigned)a.layer * 8u;
+
+  if (tid < 64) {                       // zero control words (one wave)
+    for (int i = tid; i < 64; i += 64) ctrl[i] = 0;
+  }
+  __syncthreads();                      // before ANY glds is issued
+
+  if (wave == 3) {
+    loader_wave(a, lds, ctrl, wg, lane);
+    return;
+  }
+
+  // ------------------------------------------------------------------ S1 --
+  u16* xl = (u16*)(lds + LDS_X);        // stage-input LDS region (bf16[C])
+  if (wave == 0) {
+    // norm prologue from the layer input tensor (plain loads; the input
+    // was written by the previous launch — boundary-synchronised)
+    const s16x8* xr = (const s16x8*)a.x_in;
+    const s16x8* wn = (const s16x8*)a.norm_attn;
+    float ss = 0.f;
+    const int nv = C >> 3;
+    for (int i = lane; i < nv; i += 64) {
+      s16x8 v = xr[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = bf2f((u16)v[j]);
+        ss = fmaf(f, f, ss);
+      }
+    }
+    const float inv = rsqrtf(wave_sum(ss) / (float)C + a.eps);
+    for (int i = lane; i < nv; i += 64) {
+      s16x8 v = xr[i], w = wn[i], o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = (short)f2bf(bf2f((u16)v[j]) * inv * bf2f((u16)w[j]));
+      ((s16x8*)xl)[i] = o;
+    }
+    __threadfence_block();
+    if (lane == 0) ctrl[C_STAGE] = 1;
+  } else {
+    if (!lds_wait_ge(&ctrl[C_STAGE], 1, a.fail, 20)) return;
+  }
+
+  // S1 slots: 2 qkv rows each, slot s -> wave s%3; publish one granule
+  // (f32 payload per output value, bf16-rounded like the launch GEMV)
+  for (int s = wave; s < s1; s += 3) {
+    const int slot = s % RING_SLOTS;
+    if (!lds_wait_eq(&ctrl[C_RINGF + slot], FLAG_READY, a.fail, 21)) return;
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    const float v0 = lds_row_dot(lds, LDS_RING + slot * SLOT_BYTES,
+                                 (const char*)xl, C, lane);
+    const float v1 = lds_row_dot(lds, LDS_RING + slot * SLOT_BYTES + C * 2,
+                                 (const char*)xl, C, lane);
+    if (lane == 0) {
+      const long row = (long)wg * r1 + (long)s * 2;
+      put_granule(&a.g_qkv[row], tagbase + 1, f2u(bf2f(f2bf(v0))));
+      put_granule(&a.g_qkv[row + 1], tagbase + 1, f2u(bf2f(f2bf(v1))));
+      ctrl[C_RINGF + slot] = FLAG_FREE;
+    }
+  }
+  if (lane == 0) atomicAdd((int*)&ctrl[C_DONE1], 1);
+
+  // ------------------------------------------------------------------ S2 --
+  // wave 0: this WG's attention split (kvh = wg/NSPLIT, split = wg%NSPLIT).
+  // wave 1 (wg < Hq): S2b combine for head wg. wave 2: straight to S3 wait.
+  if (wave == 0) {
+    // wait until all 3 consumer waves left S1 (xl region is reused)
+    if (!lds_wait_ge(&ctrl[C_DONE1], 3, a.fail, 22)) return;
+    const int kvh = wg / NSPLIT;
+    const int split = wg % NSPLIT;
+    const int n = pos_now + 1;
+    const int chunk = (n + NSPLIT - 1) / NSPLIT;
+    const int start = split * chunk;
+    const int end = min(start + chunk, n);
+    const int half = D / 2;
+    const int p_new = n - 1;
+    // scratch in the x-region: qs [G][D] f32, pl [G][64] f32, kn/vn bf16[D]
+    float* qs = (float*)(lds + LDS_X);                    // G*128*4 = 2 KiB
+    float* pl = qs + G * D;                               // G*64*4  = 1 KiB
+    u16* kn = (u16*)(pl + G * 64);                        // 256 B
+    u16* vn = kn + D;                                     // 256 B
+    const unsigned qtag = tagbase + 1;
+
+    if (start < end) {
+      // gather + rope q for this group's heads (sweep its qkv granules)
+      {
+        int spins = 0;
+        for (;;) {
+          bool ok = true;
+          for (int i = lane; i < G * D; i += 64) {
+            const int g = i / D, d = i % D;
+            const u64g x = get_granule(&a.g_qkv[(long)(kvh * G + g) * D + d]);
+            ok &= (unsigned)(x >> 32) == qtag;
+            qs[g * D + d] = u2f((unsigned)x);             // raw bf16-as-f32
+          }
+          if (__all(ok)) break;
+          __builtin_amdgcn_s_sleep(1);
+          if (++spins > SPIN_LIMIT) {
+            if (lane == 0) atomicExch(a.fail, 23);
+            return;
+          }
+        }
+        // rope + scale in place (pairs d, d+half)
+        for (int i = lane; i < G * half; i += 64) {
+          const int g = i / half, d = i % half;
+          const float c = a.cos_sin[((long)p_new * half + d) * 2 + 0];
+          const float sn = a.cos_sin[((long)p_new * half + d) * 2 + 1];
+          const float x1 = qs[g * D + d], x2 = qs[g * D + d + half];
+          qs[g * D + d] = (x1 * c - x2 * sn) * a.scale;
+          qs[g * D + d + half] = (x2 * c + x1 * sn) * a.scale;
+        }
+      }
+      // gather + rope the new token's k,v (qkv indices HqD.. / HqD+HkvD..)
+      {
+        int spins = 0;
+        for (;;) {
+          bool ok = true;
+          for (int i = lane; i < 2 * D; i += 64) {
+            const long idx = (i < D)
+                ? (long)Hq * D + (long)kvh * D + i
+                : (long)(Hq + Hkv) * D + (long)kvh * D + (i - D);
+            const u64g x = get_granule(&a.g_qkv[idx]);
+            ok &= (unsigned)(x >> 32) == qtag;
+            const u16 b = f2bf(u2f((unsigned)x));
+            if (i < D) kn[i] = b; else vn[i - D] = b;
+          }
+          if (__all(ok)) break;
+          __builtin_amdgcn_s_sleep(1);
+          if (++spins > SPIN_LIMIT) {
+            if (lane == 0) atomicExch(a.fail, 24);
+            return;
+          }
+        }
+        // rope k_new in LDS; owner split also appends to the caches
+        for (int d = lane; d < half; d += 64) {
+          const float c = a.cos_sin[((long)p_new * half + d) * 2 + 0];
+          const float sn = a.cos_sin[((long)p_new * half + d) * 2 + 1];
+          const float x1 = bf2f(kn[d]), x2 = bf2f(kn[d + half]);
+          const u16 k0 = f2bf(x1 * c - x2 * sn);
+          const u16 k1 = f2bf(x2 * c + x1 * sn);
+          kn[d] = k0;
+          kn[d + half] = k1;
+          if (p_new >= start && p_new < end) {
+            u16* kcp = a.k_cache + ((long)kvh * a.max_seq + p_new) * D;
+            u16* vcp = a.v_cache + ((long)kvh * a.max_seq + p_new) * D;
+            kcp[d] = k0; kcp[d + half] = k1;
+            vcp[d] = vn[d]; vcp[d + half] = vn[d + half];
+          }
+        }
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+      // online softmax over [start, end), 64-key tiles, lane-per-key
+      const u16* kbase = a.k_cache + (long)kvh * a.max_seq * D;
+      const u16* vbase = a.v_cache + (long)kvh * a.max_seq * D;
+      float m[G], l[G], o0[G], o1[G];
+#pragma unroll
+      for (int g = 0; g < G; ++g) {
+        m[g] = -1.0f / 0.0f; l[g] = 0.f; o0[g] = 0.f; o1[g] = 0.f;
+      }
+      for (int tile = start; tile < end; tile += 64) {
+        const int kk = tile + lane;
+        float sc[G];
+#pragma unroll
+        for (int g = 0; g < G; ++g) sc[g] = -1.0f / 0.0f;
+        if (kk < end) {
+          const bool is_new = (kk == p_new);
+          const s16x8* krow = is_new ? (const s16x8*)kn
+                                     : (const s16x8*)(kbase + (long)kk * D);
+#pragma unroll
+          for (int g = 0; g < G; ++g) sc[g] = 0.f;
+          for (int i = 0; i < D / 8; ++i) {
+            const s16x8 kv8 = krow[i];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              const float kf = bf2f((u16)kv8[j]);
+#pragma unroll
+              for (int g = 0; g < G; ++g)
+                sc[g] = fmaf(qs[g * D + i * 8 + j], kf, sc[g]);
+            }
+          }
+        }
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          const float tile_m = wave_max(sc[g]);
+          const float m_new = fmaxf(m[g], tile_m);
+          const float alpha = __expf(m[g] - m_new);
+          const float p = (kk < end) ? __expf(sc[g] - m_new) : 0.f;
+          pl[g * 64 + lane] = p;
+          m[g] = m_new;
+          l[g] = l[g] * alpha + wave_sum(p);
+          o0[g] *= alpha; o1[g] *= alpha;
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        // P*V: lane covers dims (2*lane, 2*lane+1); rows read coalesced
+        const int kmax = min(64, end - tile);
+        for (int kl = 0; kl < kmax; ++kl) {
+          const bool is_new = (tile + kl == p_new);
+          const u16* vrow = is_new ? vn + lane * 2
+                                   : vbase + (long)(tile + kl) * D + lane * 2;
+          const float v0 = bf2f(vrow[0]);
+          const float v1 = bf2f(vrow[1]);
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+            const float p = pl[g * 64 + kl];
+            o0[g] = fmaf(p, v0, o0[g]);
+            o1[g] = fmaf(p, v1, o1[g]);
+          }
+        }
+      }
+      // publish partials: [Hq, NSPLIT, D+2] f32 granules
+      for (int g = 0; g < G; ++g) {
+        const int hq = kvh * G + g;
+        u64g* base = a.g_part + ((long)hq * NSPLIT + split) * (D + 2);
+        put_granule(&base[2 * lane], tagbase + 2, f2u(o0[g]));
+        put_granule(&base[2 * lane + 1], tagbase + 2, f2u(o1[g]));
+        if (lane == 0) {
+          put_granule(&base[D], tagbase + 2, f2u(m[g]));
+          put_granule(&base[D + 1], tagbase + 2, f2u(l[g]));
+        }
+      }
+    } else {
+      // empty split: publish -inf/0 partials (combine sweeps all splits)
+      for (int g = 0; g < G; ++g) {
+        const int hq = kvh * G + g;
+        u64g* base = a.g_part + ((long)hq * NSPLIT + split) * (D + 2);
+        put_granule(&base[2 * lane], tagbase + 2, f2u(0.f));
+        put_granule(&base[2 * lane + 1], tagbase + 2, f2u(0.f));
+        if (lane == 0) {
+          put_granule(&base[D], tagbase + 2, f2u(-1.0f / 0.0f));
+          put_granule(&base[D + 1], tagbase + 2, f2u(0.f));
+        }
+      }
+    }
+  }
+
+  // S2b: combine head `wg` (wave 1 of WGs 0..Hq-1); mirrors
+  // k_attn_decode_combine (split-major sweep, max-shifted exp weights)
+  if (wave == 1 && wg < Hq) {
+    const u64g* base = a.g_part + (long)wg * NSPLIT * (D + 2);
+    // sweep m/l for every split first (lane s covers split s)
+    float sm = -1.0f / 0.0f, sl = 0.f;
+    {
+      int spins = 0;
+      for (;;) {
+        bool ok = true;
+        if (lane < NSPLIT) {
+          const u64g xm = get_granule(&base[(long)lane * (D + 2) + D]);
+          const u64g xlv = get_granule(&base[(long)lane * (D + 2) + D + 1]);
+          ok = ((unsigned)(xm >> 32) == tagbase + 2) &
+              ((unsigned)(xlv >> 32) == tagbase + 2);
+          sm = u2f((unsigned)xm);
+          sl = u2f((unsigned)xlv);
+        }
+        if (__all(ok)) break;
+        __builtin_amdgcn_s_sleep(1);
+        if (++spins > SPIN_LIMIT) {
+          if (lane == 0) atomicExch(a.fail, 25);
+          return;
+        }
+      }
+    }
+    const float mstar = wave_max(sm);
+    const float wexp = (lane < NSPLIT) ? __expf(sm - mstar) : 0.f;
+    const float ltot = wave_sum(wexp * sl);
+    // o: lane covers dims (2*lane, 2*lane+1); serial over splits with a
+    // bounded per-split granule wait
+    float acc0 = 0.f, acc1 = 0.f;
+    for (int s = 0; s < NSPLIT; ++s) {
+      const u64g* sb = &base[(long)s * (D + 2)];
+      u64g x0, x1v;
+      int spins = 0;
+      for (;;) {
+        x0 = get_granule(&sb[2 * lane]);
+        x1v = get_granule(&sb[2 * lane + 1]);
+        if (__all(((unsigned)(x0 >> 32) == tagbase + 2) &
+                  ((unsigned)(x1v >> 32) == tagbase + 2)))
+          break;
+        __builtin_amdgcn_s_sleep(1);
+        if (++spins > SPIN_LIMIT) {
+          if (lane == 0) atomicExch(a.fail, 26);
+          return;
+        }
+      }
+      const float w = __shfl(wexp, s);        // split s's exp weight
+      acc0 = fmaf(w, u2f((unsigned)x0), acc0);
+      acc1 = fmaf(w, u2f((unsigned)x1v), acc1);
+    }
+    const float inv_l = ltot > 0.f ? 1.f / ltot : 0.f;
+    const u16 b0 = f2bf(acc0 * inv_l);
+    const u16 b1 = f2bf(acc1 * inv_l);
+    put_granule(&a.g_att[(long)wg * (D / 2) + lane], tagbase + 3,
+                pack_bf16(b0, b1));
+  }
+
+  // ------------------------------------------------------------------ S3 --
+  // wave 0 gathers att into the x-region (all heads, bf16)
+  if (wave == 0) {
+    u64g* src = a.g_att;
+    const int n_g = Hq * D / 2;                      // granules (2 vals each)
+    int spins = 0;
+    for (;;) {
+      bool ok = true;
+      for (int i = lane; i < n_g; i += 64) {
+        const u64g x = get_granule(&src[i]);
+        ok &= (unsigned)(x >> 32) == tagbase + 3;
+        ((unsigned*)xl)[i] = (unsigned)x;            // 2 bf16 at once
+      }
+      if (__all(ok)) break;
+      __builtin_amdgcn_s_sleep(1);
+      if (++spins > SPIN_LIMIT) {
+        if (lane == 0) atomicExch(a.fail, 27);
+        return;
+      }
+    }
+    __threadfence_block();
+    if (lane == 0) ctrl[C_STAGE] = 2;
+  } else {
+    if (!lds_wait_ge(&ctrl[C_STAGE], 2, a.fail, 28)) return;
+  }
+
+  // S3 slots: 2 o rows; publish h2 granules (2 bf16) with residual add
+  for (int s = wave; s < s3; s += 3) {
+    const int slot = (s1 + s) % RING_SLOTS;
+    if (!lds_wait_eq(&ctrl[C_RINGF + slot], FLAG_READY, a.fail, 29)) return;
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    const float v0 = lds_row_dot(lds, LDS_RING + slot * SLOT_BYTES,
+                                 (const char*)xl, Hq * D, lane);
+    const float v1 = lds_row_dot(lds, LDS_RING + slot * SLOT_BYTES
+                                 + Hq * D * 2, (const char*)xl, Hq * D, lane);
+    if (lane == 0) {
+      const long row = (long)wg * r3 + (long)s * 2;
+      const u16 h0 = f2bf(bf2f(a.x_in[row]) + v0);
+      const u16 h1 = f2bf(bf2f(a.x_in[row + 1]) + v1);
+      put_granule(&a.g_h2[row / 2], tagbase + 4, pack_bf16(h0, h1));
+      ctrl[C_RINGF + slot] = FLAG_FREE;
+    }
+  }
+  if (lane == 0) atomicAdd((int*)&ctrl[C_DONE3], 1);
+
+  // ------------------------------------------------------------------ S4 --
+  // wave 0: gather h2, norm-prologue into the x-region (xl reused — wait
+  // for all consumers to leave S3)
+  if (wave == 0) {
+    if (!lds_wait_ge(&ctrl[C_DONE3], 3, a.fail, 30)) return;
+    float ss = 0.f;
+    int spins = 0;
+    for (;;) {
+      bool ok = true;
+      ss = 0.f;
+      for (int i = lane; i < C / 2; i += 64) {
+        const u64g x = get_granule(&a.g_h2[i]);
+        ok &= (unsigned)(x >> 32) == tagbase + 4;
+        const unsigned p = (unsigned)x;
+        ((unsigned*)xl)[i] = p;                      // raw h2 pair
+        const float f0 = bf2f((u16)(p & 0xffff));
+        const float f1 = bf2f((u16)(p >> 16));
+        ss = fmaf(f0, f0, fmaf(f1, f1, ss));
+      }
+      if (__all(ok)) break;
+      __builtin_amdgcn_s_sleep(1);
+      if (++spins > SPIN_LIMIT) {
+        if (lane == 0) atomicExch(a.fail, 31);
+        return;
+      }
+    }
+    const float inv = rsqrtf(wave_sum(ss) / (float)C + a.eps);
+    const s16x8* wn = (const s16x8*)a.norm_mlp;
+    for (int i = lane; i < (C >> 3); i += 64) {
+      s16x8 v = ((s16x8*)xl)[i], w = wn[i], o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = (short)f2bf(bf2f((u16)v[j]) * inv * bf2f((u16)w[j]));
+      ((s16x8*)xl)[i] = o;
+    }
+    __threadfence_block();
+    if (lane == 0) ctrl[C_STAGE] = 3;
+  } else {
+    if (!lds_wait_ge(&ctrl[C_STAGE], 3, a.fail, 32)) return;
+  }
+
+  // S4 slots: {gate row, up row}; publish act granule (f32 payload)
+  for (int s = wave; s < s4; s += 3) {
+    const int slot = (s1 + s3 + s) % RING_SLOTS;
+    if (!lds_wait_eq(&ctrl[C_RINGF + slot], FLAG_READY, a.fail, 33)) return;
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    const float g = lds_row_dot(lds, LDS_RING + slot * SLOT_BYTES,
+                                (const char*)xl, C, lane);
+    const float u = lds_row_dot(lds, LDS_RING + slot * SLOT_BYTES + C * 2,
+                                (const char*)xl, C, lane);
+    if (lane == 0) {
+      const long n = (long)wg * r4 + s;
+      const float silu = g / (1.f + __expf(-g));
+      const float act = bf2f(f2bf(silu * u));        // launch-path rounding
+      put_granule(&a.g_act[n], tagbase + 5, f2u(act));
+      ctrl[C_RINGF + slot] = FLAG_FREE;
+    }
+  }
+  if (lane == 0) atomicAdd((int*)&ctrl[C_DONE4], 1);
+
+  // ------------------------------------------------------------------ S5 --
+  // wave 0 gathers act (bf16) into the ACT region
+  u16* actl = (u16*)(lds + LDS_ACT);
+  if (wave == 0) {
+    if (!lds_wait_ge(&ctrl[C_DONE4], 3, a.fail, 34)) return;
+    int spins = 0;
+    for (;;) {
+      bool ok = true;
+      for (int i = lane; i < I; i += 64) {
+        const u64g x = get_granule(&a.g_act[i]);
+        ok &= (unsigned)(x >> 32) == tagbase + 5;
+        actl[i] = f2bf(u2f((unsigned)x));
+      }
+      if (__all(ok)) break;
+      __builtin_amdgcn_s_sleep(1);
+      if (++spins > SPIN_LIMIT) {
+        if (lane == 0) atomicExch(a.fail, 35);
+        return;
+      }
+    }
+    __threadfence_block();
+    if (lane == 0) ctrl[C_STAGE] = 4;
+  } else {
+    if (!lds_wait_ge(&ctrl[C_STAGE], 4, a.fail, 36)) return;
+  }
+
+  // S5 slots: half down-rows; row r = slots (2r, 2r+1) -> wave r%3; the
+  // wave accumulates across the pair and writes the layer output row.
+  for (int r = wave; r < r5; r += 3) {
+    float acc = 0.f;
+    for (int hs = 0; hs < 2; ++hs) {
+      const int s = 2 * r + hs;
+      const int slot = (s1 + s3 + s4 + s) % RING_SLOTS;
+      if (!lds_wait_eq(&ctrl[C_RINGF + slot], FLAG_READY, a.fail, 37)) return;
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      const s16x8* w = (const s16x8*)(lds + LDS_RING + slot * SLOT_BYTES);
+      const s16x8* x = (const s16x8*)(actl + (long)hs * (I / 2));
+      float part = 0.f;
+      const int nv = (I / 2) >> 3;
+      for (int i = lane; i < nv; i += 64) part += dot8(w[i], x[i]);
+      acc += wave_sum(part);
+      if (lane == 0) ctrl[C_RINGF + slot] = FLAG_FREE;
+    }
+    if (lane == 0) {
+      const long row = (long)wg * r5 + r;
+      // residual: re-read own h2 granule (value published this launch)
+      const u64g x = get_granule(&a.g_h2[row / 2]);
+      const unsigned p = (unsigned)x;
+      const float h2 = bf2f((u16)((row & 1) ? (p >> 16) : (p & 0xffff)));
+      a.h_out[row] = f2bf(h2 + acc);
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+// residency + shape guard; returns 0 on OK (launchable), negative on
+// refusal (caller falls back to the launch path)
+int fei_stream_layer_check(int C, int Hq, int Hkv, int D, int I) {
+  const int Nqkv = (Hq + 2 * Hkv) * D;
+  if (D != 128 || Hq / Hkv > 8 || Hq % Hkv) return -2;
+  if (Nqkv % (2 * NWG) || C % (2 * NWG) || I % NWG || (I / 2) % 1024)
+    return -3;
+  if (C * 2 > 8192 || I * 2 > 28672) return -4;   // LDS x/act regions
+  if (I > SLOT_BYTES) return -7;                  // half down-row per slot
+  if (Hq > 64 || NSPLIT != 32) return -5;
+  int max_blocks = 0;
+  hipError_t e = hipOccupancyMaxActiveBlocksPerMultiprocessor(
+      &max_blocks, (const void*)k_stream_layer<4>, 256, LDS_TOTAL);
+  if (e != hipSuccess || max_blocks < 1) return -1;
+  hipDeviceProp_t prop;
+  if (hipGetDeviceProperties(&prop, 0) != hipSuccess) return -1;
+  if (prop.multiProcessorCount < NWG) return -6;
+  return 0;
+}
+
+void fei_stream_layer(const void* x_in, void* h_out, const void* wqkv,
+                      const void* wo, const void* wgu, const void* wdown,
+                      const void* norm_attn, const void* norm_mlp,
+                      void* k_cache, void* v_cache, const float* cos_sin,
+                      const int* pos, void* g_qkv, void* g_part, void* g_att,
+                      void* g_h2, void* g_act, int* fail,
+                      int C, int Hq, int Hkv, int D, int I, int max_seq,
+                      int layer, float eps, float scale,
+                      hipStream_t stream) {
+  StreamArgs a;
+  a.x_in = (const u16*)x_in; a.h_out = (u16*)h_out;
+  a.wqkv = (const u16*)wqkv; a.wo = (const u16*)wo;
+  a.wgu = (const u16*)wgu; a.wdown = (const u16*)wdown;
+  a.norm_attn = (const u16*)norm_attn; a.norm_mlp = (const u16*)norm_mlp;
+  a.k_cache = (u16*)k_cache; a.v_cache = (u16*)v_cache;
+  a.cos_sin = cos_sin; a.pos = pos;
+  a.g_qkv = (u64g*)g_qkv; a.g_part = (u64g*)g_part; a.g_att = (u64g*)g_att;
+  a.g_h2 = (u64g*)g_h2; a.g_act = (u64g*)g_act; a.fail = fail;
+  a.C = C; a.Hq = Hq; a.Hkv = Hkv; a.D = D; a.I = I; a.max_seq = max_seq;
+  a.layer = layer; a.eps = eps; a.scale = scale;
+  const int G = Hq / Hkv;
+#define LSL(GV) hipLaunchKernelGGL((k_stream_layer<GV>), dim3(NWG), \
+    dim3(256), LDS_TOTAL, stream, a)
+  switch (G) {
+    case 1: LSL(1); break;
+    case 2: LSL(2); break;
+    case 4: LSL(4); break;
+    case 8: LSL(8); break;
+    default: break;   // fei_stream_layer_check validates
+  }
+#undef LSL
+}
+
+}  // extern "C"
