@@ -218,14 +218,121 @@ class CodeEditor:
             f.write(content)
         return {"success": True, "file": file_path, "message": "file written"}
 
-    def _validate_code(self, file_path: str, content: str) -> Optional[str]:
-        """Post-edit validation. Python via ast; other languages unchecked.
-        Returns an error message or None."""
-        if file_path.endswith(".py"):
-            try:
-                ast.parse(content)
-            except SyntaxError as e:
-                return f"python syntax error after edit: {e}"
+    # -- post-edit validators (reference code.py:827-932: ast, esprima,
+    # pylint, flake8 — each optional, skipped gracefully when absent) ------
+
+    @staticmethod
+    def _v_ast(content: str) -> Optional[str]:
+        try:
+            ast.parse(content)
+        except SyntaxError as e:
+            return f"python syntax error after edit: {e}"
+        return None
+
+    @staticmethod
+    def _v_pyflakes(content: str) -> Optional[str]:
+        try:
+            from io import StringIO
+
+            from pyflakes.api import check
+            from pyflakes.reporter import Reporter
+        except ImportError:
+            return None                     # optional linter absent: skip
+        out, err = StringIO(), StringIO()
+        n = check(content, "<edit>", Reporter(out, err))
+        hard = [ln for ln in (out.getvalue() + err.getvalue()).splitlines()
+                if "undefined name" in ln or "syntax" in ln.lower()]
+        if n and hard:
+            return "pyflakes: " + "; ".join(hard[:3])
+        return None
+
+    @staticmethod
+    def _v_esprima(content: str) -> Optional[str]:
+        try:
+            import esprima
+        except ImportError:
+            return None
+        try:
+            esprima.parseScript(content, tolerant=False)
+        except Exception as e:
+            return f"javascript syntax error after edit: {e}"
+        return None
+
+    @staticmethod
+    def _v_json(content: str) -> Optional[str]:
+        import json as _json
+        try:
+            _json.loads(content)
+        except ValueError as e:
+            return f"json syntax error after edit: {e}"
+        return None
+
+    @staticmethod
+    def _v_yaml(content: str) -> Optional[str]:
+        try:
+            import yaml
+        except ImportError:
+            return None
+        try:
+            yaml.safe_load(content)
+        except Exception as e:
+            return f"yaml syntax error after edit: {e}"
+        return None
+
+    @staticmethod
+    def _v_treesitter(content: str, ext: str) -> Optional[str]:
+        """Generic syntax gate for any language with a grammar: parse and
+        reject when the tree contains ERROR nodes. Skipped when
+        tree-sitter is unavailable (optional in this image)."""
+        from fei_amd.tools.repomap import _TS_EXT_LANG, _ts_tools
+        lang = _TS_EXT_LANG.get(ext)
+        if not lang:
+            return None
+        tools = _ts_tools(lang)
+        if tools is None:
+            return None
+        try:
+            tree = tools[0].parse(content.encode("utf-8", errors="replace"))
+            root = tree.root_node
+            if getattr(root, "has_error", False):
+                return f"{lang} syntax error after edit (tree-sitter)"
+        except Exception:
+            return None
+        return None
+
+    VALIDATORS = {
+        "ast": "_v_ast", "pyflakes": "_v_pyflakes", "esprima": "_v_esprima",
+        "json": "_v_json", "yaml": "_v_yaml", "treesitter": "_v_treesitter",
+    }
+    _EXT_VALIDATORS = {
+        ".py": ["ast", "pyflakes"],
+        ".js": ["esprima", "treesitter"], ".jsx": ["esprima", "treesitter"],
+        ".ts": ["treesitter"], ".tsx": ["treesitter"],
+        ".json": ["json"],
+        ".yaml": ["yaml"], ".yml": ["yaml"],
+        ".c": ["treesitter"], ".h": ["treesitter"], ".cpp": ["treesitter"],
+        ".hpp": ["treesitter"], ".cc": ["treesitter"], ".go": ["treesitter"],
+        ".rs": ["treesitter"], ".java": ["treesitter"], ".rb": ["treesitter"],
+    }
+
+    def _validate_code(self, file_path: str, content: str,
+                       validators: Optional[List[str]] = None
+                       ) -> Optional[str]:
+        """Post-edit validation; returns an error message or None. The
+        validator set defaults by extension; optional checkers (pyflakes,
+        esprima, yaml, tree-sitter) skip silently when not importable —
+        the reference's esprima/pylint/flake8 behavior (code.py:827-932)."""
+        ext = os.path.splitext(file_path)[1]
+        names = validators if validators is not None else \
+            self._EXT_VALIDATORS.get(ext, [])
+        for name in names:
+            attr = self.VALIDATORS.get(name)
+            if not attr:
+                continue
+            fn = getattr(self, attr)
+            err = fn(content, ext) if name == "treesitter" else fn(content)
+            if err:
+                return err
         return None
 
     def regex_replace(
@@ -251,11 +358,11 @@ class CodeEditor:
         new_content, n = rx.subn(replacement, content, count=count)
         if n == 0:
             return {"error": "pattern did not match"}
-        run_ast = validate and file_path.endswith(".py")
-        if validators is not None:
-            run_ast = "ast" in validators
-        if run_ast:
-            err = self._validate_code(file_path, new_content)
+        if validate or validators is not None:
+            # validators=None: defaults by extension; []: disabled;
+            # explicit list: exactly those (reference semantics)
+            err = self._validate_code(file_path, new_content,
+                                      validators=validators)
             if err:
                 return {"error": err, "replacements": 0}
         backup = self._backup(file_path)

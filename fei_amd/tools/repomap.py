@@ -4,10 +4,10 @@ Parity: the reference's aider-style RepoMapper (fei/tools/repomap.py:68-544):
 file walk with excludes, symbol extraction, a cross-file symbol-reference
 dependency graph, PageRank-ish file ranking, and budgeted formatting.
 
-Design difference: the reference used tree-sitter with regex fallback
-(repomap.py:160-281,324-389); we use the regex extractor as the primary
-path (tree-sitter is optional in this image) plus Python's ``ast`` for
-exact Python symbols.
+Symbol extraction mirrors the reference's ladder (repomap.py:160-281,
+324-389): Python via the exact ``ast`` parser, every other language via
+tree-sitter per-language queries WHEN ``tree_sitter_languages`` is
+importable (optional in this image), regex table as the final fallback.
 """
 
 from __future__ import annotations
@@ -45,6 +45,135 @@ _IMPORT_RE = [
     re.compile(r"""^\s*(?:import|export).*?from\s+['"]([^'"]+)['"]""", re.M),
     re.compile(r"""^\s*#include\s+[<"]([^>"]+)[>"]""", re.M),
 ]
+
+# -- tree-sitter symbol extraction (reference repomap.py:160-281) ------------
+# Used when ``tree_sitter_languages`` is importable (it is optional in this
+# image); per-language capture queries mirror the reference's set
+# (repomap.py:244-281). Python keeps the exact ``ast`` extractor; everything
+# else prefers tree-sitter and falls back to the regex table.
+
+_TS_EXT_LANG = {
+    ".js": "javascript", ".jsx": "javascript",
+    ".ts": "typescript", ".tsx": "tsx",
+    ".c": "c", ".h": "c",
+    ".cpp": "cpp", ".hpp": "cpp", ".cc": "cpp",
+    ".go": "go", ".rs": "rust", ".java": "java", ".rb": "ruby",
+}
+
+_TS_QUERIES = {
+    "javascript": """
+        (function_declaration name: (identifier) @function)
+        (generator_function_declaration name: (identifier) @function)
+        (class_declaration name: (identifier) @class)
+        (method_definition name: (property_identifier) @method)
+        (variable_declarator name: (identifier) @function
+                             value: (arrow_function))
+    """,
+    "typescript": """
+        (function_declaration name: (identifier) @function)
+        (class_declaration name: (type_identifier) @class)
+        (interface_declaration name: (type_identifier) @class)
+        (method_definition name: (property_identifier) @method)
+    """,
+    "tsx": """
+        (function_declaration name: (identifier) @function)
+        (class_declaration name: (type_identifier) @class)
+    """,
+    "c": """
+        (function_definition declarator:
+            (function_declarator declarator: (identifier) @function))
+        (struct_specifier name: (type_identifier) @class)
+        (enum_specifier name: (type_identifier) @class)
+    """,
+    "cpp": """
+        (function_definition declarator:
+            (function_declarator declarator: (identifier) @function))
+        (function_definition declarator:
+            (function_declarator declarator:
+                (qualified_identifier) @method))
+        (class_specifier name: (type_identifier) @class)
+        (struct_specifier name: (type_identifier) @class)
+    """,
+    "go": """
+        (function_declaration name: (identifier) @function)
+        (method_declaration name: (field_identifier) @method)
+        (type_declaration (type_spec name: (type_identifier) @class))
+    """,
+    "rust": """
+        (function_item name: (identifier) @function)
+        (struct_item name: (type_identifier) @class)
+        (enum_item name: (type_identifier) @class)
+        (trait_item name: (type_identifier) @class)
+    """,
+    "java": """
+        (class_declaration name: (identifier) @class)
+        (interface_declaration name: (identifier) @class)
+        (method_declaration name: (identifier) @method)
+    """,
+    "ruby": """
+        (method name: (identifier) @method)
+        (class name: (constant) @class)
+        (module name: (constant) @class)
+    """,
+}
+
+_TS_CACHE: Dict[str, Any] = {}
+
+
+def _ts_tools(lang: str):
+    """(parser, query) for a language, or None when tree-sitter (or the
+    language pack) is unavailable. Cached; failures cache as None so a
+    missing grammar is probed once."""
+    if lang in _TS_CACHE:
+        return _TS_CACHE[lang]
+    tools = None
+    try:
+        import tree_sitter_languages as tsl
+        parser = tsl.get_parser(lang)
+        language = tsl.get_language(lang)
+        query = language.query(_TS_QUERIES[lang])
+        tools = (parser, query)
+    except Exception as e:          # ImportError, unknown grammar, bad query
+        logger.debug("tree-sitter unavailable for %s: %s", lang, e)
+    _TS_CACHE[lang] = tools
+    return tools
+
+
+def _ts_captures(query, root) -> List[Tuple[Any, str]]:
+    """Normalize the two tree-sitter query APIs: ≤0.21 returns
+    [(node, name)], ≥0.22 returns {name: [nodes]}."""
+    caps = query.captures(root)
+    if isinstance(caps, dict):
+        return [(node, name) for name, nodes in caps.items()
+                for node in nodes]
+    return list(caps)
+
+
+def extract_symbols_treesitter(content: str, ext: str
+                               ) -> Optional[List[Tuple[str, str]]]:
+    """Symbols via tree-sitter for one file; None when unavailable."""
+    lang = _TS_EXT_LANG.get(ext)
+    if not lang:
+        return None
+    tools = _ts_tools(lang)
+    if tools is None:
+        return None
+    parser, query = tools
+    try:
+        tree = parser.parse(content.encode("utf-8", errors="replace"))
+        out: List[Tuple[str, str]] = []
+        seen: Set[str] = set()
+        for node, kind in _ts_captures(query, tree.root_node):
+            text = node.text
+            name = (text.decode("utf-8", errors="replace")
+                    if isinstance(text, bytes) else str(text))
+            if name and name not in seen:
+                seen.add(name)
+                out.append((name, kind))
+        return out
+    except Exception as e:
+        logger.debug("tree-sitter parse failed (%s): %s", ext, e)
+        return None
 
 
 class RepoMapper:
@@ -113,6 +242,9 @@ class RepoMapper:
                 return out
             except SyntaxError:
                 pass
+        ts = extract_symbols_treesitter(content, os.path.splitext(path)[1])
+        if ts:
+            return ts
         out = []
         seen: Set[str] = set()
         for rx, kind in _REGEX_SYMBOLS:

@@ -174,3 +174,99 @@ def test_shell_background_mode(tmp_path):
     while not marker.exists() and time.time() < deadline:
         time.sleep(0.05)
     assert marker.exists()
+
+
+# -- multi-language post-edit validators (reference code.py:827-932) ---------
+
+def test_validator_json_rejects_bad_edit(tmp_path):
+    from fei_amd.tools.code import CodeEditor
+
+    ed = CodeEditor()
+    p = tmp_path / "cfg.json"
+    p.write_text('{"a": 1}')
+    out = ed.regex_replace(str(p), r"1", "1,,")       # invalid json
+    assert "error" in out and "json" in out["error"]
+    assert p.read_text() == '{"a": 1}'                # unchanged
+    ok = ed.regex_replace(str(p), r"1", "2")
+    assert ok.get("success")
+
+
+def test_validator_yaml_rejects_bad_edit(tmp_path):
+    from fei_amd.tools.code import CodeEditor
+
+    ed = CodeEditor()
+    p = tmp_path / "c.yaml"
+    p.write_text("key: value\n")
+    out = ed.regex_replace(str(p), "key: value", "key: [unclosed")
+    assert "error" in out and "yaml" in out["error"]
+
+
+def test_validator_python_still_guards(tmp_path):
+    from fei_amd.tools.code import CodeEditor
+
+    ed = CodeEditor()
+    p = tmp_path / "m.py"
+    p.write_text("def f():\n    return 1\n")
+    out = ed.regex_replace(str(p), "return 1", "return (")
+    assert "error" in out and "syntax" in out["error"]
+
+
+def test_validator_optional_absent_skips(tmp_path):
+    """Languages whose optional checker (esprima/tree-sitter) is not
+    importable in this image pass through ungated — graceful absence."""
+    from fei_amd.tools.code import CodeEditor
+
+    ed = CodeEditor()
+    p = tmp_path / "app.js"
+    p.write_text("function f() { return 1; }\n")
+    out = ed.regex_replace(str(p), "return 1", "return (((")
+    assert out.get("success")        # no JS checker available here
+
+
+def test_validator_treesitter_gate(tmp_path, monkeypatch):
+    """With a (fake) tree-sitter present, a parse-error tree blocks the
+    edit."""
+    import sys
+
+    from fei_amd.tools import repomap as rm
+    from fei_amd.tools.code import CodeEditor
+
+    class _Node:
+        has_error = True
+
+    class _Tree:
+        root_node = _Node()
+
+    class _Parser:
+        def parse(self, data):
+            return _Tree()
+
+    class _Lang:
+        def query(self, q):
+            return object()
+
+    class _TSL:
+        def get_parser(self, lang):
+            return _Parser()
+
+        def get_language(self, lang):
+            return _Lang()
+
+    monkeypatch.setitem(sys.modules, "tree_sitter_languages", _TSL())
+    monkeypatch.setattr(rm, "_TS_CACHE", {})
+    ed = CodeEditor()
+    p = tmp_path / "lib.rs"
+    p.write_text("fn main() {}\n")
+    out = ed.regex_replace(str(p), r"\{\}", "{ let x = ; }")
+    assert "error" in out and "rust" in out["error"]
+
+
+def test_validator_explicit_list_overrides(tmp_path):
+    from fei_amd.tools.code import CodeEditor
+
+    ed = CodeEditor()
+    p = tmp_path / "m.py"
+    p.write_text("x = 1\n")
+    # empty list disables validation entirely (reference semantics)
+    out = ed.regex_replace(str(p), "x = 1", "x = (", validators=[])
+    assert out.get("success")
