@@ -46,8 +46,11 @@ constexpr int NWG = 256;          // one workgroup per CU
 constexpr int SLOT_BYTES = 16384; // ring slot capacity (16 KiB)
 constexpr int RING_SLOTS = 7;     // 7 x 16 KiB = 112 KiB of the 160
 constexpr int SPIN_LIMIT = 1 << 21;
-constexpr int FLAG_FREE = 0;
-constexpr int FLAG_READY = 1;
+// ring flag protocol: 0 = never used; s+1 = logical slot s READY;
+// -(s+1) = logical slot s consumed (physical slot free for s+RING_SLOTS).
+// The logical id in the flag is what makes reuse unambiguous: a stale
+// READY from an earlier occupant of the same physical slot can never
+// satisfy a later logical slot's wait.
 
 // dynamic-LDS carve offsets (bytes; all multiples of 16 — Guideline 17)
 constexpr int LDS_RING = 0;
@@ -141,7 +144,7 @@ __device__ __forceinline__ bool lds_wait_eq(volatile int* w, int want,
   while (__builtin_amdgcn_readfirstlane(*w) != want) {
     __builtin_amdgcn_s_sleep(1);
     if (++spins > SPIN_LIMIT) {
-      if ((threadIdx.x & 63) == 0) atomicExch(fail, code);
+      if ((threadIdx.x & 63) == 0) atomicCAS(fail, 0, code);
       return false;
     }
   }
@@ -154,7 +157,7 @@ __device__ __forceinline__ bool lds_wait_ge(volatile int* w, int want,
   while (__builtin_amdgcn_readfirstlane(*w) < want) {
     __builtin_amdgcn_s_sleep(1);
     if (++spins > SPIN_LIMIT) {
-      if ((threadIdx.x & 63) == 0) atomicExch(fail, code);
+      if ((threadIdx.x & 63) == 0) atomicCAS(fail, 0, code);
       return false;
     }
   }
@@ -187,10 +190,13 @@ __device__ void loader_wave(const StreamArgs& a, char* lds, volatile int* ctrl,
 
   int outstanding = 0;                        // slots issued, not yet gated
   int pending_slot = -1;                      // slot awaiting READY publish
+  int pending_logical = -1;
   for (int s = 0; s < n_slots; ++s) {
     const int slot = s % RING_SLOTS;
     if (s >= RING_SLOTS) {
-      if (!lds_wait_eq(&ctrl[C_RINGF + slot], FLAG_FREE, a.fail, 10)) return;
+      // previous occupant (logical s - RING_SLOTS) must be consumed
+      if (!lds_wait_eq(&ctrl[C_RINGF + slot], -(s - RING_SLOTS + 1),
+                       a.fail, 10)) return;
     }
     const unsigned ring_base = __builtin_amdgcn_readfirstlane(
         (unsigned)(unsigned long)(lds + LDS_RING + slot * SLOT_BYTES));
@@ -239,13 +245,14 @@ __device__ void loader_wave(const StreamArgs& a, char* lds, volatile int* ctrl,
         asm volatile("s_waitcnt vmcnt(14)" ::: "memory");
       else
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      if (lane == 0) ctrl[C_RINGF + pending_slot] = FLAG_READY;
+      if (lane == 0) ctrl[C_RINGF + pending_slot] = pending_logical + 1;
     }
     pending_slot = slot;
+    pending_logical = s;
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   if (pending_slot >= 0 && lane == 0)
-    ctrl[C_RINGF + pending_slot] = FLAG_READY;
+    ctrl[C_RINGF + pending_slot] = pending_logical + 1;
 }
 
 // ---------------------------------------------------------------------------
@@ -328,7 +335,7 @@ k_stream_layer(StreamArgs a) {
   // (f32 payload per output value, bf16-rounded like the launch GEMV)
   for (int s = wave; s < s1; s += 3) {
     const int slot = s % RING_SLOTS;
-    if (!lds_wait_eq(&ctrl[C_RINGF + slot], FLAG_READY, a.fail, 21)) return;
+    if (!lds_wait_eq(&ctrl[C_RINGF + slot], s + 1, a.fail, 21)) return;
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     const float v0 = lds_row_dot(lds, LDS_RING + slot * SLOT_BYTES,
                                  (const char*)xl, C, lane);
@@ -338,7 +345,7 @@ k_stream_layer(StreamArgs a) {
       const long row = (long)wg * r1 + (long)s * 2;
       put_granule(&a.g_qkv[row], tagbase + 1, f2u(bf2f(f2bf(v0))));
       put_granule(&a.g_qkv[row + 1], tagbase + 1, f2u(bf2f(f2bf(v1))));
-      ctrl[C_RINGF + slot] = FLAG_FREE;
+      ctrl[C_RINGF + slot] = -(s + 1);
     }
   }
   if (lane == 0) atomicAdd((int*)&ctrl[C_DONE1], 1);
@@ -379,7 +386,7 @@ k_stream_layer(StreamArgs a) {
           if (__all(ok)) break;
           __builtin_amdgcn_s_sleep(1);
           if (++spins > SPIN_LIMIT) {
-            if (lane == 0) atomicExch(a.fail, 23);
+            if (lane == 0) atomicCAS(a.fail, 0, 23);
             return;
           }
         }
@@ -410,7 +417,7 @@ k_stream_layer(StreamArgs a) {
           if (__all(ok)) break;
           __builtin_amdgcn_s_sleep(1);
           if (++spins > SPIN_LIMIT) {
-            if (lane == 0) atomicExch(a.fail, 24);
+            if (lane == 0) atomicCAS(a.fail, 0, 24);
             return;
           }
         }
@@ -538,7 +545,7 @@ k_stream_layer(StreamArgs a) {
         if (__all(ok)) break;
         __builtin_amdgcn_s_sleep(1);
         if (++spins > SPIN_LIMIT) {
-          if (lane == 0) atomicExch(a.fail, 25);
+          if (lane == 0) atomicCAS(a.fail, 0, 25);
           return;
         }
       }
@@ -561,7 +568,7 @@ k_stream_layer(StreamArgs a) {
           break;
         __builtin_amdgcn_s_sleep(1);
         if (++spins > SPIN_LIMIT) {
-          if (lane == 0) atomicExch(a.fail, 26);
+          if (lane == 0) atomicCAS(a.fail, 0, 26);
           return;
         }
       }
@@ -592,7 +599,7 @@ k_stream_layer(StreamArgs a) {
       if (__all(ok)) break;
       __builtin_amdgcn_s_sleep(1);
       if (++spins > SPIN_LIMIT) {
-        if (lane == 0) atomicExch(a.fail, 27);
+        if (lane == 0) atomicCAS(a.fail, 0, 27);
         return;
       }
     }
@@ -605,7 +612,7 @@ k_stream_layer(StreamArgs a) {
   // S3 slots: 2 o rows; publish h2 granules (2 bf16) with residual add
   for (int s = wave; s < s3; s += 3) {
     const int slot = (s1 + s) % RING_SLOTS;
-    if (!lds_wait_eq(&ctrl[C_RINGF + slot], FLAG_READY, a.fail, 29)) return;
+    if (!lds_wait_eq(&ctrl[C_RINGF + slot], s1 + s + 1, a.fail, 29)) return;
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     const float v0 = lds_row_dot(lds, LDS_RING + slot * SLOT_BYTES,
                                  (const char*)xl, Hq * D, lane);
@@ -616,7 +623,7 @@ k_stream_layer(StreamArgs a) {
       const u16 h0 = f2bf(bf2f(a.x_in[row]) + v0);
       const u16 h1 = f2bf(bf2f(a.x_in[row + 1]) + v1);
       put_granule(&a.g_h2[row / 2], tagbase + 4, pack_bf16(h0, h1));
-      ctrl[C_RINGF + slot] = FLAG_FREE;
+      ctrl[C_RINGF + slot] = -(s1 + s + 1);
     }
   }
   if (lane == 0) atomicAdd((int*)&ctrl[C_DONE3], 1);
@@ -643,7 +650,7 @@ k_stream_layer(StreamArgs a) {
       if (__all(ok)) break;
       __builtin_amdgcn_s_sleep(1);
       if (++spins > SPIN_LIMIT) {
-        if (lane == 0) atomicExch(a.fail, 31);
+        if (lane == 0) atomicCAS(a.fail, 0, 31);
         return;
       }
     }
@@ -665,7 +672,8 @@ k_stream_layer(StreamArgs a) {
   // S4 slots: {gate row, up row}; publish act granule (f32 payload)
   for (int s = wave; s < s4; s += 3) {
     const int slot = (s1 + s3 + s) % RING_SLOTS;
-    if (!lds_wait_eq(&ctrl[C_RINGF + slot], FLAG_READY, a.fail, 33)) return;
+    if (!lds_wait_eq(&ctrl[C_RINGF + slot], s1 + s3 + s + 1, a.fail, 33))
+      return;
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     const float g = lds_row_dot(lds, LDS_RING + slot * SLOT_BYTES,
                                 (const char*)xl, C, lane);
@@ -676,7 +684,7 @@ k_stream_layer(StreamArgs a) {
       const float silu = g / (1.f + __expf(-g));
       const float act = bf2f(f2bf(silu * u));        // launch-path rounding
       put_granule(&a.g_act[n], tagbase + 5, f2u(act));
-      ctrl[C_RINGF + slot] = FLAG_FREE;
+      ctrl[C_RINGF + slot] = -(s1 + s3 + s + 1);
     }
   }
   if (lane == 0) atomicAdd((int*)&ctrl[C_DONE4], 1);
@@ -697,7 +705,7 @@ k_stream_layer(StreamArgs a) {
       if (__all(ok)) break;
       __builtin_amdgcn_s_sleep(1);
       if (++spins > SPIN_LIMIT) {
-        if (lane == 0) atomicExch(a.fail, 35);
+        if (lane == 0) atomicCAS(a.fail, 0, 35);
         return;
       }
     }
@@ -714,7 +722,8 @@ k_stream_layer(StreamArgs a) {
     for (int hs = 0; hs < 2; ++hs) {
       const int s = 2 * r + hs;
       const int slot = (s1 + s3 + s4 + s) % RING_SLOTS;
-      if (!lds_wait_eq(&ctrl[C_RINGF + slot], FLAG_READY, a.fail, 37)) return;
+      if (!lds_wait_eq(&ctrl[C_RINGF + slot], s1 + s3 + s4 + s + 1,
+                       a.fail, 37)) return;
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       const s16x8* w = (const s16x8*)(lds + LDS_RING + slot * SLOT_BYTES);
       const s16x8* x = (const s16x8*)(actl + (long)hs * (I / 2));
@@ -722,7 +731,7 @@ k_stream_layer(StreamArgs a) {
       const int nv = (I / 2) >> 3;
       for (int i = lane; i < nv; i += 64) part += dot8(w[i], x[i]);
       acc += wave_sum(part);
-      if (lane == 0) ctrl[C_RINGF + slot] = FLAG_FREE;
+      if (lane == 0) ctrl[C_RINGF + slot] = -(s1 + s3 + s4 + s + 1);
     }
     if (lane == 0) {
       const long row = (long)wg * r5 + r;
