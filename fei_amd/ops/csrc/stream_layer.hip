@@ -129,6 +129,7 @@ struct StreamArgs {
   u64g* g_att;            // [HqD] f32 payload
   u64g* g_h2;             // [C/2] 2xbf16 payload
   u64g* g_act;            // [I] f32 payload
+  u64g* g_done;           // [6, NWG] per-stage producer-done granules
   int* fail;
   // shape
   int C, Hq, Hkv, D, I, max_seq, layer;
@@ -136,6 +137,28 @@ struct StreamArgs {
 };
 
 constexpr int NSPLIT = 32;
+
+// one wave polls n done-granules until every tag matches. The done layer
+// is what keeps the big data sweeps single-pass: polling re-reads 8 bytes
+// per producer instead of the whole payload (multi-pass payload sweeps
+// measured 2.4x the launch path in r2c3 — L2 poll traffic starved the
+// weight stream). s_sleep(16) bounds the poll rate (polling-cost row).
+__device__ __forceinline__ bool sweep_done(const u64g* base, int n,
+                                           unsigned tag, int* fail,
+                                           int code, int lane) {
+  int spins = 0;
+  for (;;) {
+    bool ok = true;
+    for (int i = lane; i < n; i += 64)
+      ok &= (unsigned)(get_granule(&base[i]) >> 32) == tag;
+    if (__all(ok)) return true;
+    __builtin_amdgcn_s_sleep(16);
+    if (++spins > (SPIN_LIMIT >> 8)) {
+      if (lane == 0) atomicCAS(fail, 0, code);
+      return false;
+    }
+  }
+}
 
 // bounded LDS-flag spin; returns false on timeout (fail stamped)
 __device__ __forceinline__ bool lds_wait_eq(volatile int* w, int want,
@@ -348,7 +371,12 @@ k_stream_layer(StreamArgs a) {
       ctrl[C_RINGF + slot] = -(s + 1);
     }
   }
-  if (lane == 0) atomicAdd((int*)&ctrl[C_DONE1], 1);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");   // granule stores out
+  if (lane == 0) {
+    const int old = atomicAdd((int*)&ctrl[C_DONE1], 1);
+    if (old == 2)        // last consumer wave: this WG's S1 outputs visible
+      put_granule(&a.g_done[1 * NWG + wg], tagbase + 1, 0);
+  }
 
   // ------------------------------------------------------------------ S2 --
   // wave 0: this WG's attention split (kvh = wg/NSPLIT, split = wg%NSPLIT).
@@ -372,6 +400,9 @@ k_stream_layer(StreamArgs a) {
     const unsigned qtag = tagbase + 1;
 
     if (start < end) {
+      // every S1 producer done -> the qkv data sweeps below are one pass
+      if (!sweep_done(&a.g_done[1 * NWG], NWG, tagbase + 1, a.fail, 40,
+                      lane)) return;
       // gather + rope q for this group's heads (sweep its qkv granules)
       {
         int spins = 0;
@@ -384,8 +415,8 @@ k_stream_layer(StreamArgs a) {
             qs[g * D + d] = u2f((unsigned)x);             // raw bf16-as-f32
           }
           if (__all(ok)) break;
-          __builtin_amdgcn_s_sleep(1);
-          if (++spins > SPIN_LIMIT) {
+          __builtin_amdgcn_s_sleep(8);
+          if (++spins > (SPIN_LIMIT >> 4)) {
             if (lane == 0) atomicCAS(a.fail, 0, 23);
             return;
           }
@@ -415,8 +446,8 @@ k_stream_layer(StreamArgs a) {
             if (i < D) kn[i] = b; else vn[i - D] = b;
           }
           if (__all(ok)) break;
-          __builtin_amdgcn_s_sleep(1);
-          if (++spins > SPIN_LIMIT) {
+          __builtin_amdgcn_s_sleep(8);
+          if (++spins > (SPIN_LIMIT >> 4)) {
             if (lane == 0) atomicCAS(a.fail, 0, 24);
             return;
           }
@@ -509,6 +540,8 @@ k_stream_layer(StreamArgs a) {
           put_granule(&base[D + 1], tagbase + 2, f2u(l[g]));
         }
       }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      if (lane == 0) put_granule(&a.g_done[2 * NWG + wg], tagbase + 2, 0);
     } else {
       // empty split: publish -inf/0 partials (combine sweeps all splits)
       for (int g = 0; g < G; ++g) {
@@ -521,6 +554,8 @@ k_stream_layer(StreamArgs a) {
           put_granule(&base[D + 1], tagbase + 2, f2u(0.f));
         }
       }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      if (lane == 0) put_granule(&a.g_done[2 * NWG + wg], tagbase + 2, 0);
     }
   }
 
@@ -528,64 +563,67 @@ k_stream_layer(StreamArgs a) {
   // k_attn_decode_combine (split-major sweep, max-shifted exp weights)
   if (wave == 1 && wg < Hq) {
     const u64g* base = a.g_part + (long)wg * NSPLIT * (D + 2);
-    // sweep m/l for every split first (lane s covers split s)
-    float sm = -1.0f / 0.0f, sl = 0.f;
+    // producers for head wg are the 32 splits of its kv-head: contiguous
+    // WGs [kvh*NSPLIT, +NSPLIT). Poll their done granules, then ONE
+    // parallel data sweep through LDS (a serial per-split dependent wait
+    // was 32 sequential L2 round trips on the critical path).
+    const int kvh_h = wg / G;
+    if (!sweep_done(&a.g_done[2 * NWG + kvh_h * NSPLIT], NSPLIT,
+                    tagbase + 2, a.fail, 41, lane)) return;
+    float* po = (float*)(lds + LDS_ACT);          // [NSPLIT][D] f32
+    float* pml = po + NSPLIT * D;                 // [NSPLIT][2]
     {
       int spins = 0;
       for (;;) {
         bool ok = true;
-        if (lane < NSPLIT) {
-          const u64g xm = get_granule(&base[(long)lane * (D + 2) + D]);
-          const u64g xlv = get_granule(&base[(long)lane * (D + 2) + D + 1]);
-          ok = ((unsigned)(xm >> 32) == tagbase + 2) &
-              ((unsigned)(xlv >> 32) == tagbase + 2);
-          sm = u2f((unsigned)xm);
-          sl = u2f((unsigned)xlv);
+        for (int sp = 0; sp < NSPLIT; ++sp) {
+          const u64g* sb = &base[(long)sp * (D + 2)];
+          const u64g x0 = get_granule(&sb[2 * lane]);
+          const u64g x1v = get_granule(&sb[2 * lane + 1]);
+          ok &= ((unsigned)(x0 >> 32) == tagbase + 2) &
+              ((unsigned)(x1v >> 32) == tagbase + 2);
+          po[sp * D + 2 * lane] = u2f((unsigned)x0);
+          po[sp * D + 2 * lane + 1] = u2f((unsigned)x1v);
+          if (lane < 2) {
+            const u64g xm = get_granule(&sb[D + lane]);
+            ok &= (unsigned)(xm >> 32) == tagbase + 2;
+            pml[sp * 2 + lane] = u2f((unsigned)xm);
+          }
         }
         if (__all(ok)) break;
-        __builtin_amdgcn_s_sleep(1);
-        if (++spins > SPIN_LIMIT) {
-          if (lane == 0) atomicCAS(a.fail, 0, 25);
-          return;
-        }
-      }
-    }
-    const float mstar = wave_max(sm);
-    const float wexp = (lane < NSPLIT) ? __expf(sm - mstar) : 0.f;
-    const float ltot = wave_sum(wexp * sl);
-    // o: lane covers dims (2*lane, 2*lane+1); serial over splits with a
-    // bounded per-split granule wait
-    float acc0 = 0.f, acc1 = 0.f;
-    for (int s = 0; s < NSPLIT; ++s) {
-      const u64g* sb = &base[(long)s * (D + 2)];
-      u64g x0, x1v;
-      int spins = 0;
-      for (;;) {
-        x0 = get_granule(&sb[2 * lane]);
-        x1v = get_granule(&sb[2 * lane + 1]);
-        if (__all(((unsigned)(x0 >> 32) == tagbase + 2) &
-                  ((unsigned)(x1v >> 32) == tagbase + 2)))
-          break;
-        __builtin_amdgcn_s_sleep(1);
-        if (++spins > SPIN_LIMIT) {
+        __builtin_amdgcn_s_sleep(16);
+        if (++spins > (SPIN_LIMIT >> 8)) {
           if (lane == 0) atomicCAS(a.fail, 0, 26);
           return;
         }
       }
-      const float w = __shfl(wexp, s);        // split s's exp weight
-      acc0 = fmaf(w, u2f((unsigned)x0), acc0);
-      acc1 = fmaf(w, u2f((unsigned)x1v), acc1);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    float sm = (lane < NSPLIT) ? pml[lane * 2] : -1.0f / 0.0f;
+    float sl = (lane < NSPLIT) ? pml[lane * 2 + 1] : 0.f;
+    const float mstar = wave_max(sm);
+    const float wexp = (lane < NSPLIT) ? __expf(sm - mstar) : 0.f;
+    const float ltot = wave_sum(wexp * sl);
+    float acc0 = 0.f, acc1 = 0.f;
+    for (int sp = 0; sp < NSPLIT; ++sp) {
+      const float w = __shfl(wexp, sp);
+      acc0 = fmaf(w, po[sp * D + 2 * lane], acc0);
+      acc1 = fmaf(w, po[sp * D + 2 * lane + 1], acc1);
     }
     const float inv_l = ltot > 0.f ? 1.f / ltot : 0.f;
     const u16 b0 = f2bf(acc0 * inv_l);
     const u16 b1 = f2bf(acc1 * inv_l);
     put_granule(&a.g_att[(long)wg * (D / 2) + lane], tagbase + 3,
                 pack_bf16(b0, b1));
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (lane == 0) put_granule(&a.g_done[3 * NWG + wg], tagbase + 3, 0);
   }
 
   // ------------------------------------------------------------------ S3 --
   // wave 0 gathers att into the x-region (all heads, bf16)
   if (wave == 0) {
+    if (!sweep_done(&a.g_done[3 * NWG], Hq, tagbase + 3, a.fail, 42, lane))
+      return;
     u64g* src = a.g_att;
     const int n_g = Hq * D / 2;                      // granules (2 vals each)
     int spins = 0;
@@ -597,8 +635,8 @@ k_stream_layer(StreamArgs a) {
         ((unsigned*)xl)[i] = (unsigned)x;            // 2 bf16 at once
       }
       if (__all(ok)) break;
-      __builtin_amdgcn_s_sleep(1);
-      if (++spins > SPIN_LIMIT) {
+      __builtin_amdgcn_s_sleep(8);
+      if (++spins > (SPIN_LIMIT >> 4)) {
         if (lane == 0) atomicCAS(a.fail, 0, 27);
         return;
       }
@@ -626,13 +664,20 @@ k_stream_layer(StreamArgs a) {
       ctrl[C_RINGF + slot] = -(s1 + s + 1);
     }
   }
-  if (lane == 0) atomicAdd((int*)&ctrl[C_DONE3], 1);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  if (lane == 0) {
+    const int old = atomicAdd((int*)&ctrl[C_DONE3], 1);
+    if (old == 2)
+      put_granule(&a.g_done[4 * NWG + wg], tagbase + 4, 0);
+  }
 
   // ------------------------------------------------------------------ S4 --
   // wave 0: gather h2, norm-prologue into the x-region (xl reused — wait
   // for all consumers to leave S3)
   if (wave == 0) {
     if (!lds_wait_ge(&ctrl[C_DONE3], 3, a.fail, 30)) return;
+    if (!sweep_done(&a.g_done[4 * NWG], NWG, tagbase + 4, a.fail, 43, lane))
+      return;
     float ss = 0.f;
     int spins = 0;
     for (;;) {
@@ -648,8 +693,8 @@ k_stream_layer(StreamArgs a) {
         ss = fmaf(f0, f0, fmaf(f1, f1, ss));
       }
       if (__all(ok)) break;
-      __builtin_amdgcn_s_sleep(1);
-      if (++spins > SPIN_LIMIT) {
+      __builtin_amdgcn_s_sleep(8);
+      if (++spins > (SPIN_LIMIT >> 4)) {
         if (lane == 0) atomicCAS(a.fail, 0, 31);
         return;
       }
@@ -687,13 +732,20 @@ k_stream_layer(StreamArgs a) {
       ctrl[C_RINGF + slot] = -(s1 + s3 + s + 1);
     }
   }
-  if (lane == 0) atomicAdd((int*)&ctrl[C_DONE4], 1);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  if (lane == 0) {
+    const int old = atomicAdd((int*)&ctrl[C_DONE4], 1);
+    if (old == 2)
+      put_granule(&a.g_done[5 * NWG + wg], tagbase + 5, 0);
+  }
 
   // ------------------------------------------------------------------ S5 --
   // wave 0 gathers act (bf16) into the ACT region
   u16* actl = (u16*)(lds + LDS_ACT);
   if (wave == 0) {
     if (!lds_wait_ge(&ctrl[C_DONE4], 3, a.fail, 34)) return;
+    if (!sweep_done(&a.g_done[5 * NWG], NWG, tagbase + 5, a.fail, 44, lane))
+      return;
     int spins = 0;
     for (;;) {
       bool ok = true;
@@ -703,8 +755,8 @@ k_stream_layer(StreamArgs a) {
         actl[i] = f2bf(u2f((unsigned)x));
       }
       if (__all(ok)) break;
-      __builtin_amdgcn_s_sleep(1);
-      if (++spins > SPIN_LIMIT) {
+      __builtin_amdgcn_s_sleep(8);
+      if (++spins > (SPIN_LIMIT >> 4)) {
         if (lane == 0) atomicCAS(a.fail, 0, 35);
         return;
       }
@@ -773,7 +825,7 @@ void fei_stream_layer(const void* x_in, void* h_out, const void* wqkv,
                       const void* norm_attn, const void* norm_mlp,
                       void* k_cache, void* v_cache, const float* cos_sin,
                       const int* pos, void* g_qkv, void* g_part, void* g_att,
-                      void* g_h2, void* g_act, int* fail,
+                      void* g_h2, void* g_act, void* g_done, int* fail,
                       int C, int Hq, int Hkv, int D, int I, int max_seq,
                       int layer, float eps, float scale,
                       hipStream_t stream) {
@@ -785,7 +837,8 @@ void fei_stream_layer(const void* x_in, void* h_out, const void* wqkv,
   a.k_cache = (u16*)k_cache; a.v_cache = (u16*)v_cache;
   a.cos_sin = cos_sin; a.pos = pos;
   a.g_qkv = (u64g*)g_qkv; a.g_part = (u64g*)g_part; a.g_att = (u64g*)g_att;
-  a.g_h2 = (u64g*)g_h2; a.g_act = (u64g*)g_act; a.fail = fail;
+  a.g_h2 = (u64g*)g_h2; a.g_act = (u64g*)g_act;
+  a.g_done = (u64g*)g_done; a.fail = fail;
   a.C = C; a.Hq = Hq; a.Hkv = Hkv; a.D = D; a.I = I; a.max_seq = max_seq;
   a.layer = layer; a.eps = eps; a.scale = scale;
   const int G = Hq / Hkv;

@@ -1,25 +1,32 @@
-"""Textual TUI: chat app with the /mem command suite.
+"""Textual TUI: chat app with message panels and the /mem command suite.
 
-Parity: reference fei/ui/textual_chat.py (1,070 LoC): FeiChatApp with
-message panels, a memory-command suggester, the /mem command set
-(help/list/search/view/save/tag/server — :557-970) calling the memory
-tool handlers directly, background assistant processing (:1002-1031), and
-``main()`` returning the App for the caller to ``.run()`` (:1044-1062).
+Parity: reference fei/ui/textual_chat.py (1,070 LoC) — message widget
+hierarchy and panels (:48-117), memory-command autocomplete suggester
+(:119-229), the /mem command set (help/list/search/view/save/tag/server —
+:557-970) calling the memory tool handlers directly, background assistant
+processing (:1002-1031), and ``main()`` returning the App for the caller
+to ``.run()`` (:1044-1062). This build adds a status sidebar (provider,
+model, turn metrics) and slash commands beyond /mem.
 """
 
 from __future__ import annotations
 
-from typing import Optional
+import time
+from typing import List, Optional
 
 from textual.app import App, ComposeResult
-from textual.containers import VerticalScroll
+from textual.containers import Horizontal, Vertical, VerticalScroll
 from textual.suggester import Suggester
 from textual.widgets import Footer, Header, Input, Static
 
 MEM_COMMANDS = [
-    "/mem help", "/mem list", "/mem search ", "/mem view ", "/mem save ",
+    "/mem help", "/mem list", "/mem search ", "/mem save ",
     "/mem tag ", "/mem server start", "/mem server stop", "/mem server status",
-    "/mem index", "/mem semantic ", "/mem keyword ",
+    "/mem index", "/mem semantic ", "/mem keyword ", "/mem view ",
+]
+
+SLASH_COMMANDS = [
+    "/help", "/clear", "/history", "/stats", "/quit",
 ]
 
 MEM_HELP = """\
@@ -34,62 +41,177 @@ MEM_HELP = """\
 /mem tag <id> <tags>      set tags on a memory
 /mem server start|stop|status"""
 
+APP_HELP = MEM_HELP + """
 
-class MemCommandSuggester(Suggester):
-    """Autocomplete for /mem commands (reference: textual_chat.py:119-229)."""
+/help                     this help
+/clear                    clear the conversation
+/history                  show the turn history
+/stats                    per-turn metrics (tokens, latency, tools)
+/quit                     exit"""
 
-    def __init__(self):
+
+class CommandSuggester(Suggester):
+    """Autocomplete for slash + /mem commands AND memory ids for
+    /mem view|tag (reference: textual_chat.py:119-229). Memory ids are
+    fetched lazily from the memory tools and cached briefly."""
+
+    def __init__(self, app: Optional["FeiChatApp"] = None):
         super().__init__(use_cache=False, case_sensitive=False)
+        self._app = app
+        self._ids: List[str] = []
+        self._ids_at = 0.0
+
+    def _memory_ids(self) -> List[str]:
+        if time.time() - self._ids_at < 5.0:
+            return self._ids
+        self._ids_at = time.time()
+        try:
+            tools = self._app._tools() if self._app else None
+            out = tools.list({"folder": ""}) if tools else {}
+            self._ids = [m["memory_id"] for m in out.get("memories", [])][:50]
+        except Exception:
+            self._ids = []
+        return self._ids
 
     async def get_suggestion(self, value: str) -> Optional[str]:
         if not value.startswith("/"):
             return None
-        for cmd in MEM_COMMANDS:
+        # memory-id completion for view/tag
+        for stem in ("/mem view ", "/mem tag "):
+            if value.startswith(stem):
+                frag = value[len(stem):]
+                for mid in self._memory_ids():
+                    if mid.startswith(frag) and mid != frag:
+                        return stem + mid
+                return None
+        for cmd in MEM_COMMANDS + SLASH_COMMANDS:
             if cmd.startswith(value) and cmd != value:
                 return cmd
         return None
 
 
+# -- message widget hierarchy (reference: textual_chat.py:48-117) ------------
+
 class ChatMessage(Static):
-    pass
+    """Base message panel; subclasses set the border title and style."""
+
+    role = "system"
+    prefix = ""
+
+    def __init__(self, text: str):
+        super().__init__(text)
+        self.add_class(self.role)
+        self.border_title = self.prefix + time.strftime("%H:%M:%S")
+
+
+class UserMessage(ChatMessage):
+    role = "user"
+    prefix = "you · "
+
+
+class AssistantMessage(ChatMessage):
+    role = "assistant"
+    prefix = "fei · "
+
+
+class ToolMessage(ChatMessage):
+    role = "tool"
+    prefix = "tool · "
+
+
+class SystemMessage(ChatMessage):
+    role = "system"
+
+
+class StatusPanel(Static):
+    """Sidebar: provider/model and rolling turn metrics."""
+
+    def update_from(self, assistant) -> None:
+        lines = ["fei_amd", ""]
+        if assistant is not None:
+            prov = getattr(assistant, "provider", None) or "?"
+            model = getattr(assistant, "model", None) or ""
+            lines += [f"provider  {prov}", f"model     {model}", ""]
+            metrics = getattr(assistant, "turn_metrics", [])
+            if metrics:
+                m = metrics[-1]
+                lines += ["last turn:",
+                          f"  {m.get('latency_s', 0):.2f}s",
+                          f"  tools: {len(m.get('tools', []))}",
+                          f"  in/out: {m.get('input_tokens', 0)}"
+                          f"/{m.get('output_tokens', 0)}"]
+            lines += ["", f"turns     {len(metrics)}"]
+        self.update("\n".join(lines))
 
 
 class FeiChatApp(App):
     CSS = """
+    #body { height: 1fr; }
     #chat { height: 1fr; }
-    ChatMessage { padding: 0 1; margin-bottom: 1; }
-    .user { background: $boost; }
-    .assistant { }
-    .system { color: $text-muted; }
+    #status { width: 26; border-left: solid $primary; padding: 0 1;
+              display: none; }
+    #status.visible { display: block; }
+    ChatMessage { padding: 0 1; margin-bottom: 1; border: round $primary 30%;
+                  border-title-align: left; }
+    .user { background: $boost; border: round $secondary; }
+    .assistant { border: round $primary; }
+    .tool { color: $text-muted; border: round $primary 30%; }
+    .system { color: $text-muted; border: none; }
+    #thinking { color: $text-muted; padding: 0 1; display: none; }
+    #thinking.active { display: block; }
     """
-    BINDINGS = [("ctrl+c", "quit", "Quit"), ("ctrl+l", "clear", "Clear")]
+    BINDINGS = [
+        ("ctrl+c", "quit", "Quit"),
+        ("ctrl+l", "clear", "Clear"),
+        ("f2", "toggle_status", "Status"),
+    ]
 
     def __init__(self, assistant=None, memory_tools=None, **kwargs):
         super().__init__(**kwargs)
         self.assistant = assistant
         self.memory_tools = memory_tools
 
+    def _tools(self):
+        if self.memory_tools is None:
+            from fei_amd.tools.memory_tools import MemoryTools
+            self.memory_tools = MemoryTools()
+        return self.memory_tools
+
     # -- layout --------------------------------------------------------------
 
     def compose(self) -> ComposeResult:
         yield Header(show_clock=True)
-        yield VerticalScroll(id="chat")
-        yield Input(placeholder="Message (or /mem ...)",
-                    suggester=MemCommandSuggester(), id="input")
+        with Horizontal(id="body"):
+            with Vertical():
+                yield VerticalScroll(id="chat")
+                yield Static("thinking…", id="thinking")
+            yield StatusPanel(id="status")
+        yield Input(placeholder="Message (or /help, /mem ...)",
+                    suggester=CommandSuggester(self), id="input")
         yield Footer()
 
     def add_message(self, text: str, kind: str = "assistant") -> None:
         chat = self.query_one("#chat", VerticalScroll)
-        prefix = {"user": "you> ", "assistant": "fei> ", "system": ""}[kind]
-        msg = ChatMessage(prefix + text)
-        msg.add_class(kind)
-        chat.mount(msg)
+        cls = {"user": UserMessage, "assistant": AssistantMessage,
+               "tool": ToolMessage, "system": SystemMessage}[kind]
+        chat.mount(cls(text))
         chat.scroll_end(animate=False)
 
     def action_clear(self) -> None:
         if self.assistant is not None:
             self.assistant.reset()
         self.query_one("#chat", VerticalScroll).remove_children()
+
+    def action_toggle_status(self) -> None:
+        panel = self.query_one("#status", StatusPanel)
+        panel.toggle_class("visible")
+        panel.update_from(self.assistant)
+
+    def _set_thinking(self, on: bool) -> None:
+        try:
+            self.query_one("#thinking", Static).set_class(on, "active")
+        except Exception:
+            pass
 
     # -- input handling ------------------------------------------------------
 
@@ -102,7 +224,49 @@ class FeiChatApp(App):
         if text.startswith("/mem"):
             self.add_message(self.handle_memory_command(text), "system")
             return
+        if text.startswith("/"):
+            out = self.handle_slash_command(text)
+            if out is not None:
+                self.add_message(out, "system")
+            return
         self.run_worker(self.process_with_assistant(text), exclusive=True)
+
+    def handle_slash_command(self, text: str) -> Optional[str]:
+        cmd = text.split()[0]
+        if cmd == "/help":
+            return APP_HELP
+        if cmd == "/clear":
+            self.action_clear()
+            return "(cleared)"
+        if cmd == "/quit":
+            self.exit()
+            return None
+        if cmd == "/history":
+            if self.assistant is None:
+                return "(no assistant)"
+            msgs = self.assistant.conversation.messages
+            lines = []
+            for m in msgs[-20:]:
+                content = m.get("content", "")
+                if isinstance(content, list):
+                    content = " ".join(
+                        str(b.get("text", b.get("content", "")))[:40]
+                        for b in content if isinstance(b, dict))
+                lines.append(f"{m.get('role', '?'):<10} {str(content)[:70]}")
+            return "\n".join(lines) or "(empty)"
+        if cmd == "/stats":
+            if self.assistant is None:
+                return "(no assistant)"
+            lines = []
+            for i, m in enumerate(self.assistant.turn_metrics[-10:]):
+                tools = ",".join(t.get("name", "?")
+                                 for t in m.get("tools", []))
+                lines.append(f"turn {i}: {m.get('latency_s', 0):.2f}s "
+                             f"in {m.get('input_tokens', 0)} "
+                             f"out {m.get('output_tokens', 0)} "
+                             f"[{tools}]")
+            return "\n".join(lines) or "(no turns yet)"
+        return f"unknown command {cmd} (try /help)"
 
     async def process_with_assistant(self, text: str) -> None:
         """Background turn (reference: textual_chat.py:1002-1031)."""
@@ -110,17 +274,29 @@ class FeiChatApp(App):
             self.add_message("(no assistant configured)", "system")
             return
         import asyncio
-        loop = asyncio.get_running_loop()
-        response = await loop.run_in_executor(None, self.assistant.ask, text)
+        self._set_thinking(True)
+        try:
+            loop = asyncio.get_running_loop()
+            response = await loop.run_in_executor(None, self.assistant.ask,
+                                                  text)
+        finally:
+            self._set_thinking(False)
+        # surface the turn's tool calls like the reference's tool panels
+        metrics = getattr(self.assistant, "turn_metrics", [])
+        if metrics:
+            for t in metrics[-1].get("tools", []):
+                self.add_message(
+                    f"{t.get('name', '?')}({t.get('latency_s', 0):.2f}s)",
+                    "tool")
         self.add_message(response or "(no response)")
+        panel = self.query_one("#status", StatusPanel)
+        if panel.has_class("visible"):
+            panel.update_from(self.assistant)
 
     # -- /mem commands (reference: textual_chat.py:557-970) -------------------
 
     def handle_memory_command(self, text: str) -> str:
-        tools = self.memory_tools
-        if tools is None:
-            from fei_amd.tools.memory_tools import MemoryTools
-            tools = self.memory_tools = MemoryTools()
+        tools = self._tools()
         parts = text.split()
         cmd = parts[1] if len(parts) > 1 else "help"
         rest = parts[2:]
@@ -175,14 +351,14 @@ class FeiChatApp(App):
                 mem = tools.view({"memory_id": rest[0]})
                 if "error" in mem:
                     return mem["error"]
-                from fei_amd.memdir import utils as mu
                 import os
+
+                from fei_amd.memdir import utils as mu
                 root = mu.get_memdir_base(tools.base)
                 headers = dict(mem["headers"])
                 headers["Tags"] = ",".join(rest[1:])
-                path = os.path.join(root, mem["folder"], mem["status"],
-                                    mem["filename"]) if mem["folder"] else \
-                    os.path.join(root, mem["status"], mem["filename"])
+                path = os.path.join(mu.folder_path(mem["folder"], tools.base),
+                                    mem["status"], mem["filename"])
                 with open(path, "w", encoding="utf-8") as f:
                     f.write(mu.format_memory_content(headers, mem["content"]))
                 return f"tagged {rest[0]}"
