@@ -67,8 +67,8 @@ def _try_load() -> None:
     lib.fei_stream_layer_check.restype = _i
     lib.fei_stream_layer.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp, _vp,
                                      _vp, _vp, _vp, _vp, _vp, _vp, _vp, _vp,
-                                     _vp, _vp, _vp, _i, _i, _i, _i, _i, _i,
-                                     _i, _f, _f, _vp]
+                                     _vp, _vp, _vp, _vp, _i, _i, _i, _i, _i,
+                                     _i, _i, _f, _f, _vp]
     lib.fei_attn_decode_paged.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp,
                                           _i, _i, _i, _i, _i, _i, _i, _f,
                                           _l, _vp]
@@ -435,6 +435,7 @@ def stream_layer(x_in: torch.Tensor, h_out: torch.Tensor, lw, spec,
         _ptr(k_cache), _ptr(v_cache), _ptr(cos_sin), _ptr(pos),
         _ptr(ws["g_qkv"]), _ptr(ws["g_part"]), _ptr(ws["g_att"]),
         _ptr(ws["g_h2"]), _ptr(ws["g_act"]), _ptr(ws["g_done"]),
+        _ptr(ws["dbg"]) if "dbg" in ws else None,
         _ptr(ws["fail"]),
         spec.hidden_size, spec.num_heads, spec.num_kv_heads, spec.head_dim,
         spec.intermediate_size, k_cache.shape[-2], layer, spec.norm_eps,

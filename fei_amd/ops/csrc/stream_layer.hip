@@ -130,6 +130,7 @@ struct StreamArgs {
   u64g* g_h2;             // [C/2] 2xbf16 payload
   u64g* g_act;            // [I] f32 payload
   u64g* g_done;           // [6, NWG] per-stage producer-done granules
+  u64g* dbg;              // nullable: [NWG, 16] s_memtime phase stamps
   int* fail;
   // shape
   int C, Hq, Hkv, D, I, max_seq, layer;
@@ -137,6 +138,12 @@ struct StreamArgs {
 };
 
 constexpr int NSPLIT = 32;
+
+__device__ __forceinline__ void stamp(const StreamArgs& a, int wg, int slot,
+                                      int lane) {
+  if (a.dbg && lane == 0)
+    a.dbg[wg * 16 + slot] = __builtin_amdgcn_s_memtime();
+}
 
 // one wave polls n done-granules until every tag matches. The done layer
 // is what keeps the big data sweeps single-pass: polling re-reads 8 bytes
@@ -319,9 +326,12 @@ k_stream_layer(StreamArgs a) {
   __syncthreads();                      // before ANY glds is issued
 
   if (wave == 3) {
+    stamp(a, wg, 14, lane);
     loader_wave(a, lds, ctrl, wg, lane);
+    stamp(a, wg, 15, lane);
     return;
   }
+  if (wave == 0) stamp(a, wg, 0, lane);
 
   // ------------------------------------------------------------------ S1 --
   u16* xl = (u16*)(lds + LDS_X);        // stage-input LDS region (bf16[C])
@@ -350,6 +360,7 @@ k_stream_layer(StreamArgs a) {
     }
     __threadfence_block();
     if (lane == 0) ctrl[C_STAGE] = 1;
+    stamp(a, wg, 1, lane);
   } else {
     if (!lds_wait_ge(&ctrl[C_STAGE], 1, a.fail, 20)) return;
   }
@@ -377,6 +388,8 @@ k_stream_layer(StreamArgs a) {
     if (old == 2)        // last consumer wave: this WG's S1 outputs visible
       put_granule(&a.g_done[1 * NWG + wg], tagbase + 1, 0);
   }
+
+  if (wave == 0) stamp(a, wg, 2, lane);          // S1 slots done
 
   // ------------------------------------------------------------------ S2 --
   // wave 0: this WG's attention split (kvh = wg/NSPLIT, split = wg%NSPLIT).
@@ -542,6 +555,7 @@ k_stream_layer(StreamArgs a) {
       }
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       if (lane == 0) put_granule(&a.g_done[2 * NWG + wg], tagbase + 2, 0);
+      stamp(a, wg, 3, lane);                     // partials published
     } else {
       // empty split: publish -inf/0 partials (combine sweeps all splits)
       for (int g = 0; g < G; ++g) {
@@ -643,6 +657,7 @@ k_stream_layer(StreamArgs a) {
     }
     __threadfence_block();
     if (lane == 0) ctrl[C_STAGE] = 2;
+    stamp(a, wg, 4, lane);                       // att gathered
   } else {
     if (!lds_wait_ge(&ctrl[C_STAGE], 2, a.fail, 28)) return;
   }
@@ -670,6 +685,8 @@ k_stream_layer(StreamArgs a) {
     if (old == 2)
       put_granule(&a.g_done[4 * NWG + wg], tagbase + 4, 0);
   }
+
+  if (wave == 0) stamp(a, wg, 5, lane);          // S3 slots done
 
   // ------------------------------------------------------------------ S4 --
   // wave 0: gather h2, norm-prologue into the x-region (xl reused — wait
@@ -710,6 +727,7 @@ k_stream_layer(StreamArgs a) {
     }
     __threadfence_block();
     if (lane == 0) ctrl[C_STAGE] = 3;
+    stamp(a, wg, 6, lane);                       // x2 ready
   } else {
     if (!lds_wait_ge(&ctrl[C_STAGE], 3, a.fail, 32)) return;
   }
@@ -739,6 +757,8 @@ k_stream_layer(StreamArgs a) {
       put_granule(&a.g_done[5 * NWG + wg], tagbase + 5, 0);
   }
 
+  if (wave == 0) stamp(a, wg, 7, lane);          // S4 slots done
+
   // ------------------------------------------------------------------ S5 --
   // wave 0 gathers act (bf16) into the ACT region
   u16* actl = (u16*)(lds + LDS_ACT);
@@ -763,6 +783,7 @@ k_stream_layer(StreamArgs a) {
     }
     __threadfence_block();
     if (lane == 0) ctrl[C_STAGE] = 4;
+    stamp(a, wg, 8, lane);                       // act gathered
   } else {
     if (!lds_wait_ge(&ctrl[C_STAGE], 4, a.fail, 36)) return;
   }
@@ -785,6 +806,7 @@ k_stream_layer(StreamArgs a) {
       acc += wave_sum(part);
       if (lane == 0) ctrl[C_RINGF + slot] = -(s1 + s3 + s4 + s + 1);
     }
+    if (lane == 0 && r + 3 >= r5) stamp(a, wg, 9, lane);  // last S5 row
     if (lane == 0) {
       const long row = (long)wg * r5 + r;
       // residual: re-read own h2 granule (value published this launch)
@@ -825,7 +847,8 @@ void fei_stream_layer(const void* x_in, void* h_out, const void* wqkv,
                       const void* norm_attn, const void* norm_mlp,
                       void* k_cache, void* v_cache, const float* cos_sin,
                       const int* pos, void* g_qkv, void* g_part, void* g_att,
-                      void* g_h2, void* g_act, void* g_done, int* fail,
+                      void* g_h2, void* g_act, void* g_done, void* dbg,
+                      int* fail,
                       int C, int Hq, int Hkv, int D, int I, int max_seq,
                       int layer, float eps, float scale,
                       hipStream_t stream) {
@@ -838,7 +861,7 @@ void fei_stream_layer(const void* x_in, void* h_out, const void* wqkv,
   a.cos_sin = cos_sin; a.pos = pos;
   a.g_qkv = (u64g*)g_qkv; a.g_part = (u64g*)g_part; a.g_att = (u64g*)g_att;
   a.g_h2 = (u64g*)g_h2; a.g_act = (u64g*)g_act;
-  a.g_done = (u64g*)g_done; a.fail = fail;
+  a.g_done = (u64g*)g_done; a.dbg = (u64g*)dbg; a.fail = fail;
   a.C = C; a.Hq = Hq; a.Hkv = Hkv; a.D = D; a.I = I; a.max_seq = max_seq;
   a.layer = layer; a.eps = eps; a.scale = scale;
   const int G = Hq / Hkv;

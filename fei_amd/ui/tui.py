@@ -90,6 +90,9 @@ class CommandSuggester(Suggester):
         return None
 
 
+MemCommandSuggester = CommandSuggester        # back-compat alias
+
+
 # -- message widget hierarchy (reference: textual_chat.py:48-117) ------------
 
 class ChatMessage(Static):

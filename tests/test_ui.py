@@ -86,8 +86,8 @@ def test_cli_search_subcommand(home, tmp_path, monkeypatch, capsys):
 
 def test_tui_construction(home):
     """Construct the app + exercise /mem handlers without running it."""
-    from fei_amd.ui.tui import FeiChatApp, MemCommandSuggester
     from fei_amd.tools.memory_tools import MemoryTools
+    from fei_amd.ui.tui import FeiChatApp
 
     tools = MemoryTools(base=str(home / "Memdir"))
     tools.create({"subject": "tui memory", "tags": "t"})
@@ -167,3 +167,121 @@ def test_tui_mem_keyword(home):
     app = FeiChatApp(assistant=None, memory_tools=tools)
     out = app.handle_memory_command("/mem keyword giraffe")
     assert "keyword target" in out
+
+
+# -- TUI depth (reference textual_chat.py:48-229, 557-970) -------------------
+
+def test_tui_pilot_user_message_and_mem(home):
+    """Run the app headless with Textual's Pilot: submitting input mounts a
+    UserMessage panel; a /mem command mounts its SystemMessage reply."""
+    import asyncio
+
+    from fei_amd.tools.memory_tools import MemoryTools
+    from fei_amd.ui.tui import (FeiChatApp, SystemMessage, UserMessage)
+
+    tools = MemoryTools(base=str(home / "Memdir"))
+    tools.create({"subject": "pilot memory", "tags": "t"})
+
+    async def run():
+        app = FeiChatApp(assistant=None, memory_tools=tools)
+        async with app.run_test() as pilot:
+            inp = app.query_one("#input")
+            inp.value = "/mem list"
+            await pilot.press("enter")
+            await pilot.pause()
+            users = app.query(UserMessage)
+            systems = app.query(SystemMessage)
+            assert len(users) == 1
+            assert any("pilot memory" in str(m.content) for m in systems)
+
+    asyncio.run(run())
+
+
+def test_tui_pilot_slash_commands(home):
+    import asyncio
+
+    from fei_amd.ui.tui import FeiChatApp, StatusPanel, SystemMessage
+
+    async def run():
+        app = FeiChatApp(assistant=None)
+        async with app.run_test() as pilot:
+            inp = app.query_one("#input")
+            inp.value = "/help"
+            await pilot.press("enter")
+            await pilot.pause()
+            assert any("/mem save" in str(m.content)
+                       for m in app.query(SystemMessage))
+            # F2 toggles the status sidebar
+            panel = app.query_one("#status", StatusPanel)
+            assert not panel.has_class("visible")
+            await pilot.press("f2")
+            assert panel.has_class("visible")
+            # /clear empties the chat
+            inp.value = "/clear"
+            await pilot.press("enter")
+            await pilot.pause()
+            texts = [str(m.content) for m in app.query(SystemMessage)]
+            assert "(cleared)" in texts
+
+    asyncio.run(run())
+
+
+def test_tui_pilot_assistant_turn_with_tool_panels(home):
+    """A full background assistant turn through the stub backend renders
+    the assistant panel (and tool panels when the turn ran tools)."""
+    import asyncio
+
+    from fei_amd.core.assistant import Assistant
+    from fei_amd.tools.code import create_code_tools
+    from fei_amd.tools.registry import ToolRegistry
+    from fei_amd.ui.tui import AssistantMessage, FeiChatApp
+
+    registry = ToolRegistry()
+    create_code_tools(registry)
+    assistant = Assistant(provider="stub", tool_registry=registry)
+
+    async def run():
+        app = FeiChatApp(assistant=assistant)
+        async with app.run_test() as pilot:
+            inp = app.query_one("#input")
+            inp.value = "hello there"
+            await pilot.press("enter")
+            for _ in range(50):
+                await pilot.pause(0.1)
+                if app.query(AssistantMessage):
+                    break
+            msgs = app.query(AssistantMessage)
+            assert len(msgs) == 1
+            assert "hello there" in str(msgs.first().content)
+
+    asyncio.run(run())
+
+
+def test_tui_suggester_completes_slash_and_ids(home):
+    import asyncio
+
+    from fei_amd.tools.memory_tools import MemoryTools
+    from fei_amd.ui.tui import CommandSuggester, FeiChatApp
+
+    tools = MemoryTools(base=str(home / "Memdir"))
+    out = tools.create({"subject": "sugg", "tags": ""})
+    mid = out["memory_id"]
+    app = FeiChatApp(assistant=None, memory_tools=tools)
+    s = CommandSuggester(app)
+    assert asyncio.run(s.get_suggestion("/he")) == "/help"
+    assert asyncio.run(s.get_suggestion("/mem se")) == "/mem search "
+    got = asyncio.run(s.get_suggestion("/mem view " + mid[:3]))
+    assert got == "/mem view " + mid
+
+
+def test_tui_history_and_stats_commands(home):
+    from fei_amd.core.assistant import Assistant
+    from fei_amd.ui.tui import FeiChatApp
+
+    assistant = Assistant(provider="stub")
+    assistant.chat("first message")
+    app = FeiChatApp(assistant=assistant)
+    out = app.handle_slash_command("/history")
+    assert "first message" in out
+    out = app.handle_slash_command("/stats")
+    assert "turn 0" in out
