@@ -39,7 +39,6 @@ for i, nm in enumerate(names[1:], start=1):
     dt = (cur - prev) / clk
     out[nm] = round(float(dt.mean()), 2)
     prev = cur
-out["wave0_total"] = round(float((d[:, 9].astype("int64") - t0) / clk).mean() if hasattr((d[:,9]-t0), 'mean') else 0, 2)
 out["wave0_total"] = round(float(((d[:, 9].astype("int64") - t0) / clk).mean()), 2)
 out["loader_total"] = round(float(((d[:, 15].astype("int64") - d[:, 14].astype("int64")) / clk).mean()), 2)
 print(json.dumps(out))
