@@ -65,6 +65,9 @@ constexpr int C_STAGE = C_RINGF + RING_SLOTS;  // stage-input-ready counter
 constexpr int C_DONE1 = C_STAGE + 1;       // consumers done with S1 slots
 constexpr int C_DONE3 = C_DONE1 + 1;
 constexpr int C_DONE4 = C_DONE3 + 1;
+constexpr int C_GATH3 = C_DONE4 + 1;       // 3-wave parallel-gather counters
+constexpr int C_GATH4 = C_GATH3 + 1;
+constexpr int C_GATH5 = C_GATH4 + 1;
 
 #define dot8 dot8_bf16
 #define wave_sum wave_reduce_sum
@@ -218,9 +221,13 @@ __device__ void loader_wave(const StreamArgs& a, char* lds, volatile int* ctrl,
   const long rowb = (long)C * 2;              // 8 KiB row bytes
   const long halfb = (long)I;                 // half down-row bytes (I*2/2)
 
-  int outstanding = 0;                        // slots issued, not yet gated
-  int pending_slot = -1;                      // slot awaiting READY publish
-  int pending_logical = -1;
+  // two-deep publish pipeline: slot s issues while s-1 lands; publish
+  // s-2 once outstanding <= pieces(s)+pieces(s-1). Depth-1 gating made
+  // the cadence the LANDING latency (~1.2 us/slot measured) instead of
+  // the issue rate (ldsdma-fill prices 0.64 us with overlap).
+  int pend_slot[2] = {-1, -1};
+  int pend_log[2] = {-1, -1};
+  int pend_pieces = 0;                        // pieces of slot s-1
   for (int s = 0; s < n_slots; ++s) {
     const int slot = s % RING_SLOTS;
     if (s >= RING_SLOTS) {
@@ -263,26 +270,30 @@ __device__ void loader_wave(const StreamArgs& a, char* lds, volatile int* ctrl,
       for (int j = 0; j < pieces; ++j)
         glds16_nt(src + (long)j * 1024 + lane * 16, ring_base + j * 1024);
     }
-    // gate: leave only the just-issued slot's pieces in flight (counted
-    // wait = its piece count: vmcnt retires in issue order, so everything
-    // older — the previous slot — has landed), then publish the PREVIOUS
-    // slot. hipcc does not track asm loads; the counts are ours (§5.7).
-    (void)outstanding;
-    if (pending_slot >= 0) {
-      if (pieces >= 16)
-        asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
-      else if (pieces >= 14)
-        asm volatile("s_waitcnt vmcnt(14)" ::: "memory");
+    // counted gate (hipcc does not track asm loads — §5.7): allow the
+    // newest TWO slots' pieces to stay in flight; everything older (slot
+    // s-2) has landed, publish it.
+    if (pend_slot[0] >= 0) {
+      const int allowed = pieces + pend_pieces;
+      if (allowed >= 32)
+        asm volatile("s_waitcnt vmcnt(32)" ::: "memory");
+      else if (allowed >= 30)
+        asm volatile("s_waitcnt vmcnt(30)" ::: "memory");
+      else if (allowed >= 28)
+        asm volatile("s_waitcnt vmcnt(28)" ::: "memory");
       else
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      if (lane == 0) ctrl[C_RINGF + pending_slot] = pending_logical + 1;
+      if (lane == 0) ctrl[C_RINGF + pend_slot[0]] = pend_log[0] + 1;
     }
-    pending_slot = slot;
-    pending_logical = s;
+    pend_slot[0] = pend_slot[1]; pend_log[0] = pend_log[1];
+    pend_slot[1] = slot; pend_log[1] = s;
+    pend_pieces = pieces;
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  if (pending_slot >= 0 && lane == 0)
-    ctrl[C_RINGF + pending_slot] = pending_logical + 1;
+  if (lane == 0) {
+    if (pend_slot[0] >= 0) ctrl[C_RINGF + pend_slot[0]] = pend_log[0] + 1;
+    if (pend_slot[1] >= 0) ctrl[C_RINGF + pend_slot[1]] = pend_log[1] + 1;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -413,9 +424,19 @@ k_stream_layer(StreamArgs a) {
     const unsigned qtag = tagbase + 1;
 
     if (start < end) {
-      // every S1 producer done -> the qkv data sweeps below are one pass
-      if (!sweep_done(&a.g_done[1 * NWG], NWG, tagbase + 1, a.fail, 40,
-                      lane)) return;
+      // poll only the WGs producing THIS kv-head's qkv slice (rows
+      // [kvh*G*D, +G*D) and the k/v-new rows) instead of all 256 — the
+      // attention starts the moment its own producers finish
+      {
+        const int q_lo = (kvh * G * D) / r1;
+        const int q_hi = (kvh * G * D + G * D - 1) / r1;
+        const int k_lo = (Hq * D + kvh * D) / r1;
+        const int k_hi = ((Hq + Hkv) * D + kvh * D + D - 1) / r1;
+        if (!sweep_done(&a.g_done[1 * NWG + q_lo], q_hi - q_lo + 1,
+                        tagbase + 1, a.fail, 40, lane)) return;
+        if (!sweep_done(&a.g_done[1 * NWG + k_lo], k_hi - k_lo + 1,
+                        tagbase + 1, a.fail, 40, lane)) return;
+      }
       // gather + rope q for this group's heads (sweep its qkv granules)
       {
         int spins = 0;
@@ -634,32 +655,36 @@ k_stream_layer(StreamArgs a) {
   }
 
   // ------------------------------------------------------------------ S3 --
-  // wave 0 gathers att into the x-region (all heads, bf16)
+  // att gather: done-poll (wave 0) -> ONE agent acquire (drops this CU's
+  // stale L1 lines of the granule buffers) -> flag -> ALL THREE consumer
+  // waves plain-load their third 16 B/lane wide (Guideline 16 R1 consumer
+  // form; the producers stored sc1/write-through). A one-wave 8-byte sc1
+  // sweep measured 87 us on the act edge (r2c5 probe).
   if (wave == 0) {
     if (!sweep_done(&a.g_done[3 * NWG], Hq, tagbase + 3, a.fail, 42, lane))
       return;
-    u64g* src = a.g_att;
-    const int n_g = Hq * D / 2;                      // granules (2 vals each)
-    int spins = 0;
-    for (;;) {
-      bool ok = true;
-      for (int i = lane; i < n_g; i += 64) {
-        const u64g x = get_granule(&src[i]);
-        ok &= (unsigned)(x >> 32) == tagbase + 3;
-        ((unsigned*)xl)[i] = (unsigned)x;            // 2 bf16 at once
-      }
-      if (__all(ok)) break;
-      __builtin_amdgcn_s_sleep(8);
-      if (++spins > (SPIN_LIMIT >> 4)) {
-        if (lane == 0) atomicCAS(a.fail, 0, 27);
-        return;
-      }
-    }
+    if (lane == 0)
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     __threadfence_block();
     if (lane == 0) ctrl[C_STAGE] = 2;
-    stamp(a, wg, 4, lane);                       // att gathered
   } else {
     if (!lds_wait_ge(&ctrl[C_STAGE], 2, a.fail, 28)) return;
+  }
+  {
+    // granule PAIRS: 16 B = 2x{tag, 2xbf16}; each wave a contiguous third
+    const int n_pair = Hq * D / 4;                   // 1024 pairs
+    const int per = (n_pair + 2) / 3;
+    const int p_lo = wave * per, p_hi = min(p_lo + per, n_pair);
+    const ulonglong2* src2 = (const ulonglong2*)a.g_att;
+    for (int i = p_lo + lane; i < p_hi; i += 64) {
+      const ulonglong2 v = src2[i];
+      ((unsigned*)xl)[2 * i] = (unsigned)v.x;        // 2 bf16
+      ((unsigned*)xl)[2 * i + 1] = (unsigned)v.y;
+    }
+    __threadfence_block();
+    if (lane == 0) atomicAdd((int*)&ctrl[C_GATH3], 1);
+    if (!lds_wait_ge(&ctrl[C_GATH3], 3, a.fail, 27)) return;
+    if (wave == 0) stamp(a, wg, 4, lane);            // att gathered
   }
 
   // S3 slots: 2 o rows; publish h2 granules (2 bf16) with residual add
@@ -695,30 +720,42 @@ k_stream_layer(StreamArgs a) {
     if (!lds_wait_ge(&ctrl[C_DONE3], 3, a.fail, 30)) return;
     if (!sweep_done(&a.g_done[4 * NWG], NWG, tagbase + 4, a.fail, 43, lane))
       return;
+    if (lane == 0)
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    __threadfence_block();
+    if (lane == 0) ctrl[C_STAGE] = 3;
+  } else {
+    if (!lds_wait_ge(&ctrl[C_STAGE], 3, a.fail, 32)) return;
+  }
+  {
+    // parallel gather of raw h2 + per-wave sumsq partials, then every
+    // wave normalizes its third (bf16-rounded like k_gemv_norm)
+    float* ssred = (float*)(ctrl + 32);              // 3 f32 partials
+    const int n_pair = C / 4;                        // 1024 pairs
+    const int per = (n_pair + 2) / 3;
+    const int p_lo = wave * per, p_hi = min(p_lo + per, n_pair);
+    const ulonglong2* src2 = (const ulonglong2*)a.g_h2;
     float ss = 0.f;
-    int spins = 0;
-    for (;;) {
-      bool ok = true;
-      ss = 0.f;
-      for (int i = lane; i < C / 2; i += 64) {
-        const u64g x = get_granule(&a.g_h2[i]);
-        ok &= (unsigned)(x >> 32) == tagbase + 4;
-        const unsigned p = (unsigned)x;
-        ((unsigned*)xl)[i] = p;                      // raw h2 pair
-        const float f0 = bf2f((u16)(p & 0xffff));
-        const float f1 = bf2f((u16)(p >> 16));
-        ss = fmaf(f0, f0, fmaf(f1, f1, ss));
-      }
-      if (__all(ok)) break;
-      __builtin_amdgcn_s_sleep(8);
-      if (++spins > (SPIN_LIMIT >> 4)) {
-        if (lane == 0) atomicCAS(a.fail, 0, 31);
-        return;
-      }
+    for (int i = p_lo + lane; i < p_hi; i += 64) {
+      const ulonglong2 v = src2[i];
+      const unsigned pa = (unsigned)v.x, pb = (unsigned)v.y;
+      ((unsigned*)xl)[2 * i] = pa;
+      ((unsigned*)xl)[2 * i + 1] = pb;
+      const float f0 = bf2f((u16)(pa & 0xffff)), f1 = bf2f((u16)(pa >> 16));
+      const float f2 = bf2f((u16)(pb & 0xffff)), f3 = bf2f((u16)(pb >> 16));
+      ss = fmaf(f0, f0, fmaf(f1, f1, fmaf(f2, f2, fmaf(f3, f3, ss))));
     }
-    const float inv = rsqrtf(wave_sum(ss) / (float)C + a.eps);
+    ss = wave_sum(ss);
+    if (lane == 0) ssred[wave] = ss;
+    __threadfence_block();
+    if (lane == 0) atomicAdd((int*)&ctrl[C_GATH4], 1);
+    if (!lds_wait_ge(&ctrl[C_GATH4], 3, a.fail, 31)) return;
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    const float inv = rsqrtf((ssred[0] + ssred[1] + ssred[2]) / (float)C
+                             + a.eps);
     const s16x8* wn = (const s16x8*)a.norm_mlp;
-    for (int i = lane; i < (C >> 3); i += 64) {
+    const int v_lo = p_lo / 2, v_hi = p_hi / 2;      // vec8 range (C>>3)
+    for (int i = v_lo + lane; i < v_hi; i += 64) {
       s16x8 v = ((s16x8*)xl)[i], w = wn[i], o;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
@@ -726,10 +763,9 @@ k_stream_layer(StreamArgs a) {
       ((s16x8*)xl)[i] = o;
     }
     __threadfence_block();
-    if (lane == 0) ctrl[C_STAGE] = 3;
-    stamp(a, wg, 6, lane);                       // x2 ready
-  } else {
-    if (!lds_wait_ge(&ctrl[C_STAGE], 3, a.fail, 32)) return;
+    if (lane == 0) atomicAdd((int*)&ctrl[C_GATH4], 1);
+    if (!lds_wait_ge(&ctrl[C_GATH4], 6, a.fail, 31)) return;
+    if (wave == 0) stamp(a, wg, 6, lane);            // x2 ready
   }
 
   // S4 slots: {gate row, up row}; publish act granule (f32 payload)
@@ -766,26 +802,27 @@ k_stream_layer(StreamArgs a) {
     if (!lds_wait_ge(&ctrl[C_DONE4], 3, a.fail, 34)) return;
     if (!sweep_done(&a.g_done[5 * NWG], NWG, tagbase + 5, a.fail, 44, lane))
       return;
-    int spins = 0;
-    for (;;) {
-      bool ok = true;
-      for (int i = lane; i < I; i += 64) {
-        const u64g x = get_granule(&a.g_act[i]);
-        ok &= (unsigned)(x >> 32) == tagbase + 5;
-        actl[i] = f2bf(u2f((unsigned)x));
-      }
-      if (__all(ok)) break;
-      __builtin_amdgcn_s_sleep(8);
-      if (++spins > (SPIN_LIMIT >> 4)) {
-        if (lane == 0) atomicCAS(a.fail, 0, 35);
-        return;
-      }
-    }
+    if (lane == 0)
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     __threadfence_block();
     if (lane == 0) ctrl[C_STAGE] = 4;
-    stamp(a, wg, 8, lane);                       // act gathered
   } else {
     if (!lds_wait_ge(&ctrl[C_STAGE], 4, a.fail, 36)) return;
+  }
+  {
+    const int n_pair = I / 2;                        // f32-payload pairs
+    const int per = (n_pair + 2) / 3;
+    const int p_lo = wave * per, p_hi = min(p_lo + per, n_pair);
+    const ulonglong2* src2 = (const ulonglong2*)a.g_act;
+    for (int i = p_lo + lane; i < p_hi; i += 64) {
+      const ulonglong2 v = src2[i];
+      actl[2 * i] = f2bf(u2f((unsigned)v.x));
+      actl[2 * i + 1] = f2bf(u2f((unsigned)v.y));
+    }
+    __threadfence_block();
+    if (lane == 0) atomicAdd((int*)&ctrl[C_GATH5], 1);
+    if (!lds_wait_ge(&ctrl[C_GATH5], 3, a.fail, 35)) return;
+    if (wave == 0) stamp(a, wg, 8, lane);            // act gathered
   }
 
   // S5 slots: half down-rows; row r = slots (2r, 2r+1) -> wave r%3; the
