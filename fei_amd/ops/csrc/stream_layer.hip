@@ -69,9 +69,23 @@ constexpr int C_GATH3 = C_DONE4 + 1;       // 3-wave parallel-gather counters
 constexpr int C_GATH4 = C_GATH3 + 1;
 constexpr int C_GATH5 = C_GATH4 + 1;
 
-#define dot8 dot8_bf16
 #define wave_sum wave_reduce_sum
 #define wave_max wave_reduce_max
+
+typedef short s16x2 __attribute__((ext_vector_type(2)));
+
+// packed-bf16 dot via v_dot2_f32_bf16 (2 MACs/instruction; the consumer
+// retire instruction the ldsdma-fill row prices at 0.5 us/slot/wave) —
+// ~6x fewer VALU ops than the fmaf+cvt chain
+__device__ __forceinline__ float dot8(s16x8 a, s16x8 b) {
+  float acc = 0.f;
+  const s16x2* ap = (const s16x2*)&a;
+  const s16x2* bp = (const s16x2*)&b;
+#pragma unroll
+  for (int j = 0; j < 4; ++j)
+    acc = __builtin_amdgcn_fdot2_f32_bf16(ap[j], bp[j], acc, false);
+  return acc;
+}
 
 __device__ __forceinline__ u32 f2u(float f) {
   union { float f; u32 i; } cv; cv.f = f; return cv.i;
@@ -221,13 +235,13 @@ __device__ void loader_wave(const StreamArgs& a, char* lds, volatile int* ctrl,
   const long rowb = (long)C * 2;              // 8 KiB row bytes
   const long halfb = (long)I;                 // half down-row bytes (I*2/2)
 
-  // two-deep publish pipeline: slot s issues while s-1 lands; publish
-  // s-2 once outstanding <= pieces(s)+pieces(s-1). Depth-1 gating made
-  // the cadence the LANDING latency (~1.2 us/slot measured) instead of
-  // the issue rate (ldsdma-fill prices 0.64 us with overlap).
-  int pend_slot[2] = {-1, -1};
-  int pend_log[2] = {-1, -1};
-  int pend_pieces = 0;                        // pieces of slot s-1
+  // three-deep publish pipeline: slots s, s-1, s-2 in flight; publish
+  // s-3 once outstanding <= their piece sum. Depth-1 gating made the
+  // cadence the LANDING latency (~1.2 us/slot measured) instead of the
+  // issue rate (ldsdma-fill prices 0.64 us with overlap).
+  int pend_slot[3] = {-1, -1, -1};
+  int pend_log[3] = {-1, -1, -1};
+  int pp1 = 0, pp2 = 0;                       // pieces of s-1, s-2
   for (int s = 0; s < n_slots; ++s) {
     const int slot = s % RING_SLOTS;
     if (s >= RING_SLOTS) {
@@ -271,28 +285,31 @@ __device__ void loader_wave(const StreamArgs& a, char* lds, volatile int* ctrl,
         glds16_nt(src + (long)j * 1024 + lane * 16, ring_base + j * 1024);
     }
     // counted gate (hipcc does not track asm loads — §5.7): allow the
-    // newest TWO slots' pieces to stay in flight; everything older (slot
-    // s-2) has landed, publish it.
+    // newest THREE slots' pieces in flight; slot s-3 has landed, publish.
     if (pend_slot[0] >= 0) {
-      const int allowed = pieces + pend_pieces;
-      if (allowed >= 32)
-        asm volatile("s_waitcnt vmcnt(32)" ::: "memory");
-      else if (allowed >= 30)
-        asm volatile("s_waitcnt vmcnt(30)" ::: "memory");
-      else if (allowed >= 28)
-        asm volatile("s_waitcnt vmcnt(28)" ::: "memory");
+      const int allowed = pieces + pp1 + pp2;
+      if (allowed >= 48)
+        asm volatile("s_waitcnt vmcnt(48)" ::: "memory");
+      else if (allowed >= 46)
+        asm volatile("s_waitcnt vmcnt(46)" ::: "memory");
+      else if (allowed >= 44)
+        asm volatile("s_waitcnt vmcnt(44)" ::: "memory");
+      else if (allowed >= 42)
+        asm volatile("s_waitcnt vmcnt(42)" ::: "memory");
       else
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       if (lane == 0) ctrl[C_RINGF + pend_slot[0]] = pend_log[0] + 1;
     }
     pend_slot[0] = pend_slot[1]; pend_log[0] = pend_log[1];
-    pend_slot[1] = slot; pend_log[1] = s;
-    pend_pieces = pieces;
+    pend_slot[1] = pend_slot[2]; pend_log[1] = pend_log[2];
+    pend_slot[2] = slot; pend_log[2] = s;
+    pp2 = pp1; pp1 = pieces;
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   if (lane == 0) {
-    if (pend_slot[0] >= 0) ctrl[C_RINGF + pend_slot[0]] = pend_log[0] + 1;
-    if (pend_slot[1] >= 0) ctrl[C_RINGF + pend_slot[1]] = pend_log[1] + 1;
+#pragma unroll
+    for (int q = 0; q < 3; ++q)
+      if (pend_slot[q] >= 0) ctrl[C_RINGF + pend_slot[q]] = pend_log[q] + 1;
   }
 }
 
@@ -345,15 +362,19 @@ k_stream_layer(StreamArgs a) {
   if (wave == 0) stamp(a, wg, 0, lane);
 
   // ------------------------------------------------------------------ S1 --
+  // norm prologue from the layer input tensor, all three consumer waves
+  // in parallel (plain loads: written by the previous launch, boundary-
+  // synchronised). Sumsq partials combine through LDS.
   u16* xl = (u16*)(lds + LDS_X);        // stage-input LDS region (bf16[C])
-  if (wave == 0) {
-    // norm prologue from the layer input tensor (plain loads; the input
-    // was written by the previous launch — boundary-synchronised)
+  {
+    float* ssred = (float*)(ctrl + 36);
     const s16x8* xr = (const s16x8*)a.x_in;
     const s16x8* wn = (const s16x8*)a.norm_attn;
-    float ss = 0.f;
     const int nv = C >> 3;
-    for (int i = lane; i < nv; i += 64) {
+    const int per = (nv + 2) / 3;
+    const int v_lo = wave * per, v_hi = min(v_lo + per, nv);
+    float ss = 0.f;
+    for (int i = v_lo + lane; i < v_hi; i += 64) {
       s16x8 v = xr[i];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -361,8 +382,15 @@ k_stream_layer(StreamArgs a) {
         ss = fmaf(f, f, ss);
       }
     }
-    const float inv = rsqrtf(wave_sum(ss) / (float)C + a.eps);
-    for (int i = lane; i < nv; i += 64) {
+    ss = wave_sum(ss);
+    if (lane == 0) ssred[wave] = ss;
+    __threadfence_block();
+    if (lane == 0) atomicAdd((int*)&ctrl[C_STAGE], 1);
+    if (!lds_wait_ge(&ctrl[C_STAGE], 3, a.fail, 20)) return;
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    const float inv = rsqrtf((ssred[0] + ssred[1] + ssred[2]) / (float)C
+                             + a.eps);
+    for (int i = v_lo + lane; i < v_hi; i += 64) {
       s16x8 v = xr[i], w = wn[i], o;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
@@ -370,10 +398,9 @@ k_stream_layer(StreamArgs a) {
       ((s16x8*)xl)[i] = o;
     }
     __threadfence_block();
-    if (lane == 0) ctrl[C_STAGE] = 1;
-    stamp(a, wg, 1, lane);
-  } else {
-    if (!lds_wait_ge(&ctrl[C_STAGE], 1, a.fail, 20)) return;
+    if (lane == 0) atomicAdd((int*)&ctrl[C_STAGE], 1);
+    if (!lds_wait_ge(&ctrl[C_STAGE], 6, a.fail, 20)) return;
+    if (wave == 0) stamp(a, wg, 1, lane);
   }
 
   // S1 slots: 2 qkv rows each, slot s -> wave s%3; publish one granule
@@ -666,9 +693,9 @@ k_stream_layer(StreamArgs a) {
     if (lane == 0)
       __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     __threadfence_block();
-    if (lane == 0) ctrl[C_STAGE] = 2;
+    if (lane == 0) ctrl[C_STAGE] = 7;
   } else {
-    if (!lds_wait_ge(&ctrl[C_STAGE], 2, a.fail, 28)) return;
+    if (!lds_wait_ge(&ctrl[C_STAGE], 7, a.fail, 28)) return;
   }
   {
     // granule PAIRS: 16 B = 2x{tag, 2xbf16}; each wave a contiguous third
@@ -723,9 +750,9 @@ k_stream_layer(StreamArgs a) {
     if (lane == 0)
       __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     __threadfence_block();
-    if (lane == 0) ctrl[C_STAGE] = 3;
+    if (lane == 0) ctrl[C_STAGE] = 8;
   } else {
-    if (!lds_wait_ge(&ctrl[C_STAGE], 3, a.fail, 32)) return;
+    if (!lds_wait_ge(&ctrl[C_STAGE], 8, a.fail, 32)) return;
   }
   {
     // parallel gather of raw h2 + per-wave sumsq partials, then every
@@ -805,9 +832,9 @@ k_stream_layer(StreamArgs a) {
     if (lane == 0)
       __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     __threadfence_block();
-    if (lane == 0) ctrl[C_STAGE] = 4;
+    if (lane == 0) ctrl[C_STAGE] = 9;
   } else {
-    if (!lds_wait_ge(&ctrl[C_STAGE], 4, a.fail, 36)) return;
+    if (!lds_wait_ge(&ctrl[C_STAGE], 9, a.fail, 36)) return;
   }
   {
     const int n_pair = I / 2;                        // f32-payload pairs
