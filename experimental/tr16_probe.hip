@@ -23,7 +23,7 @@ __global__ void k_tr16_probe(u16* __restrict__ out, int scheme) {
     case 4: elem = (lane & 3) * 4; break;            // 8B-aligned var
     default: elem = lane; break;
   }
-  const u16* addr = &buf[elem];
+  const unsigned addr = (unsigned)(unsigned long)&buf[elem];   // LDS byte addr
   s16x4 r0, r4;
   asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
                "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"

@@ -67,7 +67,7 @@ def _try_load() -> None:
     lib.fei_stream_layer_check.restype = _i
     lib.fei_stream_layer.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp, _vp,
                                      _vp, _vp, _vp, _vp, _vp, _vp, _vp, _vp,
-                                     _vp, _vp, _vp, _vp, _i, _i, _i, _i, _i,
+                                     _vp, _vp, _vp, _i, _i, _i, _i, _i,
                                      _i, _i, _f, _f, _vp]
     lib.fei_attn_decode_paged.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp,
                                           _i, _i, _i, _i, _i, _i, _i, _f,
@@ -411,7 +411,6 @@ def stream_workspace(spec, device) -> dict:
     mk = lambda n: torch.zeros(n, dtype=torch.int64, device=device)
     return {
         "g_qkv": mk((Hq + 2 * Hkv) * D),
-        "g_part": mk(Hq * STREAM_NSPLIT * (D + 2)),
         "g_att": mk(Hq * D // 2),
         "g_h2": mk(C // 2),
         "g_act": mk(I),
@@ -433,7 +432,7 @@ def stream_layer(x_in: torch.Tensor, h_out: torch.Tensor, lw, spec,
         _ptr(x_in), _ptr(h_out), _ptr(lw.wqkv), _ptr(lw.wo), _ptr(lw.wgu),
         _ptr(lw.wdown), _ptr(lw.norm_attn), _ptr(lw.norm_mlp),
         _ptr(k_cache), _ptr(v_cache), _ptr(cos_sin), _ptr(pos),
-        _ptr(ws["g_qkv"]), _ptr(ws["g_part"]), _ptr(ws["g_att"]),
+        _ptr(ws["g_qkv"]), _ptr(ws["g_att"]),
         _ptr(ws["g_h2"]), _ptr(ws["g_act"]), _ptr(ws["g_done"]),
         _ptr(ws["dbg"]) if "dbg" in ws else None,
         _ptr(ws["fail"]),

@@ -68,6 +68,8 @@ constexpr int C_DONE4 = C_DONE3 + 1;
 constexpr int C_GATH3 = C_DONE4 + 1;       // 3-wave parallel-gather counters
 constexpr int C_GATH4 = C_GATH3 + 1;
 constexpr int C_GATH5 = C_GATH4 + 1;
+constexpr int C_S2Q = C_GATH5 + 1;         // attention q/kn/vn staged
+constexpr int C_S2C = C_S2Q + 1;           // attention wave partials done
 
 #define wave_sum wave_reduce_sum
 #define wave_max wave_reduce_max
@@ -142,8 +144,7 @@ struct StreamArgs {
   const int* pos;         // [1]
   // granule workspace (u64 each)
   u64g* g_qkv;            // [Nqkv] f32 payload
-  u64g* g_part;           // [Hq, NSPLIT, D+2] f32 payload (o..., m, l)
-  u64g* g_att;            // [HqD] f32 payload
+  u64g* g_att;            // [Hq*D/2] 2xbf16 payload
   u64g* g_h2;             // [C/2] 2xbf16 payload
   u64g* g_act;            // [I] f32 payload
   u64g* g_done;           // [6, NWG] per-stage producer-done granules
@@ -430,30 +431,30 @@ k_stream_layer(StreamArgs a) {
   if (wave == 0) stamp(a, wg, 2, lane);          // S1 slots done
 
   // ------------------------------------------------------------------ S2 --
-  // wave 0: this WG's attention split (kvh = wg/NSPLIT, split = wg%NSPLIT).
-  // wave 1 (wg < Hq): S2b combine for head wg. wave 2: straight to S3 wait.
-  if (wave == 0) {
-    // wait until all 3 consumer waves left S1 (xl region is reused)
-    if (!lds_wait_ge(&ctrl[C_DONE1], 3, a.fail, 22)) return;
-    const int kvh = wg / NSPLIT;
-    const int split = wg % NSPLIT;
+  // Attention v2: ONE workgroup per kv-head (wg < Hkv); its three
+  // consumer waves split the keys, combine through LDS, and wave 0
+  // publishes the head-group's att directly. v1 (32 splits x 8 kvh over
+  // all 256 WGs + a cross-WG combine) spent ~46 us/layer in its three
+  // global dependency hops (r2c6 probe); the WG-local form has ONE
+  // producer hop and an 8-entry done poll.
+  if (wg < Hkv) {
+    const int kvh = wg;
     const int n = pos_now + 1;
-    const int chunk = (n + NSPLIT - 1) / NSPLIT;
-    const int start = split * chunk;
-    const int end = min(start + chunk, n);
     const int half = D / 2;
     const int p_new = n - 1;
-    // scratch in the x-region: qs [G][D] f32, pl [G][64] f32, kn/vn bf16[D]
-    float* qs = (float*)(lds + LDS_X);                    // G*128*4 = 2 KiB
-    float* pl = qs + G * D;                               // G*64*4  = 1 KiB
-    u16* kn = (u16*)(pl + G * 64);                        // 256 B
-    u16* vn = kn + D;                                     // 256 B
+    // scratch: qs [G][D] f32 + kn/vn bf16[D] in the x-region (dead after
+    // S1: C_DONE1 == 3); per-wave partials in the ACT region (free until
+    // S5): po3 [3][G][D] f32, ml3 [3][G][2] f32.
+    float* qs = (float*)(lds + LDS_X);
+    u16* kn = (u16*)(qs + G * D);
+    u16* vn = kn + D;
+    float* po3 = (float*)(lds + LDS_ACT);
+    float* ml3 = po3 + 3 * G * D;
     const unsigned qtag = tagbase + 1;
 
-    if (start < end) {
-      // poll only the WGs producing THIS kv-head's qkv slice (rows
-      // [kvh*G*D, +G*D) and the k/v-new rows) instead of all 256 — the
-      // attention starts the moment its own producers finish
+    if (wave == 0) {
+      if (!lds_wait_ge(&ctrl[C_DONE1], 3, a.fail, 22)) return;
+      // poll only the WGs producing THIS kv-head's qkv slice
       {
         const int q_lo = (kvh * G * D) / r1;
         const int q_hi = (kvh * G * D + G * D - 1) / r1;
@@ -464,7 +465,7 @@ k_stream_layer(StreamArgs a) {
         if (!sweep_done(&a.g_done[1 * NWG + k_lo], k_hi - k_lo + 1,
                         tagbase + 1, a.fail, 40, lane)) return;
       }
-      // gather + rope q for this group's heads (sweep its qkv granules)
+      // gather + rope + scale q (sc1 sweeps, small and already-ready)
       {
         int spins = 0;
         for (;;) {
@@ -473,7 +474,7 @@ k_stream_layer(StreamArgs a) {
             const int g = i / D, d = i % D;
             const u64g x = get_granule(&a.g_qkv[(long)(kvh * G + g) * D + d]);
             ok &= (unsigned)(x >> 32) == qtag;
-            qs[g * D + d] = u2f((unsigned)x);             // raw bf16-as-f32
+            qs[g * D + d] = u2f((unsigned)x);
           }
           if (__all(ok)) break;
           __builtin_amdgcn_s_sleep(8);
@@ -482,7 +483,6 @@ k_stream_layer(StreamArgs a) {
             return;
           }
         }
-        // rope + scale in place (pairs d, d+half)
         for (int i = lane; i < G * half; i += 64) {
           const int g = i / half, d = i % half;
           const float c = a.cos_sin[((long)p_new * half + d) * 2 + 0];
@@ -492,7 +492,8 @@ k_stream_layer(StreamArgs a) {
           qs[g * D + d + half] = (x2 * c + x1 * sn) * a.scale;
         }
       }
-      // gather + rope the new token's k,v (qkv indices HqD.. / HqD+HkvD..)
+      // gather + rope the new token's k,v; append to the caches (readers
+      // in THIS WG use the LDS copy; other WGs never touch key p_new)
       {
         int spins = 0;
         for (;;) {
@@ -513,7 +514,6 @@ k_stream_layer(StreamArgs a) {
             return;
           }
         }
-        // rope k_new in LDS; owner split also appends to the caches
         for (int d = lane; d < half; d += 64) {
           const float c = a.cos_sin[((long)p_new * half + d) * 2 + 0];
           const float sn = a.cos_sin[((long)p_new * half + d) * 2 + 1];
@@ -522,163 +522,121 @@ k_stream_layer(StreamArgs a) {
           const u16 k1 = f2bf(x2 * c + x1 * sn);
           kn[d] = k0;
           kn[d + half] = k1;
-          if (p_new >= start && p_new < end) {
-            u16* kcp = a.k_cache + ((long)kvh * a.max_seq + p_new) * D;
-            u16* vcp = a.v_cache + ((long)kvh * a.max_seq + p_new) * D;
-            kcp[d] = k0; kcp[d + half] = k1;
-            vcp[d] = vn[d]; vcp[d + half] = vn[d + half];
+          u16* kcp = a.k_cache + ((long)kvh * a.max_seq + p_new) * D;
+          u16* vcp = a.v_cache + ((long)kvh * a.max_seq + p_new) * D;
+          kcp[d] = k0; kcp[d + half] = k1;
+          vcp[d] = vn[d]; vcp[d + half] = vn[d + half];
+        }
+      }
+      __threadfence_block();
+      if (lane == 0) atomicAdd((int*)&ctrl[C_S2Q], 1);
+    } else {
+      if (!lds_wait_ge(&ctrl[C_S2Q], 1, a.fail, 25)) return;
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    }
+
+    // per-wave key range; online softmax (lane covers dims 2l, 2l+1)
+    const int per_k = (n + 2) / 3;
+    const int start = wave * per_k;
+    const int end = min(start + per_k, n);
+    const u16* kbase = a.k_cache + (long)kvh * a.max_seq * D;
+    const u16* vbase = a.v_cache + (long)kvh * a.max_seq * D;
+    float* pl = (float*)(ml3 + 3 * G * 2);         // [G][64] tile P scratch
+    pl += wave * G * 64;                           // per-wave slice
+    float m[G], l[G], o0[G], o1[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      m[g] = -1.0f / 0.0f; l[g] = 0.f; o0[g] = 0.f; o1[g] = 0.f;
+    }
+    for (int tile = start; tile < end; tile += 64) {
+      const int kk = tile + lane;
+      float sc[G];
+#pragma unroll
+      for (int g = 0; g < G; ++g) sc[g] = -1.0f / 0.0f;
+      if (kk < end) {
+        const bool is_new = (kk == p_new);
+        const s16x8* krow = is_new ? (const s16x8*)kn
+                                   : (const s16x8*)(kbase + (long)kk * D);
+#pragma unroll
+        for (int g = 0; g < G; ++g) sc[g] = 0.f;
+        for (int i = 0; i < D / 8; ++i) {
+          const s16x8 kv8 = krow[i];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const float kf = bf2f((u16)kv8[j]);
+#pragma unroll
+            for (int g = 0; g < G; ++g)
+              sc[g] = fmaf(qs[g * D + i * 8 + j], kf, sc[g]);
           }
         }
+      }
+#pragma unroll
+      for (int g = 0; g < G; ++g) {
+        const float tile_m = wave_max(sc[g]);
+        const float m_new = fmaxf(m[g], tile_m);
+        const float alpha = __expf(m[g] - m_new);
+        const float pv = (kk < end) ? __expf(sc[g] - m_new) : 0.f;
+        pl[g * 64 + lane] = pv;
+        m[g] = m_new;
+        l[g] = l[g] * alpha + wave_sum(pv);
+        o0[g] *= alpha; o1[g] *= alpha;
       }
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-
-      // online softmax over [start, end), 64-key tiles, lane-per-key
-      const u16* kbase = a.k_cache + (long)kvh * a.max_seq * D;
-      const u16* vbase = a.v_cache + (long)kvh * a.max_seq * D;
-      float m[G], l[G], o0[G], o1[G];
-#pragma unroll
-      for (int g = 0; g < G; ++g) {
-        m[g] = -1.0f / 0.0f; l[g] = 0.f; o0[g] = 0.f; o1[g] = 0.f;
-      }
-      for (int tile = start; tile < end; tile += 64) {
-        const int kk = tile + lane;
-        float sc[G];
-#pragma unroll
-        for (int g = 0; g < G; ++g) sc[g] = -1.0f / 0.0f;
-        if (kk < end) {
-          const bool is_new = (kk == p_new);
-          const s16x8* krow = is_new ? (const s16x8*)kn
-                                     : (const s16x8*)(kbase + (long)kk * D);
-#pragma unroll
-          for (int g = 0; g < G; ++g) sc[g] = 0.f;
-          for (int i = 0; i < D / 8; ++i) {
-            const s16x8 kv8 = krow[i];
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              const float kf = bf2f((u16)kv8[j]);
-#pragma unroll
-              for (int g = 0; g < G; ++g)
-                sc[g] = fmaf(qs[g * D + i * 8 + j], kf, sc[g]);
-            }
-          }
-        }
+      const int kmax = min(64, end - tile);
+      for (int kl = 0; kl < kmax; ++kl) {
+        const bool is_new = (tile + kl == p_new);
+        const u16* vrow = is_new ? vn + lane * 2
+                                 : vbase + (long)(tile + kl) * D + lane * 2;
+        const float v0 = bf2f(vrow[0]);
+        const float v1 = bf2f(vrow[1]);
 #pragma unroll
         for (int g = 0; g < G; ++g) {
-          const float tile_m = wave_max(sc[g]);
-          const float m_new = fmaxf(m[g], tile_m);
-          const float alpha = __expf(m[g] - m_new);
-          const float p = (kk < end) ? __expf(sc[g] - m_new) : 0.f;
-          pl[g * 64 + lane] = p;
-          m[g] = m_new;
-          l[g] = l[g] * alpha + wave_sum(p);
-          o0[g] *= alpha; o1[g] *= alpha;
+          const float pv = pl[g * 64 + kl];
+          o0[g] = fmaf(pv, v0, o0[g]);
+          o1[g] = fmaf(pv, v1, o1[g]);
         }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        // P*V: lane covers dims (2*lane, 2*lane+1); rows read coalesced
-        const int kmax = min(64, end - tile);
-        for (int kl = 0; kl < kmax; ++kl) {
-          const bool is_new = (tile + kl == p_new);
-          const u16* vrow = is_new ? vn + lane * 2
-                                   : vbase + (long)(tile + kl) * D + lane * 2;
-          const float v0 = bf2f(vrow[0]);
-          const float v1 = bf2f(vrow[1]);
+      }
+    }
+    // stage this wave's partials in LDS; wave 0 combines
 #pragma unroll
-          for (int g = 0; g < G; ++g) {
-            const float p = pl[g * 64 + kl];
-            o0[g] = fmaf(p, v0, o0[g]);
-            o1[g] = fmaf(p, v1, o1[g]);
-          }
-        }
+    for (int g = 0; g < G; ++g) {
+      float* po = po3 + ((long)wave * G + g) * D;
+      po[2 * lane] = o0[g];
+      po[2 * lane + 1] = o1[g];
+      if (lane == 0) {
+        ml3[(wave * G + g) * 2] = (start < end) ? m[g] : -1.0f / 0.0f;
+        ml3[(wave * G + g) * 2 + 1] = (start < end) ? l[g] : 0.f;
       }
-      // publish partials: [Hq, NSPLIT, D+2] f32 granules
+    }
+    __threadfence_block();
+    if (lane == 0) atomicAdd((int*)&ctrl[C_S2C], 1);
+    if (wave == 0) {
+      if (!lds_wait_ge(&ctrl[C_S2C], 3, a.fail, 26)) return;
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
       for (int g = 0; g < G; ++g) {
-        const int hq = kvh * G + g;
-        u64g* base = a.g_part + ((long)hq * NSPLIT + split) * (D + 2);
-        put_granule(&base[2 * lane], tagbase + 2, f2u(o0[g]));
-        put_granule(&base[2 * lane + 1], tagbase + 2, f2u(o1[g]));
-        if (lane == 0) {
-          put_granule(&base[D], tagbase + 2, f2u(m[g]));
-          put_granule(&base[D + 1], tagbase + 2, f2u(l[g]));
+        float mm = -1.0f / 0.0f;
+#pragma unroll
+        for (int wv = 0; wv < 3; ++wv)
+          mm = fmaxf(mm, ml3[(wv * G + g) * 2]);
+        float acc0 = 0.f, acc1 = 0.f, lt = 0.f;
+#pragma unroll
+        for (int wv = 0; wv < 3; ++wv) {
+          const float w = __expf(ml3[(wv * G + g) * 2] - mm);
+          lt = fmaf(w, ml3[(wv * G + g) * 2 + 1], lt);
+          const float* po = po3 + ((long)wv * G + g) * D;
+          acc0 = fmaf(w, po[2 * lane], acc0);
+          acc1 = fmaf(w, po[2 * lane + 1], acc1);
         }
+        const float inv_l = lt > 0.f ? 1.f / lt : 0.f;
+        const int hq = kvh * G + g;
+        put_granule(&a.g_att[(long)hq * (D / 2) + lane], tagbase + 3,
+                    pack_bf16(f2bf(acc0 * inv_l), f2bf(acc1 * inv_l)));
       }
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      if (lane == 0) put_granule(&a.g_done[2 * NWG + wg], tagbase + 2, 0);
-      stamp(a, wg, 3, lane);                     // partials published
-    } else {
-      // empty split: publish -inf/0 partials (combine sweeps all splits)
-      for (int g = 0; g < G; ++g) {
-        const int hq = kvh * G + g;
-        u64g* base = a.g_part + ((long)hq * NSPLIT + split) * (D + 2);
-        put_granule(&base[2 * lane], tagbase + 2, f2u(0.f));
-        put_granule(&base[2 * lane + 1], tagbase + 2, f2u(0.f));
-        if (lane == 0) {
-          put_granule(&base[D], tagbase + 2, f2u(-1.0f / 0.0f));
-          put_granule(&base[D + 1], tagbase + 2, f2u(0.f));
-        }
-      }
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      if (lane == 0) put_granule(&a.g_done[2 * NWG + wg], tagbase + 2, 0);
+      if (lane == 0) put_granule(&a.g_done[3 * NWG + kvh], tagbase + 3, 0);
+      stamp(a, wg, 3, lane);                     // att published
     }
-  }
-
-  // S2b: combine head `wg` (wave 1 of WGs 0..Hq-1); mirrors
-  // k_attn_decode_combine (split-major sweep, max-shifted exp weights)
-  if (wave == 1 && wg < Hq) {
-    const u64g* base = a.g_part + (long)wg * NSPLIT * (D + 2);
-    // producers for head wg are the 32 splits of its kv-head: contiguous
-    // WGs [kvh*NSPLIT, +NSPLIT). Poll their done granules, then ONE
-    // parallel data sweep through LDS (a serial per-split dependent wait
-    // was 32 sequential L2 round trips on the critical path).
-    const int kvh_h = wg / G;
-    if (!sweep_done(&a.g_done[2 * NWG + kvh_h * NSPLIT], NSPLIT,
-                    tagbase + 2, a.fail, 41, lane)) return;
-    float* po = (float*)(lds + LDS_ACT);          // [NSPLIT][D] f32
-    float* pml = po + NSPLIT * D;                 // [NSPLIT][2]
-    {
-      int spins = 0;
-      for (;;) {
-        bool ok = true;
-        for (int sp = 0; sp < NSPLIT; ++sp) {
-          const u64g* sb = &base[(long)sp * (D + 2)];
-          const u64g x0 = get_granule(&sb[2 * lane]);
-          const u64g x1v = get_granule(&sb[2 * lane + 1]);
-          ok &= ((unsigned)(x0 >> 32) == tagbase + 2) &
-              ((unsigned)(x1v >> 32) == tagbase + 2);
-          po[sp * D + 2 * lane] = u2f((unsigned)x0);
-          po[sp * D + 2 * lane + 1] = u2f((unsigned)x1v);
-          if (lane < 2) {
-            const u64g xm = get_granule(&sb[D + lane]);
-            ok &= (unsigned)(xm >> 32) == tagbase + 2;
-            pml[sp * 2 + lane] = u2f((unsigned)xm);
-          }
-        }
-        if (__all(ok)) break;
-        __builtin_amdgcn_s_sleep(16);
-        if (++spins > (SPIN_LIMIT >> 8)) {
-          if (lane == 0) atomicCAS(a.fail, 0, 26);
-          return;
-        }
-      }
-    }
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    float sm = (lane < NSPLIT) ? pml[lane * 2] : -1.0f / 0.0f;
-    float sl = (lane < NSPLIT) ? pml[lane * 2 + 1] : 0.f;
-    const float mstar = wave_max(sm);
-    const float wexp = (lane < NSPLIT) ? __expf(sm - mstar) : 0.f;
-    const float ltot = wave_sum(wexp * sl);
-    float acc0 = 0.f, acc1 = 0.f;
-    for (int sp = 0; sp < NSPLIT; ++sp) {
-      const float w = __shfl(wexp, sp);
-      acc0 = fmaf(w, po[sp * D + 2 * lane], acc0);
-      acc1 = fmaf(w, po[sp * D + 2 * lane + 1], acc1);
-    }
-    const float inv_l = ltot > 0.f ? 1.f / ltot : 0.f;
-    const u16 b0 = f2bf(acc0 * inv_l);
-    const u16 b1 = f2bf(acc1 * inv_l);
-    put_granule(&a.g_att[(long)wg * (D / 2) + lane], tagbase + 3,
-                pack_bf16(b0, b1));
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    if (lane == 0) put_granule(&a.g_done[3 * NWG + wg], tagbase + 3, 0);
   }
 
   // ------------------------------------------------------------------ S3 --
@@ -688,7 +646,7 @@ k_stream_layer(StreamArgs a) {
   // form; the producers stored sc1/write-through). A one-wave 8-byte sc1
   // sweep measured 87 us on the act edge (r2c5 probe).
   if (wave == 0) {
-    if (!sweep_done(&a.g_done[3 * NWG], Hq, tagbase + 3, a.fail, 42, lane))
+    if (!sweep_done(&a.g_done[3 * NWG], Hkv, tagbase + 3, a.fail, 42, lane))
       return;
     if (lane == 0)
       __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
@@ -910,7 +868,7 @@ void fei_stream_layer(const void* x_in, void* h_out, const void* wqkv,
                       const void* wo, const void* wgu, const void* wdown,
                       const void* norm_attn, const void* norm_mlp,
                       void* k_cache, void* v_cache, const float* cos_sin,
-                      const int* pos, void* g_qkv, void* g_part, void* g_att,
+                      const int* pos, void* g_qkv, void* g_att,
                       void* g_h2, void* g_act, void* g_done, void* dbg,
                       int* fail,
                       int C, int Hq, int Hkv, int D, int I, int max_seq,
@@ -923,7 +881,7 @@ void fei_stream_layer(const void* x_in, void* h_out, const void* wqkv,
   a.norm_attn = (const u16*)norm_attn; a.norm_mlp = (const u16*)norm_mlp;
   a.k_cache = (u16*)k_cache; a.v_cache = (u16*)v_cache;
   a.cos_sin = cos_sin; a.pos = pos;
-  a.g_qkv = (u64g*)g_qkv; a.g_part = (u64g*)g_part; a.g_att = (u64g*)g_att;
+  a.g_qkv = (u64g*)g_qkv; a.g_att = (u64g*)g_att;
   a.g_h2 = (u64g*)g_h2; a.g_act = (u64g*)g_act;
   a.g_done = (u64g*)g_done; a.dbg = (u64g*)dbg; a.fail = fail;
   a.C = C; a.Hq = Hq; a.Hkv = Hkv; a.D = D; a.I = I; a.max_seq = max_seq;
