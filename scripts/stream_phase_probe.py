@@ -36,11 +36,8 @@ out = {}
 prev = t0
 for i, nm in enumerate(names[1:], start=1):
     cur = d[:, i].astype("int64")
-    if i in (2, 3):            # attention stamps exist only in WGs 0..7
+    if i in (2, 3, 4):         # attention stamp chain: WGs 0..7 only
         dt = (cur[:8] - prev[:8]) / clk
-    elif i == 4:
-        dt = (cur - prev) / clk
-        dt = dt[dt > 0]        # skip WGs without the s2 stamp chain
     else:
         dt = (cur - prev) / clk
     out[nm] = round(float(dt.mean()), 2)
