@@ -47,11 +47,10 @@ extern "C" int tr16_probe_run() {
   hipMemcpy(h, d, sizeof(h), hipMemcpyDeviceToHost);
   for (int s = 0; s < 6; ++s) {
     printf("scheme %d:\n", s);
-    for (int lane = 0; lane < 64; lane += (lane == 3 ? 13 : 1)) {
+    for (int lane = 0; lane < 32; ++lane) {
       printf("  lane %2d:", lane);
       for (int j = 0; j < 8; ++j) printf(" %4d", h[s * 512 + lane * 8 + j]);
       printf("\n");
-      if (lane > 40) break;
     }
   }
   hipFree(d);

@@ -305,6 +305,14 @@ __device__ void loader_wave(const StreamArgs& a, char* lds, volatile int* ctrl,
     pend_slot[1] = pend_slot[2]; pend_log[1] = pend_log[2];
     pend_slot[2] = slot; pend_log[2] = s;
     pp2 = pp1; pp1 = pieces;
+    // loader stage-issue stamps (s1/s3/s4/s5 ends -> dbg 10..13)
+    if (a.dbg && lane == 0) {
+      const int s1e = r1 / 2, s3e = s1e + r3 / 2, s4e = s3e + r4;
+      if (s == s1e - 1) a.dbg[wg * 16 + 10] = __builtin_amdgcn_s_memtime();
+      else if (s == s3e - 1) a.dbg[wg * 16 + 11] = __builtin_amdgcn_s_memtime();
+      else if (s == s4e - 1) a.dbg[wg * 16 + 12] = __builtin_amdgcn_s_memtime();
+      else if (s == n_slots - 1) a.dbg[wg * 16 + 13] = __builtin_amdgcn_s_memtime();
+    }
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   if (lane == 0) {
@@ -554,11 +562,13 @@ k_stream_layer(StreamArgs a) {
 #pragma unroll
       for (int g = 0; g < G; ++g) sc[g] = -1.0f / 0.0f;
       if (kk < end) {
-        const bool is_new = (kk == p_new);
-        const s16x8* krow = is_new ? (const s16x8*)kn
-                                   : (const s16x8*)(kbase + (long)kk * D);
+        // p_new's K row comes from the LDS copy; all other rows from the
+        // cache (pointer select only — no flow divergence, loop unrolls)
+        const s16x8* krow = (kk == p_new)
+            ? (const s16x8*)kn : (const s16x8*)(kbase + (long)kk * D);
 #pragma unroll
         for (int g = 0; g < G; ++g) sc[g] = 0.f;
+#pragma unroll 4
         for (int i = 0; i < D / 8; ++i) {
           const s16x8 kv8 = krow[i];
 #pragma unroll
@@ -582,16 +592,30 @@ k_stream_layer(StreamArgs a) {
         o0[g] *= alpha; o1[g] *= alpha;
       }
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      // P*V. p_new is always the LAST key, so the cache-row loop below is
+      // branch-free and unrolls — a data-dependent branch per key
+      // serialised it into dependent L2 round trips (~39 us of S2 chain,
+      // r2c8 probe; same lesson as the launch kernel's break-free loop).
       const int kmax = min(64, end - tile);
-      for (int kl = 0; kl < kmax; ++kl) {
-        const bool is_new = (tile + kl == p_new);
-        const u16* vrow = is_new ? vn + lane * 2
-                                 : vbase + (long)(tile + kl) * D + lane * 2;
+      const int kmax_c = (tile + kmax == n) ? kmax - 1 : kmax;  // cache keys
+#pragma unroll 4
+      for (int kl = 0; kl < kmax_c; ++kl) {
+        const u16* vrow = vbase + (long)(tile + kl) * D + lane * 2;
         const float v0 = bf2f(vrow[0]);
         const float v1 = bf2f(vrow[1]);
 #pragma unroll
         for (int g = 0; g < G; ++g) {
           const float pv = pl[g * 64 + kl];
+          o0[g] = fmaf(pv, v0, o0[g]);
+          o1[g] = fmaf(pv, v1, o1[g]);
+        }
+      }
+      if (kmax_c < kmax) {                       // the new token's V (LDS)
+        const float v0 = bf2f(vn[lane * 2]);
+        const float v1 = bf2f(vn[lane * 2 + 1]);
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          const float pv = pl[g * 64 + kmax_c];
           o0[g] = fmaf(pv, v0, o0[g]);
           o1[g] = fmaf(pv, v1, o1[g]);
         }

@@ -44,4 +44,9 @@ for i, nm in enumerate(names[1:], start=1):
     prev = cur
 out["wave0_total"] = round(float(((d[:, 9].astype("int64") - t0) / clk).mean()), 2)
 out["loader_total"] = round(float(((d[:, 15].astype("int64") - d[:, 14].astype("int64")) / clk).mean()), 2)
+ld = d[:, [14, 10, 11, 12, 13]].astype("int64")
+for nm, a_, b_, ns in (("ld_s1", 0, 1, 12), ("ld_s3", 1, 2, 8),
+                       ("ld_s4", 2, 3, 56), ("ld_s5", 3, 4, 32)):
+    out[nm + "_per_slot"] = round(float(((ld[:, b_] - ld[:, a_]) / clk).mean())
+                                  / ns, 3)
 print(json.dumps(out))
