@@ -169,10 +169,13 @@ int main() {
   run_variant<3, false, true>(w, sink, cyc, "depth3 DEFAULT cons");
   run_variant<3, true, false>(w, sink, cyc, "depth3 nt FREE(no dots)");
   run_variant<4, true, false>(w, sink, cyc, "depth4 nt FREE");
-  run_variant<2, true, true, true>(w, sink, cyc, "BATCH depth2 nt cons");
-  run_variant<3, true, true, true>(w, sink, cyc, "BATCH depth3 nt cons");
-  run_variant<4, true, true, true>(w, sink, cyc, "BATCH depth4 nt cons");
-  run_variant<3, true, false, true>(w, sink, cyc, "BATCH depth3 nt FREE");
-  // correctness spot-check of the batched path: CONS dots nonzero pattern
+  // The BATCH variants are DISABLED: measured as a HANG on MI355X —
+  // evidence that global_load_lds's offset: immediate does not combine
+  // with an s_add-walked M0 the way a global-side offset would (the DMA
+  // appears to land outside the ring and corrupt the flag words). Kept
+  // compiled for the record; do not re-enable without an LDS-dest probe.
+  if (0) {
+    run_variant<3, true, true, true>(w, sink, cyc, "BATCH depth3 nt cons");
+  }
   return 0;
 }
