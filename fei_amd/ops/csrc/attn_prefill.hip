@@ -42,15 +42,23 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
   const int l15 = lane & 15;
   const int lg = lane >> 4;        // 16-lane group 0..3
 
-  // K/V tiles are stored with a T2 XOR swizzle (guide Guideline 4): a
-  // row-major [32][D] bf16 tile read column-slice-wise with ds_read_b128
-  // puts each 16-lane group on ONE bank (16-way conflict). Swizzling the
-  // 16-byte slot index by (row&7) spreads the group over 8 slots. Both
-  // the staging writes and all reads apply the same XOR.
+  // K tile: row-major with a T2 XOR swizzle (guide Guideline 4): read
+  // column-slice-wise with ds_read_b128, swizzling the 16-byte slot index
+  // by (row&7) spreads each 16-lane group over 8 slots.
+  // V tile: stored TRANSPOSED [col][key] in 8-key blocks so the PV
+  // B-fragment (8 consecutive keys of one column) is ONE 16-byte vector
+  // read — the r02 ablation (experimental/prefill_ablate.hip) measured
+  // the per-element PV gather at 56% of the kernel. The 8-key block
+  // index is XOR-swizzled by (col&7) to break the 128-byte column
+  // stride's bank pattern (<=2-way). The transpose cost moves to the
+  // staging scatter: 8 ds_write_b16 per global vec8 — 4x fewer LDS ops
+  // than the per-element PV reads they replace.
   __shared__ u16 kt[64][D];
-  __shared__ u16 vt[64][D];
+  __shared__ u16 vtT[D][64];
   __shared__ u16 p_lds[4][16][64];
 #define SWZ16(row, col8) ((col8) ^ ((row) & 7))
+#define VT_OFF(col, key) \
+  ((col) * 64 + ((((key) >> 3) ^ ((col) & 7)) << 3) + ((key) & 7))
 
   const int p0 = pos0[b];
   const int q_hi = min(qt * 64 + 64, S);           // exclusive rel row bound
@@ -90,13 +98,17 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
         const int kk = t * KV + key;
         const int dst = key * (D / 8) + SWZ16(key, col8);
         s16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+        s16x8 v8 = z;
         if (kk < kv_end) {
           ((s16x8*)kt)[dst] = *(const s16x8*)(kbase + (long)kk * D + col8 * 8);
-          ((s16x8*)vt)[dst] = *(const s16x8*)(vbase + (long)kk * D + col8 * 8);
+          v8 = *(const s16x8*)(vbase + (long)kk * D + col8 * 8);
         } else {
           ((s16x8*)kt)[dst] = z;
-          ((s16x8*)vt)[dst] = z;
         }
+        // V transposed scatter (VT_OFF block swizzle)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          ((u16*)vtT)[VT_OFF(col8 * 8 + j, key)] = (u16)v8[j];
       }
     }
     __syncthreads();
@@ -167,16 +179,12 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
       a_p[ka] = *(const s16x8*)(&p_lds[w][l15][ka * 32 + 8 * lg]);
 #pragma unroll
     for (int dc = 0; dc < DC; ++dc) {
+      const int col = dc * 16 + l15;
 #pragma unroll
       for (int ka = 0; ka < KA; ++ka) {
-        s16x8 b_v;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int vrow = ka * 32 + 8 * lg + j;
-          const int col = dc * 16 + l15;
-          const int vcol8 = SWZ16(vrow, col >> 3);
-          b_v[j] = (short)((u16*)vt)[vrow * D + vcol8 * 8 + (col & 7)];
-        }
+        // 8 consecutive keys of this column = one 16-byte read
+        const s16x8 b_v = *(const s16x8*)(
+            &((u16*)vtT)[VT_OFF(col, ka * 32 + 8 * lg)]);
         o_acc[dc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p[ka], b_v,
                                                             o_acc[dc], 0, 0, 0);
       }
