@@ -1,3 +1,5 @@
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import time, torch
 from fei_amd.engine.engine import LocalEngine
 eng = LocalEngine.create("llama3-8b", max_seq_len=4096, seed=7, use_hip_graph=False)
