@@ -192,6 +192,162 @@ k_ablate(const u16* __restrict__ q, const u16* __restrict__ kc,
   }
 }
 
+
+// v2: V stored transposed [col][key] with 8-key block swizzle; PV
+// B-fragment = ONE 16-byte vector read (the shipped r02 layout)
+__global__ void __launch_bounds__(256)
+k_ablate_v2(const u16* __restrict__ q, const u16* __restrict__ kc,
+            const u16* __restrict__ vc, u16* __restrict__ out,
+            int S, int Hq, int Hkv, int max_seq, float scale, long q_ts) {
+  const int qt = blockIdx.x, hq = blockIdx.y;
+  const int hkv = hq / (Hq / Hkv);
+  const int tid = threadIdx.x, w = tid >> 6, lane = tid & 63;
+  const int l15 = lane & 15, lg = lane >> 4;
+  __shared__ u16 kt[64][D];
+  __shared__ u16 vtT[D][64];
+  __shared__ u16 p_lds[4][16][64];
+#define VT_OFF(col, key) \
+  ((col) * 64 + ((((key) >> 3) ^ ((col) & 7)) << 3) + ((key) & 7))
+  const int q_hi = min(qt * 64 + 64, S);
+  const int kv_end = q_hi;
+  const int qrow_ld = min(qt * 64 + w * 16 + l15, S - 1);
+  s16x8 a_q[KS];
+  {
+    const u16* qp = q + (long)qrow_ld * q_ts + (long)hq * D + 8 * lg;
+#pragma unroll
+    for (int ks = 0; ks < KS; ++ks) a_q[ks] = *(const s16x8*)(qp + ks * 32);
+  }
+  float m_row[4], l_row[4];
+  f32x4 o_acc[DC];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m_row[r] = -1.0f / 0.0f; l_row[r] = 0.f; }
+#pragma unroll
+  for (int dc = 0; dc < DC; ++dc) o_acc[dc] = f32x4{0.f, 0.f, 0.f, 0.f};
+  const u16* kbase = kc + (long)hkv * max_seq * D;
+  const u16* vbase = vc + (long)hkv * max_seq * D;
+  const int ntiles = (kv_end + KVT - 1) / KVT;
+  for (int t = 0; t < ntiles; ++t) {
+    __syncthreads();
+    {
+      const int nv8 = KVT * D / 8;
+      for (int i = tid; i < nv8; i += 256) {
+        const int key = i / (D / 8), col8 = i % (D / 8);
+        const int kk = t * KVT + key;
+        const int dst = key * (D / 8) + SWZ16(key, col8);
+        s16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+        s16x8 v8 = z;
+        if (kk < kv_end) {
+          ((s16x8*)kt)[dst] = *(const s16x8*)(kbase + (long)kk * D + col8 * 8);
+          v8 = *(const s16x8*)(vbase + (long)kk * D + col8 * 8);
+        } else { ((s16x8*)kt)[dst] = z; }
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          ((u16*)vtT)[VT_OFF(col8 * 8 + j, key)] = (u16)v8[j];
+      }
+    }
+    __syncthreads();
+    f32x4 sfrag[NC];
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < KS; ++ks) {
+        const int krow = c * 16 + l15;
+        const int kcol8 = SWZ16(krow, ks * 4 + lg);
+        s16x8 b_k = *(const s16x8*)(&((s16x8*)kt)[krow * (D / 8) + kcol8]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[ks], b_k, acc,
+                                                      0, 0, 0);
+      }
+      sfrag[c] = acc;
+    }
+    float p_val[NC][4];
+    float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qrow_abs = qt * 64 + w * 16 + 4 * lg + r;
+      float sv[NC]; float mx = -1.0f / 0.0f;
+#pragma unroll
+      for (int c = 0; c < NC; ++c) {
+        sv[c] = sfrag[c][r] * scale;
+        const int key = t * KVT + c * 16 + l15;
+        if (key >= kv_end || key > qrow_abs) sv[c] = -1.0f / 0.0f;
+        mx = fmaxf(mx, sv[c]);
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        mx = fmaxf(mx, __shfl_xor(mx, off));
+      const float m_new = fmaxf(m_row[r], mx);
+      alpha[r] = __expf(m_row[r] - m_new);
+      m_row[r] = m_new;
+      float psum = 0.f;
+#pragma unroll
+      for (int c = 0; c < NC; ++c) {
+        p_val[c][r] = (sv[c] == -1.0f / 0.0f) ? 0.f : __expf(sv[c] - m_new);
+        psum += p_val[c][r];
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) psum += __shfl_xor(psum, off);
+      l_row[r] = l_row[r] * alpha[r] + psum;
+    }
+#pragma unroll
+    for (int dc = 0; dc < DC; ++dc)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) o_acc[dc][r] *= alpha[r];
+#pragma unroll
+    for (int c = 0; c < NC; ++c)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        p_lds[w][4 * lg + r][c * 16 + l15] = f2bf(p_val[c][r]);
+    s16x8 a_p[KA];
+#pragma unroll
+    for (int ka = 0; ka < KA; ++ka)
+      a_p[ka] = *(const s16x8*)(&p_lds[w][l15][ka * 32 + 8 * lg]);
+#pragma unroll
+    for (int dc = 0; dc < DC; ++dc) {
+      const int col = dc * 16 + l15;
+#pragma unroll
+      for (int ka = 0; ka < KA; ++ka) {
+        const s16x8 b_v = *(const s16x8*)(
+            &((u16*)vtT)[VT_OFF(col, ka * 32 + 8 * lg)]);
+        o_acc[dc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_p[ka], b_v, o_acc[dc], 0, 0, 0);
+      }
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int srow = qt * 64 + w * 16 + 4 * lg + r;
+    if (srow >= S) continue;
+    const float inv_l = l_row[r] > 0.f ? 1.f / l_row[r] : 1.f;
+    u16* orow = out + ((long)srow * Hq + hq) * D;
+#pragma unroll
+    for (int dc = 0; dc < DC; ++dc)
+      orow[dc * 16 + l15] = f2bf(o_acc[dc][r] * inv_l);
+  }
+#undef VT_OFF
+}
+
+static double run_v2(const u16* q, const u16* kc, const u16* vc, u16* out,
+                     int S, int Hq, int Hkv, int max_seq, long q_ts) {
+  dim3 grid((S + 63) / 64, Hq, 1);
+  hipEvent_t a, b;
+  (void)hipEventCreate(&a);
+  (void)hipEventCreate(&b);
+  for (int r = 0; r < 2; ++r)
+    hipLaunchKernelGGL(k_ablate_v2, grid, dim3(256), 0, 0, q, kc, vc, out,
+                       S, Hq, Hkv, max_seq, 0.0883883f, q_ts);
+  (void)hipEventRecord(a);
+  for (int r = 0; r < 10; ++r)
+    hipLaunchKernelGGL(k_ablate_v2, grid, dim3(256), 0, 0, q, kc, vc, out,
+                       S, Hq, Hkv, max_seq, 0.0883883f, q_ts);
+  (void)hipEventRecord(b);
+  (void)hipDeviceSynchronize();
+  float ms = 0;
+  (void)hipEventElapsedTime(&ms, a, b);
+  printf("%-10s %8.3f ms\n", "FULL_v2", ms / 10.0);
+  return ms / 10.0;
+}
+
 template <int CUT>
 static double run(const u16* q, const u16* kc, const u16* vc, u16* out,
                   int S, int Hq, int Hkv, int max_seq, long q_ts,
@@ -228,6 +384,8 @@ int main() {
   (void)hipMemset(kc, 0x31, (long)Hkv * max_seq * D * 2);
   (void)hipMemset(vc, 0x33, (long)Hkv * max_seq * D * 2);
   const double full = run<0>(q, kc, vc, out, S, Hq, Hkv, max_seq, q_ts, "FULL");
+  run_v2(q, kc, vc, out, S, Hq, Hkv, max_seq, q_ts);
+  run_v2(q, kc, vc, out, S, Hq, Hkv, max_seq, q_ts);
   const double nosm = run<1>(q, kc, vc, out, S, Hq, Hkv, max_seq, q_ts, "NOSM");
   const double nopv = run<2>(q, kc, vc, out, S, Hq, Hkv, max_seq, q_ts, "NOPV");
   const double noqk = run<3>(q, kc, vc, out, S, Hq, Hkv, max_seq, q_ts, "NOQK");
