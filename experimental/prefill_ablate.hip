@@ -207,7 +207,8 @@ k_ablate_v2(const u16* __restrict__ q, const u16* __restrict__ kc,
   __shared__ u16 vtT[D][64];
   __shared__ u16 p_lds[4][16][64];
 #define VT_OFF(col, key) \
-  ((col) * 64 + ((((key) >> 3) ^ ((col) & 7)) << 3) + ((key) & 7))
+  ((col) * 64 + \
+   ((((key) >> 3) ^ ((col) & 7) ^ (((col) >> 3) & 7)) << 3) + ((key) & 7))
   const int q_hi = min(qt * 64 + 64, S);
   const int kv_end = q_hi;
   const int qrow_ld = min(qt * 64 + w * 16 + l15, S - 1);

@@ -58,7 +58,8 @@ k_attn_prefill(const u16* __restrict__ q, const u16* __restrict__ kc,
   __shared__ u16 p_lds[4][16][64];
 #define SWZ16(row, col8) ((col8) ^ ((row) & 7))
 #define VT_OFF(col, key) \
-  ((col) * 64 + ((((key) >> 3) ^ ((col) & 7)) << 3) + ((key) & 7))
+  ((col) * 64 + \
+   ((((key) >> 3) ^ ((col) & 7) ^ (((col) >> 3) & 7)) << 3) + ((key) & 7))
 
   const int p0 = pos0[b];
   const int q_hi = min(qt * 64 + 64, S);           // exclusive rel row bound
