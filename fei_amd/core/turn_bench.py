@@ -31,8 +31,16 @@ class TurnBackend(LocalBackend):
         super().__init__(engine=engine, stop_on_eos=False)
         self.max_new = max_new
         self.tool_name = tool_name
-        self.tool_input = tool_input or {"pattern": "**/*.py",
-                                         "path": os.getcwd()}
+        # BOUNDED tool output: a repo-wide '**/*.py' glob returns hundreds
+        # of paths, overflowing the engine context so the continuation
+        # round's decode budget collapses to 1 token — the turn latency
+        # then measures a degenerate turn. A package-dir glob (~10 files)
+        # keeps both rounds decoding their full budget.
+        if tool_input is None:
+            import fei_amd.core as _core
+            tool_input = {"pattern": "*.py",
+                          "path": os.path.dirname(_core.__file__)}
+        self.tool_input = tool_input
         self._round = 0
 
     def complete(self, messages, tools=None, system=None, max_tokens=4000,
