@@ -42,7 +42,7 @@ class LocalEngine:
         spec: ModelSpec,
         device: Optional[torch.device] = None,
         batch_size: int = 1,
-        max_seq_len: Optional[int] = None,
+        max_seq_len: Union[int, str, None] = None,   # int | "hbm" | None
         use_hip_graph: Optional[bool] = None,
         tp: Optional[ParallelContext] = None,
         seed: int = 1234,
@@ -61,6 +61,12 @@ class LocalEngine:
             dtype = torch.bfloat16 if self.is_gpu else torch.float32
         self.dtype = dtype
         self.B = batch_size
+        # max_seq_len="hbm": size the KV caches toward the 288 GB of HBM3E
+        # instead of the model's nominal window (SURVEY §5 long-context
+        # note) — resolved after model init, when free memory is known
+        self._hbm_seq = (max_seq_len == "hbm")
+        if self._hbm_seq:
+            max_seq_len = spec.max_seq_len          # placeholder until init
         self.max_seq_len = max_seq_len or spec.max_seq_len
         self.tp = tp or ParallelContext()
         # hipGraph decode is on under TP too: RCCL collectives are
@@ -134,6 +140,18 @@ class LocalEngine:
             self.model.load_weights(w_path)
             logger.info("loaded weights from %s in %.1fs", w_path,
                         time.perf_counter() - t1)
+        if self._hbm_seq:
+            self.max_seq_len = self._resolve_hbm_seq()
+            # RoPE table must cover the enlarged window
+            from fei_amd.ops import reference as _ref
+            self.model.max_seq_len = self.max_seq_len
+            self.model.rope = _ref.rope_table(
+                self.max_seq_len, self.model.D, spec.rope_theta,
+                device=device)
+            logger.info("max_seq_len='hbm' -> %d tokens (%.1f GB KV)",
+                        self.max_seq_len,
+                        self._kv_bytes_per_token() * self.B *
+                        self.max_seq_len / 2 ** 30)
         self.k_caches, self.v_caches = self.model.new_kv_cache(self.B, self.max_seq_len)
         logger.info("model %s init in %.1fs (%.2f GB params)", spec.name,
                     time.perf_counter() - t0, self.model.param_bytes() / 2**30)
@@ -208,6 +226,28 @@ class LocalEngine:
             torch.cuda.synchronize(device)
 
     # -- construction helpers ------------------------------------------------
+
+    def _kv_bytes_per_token(self) -> int:
+        s = self.spec
+        el = 2 if self.dtype == torch.bfloat16 else 4
+        return (s.num_layers * (s.num_kv_heads //
+                                (self.tp.world_size
+                                 if self.tp.is_distributed else 1))
+                * s.head_dim * 2 * el)
+
+    def _resolve_hbm_seq(self) -> int:
+        """Largest KV window that fits the remaining device memory with
+        20% headroom (activations, graph pool, workspace), in 1k steps,
+        capped at 256k (the RoPE table and position math stay exact far
+        beyond any trained window; weights here are synthetic anyway)."""
+        if not self.is_gpu:
+            return self.spec.max_seq_len
+        free, _total = torch.cuda.mem_get_info(self.device)
+        budget = int(free * 0.8)
+        per_tok = self._kv_bytes_per_token() * self.B
+        seq = min(budget // per_tok, 262144)
+        seq = max((seq // 1024) * 1024, self.spec.max_seq_len)
+        return int(seq)
 
     @classmethod
     def create(cls, model: Union[str, ModelSpec] = "llama3-8b", **kwargs) -> "LocalEngine":

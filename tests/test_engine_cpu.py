@@ -562,3 +562,12 @@ def test_packaged_spm_tokenizer_default():
 
     tiny = LocalEngine.create("llama3-tiny", max_seq_len=64, seed=7)
     assert isinstance(tiny.tokenizer, ByteTokenizer)
+
+
+def test_hbm_seq_plumbing_cpu():
+    """max_seq_len='hbm' resolves to the spec window on CPU (the sizing
+    toward 288 GB happens on GPU; tests/test_engine_gpu.py covers it)."""
+    eng = LocalEngine.create("llama3-tiny", max_seq_len="hbm", seed=7)
+    assert eng.max_seq_len == eng.spec.max_seq_len
+    out = eng.generate("abc", max_new_tokens=4, stop_on_eos=False)
+    assert len(out["token_ids"]) == 4

@@ -210,3 +210,28 @@ def test_recapture_preserves_kv_state():
         return o2["token_ids"]
 
     assert run(True) == run(False)
+
+
+def test_hbm_sized_kv_decodes_past_nominal_window():
+    """max_seq_len='hbm' sizes the KV caches toward the 288 GB HBM3E
+    (SURVEY §5 long-context): an 8B engine gets a >=64k window and
+    decodes correctly PAST the nominal 8192 context (chunked prefill of
+    9k tokens + decode; RoPE table covers the full window)."""
+    import torch
+
+    from fei_amd.engine.engine import LocalEngine
+
+    eng = LocalEngine.create("llama3-8b", max_seq_len="hbm", seed=7,
+                             use_hip_graph=False)
+    try:
+        assert eng.max_seq_len >= 65536, eng.max_seq_len
+        assert eng.k_caches[0].shape[2] == eng.max_seq_len
+        rng = torch.Generator().manual_seed(5)
+        ids = torch.randint(4, 16000, (9000,), generator=rng).tolist()
+        out = eng.generate(ids, max_new_tokens=8, stop_on_eos=False)
+        assert len(out["token_ids"]) == 8
+        assert int(eng.pos[0]) == 9008        # past the nominal window
+        assert int(eng._stream_ws["fail"][0]) == 0 if eng.stream_decode \
+            else True
+    finally:
+        eng.shutdown()
