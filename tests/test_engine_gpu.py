@@ -230,7 +230,8 @@ def test_hbm_sized_kv_decodes_past_nominal_window():
         ids = torch.randint(4, 16000, (9000,), generator=rng).tolist()
         out = eng.generate(ids, max_new_tokens=8, stop_on_eos=False)
         assert len(out["token_ids"]) == 8
-        assert int(eng.pos[0]) == 9008        # past the nominal window
+        # pos counts context tokens: 9000 prompt + 7 fed-back generations
+        assert int(eng.pos[0]) == 9007        # past the nominal window
         assert int(eng._stream_ws["fail"][0]) == 0 if eng.stream_decode \
             else True
     finally:
