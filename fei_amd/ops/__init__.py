@@ -81,13 +81,13 @@ def _try_load() -> None:
     lib.fei_gemv_swiglu_norm_fp8.argtypes = [_vp, _vp, _vp, _vp, _vp, _i, _i,
                                              _i, _f, _vp]
     lib.fei_quant_fp8_rows.argtypes = [_vp, _vp, _vp, _i, _i, _vp]
-    lib.fei_gemv_swiglu.argtypes = [_vp, _vp, _vp, _i, _i, _i, _vp]
+    lib.fei_gemv_swiglu.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
     lib.fei_prefetch.argtypes = [_vp, _l, _vp, _i, _vp]
     lib.fei_gemv_res.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp, _vp]
     lib.fei_gemv_norm.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _i, _f, _i, _vp,
                                   _vp]
     lib.fei_gemv_swiglu_norm.argtypes = [_vp, _vp, _vp, _vp, _i, _i, _i, _f,
-                                         _vp, _vp]
+                                         _i, _vp, _vp]
     _LIB = lib
 
 
@@ -511,7 +511,9 @@ def gemv_swiglu(x: torch.Tensor, wgu: torch.Tensor,
     x2 = x.contiguous().view(M, K)
     if out is None:
         out = torch.empty(*x.shape[:-1], I, dtype=x.dtype, device=x.device)
-    lib.fei_gemv_swiglu(_ptr(out), _ptr(x2), _ptr(wgu), M, I, K, _stream())
+    nt = 1 if (2 * I * K * 2 >= _GEMV_NT_MIN_BYTES) else 0
+    lib.fei_gemv_swiglu(_ptr(out), _ptr(x2), _ptr(wgu), M, I, K, nt,
+                        _stream())
     return out
 
 
@@ -579,8 +581,9 @@ def gemv_swiglu_norm(res: torch.Tensor, wnorm: torch.Tensor,
     r2 = res.contiguous().view(M, K)
     if out is None:
         out = torch.empty(*res.shape[:-1], I, dtype=res.dtype, device=res.device)
+    nt = 1 if (wgu.shape[0] * K * 2 >= _GEMV_NT_MIN_BYTES) else 0
     lib.fei_gemv_swiglu_norm(_ptr(out), _ptr(r2), _ptr(wnorm), _ptr(wgu),
-                             M, I, K, eps,
+                             M, I, K, eps, nt,
                              _ptr(ssq) if ssq is not None else None,
                              _stream())
     return out

@@ -66,7 +66,7 @@ k_gemv(u16* __restrict__ out, const u16* __restrict__ x,
 }
 
 // gate/up + SwiGLU fused: W = [gate rows (N) ; up rows (N)] stacked, out[M,N]
-template <int M>
+template <int M, bool NT>
 __global__ void __launch_bounds__(256)
 k_gemv_swiglu(u16* __restrict__ out, const u16* __restrict__ x,
               const u16* __restrict__ w, int N, int K) {
@@ -84,8 +84,8 @@ k_gemv_swiglu(u16* __restrict__ out, const u16* __restrict__ x,
   for (int m = 0; m < M; ++m) { accg[m] = 0.f; accu[m] = 0.f; }
   const int nv = K >> 3;
   for (int i = lane; i < nv; i += 64) {
-    s16x8 gv = grow[i];
-    s16x8 uv = urow[i];
+    s16x8 gv = NT ? __builtin_nontemporal_load(&grow[i]) : grow[i];
+    s16x8 uv = NT ? __builtin_nontemporal_load(&urow[i]) : urow[i];
 #pragma unroll
     for (int m = 0; m < M; ++m) {
       s16x8 xv = ((const s16x8*)(xs + m * K))[i];
@@ -128,19 +128,21 @@ void fei_gemv(void* out, const void* x, const void* w, int M, int N, int K,
 }
 
 void fei_gemv_swiglu(void* out, const void* x, const void* w, int M, int N,
-                     int K, hipStream_t stream) {
+                     int K, int nontemporal, hipStream_t stream) {
   dim3 grid((N + 3) / 4);
   const size_t lds = 0;
-#define LG(MV) hipLaunchKernelGGL(k_gemv_swiglu<MV>, grid, dim3(256), lds, \
-                                  stream, (u16*)out, (const u16*)x, \
-                                  (const u16*)w, N, K)
+#define LG(MV, NTV) hipLaunchKernelGGL((k_gemv_swiglu<MV, NTV>), grid, \
+                                  dim3(256), lds, stream, (u16*)out, \
+                                  (const u16*)x, (const u16*)w, N, K)
+#define LGD(MV) do { if (nontemporal) LG(MV, true); else LG(MV, false); } while (0)
   switch (M) {
-    case 1: LG(1); break;
-    case 2: LG(2); break;
-    case 4: LG(4); break;
-    case 8: LG(8); break;
+    case 1: LGD(1); break;
+    case 2: LGD(2); break;
+    case 4: LGD(4); break;
+    case 8: LGD(8); break;
     default: break;
   }
+#undef LGD
 #undef LG
 }
 
@@ -302,7 +304,7 @@ k_gemv_norm(u16* __restrict__ out, const u16* __restrict__ res,
   }
 }
 
-template <int M>
+template <int M, bool NT>
 __global__ void __launch_bounds__(256)
 k_gemv_swiglu_norm(u16* __restrict__ out, const u16* __restrict__ res,
                    const u16* __restrict__ wn, const u16* __restrict__ w,
@@ -327,8 +329,8 @@ k_gemv_swiglu_norm(u16* __restrict__ out, const u16* __restrict__ res,
   for (int m = 0; m < M; ++m) { accg[m] = 0.f; accu[m] = 0.f; }
   const int nv = K >> 3;
   for (int i = lane; i < nv; i += 64) {
-    s16x8 gv = grow[i];
-    s16x8 uv = urow[i];
+    s16x8 gv = NT ? __builtin_nontemporal_load(&grow[i]) : grow[i];
+    s16x8 uv = NT ? __builtin_nontemporal_load(&urow[i]) : urow[i];
     s16x8 wnv = ((const s16x8*)wn)[i];
 #pragma unroll
     for (int m = 0; m < M; ++m) {
@@ -414,12 +416,15 @@ void fei_gemv_norm(void* out, const void* res, const void* wn, const void* w,
 
 void fei_gemv_swiglu_norm(void* out, const void* res, const void* wn,
                           const void* w, int M, int N, int K, float eps,
-                          const void* ssq, hipStream_t stream) {
+                          int nontemporal, const void* ssq,
+                          hipStream_t stream) {
   dim3 grid((N + 3) / 4);
-#define LS(MV, _ignored) hipLaunchKernelGGL(k_gemv_swiglu_norm<MV>, grid, \
+#define LS(MV, NTV) hipLaunchKernelGGL((k_gemv_swiglu_norm<MV, NTV>), grid, \
     dim3(256), 0, stream, (u16*)out, (const u16*)res, (const u16*)wn, \
     (const u16*)w, N, K, eps, (const float*)ssq)
-  DISPATCH_M(LS, 0)
+#define LSD(MV, _ignored) do { if (nontemporal) LS(MV, true); else LS(MV, false); } while (0)
+  DISPATCH_M(LSD, 0)
+#undef LSD
 #undef LS
 }
 
