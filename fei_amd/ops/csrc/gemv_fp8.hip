@@ -122,8 +122,8 @@ k_gemv_norm_fp8(u16* __restrict__ out, const u16* __restrict__ res,
   for (int m = 0; m < M; ++m) { acc0[m] = 0.f; acc1[m] = 0.f; }
   const int nv = K >> 4;                    // 16 elems per 16-B load
   for (int i = lane; i < nv; i += 64) {
-    const u32x4 wv0 = wrow0[i];
-    const u32x4 wv1 = wrow1[i];
+    const u32x4 wv0 = __builtin_nontemporal_load(&wrow0[i]);
+    const u32x4 wv1 = __builtin_nontemporal_load(&wrow1[i]);
 #pragma unroll
     for (int m = 0; m < M; ++m) {
       acc0[m] += dot16_fp8(nullptr, xs, (long)m * K + i * 16, wv0);
@@ -160,8 +160,8 @@ k_gemv_res_fp8(u16* __restrict__ res, const u16* __restrict__ x,
   for (int m = 0; m < M; ++m) { acc0[m] = 0.f; acc1[m] = 0.f; }
   const int nv = K >> 4;
   for (int i = lane; i < nv; i += 64) {
-    const u32x4 wv0 = wrow0[i];
-    const u32x4 wv1 = wrow1[i];
+    const u32x4 wv0 = __builtin_nontemporal_load(&wrow0[i]);
+    const u32x4 wv1 = __builtin_nontemporal_load(&wrow1[i]);
 #pragma unroll
     for (int m = 0; m < M; ++m) {
       acc0[m] += dot16_fp8(nullptr, x, (long)m * K + i * 16, wv0);
@@ -209,8 +209,8 @@ k_gemv_swiglu_norm_fp8(u16* __restrict__ out, const u16* __restrict__ res,
   for (int m = 0; m < M; ++m) { accg[m] = 0.f; accu[m] = 0.f; }
   const int nv = K >> 4;
   for (int i = lane; i < nv; i += 64) {
-    const u32x4 gv = grow[i];
-    const u32x4 uv = urow[i];
+    const u32x4 gv = __builtin_nontemporal_load(&grow[i]);
+    const u32x4 uv = __builtin_nontemporal_load(&urow[i]);
 #pragma unroll
     for (int m = 0; m < M; ++m) {
       accg[m] += dot16_fp8(nullptr, xs, (long)m * K + i * 16, gv);
