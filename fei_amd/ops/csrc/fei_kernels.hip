@@ -9,6 +9,7 @@
 //    decode step is hipGraph-capturable with a static grid
 //  - block sizes are multiples of 64 (wave64)
 #include "fei_common.h"
+#include <cstdlib>  // getenv: FEI_ATTN_KV_NT (decode KV cache policy)
 
 extern "C" {
 
@@ -261,7 +262,7 @@ __device__ __forceinline__ void block_reduce_vec(float* v,
 // fall in any split's chunk) and the split that OWNS key n-1 writes the
 // caches. Readers never read cache[n-1]; they use the LDS copy, so there is
 // no cross-workgroup ordering (placement-independent by construction).
-template <int G, bool ROPE>
+template <int G, bool ROPE, bool KVNT = false>
 __global__ void __launch_bounds__(256)
 k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
               u16* __restrict__ vc, float* __restrict__ part_o,
@@ -364,7 +365,7 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
       #pragma unroll
     for (int g = 0; g < G; ++g) sc[g] = 0.f;
       for (int i = 0; i < D / 8; ++i) {
-        s16x8 kv8 = krow[i];
+        s16x8 kv8 = KVNT ? __builtin_nontemporal_load(&krow[i]) : krow[i];
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const float kf = bf2f((u16)kv8[j]);
@@ -410,8 +411,11 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
     for (int j = 0; j < iters; ++j) {
       const int kl = kbase_local + j;
       const u16* vrow = vbase + (long)(tile + kl) * D + dp * 2;
-      const float v0 = bf2f(vrow[0]);
-      const float v1 = bf2f(vrow[1]);
+      const u32 vpair = KVNT
+          ? __builtin_nontemporal_load((const u32*)vrow)
+          : *(const u32*)vrow;
+      const float v0 = bf2f((u16)(vpair & 0xffff));
+      const float v1 = bf2f((u16)(vpair >> 16));
       #pragma unroll
     for (int g = 0; g < G; ++g) {
         const float p = pl[g][kl];
@@ -460,14 +464,22 @@ void fei_attn_decode(const void* q, const void* k_cache, const void* v_cache,
   const int G = Hq / Hkv;
   dim3 grid(splits, Hkv, B);
   const int rope = cos_sin != nullptr;
-#define LAUNCH_DEC(GV, RV) \
-  hipLaunchKernelGGL((k_attn_decode<GV, RV>), grid, dim3(256), 0, stream, \
-                     (const u16*)q, (u16*)k_cache, (u16*)v_cache, \
+  static int kvnt = -1;
+  if (kvnt < 0) {
+    const char* e = getenv("FEI_ATTN_KV_NT");
+    kvnt = (e && e[0] == '1') ? 1 : 0;
+  }
+#define LAUNCH_DEC(GV, RV, NTV) \
+  hipLaunchKernelGGL((k_attn_decode<GV, RV, NTV>), grid, dim3(256), 0, \
+                     stream, (const u16*)q, (u16*)k_cache, (u16*)v_cache, \
                      part_o, part_ml, pos, B, Hq, Hkv, D, max_seq, splits, \
                      scale, q_bs, (const u16*)kin, (const u16*)vin, cos_sin, \
                      kv_bs)
-#define LAUNCH_DEC_R(GV) do { if (rope) LAUNCH_DEC(GV, true); \
-                              else LAUNCH_DEC(GV, false); } while (0)
+#define LAUNCH_DEC_R(GV) do { \
+    if (rope) { if (kvnt) LAUNCH_DEC(GV, true, true); \
+                else LAUNCH_DEC(GV, true, false); } \
+    else { if (kvnt) LAUNCH_DEC(GV, false, true); \
+           else LAUNCH_DEC(GV, false, false); } } while (0)
   switch (G) {
     case 1: LAUNCH_DEC_R(1); break;
     case 2: LAUNCH_DEC_R(2); break;
