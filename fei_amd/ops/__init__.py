@@ -222,7 +222,7 @@ def attn_decode(q, k_cache, v_cache, pos, splits: int = 32,
         out = torch.empty_like(q)
     Hkv = k_cache.shape[1]
     assert q.stride(2) == 1 and q.stride(1) == D
-    assert splits <= 32, "combine kernel stages at most 32 split partials"
+    assert splits <= 64, "combine kernel stages at most 64 split partials"
     if table is not None:
         assert k is not None and v is not None
         assert k.stride(1) == D and k.stride(0) == v.stride(0)
@@ -636,7 +636,7 @@ def attn_decode_paged(q, k_pool, v_pool, block_table, pos, splits: int = 32,
     if out is None:
         out = torch.empty_like(q)
     assert q.stride(2) == 1 and q.stride(1) == D
-    assert splits <= 32
+    assert splits <= 64
     lib.fei_attn_decode_paged(_ptr(q), _ptr(k_pool), _ptr(v_pool),
                               _ptr(block_table), _ptr(part_o), _ptr(part_ml),
                               _ptr(pos), B, Hq, Hkv, D, bs_log,

@@ -495,7 +495,7 @@ void fei_attn_decode(const void* q, const void* k_cache, const void* v_cache,
 // Grid (Hq, B), block D threads. All split partials are loaded in PARALLEL
 // (LDS-staged m/l, unrolled independent o loads) — a serial per-split loop
 // of dependent global loads cost ~7 us at splits=16 (profiles/r01).
-#define CMB_SMAX 32
+#define CMB_SMAX 64
 
 __global__ void k_attn_decode_combine(u16* __restrict__ out,
                                       const float* __restrict__ part_o,
