@@ -78,11 +78,19 @@ class LocalEngine:
         import os as _os
         # split-K decode attention: splits x Hkv x B workgroups. 4 splits
         # (32 WGs on a 256-CU chip) starved KV bandwidth at agent context
-        # lengths: 4 -> 32 splits measured +3% at seq 512 and +41% at seq
-        # 3400 on 8B (docs/BENCHMARKS.md). Empty splits at short lengths
-        # write -inf partials and cost nothing.
+        # lengths (r01: 4 -> 32 measured +41% at seq 3400). After the r02
+        # non-temporal weight fix the optimum moved: the r2c27 sweep
+        # measured splits=16 fastest up to ~4k context (+2% at 512) and
+        # splits=32 clearly best at 8k (245.6 vs 217.7 tok/s) — the
+        # captured graph freezes the split count, so pick by the engine's
+        # window. FEI_ATTN_SPLITS still overrides.
         env_splits = _os.environ.get("FEI_ATTN_SPLITS")
-        self.attn_splits = int(env_splits) if env_splits else attn_splits
+        if env_splits:
+            self.attn_splits = int(env_splits)
+        elif attn_splits != 32:
+            self.attn_splits = attn_splits        # explicit ctor choice
+        else:
+            self.attn_splits = 16 if self.max_seq_len <= 4096 else 32
         # Fused single-pass attention (rope+append+attn in one kernel) runs
         # grid (Hkv x B) — too few workgroups to pull KV bandwidth at B=1
         # (measured 194 vs 244 tok/s on 8B); it pays only when B*Hkv fills
