@@ -101,9 +101,13 @@ class LocalEngine:
             self.fused_attn = env not in ("0", "false")
         else:
             self.fused_attn = batch_size * spec.num_kv_heads >= 64
-        env = _os.environ.get("FEI_FUSED_NORM")
-        self.fused_norm = (env not in ("0", "false")) if env is not None else \
-            (self.is_gpu and not self.tp.is_distributed)
+        # fused-norm chain default flipped in r02: after the non-temporal
+        # weight fix the PLAIN path (separate rmsnorm + nt GEMVs) measured
+        # FASTER (271.8 vs 266.4 tok/s, same-box A/B r2c30) — the norm
+        # prologue's per-element VALU chain throttles the GEMV stream to
+        # ~3-5 TB/s vs 6.6 plain. The fused chain stays for fp8 decode
+        # (its kernels are norm-fused) and via FEI_FUSED_NORM=1.
+        self._fused_norm_env = _os.environ.get("FEI_FUSED_NORM")
         self.seed = seed
         self.temperature = 0.0       # graph-captured; set before capture
         if tokenizer is None:
@@ -164,6 +168,11 @@ class LocalEngine:
         logger.info("model %s init in %.1fs (%.2f GB params)", spec.name,
                     time.perf_counter() - t0, self.model.param_bytes() / 2**30)
         self.weight_quant = weight_quant or _os.environ.get("FEI_WEIGHT_QUANT")
+        if self._fused_norm_env is not None:
+            self.fused_norm = self._fused_norm_env not in ("0", "false")
+        else:
+            self.fused_norm = (self.is_gpu and not self.tp.is_distributed
+                               and self.weight_quant == "fp8")
         if self.weight_quant == "fp8":
             t0 = time.perf_counter()
             self.model.quantize_fp8()
