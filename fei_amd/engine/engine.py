@@ -78,19 +78,13 @@ class LocalEngine:
         import os as _os
         # split-K decode attention: splits x Hkv x B workgroups. 4 splits
         # (32 WGs on a 256-CU chip) starved KV bandwidth at agent context
-        # lengths (r01: 4 -> 32 measured +41% at seq 3400). After the r02
-        # non-temporal weight fix the optimum moved: the r2c27 sweep
-        # measured splits=16 fastest up to ~4k context (+2% at 512) and
-        # splits=32 clearly best at 8k (245.6 vs 217.7 tok/s) — the
-        # captured graph freezes the split count, so pick by the engine's
-        # window. FEI_ATTN_SPLITS still overrides.
+        # lengths (r01: 4 -> 32 measured +41% at seq 3400). Two r02
+        # post-nt sweeps (r2c27, r2c32): 16 vs 32 differ only within
+        # run-to-run noise (±2%) below ~4k context, while 32 is robustly
+        # ahead at 8k (249 vs 217 tok/s) — 32 stays the flat default.
+        # The combine kernel now stages up to 64 splits for experiments.
         env_splits = _os.environ.get("FEI_ATTN_SPLITS")
-        if env_splits:
-            self.attn_splits = int(env_splits)
-        elif attn_splits != 32:
-            self.attn_splits = attn_splits        # explicit ctor choice
-        else:
-            self.attn_splits = 16 if self.max_seq_len <= 4096 else 32
+        self.attn_splits = int(env_splits) if env_splits else attn_splits
         # Fused single-pass attention (rope+append+attn in one kernel) runs
         # grid (Hkv x B) — too few workgroups to pull KV bandwidth at B=1
         # (measured 194 vs 244 tok/s on 8B); it pays only when B*Hkv fills
