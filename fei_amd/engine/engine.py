@@ -1,12 +1,15 @@
 """LocalEngine: the MI355X inference engine behind the Assistant.
 
-The decode step (the agent-tok/s hot path) is fully device-resident. At
-tp=1 with the fused-norm chain it is 5 kernels per layer:
-  norm-prologue QKV GEMV -> split-K decode attention (in-kernel RoPE +
-  KV-append) -> residual-epilogue O GEMV -> norm-prologue gate/up GEMV
-  with fused SwiGLU -> residual-epilogue down GEMV
-then a norm-prologue lm_head GEMV, the sampling kernel (writes the next
-input token + the out_tokens ring ON DEVICE) and the position advance —
+The decode step (the agent-tok/s hot path) is fully device-resident.
+The bf16 default is the PLAIN chain (r02: non-temporal weight loads made
+it +2% over the fused-norm chain, profiles/r02_decode_nt.md):
+  rmsnorm -> QKV GEMV -> split-K decode attention (in-kernel RoPE +
+  KV-append) -> combine -> O GEMV -> fused add+rmsnorm -> gate/up GEMV
+  with fused SwiGLU -> down GEMV -> fused add+rmsnorm
+(9 nt-streaming kernels/layer; FEI_FUSED_NORM=1 restores the 5-kernel
+norm-prologue chain, which remains the fp8 decode path), then the
+lm_head GEMV, the sampling kernel (writes the next input token + the
+out_tokens ring ON DEVICE) and the position advance —
 the whole step is captured once into a hipGraph (torch.cuda.CUDAGraph)
 and replayed with zero host work per token. EOS is checked every
 `eos_check_every` replays (one device->host copy per chunk, not per
