@@ -204,3 +204,47 @@ def test_sample_shard_matches_full_sampler():
             noisy = logits / temp + ref.hash_gumbel(B, V, 0, 17, 5)
             expect = noisy.argmax(dim=-1).to(T.int32)
         assert tok.tolist() == expect.tolist(), temp
+
+
+SAMPLED_WORKER = r"""
+import os, json, torch
+from fei_amd.engine.engine import LocalEngine
+from fei_amd.parallel.pg import init_from_env
+
+ctx = init_from_env(backend="gloo")
+eng = LocalEngine.create("llama3-tiny", max_seq_len=96, seed=11, tp=ctx,
+                         use_hip_graph=False)
+out = eng.generate([1, 5, 9, 13], max_new_tokens=8, temperature=0.8,
+                   stop_on_eos=False)
+with open(os.environ["TP_OUT"] + f".rank{ctx.rank}", "w") as f:
+    json.dump({"token_ids": out["token_ids"]}, f)
+"""
+
+
+def test_tp2_sampled_ranks_agree(tmp_path):
+    """Temperature>0 under TP: every rank must sample the IDENTICAL token
+    each step (the winner combine is deterministic and the Gumbel noise is
+    keyed by seed/step/global-index) — divergent ranks would silently
+    corrupt the sharded KV caches."""
+    out_file = tmp_path / "tp_sampled"
+    script = tmp_path / "sampled_worker.py"
+    script.write_text(SAMPLED_WORKER)
+    env = dict(os.environ)
+    env["TP_OUT"] = str(out_file)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    env.setdefault("PYTHONPATH", os.getcwd())
+    for attempt in range(3):
+        proc = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+             "--master-port", str(29561 + attempt * 7), str(script)],
+            env=env, capture_output=True, text=True, timeout=300,
+        )
+        if proc.returncode == 0:
+            break
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    r0 = json.loads((tmp_path / "tp_sampled.rank0").read_text())["token_ids"]
+    r1 = json.loads((tmp_path / "tp_sampled.rank1").read_text())["token_ids"]
+    assert r0 == r1
+    assert len(r0) == 8
+    assert all(0 <= t < 512 for t in r0)
