@@ -32,5 +32,6 @@ setup(
             "fei-api=fei_amd.serve.api:main",
         ],
     },
-    package_data={"fei_amd.ops": ["csrc/*.hip", "csrc/*.h", "*.so"]},
+    package_data={"fei_amd.ops": ["csrc/*.hip", "csrc/*.h", "*.so"],
+                  "fei_amd.engine": ["fei16k.model", "fei16k.vocab"]},
 )
