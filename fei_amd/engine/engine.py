@@ -188,9 +188,15 @@ class LocalEngine:
         # sized from self.attn_splits (NOT the ctor arg: FEI_ATTN_SPLITS
         # overrides it, and an undersized workspace is an out-of-bounds
         # write in the split kernel)
+        # 3rd element: zeroed per-split done-flag buffer [B*Hkv_l*splits]
+        # selecting the FUSED split-K combine (split 0 of each (b, hkv)
+        # pair polls the flags and reduces the partials in-kernel — no
+        # separate combine launch per layer).
         self.attn_ws = (
             torch.zeros(self.B, Hq_l, self.attn_splits, D, dtype=torch.float32, device=device),
             torch.zeros(self.B, Hq_l, self.attn_splits, 2, dtype=torch.float32, device=device),
+            torch.zeros(self.B * self.model.hkv_l * self.attn_splits,
+                        dtype=torch.int32, device=device),
         )
         self.attn_out = torch.zeros(self.B, Hq_l, D, dtype=self.dtype,
                                     device=device)
@@ -365,6 +371,12 @@ class LocalEngine:
             k[:, :, kv_rows].copy_(sk)
         for v, sv in zip(self.v_caches, saved_v):
             v[:, :, kv_rows].copy_(sv)
+        # fused-combine done-flags are (pos, layer)-tagged: the warmup ran
+        # at pos 1..2, and the first real replay may run at those same
+        # positions — stale matching tags would let the in-kernel reducer
+        # read partials before the current step's splits wrote them.
+        if len(self.attn_ws) >= 3:
+            self.attn_ws[2].zero_()
         torch.cuda.synchronize()
         self._graph = g
         self._graph_params = (self.temperature, self.seed)
@@ -414,6 +426,11 @@ class LocalEngine:
             # granule tags are pos-keyed, so stale tags must be cleared
             for t in self._stream_ws.values():
                 t.zero_()
+        # same pos-keyed-tag rule for the fused-combine done flags: a new
+        # request replays earlier positions, and a stale matching tag would
+        # release the in-kernel reducer before this step's splits published
+        if self.is_gpu and len(self.attn_ws) >= 3:
+            self.attn_ws[2].zero_()
         # Cap to the context budget BEFORE chunking (mirrors _prep_prompt's
         # cap for the from_pos>0 prefix-cached path, which bypasses it):
         # keep the TAIL of the new tokens — a mid-loop break would silently

@@ -231,7 +231,7 @@ class LlamaModel:
                 att = ops.attn_decode(q, k_caches[li], v_caches[li], pos,
                                       splits=attn_splits, scale=scale,
                                       workspace=workspace, out=attn_out,
-                                      k=k, v=v, table=self.rope)
+                                      k=k, v=v, table=self.rope, layer=li)
             o = ops.linear_decode(att.reshape(B, -1), lw.wo)
             all_reduce_sum(o, self.tp)
             x, h = ops.fused_add_rmsnorm(o, h, lw.norm_mlp, s.norm_eps)
@@ -332,7 +332,7 @@ class LlamaModel:
                 att = ops.attn_decode(q, k_caches[li], v_caches[li], pos,
                                       splits=attn_splits, scale=scale,
                                       workspace=workspace, out=attn_out,
-                                      k=k, v=v, table=self.rope)
+                                      k=k, v=v, table=self.rope, layer=li)
             if fp8 is not None:
                 ops.gemv_res_fp8(att.reshape(B, -1), *q8["wo"], h)
                 act = ops.gemv_swiglu_norm_fp8(h, lw.norm_mlp, *q8["wgu"],

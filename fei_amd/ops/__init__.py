@@ -47,7 +47,7 @@ def _try_load() -> None:
                                         _i, _i, _i, _i, _i, _i, _l, _l, _vp]
     lib.fei_attn_decode.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp,
                                     _i, _i, _i, _i, _i, _i, _f, _l,
-                                    _vp, _vp, _vp, _l, _vp]
+                                    _vp, _vp, _vp, _l, _vp, _vp, _i, _vp]
     lib.fei_attn_decode_combine.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
     lib.fei_swiglu.argtypes = [_vp, _vp, _l, _i, _vp]
     lib.fei_sample.argtypes = [_vp, _vp, _vp, _vp, _vp, _i, _i, _i, _f, _u64,
@@ -198,7 +198,8 @@ def attn_decode(q, k_cache, v_cache, pos, splits: int = 32,
                 out: Optional[torch.Tensor] = None,
                 k: Optional[torch.Tensor] = None,
                 v: Optional[torch.Tensor] = None,
-                table: Optional[torch.Tensor] = None) -> torch.Tensor:
+                table: Optional[torch.Tensor] = None,
+                layer: int = 0) -> torch.Tensor:
     """Single-token GQA attention over n = pos[b]+1 cache entries.
     q [B,Hq,D] -> out [B,Hq,D]. The length is read on DEVICE (hipGraph).
     When (k, v, table) are given, the kernel also fuses the step's RoPE +
@@ -211,11 +212,20 @@ def attn_decode(q, k_cache, v_cache, pos, splits: int = 32,
         seqlen = pos + 1
         return ref.attn_decode(q, k_cache, v_cache, seqlen, scale)
     lib = require_lib()
+    arrive = None
     if workspace is None:
         part_o = torch.empty(B, Hq, splits, D, dtype=torch.float32, device=q.device)
         part_ml = torch.empty(B, Hq, splits, 2, dtype=torch.float32, device=q.device)
     else:
-        part_o, part_ml = workspace
+        part_o, part_ml = workspace[0], workspace[1]
+        # a 3rd workspace element (zeroed int32 [B*Hkv*splits] per-split
+        # done-flag buffer) selects the FUSED combine: split 0 of each
+        # (b, hkv) pair polls the flags and reduces the partials in-kernel,
+        # and the separate combine launch is skipped (FEI_FUSED_CMB=0 opts
+        # out for A/B; pass `layer` so the flag tags stay unique across the
+        # layers of one step).
+        if len(workspace) >= 3 and _FUSED_CMB:
+            arrive = workspace[2]
         assert part_o.shape[2] == splits and part_ml.shape[2] == splits, \
             f"workspace sized for {part_o.shape[2]} splits, kernel asked {splits}"
     if out is None:
@@ -230,12 +240,19 @@ def attn_decode(q, k_cache, v_cache, pos, splits: int = 32,
     else:
         kin = vin = cs = None
         kv_bs = 0
+    if arrive is not None:
+        assert arrive.dtype == torch.int32 and \
+            arrive.numel() >= B * Hkv * splits
     lib.fei_attn_decode(_ptr(q), _ptr(k_cache), _ptr(v_cache), _ptr(part_o),
                         _ptr(part_ml), _ptr(pos), B, Hq, Hkv, D,
                         k_cache.shape[2], splits, scale, q.stride(0),
-                        kin, vin, cs, kv_bs, _stream())
-    lib.fei_attn_decode_combine(_ptr(out), _ptr(part_o), _ptr(part_ml),
-                                B, Hq, D, splits, _stream())
+                        kin, vin, cs, kv_bs,
+                        _ptr(out) if arrive is not None else None,
+                        _ptr(arrive) if arrive is not None else None,
+                        layer, _stream())
+    if arrive is None:
+        lib.fei_attn_decode_combine(_ptr(out), _ptr(part_o), _ptr(part_ml),
+                                    B, Hq, D, splits, _stream())
     return out
 
 
@@ -474,6 +491,15 @@ def mfma_probe(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
 # microbench that suggested a 512 MB threshold was L3-warm across runs.
 # FEI_GEMV_NT_MIN (bytes) restores a threshold.
 _GEMV_NT_MIN_BYTES = int(os.environ.get("FEI_GEMV_NT_MIN", 0))
+# Fused split-K combine in decode attention (per-head reducer splits poll
+# sc1 done-flags and reduce the partials in-kernel; needs the 3-element
+# workspace). Verified token-identical and measured at PARITY with the
+# separate k_attn_decode_combine launch (275.5 vs 275.8 tok/s on the 8B
+# chain — in a captured graph the combine launch already overlaps the
+# attention grid's drain, so there was no boundary to win; the full
+# 4-protocol measurement ladder is profiles/r02_fused_combine.md).
+# Opt-in via FEI_FUSED_CMB=1; the launch form stays the default.
+_FUSED_CMB = os.environ.get("FEI_FUSED_CMB", "0") == "1"
 
 
 def _gemv_ok(M: int, K: int) -> bool:

@@ -262,6 +262,42 @@ __device__ __forceinline__ void block_reduce_vec(float* v,
 // fall in any split's chunk) and the split that OWNS key n-1 writes the
 // caches. Readers never read cache[n-1]; they use the LDS copy, so there is
 // no cross-workgroup ordering (placement-independent by construction).
+// sc1 write-through hand-off helpers for the fused split-K combine (guide
+// §6 G16 cheap variant): relaxed agent-scope stores/loads lower to sc1 at
+// 4-8 B widths — visibility comes from the write-through to the coherence
+// point, so NO release/acquire fences are needed (the fence-based recipe's
+// per-WG buffer_wbl2 + reducer-side L2 invalidate measured 9% SLOWER
+// end-to-end than the separate combine launch it replaced).
+__device__ inline void store2_sc1(float* p, float a, float b) {
+  float2 v = make_float2(a, b);
+  unsigned long long bits;
+  __builtin_memcpy(&bits, &v, 8);
+  __hip_atomic_store((unsigned long long*)p, bits, __ATOMIC_RELAXED,
+                     __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ inline float load_sc1(const float* p) {
+  return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ inline void store_i_sc1(int* p, int v) {
+  __hip_atomic_store(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ inline int load_i_sc1(const int* p) {
+  return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+// When `flags` is non-null the split-K combine is FUSED: every split
+// workgroup of a (b, hkv) pair publishes its partials with sc1
+// write-through stores, drains them (vmcnt), then lane 0 writes a per-split
+// DONE flag (sc1, DISTINCT address per split — a shared arrival counter's
+// same-address fetch_adds serialize at the coherence point, ~300 ns each x
+// 32 splits ≈ 10 us/layer, which made the counter form 8-9% slower
+// end-to-end). Split 0 (never empty: its chunk starts at key 0) is the
+// fixed reducer: it polls the 32 flags in parallel (one lane each), then
+// reduces every split's partials for its G heads into `outf` (bf16) with
+// sc1 loads. Flags carry a (step, layer) tag — no reset, no poisoning on
+// replay: a flag matches only when THIS step's layer instance wrote it.
+// This removes the separate k_attn_decode_combine launch (32/step, ~4.9 us
+// + ~1.2 us boundary each — profiles/r02_plainchain_kernel_shares).
 template <int G, bool ROPE, bool KVNT = false>
 __global__ void __launch_bounds__(256)
 k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
@@ -270,7 +306,8 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
               int B, int Hq, int Hkv, int D, int max_seq, int splits,
               float scale, long q_bs,
               const u16* __restrict__ kin, const u16* __restrict__ vin,
-              const float* __restrict__ cos_sin, long kv_bs) {
+              const float* __restrict__ cos_sin, long kv_bs,
+              u16* __restrict__ outf, int* __restrict__ arrive, int layer) {
   const int split = blockIdx.x;
   const int hkv = blockIdx.y;
   const int b = blockIdx.z;
@@ -292,11 +329,17 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
       const int hq = hkv * G + g;
       float* po = part_o + (((long)b * Hq + hq) * splits + split) * D;
       float* pml = part_ml + (((long)b * Hq + hq) * splits + split) * 2;
-      for (int d = tid; d < D; d += blockDim.x) po[d] = 0.f;
-      if (tid == 0) { pml[0] = -1.0f / 0.0f; pml[1] = 0.f; }
+      if (arrive != nullptr) {
+        for (int d = tid * 2; d < D; d += blockDim.x * 2)
+          store2_sc1(&po[d], 0.f, 0.f);
+        if (tid == 0) store2_sc1(pml, -1.0f / 0.0f, 0.f);
+      } else {
+        for (int d = tid; d < D; d += blockDim.x) po[d] = 0.f;
+        if (tid == 0) { pml[0] = -1.0f / 0.0f; pml[1] = 0.f; }
+      }
     }
-    return;
-  }
+    if (arrive == nullptr) return;
+  } else {
 
   const int half = D / 2;
   const int p_new = n - 1;
@@ -443,11 +486,71 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
         s0 += osh[gg][dp][0];
         s1 += osh[gg][dp][1];
       }
-      po[dp * 2] = s0;
-      po[dp * 2 + 1] = s1;
+      if (arrive != nullptr) store2_sc1(&po[dp * 2], s0, s1);
+      else { po[dp * 2] = s0; po[dp * 2 + 1] = s1; }
     }
-    if (tid == 0) { pml[0] = m[g]; pml[1] = l[g]; }
+    if (tid == 0) {
+      if (arrive != nullptr) store2_sc1(pml, m[g], l[g]);
+      else { pml[0] = m[g]; pml[1] = l[g]; }
+    }
     __syncthreads();
+  }
+  }  // start < end
+
+  if (arrive == nullptr) return;
+  // --- fused combine hand-off: partials are at the coherence point (sc1
+  // write-through stores above); drain them (vmcnt), barrier, then lane 0
+  // publishes this split's DONE flag. Tag (step, layer)-unique => no reset.
+  {
+    const int tag = 0x40000000 | (n * 64 + (layer & 63));
+    int* flags = arrive + ((long)b * Hkv + hkv) * splits;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (tid == 0) store_i_sc1(&flags[split], tag);
+    if (split >= G) return;
+    // splits 0..G-1 are reducers, ONE HEAD EACH (same combine parallelism
+    // as the separate kernel — a single split-0 reducer walking G heads
+    // read 4x the bytes per WG and lost ~3.6 us/layer). Each polls the
+    // flags one lane per split (distinct addresses — parallel sc1 loads),
+    // bounded spin (a lost producer makes this give up and produce stale
+    // output rather than wedge the GPU).
+    int ready = 0;
+    for (int spin = 0; spin < (1 << 18) && !ready; ++spin) {
+      const int mine =
+          (tid < splits) ? (load_i_sc1(&flags[tid]) == tag) : 1;
+      ready = __syncthreads_and(mine);
+    }
+    // One agent-scope ACQUIRE, then PLAIN cached loads: relaxed-atomic
+    // (sc1) reducer loads stay in program order and bypass the caches, so
+    // per-thread re-reads of the same 2*splits m/l values would serialize
+    // ~192 uncached round trips per thread (the first reducer forms lost
+    // ~10 us/layer exactly here). The fence orders the loads after the
+    // observed flags and invalidates the local caches, so plain loads see
+    // the written-through partials (placement-independent, guide G16).
+    if (tid == 0) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    __syncthreads();
+    float* pmlsh = &pl[0][0];                     // dead after main loop
+    for (int g = split; g < G; g += splits) {     // usually exactly one
+      const int hq = hkv * G + g;
+      const float* pml0 = part_ml + ((long)b * Hq + hq) * splits * 2;
+      for (int i = tid; i < splits * 2; i += blockDim.x)
+        pmlsh[i] = pml0[i];
+      __syncthreads();
+      const float* po = part_o + ((long)b * Hq + hq) * splits * D;
+      for (int d = tid; d < D; d += blockDim.x) {
+        float mm = -1.0f / 0.0f;
+        for (int s = 0; s < splits; ++s) mm = fmaxf(mm, pmlsh[s * 2]);
+        float ll = 0.f, oo = 0.f;
+#pragma unroll 8
+        for (int s = 0; s < splits; ++s) {
+          const float w = __expf(pmlsh[s * 2] - mm);
+          ll = fmaf(w, pmlsh[s * 2 + 1], ll);
+          oo = fmaf(w, po[s * D + d], oo);
+        }
+        outf[((long)b * Hq + hq) * D + d] = f2bf(ll > 0.f ? oo / ll : 0.f);
+      }
+      __syncthreads();                            // pmlsh reuse next g
+    }
   }
 }
 
@@ -460,7 +563,8 @@ void fei_attn_decode(const void* q, const void* k_cache, const void* v_cache,
                      int B, int Hq, int Hkv, int D, int max_seq, int splits,
                      float scale, long q_bs,
                      const void* kin, const void* vin, const float* cos_sin,
-                     long kv_bs, hipStream_t stream) {
+                     long kv_bs, void* out_fused, int* arrive, int layer,
+                     hipStream_t stream) {
   const int G = Hq / Hkv;
   dim3 grid(splits, Hkv, B);
   const int rope = cos_sin != nullptr;
@@ -474,7 +578,7 @@ void fei_attn_decode(const void* q, const void* k_cache, const void* v_cache,
                      stream, (const u16*)q, (u16*)k_cache, (u16*)v_cache, \
                      part_o, part_ml, pos, B, Hq, Hkv, D, max_seq, splits, \
                      scale, q_bs, (const u16*)kin, (const u16*)vin, cos_sin, \
-                     kv_bs)
+                     kv_bs, (u16*)out_fused, arrive, layer)
 #define LAUNCH_DEC_R(GV) do { \
     if (rope) { if (kvnt) LAUNCH_DEC(GV, true, true); \
                 else LAUNCH_DEC(GV, true, false); } \

@@ -120,6 +120,46 @@ def test_attn_decode(lib, n, splits):
     assert err < 2e-2, f"max err {err}"
 
 
+@pytest.mark.parametrize("n,splits", [(1, 1), (7, 4), (300, 16), (1000, 32),
+                                      (20, 32)])
+def test_attn_decode_fused_combine(lib, n, splits, monkeypatch):
+    """3-element workspace + FEI_FUSED_CMB => in-kernel per-head-reducer
+    combine; must match the fp32 reference AND the separate-combine path
+    (the partials are computed identically — only the reduction site
+    moves). Opt-in at parity (profiles/r02_fused_combine.md), so the gate
+    is forced on here."""
+    monkeypatch.setattr(lib, "_FUSED_CMB", True)
+    from fei_amd.ops import reference as ref
+    B, Hq, Hkv, D, MS = 2, 8, 2, 128, 1024
+    q = randbf(B, Hq, D, seed=50 + n)
+    kc = _mk_cache(B, Hkv, MS, D, seed=51 + n)
+    vc = _mk_cache(B, Hkv, MS, D, seed=52 + n)
+    pos = torch.tensor([n - 1, max(n // 2 - 1, 0)], dtype=torch.int32,
+                       device=DEV)
+    ws = (torch.zeros(B, Hq, splits, D, dtype=torch.float32, device=DEV),
+          torch.zeros(B, Hq, splits, 2, dtype=torch.float32, device=DEV),
+          torch.zeros(B * Hkv * splits, dtype=torch.int32, device=DEV))
+    out_f = lib.attn_decode(q, kc, vc, pos, splits=splits, workspace=ws,
+                            layer=3)
+    out_u = lib.attn_decode(q, kc, vc, pos, splits=splits,
+                            workspace=(ws[0], ws[1]))
+    expected = ref.attn_decode(q, kc, vc, pos + 1)
+    assert (out_f.float() - expected.float()).abs().max() < 2e-2
+    assert (out_f.float() - out_u.float()).abs().max() < 1e-2
+    # distinct layer => distinct tags: back-to-back same-pos calls are safe
+    # without zeroing (the per-step layer walk the engine actually does)
+    out_l = lib.attn_decode(q, kc, vc, pos, splits=splits, workspace=ws,
+                            layer=4)
+    assert (out_l.float() - out_f.float()).abs().max() == 0.0
+    # same (pos, layer) repeats only across requests; the engine zeroes the
+    # flag buffer at each prefill — replay under that contract is stable
+    for _ in range(25):
+        ws[2].zero_()
+        out_r = lib.attn_decode(q, kc, vc, pos, splits=splits, workspace=ws,
+                                layer=3)
+    assert (out_r.float() - out_f.float()).abs().max() == 0.0
+
+
 @pytest.mark.parametrize("D", [64, 128])
 @pytest.mark.parametrize("S,p0", [(16, 0), (64, 0), (129, 0), (64, 37)])
 def test_attn_prefill_causal(lib, D, S, p0):
