@@ -84,20 +84,29 @@ class LocalEngine:
         # lengths (r01: 4 -> 32 measured +41% at seq 3400). Two r02
         # post-nt sweeps (r2c27, r2c32): 16 vs 32 differ only within
         # run-to-run noise (±2%) below ~4k context, while 32 is robustly
-        # ahead at 8k (249 vs 217 tok/s) — 32 stays the flat default.
+        # ahead at 8k (249 vs 217 tok/s) — 32 stays the B<=2 default.
+        # Batches multiply the grid by B, so B>=4 halves the split count
+        # (batch-8: splits 32 -> 16 measured 6.34 -> 5.85 ms/step; 4/8/16
+        # within noise — the grid already fills the chip at 16).
         # The combine kernel now stages up to 64 splits for experiments.
         env_splits = _os.environ.get("FEI_ATTN_SPLITS")
-        self.attn_splits = int(env_splits) if env_splits else attn_splits
+        if env_splits:
+            self.attn_splits = int(env_splits)
+        elif attn_splits == 32 and batch_size >= 4:
+            self.attn_splits = 16
+        else:
+            self.attn_splits = attn_splits
         # Fused single-pass attention (rope+append+attn in one kernel) runs
-        # grid (Hkv x B) — too few workgroups to pull KV bandwidth at B=1
-        # (measured 194 vs 244 tok/s on 8B); it pays only when B*Hkv fills
-        # the chip. The split-K path stays the default.
+        # grid (Hkv x B) — too few workgroups to pull KV bandwidth: 194 vs
+        # 244 tok/s at B=1 (r01), and the r01 "pays when B*Hkv >= 64"
+        # heuristic was also wrong — 64 workgroups still leaves 3/4 of the
+        # 256-CU chip idle (batch-8: split-K 1363-1368 tok/s vs fused 1159,
+        # r02 late sweep). Split-K is the default at EVERY batch; the fused
+        # kernel stays available via FEI_FUSED_ATTN=1.
         import os as _os
         env = _os.environ.get("FEI_FUSED_ATTN")
-        if env is not None:
-            self.fused_attn = env not in ("0", "false")
-        else:
-            self.fused_attn = batch_size * spec.num_kv_heads >= 64
+        self.fused_attn = env not in ("0", "false") if env is not None \
+            else False
         # fused-norm chain default flipped in r02: after the non-temporal
         # weight fix the PLAIN path (separate rmsnorm + nt GEMVs) measured
         # FASTER (271.8 vs 266.4 tok/s, same-box A/B r2c30) — the norm

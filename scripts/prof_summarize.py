@@ -6,5 +6,6 @@ print(f"{'kernel':<60} {'calls':>6} {'total_ms':>10} {'avg_us':>8} {'pct':>6}")
 for name, calls, tot, avg, pct in cur.execute(
         "SELECT name, total_calls, total_duration, average, percentage "
         "FROM top_kernels LIMIT 20"):
-    short = name.split("(")[0].replace("void ", "")[:58]
+    clean = name.replace("(anonymous namespace)::", "").replace("void ", "")
+    short = clean.split("(")[0][:58]
     print(f"{short:<60} {calls:>6} {tot/1000:>10.2f} {avg:>8.1f} {pct:>6.2f}")
