@@ -55,10 +55,17 @@ class PagedSessionManager:
         self.eos = engine.tokenizer.eos_id
         self._ws: Dict[int, tuple] = {}     # split-K workspace per batch size
 
+    def _splits(self, B: int) -> int:
+        # batch-aware split count (engine ctor note): the B-wide grid
+        # already fills the chip at 16 splits once B >= 4 — 32 over-splits
+        # (batch-8 plain path measured 6.34 -> 5.85 ms/step at 16)
+        return min(self.engine.attn_splits, 16) if B >= 4 \
+            else self.engine.attn_splits
+
     def _workspace(self, B: int):
         if B not in self._ws:
             m = self.model
-            sp = self.engine.attn_splits
+            sp = self._splits(B)
             self._ws[B] = (
                 torch.zeros(B, m.hq_l, sp, m.D, dtype=torch.float32,
                             device=m.device),
@@ -189,7 +196,7 @@ class PagedSessionManager:
                 kp[blk_of[i], :, off_of[i], :] = k_r[i]
                 vp[blk_of[i], :, off_of[i], :] = v[i]
             att = ops.attn_decode_paged(q_r, kp, vp, table, pos,
-                                        splits=self.engine.attn_splits,
+                                        splits=self._splits(B),
                                         scale=scale,
                                         workspace=self._workspace(B))
             o = ops.linear_decode(att.reshape(B, -1), lw.wo)
