@@ -407,6 +407,9 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
       const s16x8* krow = (const s16x8*)(kbase + (long)kk * D);
       #pragma unroll
     for (int g = 0; g < G; ++g) sc[g] = 0.f;
+      // (a #pragma unroll 4 here measured exactly neutral at every
+      // context length — the backend already pipelines the key-row loads
+      // despite the runtime D trip count)
       for (int i = 0; i < D / 8; ++i) {
         s16x8 kv8 = KVNT ? __builtin_nontemporal_load(&krow[i]) : krow[i];
 #pragma unroll
