@@ -316,7 +316,9 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
   __shared__ float qs[DEC_GMAX][DEC_DMAX];
   __shared__ float pl[DEC_GMAX][DEC_TILE];
   __shared__ float red[DEC_GMAX][4];
-  __shared__ float osh[8][DEC_DMAX / 2][2];
+  // key-group partial-o staging: [kg * dvecs + dv][element] — kgroups *
+  // dvecs == blockDim == DEC_TILE for every D
+  __shared__ float osh[DEC_TILE][8];
 
   const int n = pos[b] + 1;
   const int chunk = (n + splits - 1) / splits;
@@ -387,17 +389,31 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
   const u16* kbase = kc + ((long)b * Hkv + hkv) * max_seq * D;
   const u16* vbase = vc + ((long)b * Hkv + hkv) * max_seq * D;
 
-  const int dpairs = D / 2;                       // 64 for D=128
-  const int kgroups = blockDim.x / dpairs;        // 4 for D=128
-  const int keys_per_group = DEC_TILE / kgroups;  // 64
+  // V pass lane maps. WIDE (chunk >= 64): 16 B of one V row per lane —
+  // fewer, full-width loads win once every key-group has work (+1% at
+  // 2k-8k context). PAIR (short chunks): 4 B per lane over D/2 lanes —
+  // 4x more active lanes when the chunk covers few keys (the wide map
+  // measured -3% on the 512-ctx headline: only 16 of 256 lanes active).
+  // Ablation: experimental/decode_ablate.hip, profiles/r02.
+  const int dvecs = D / 8;                        // 16 for D=128
+  const int kgroups = blockDim.x / dvecs;         // 16
+  const int keys_per_group = DEC_TILE / kgroups;  // 16
+  const int dv = tid % dvecs;
+  const int kg = tid / dvecs;
+  const bool wide = chunk >= 64;
+  const int dpairs = D / 2;                       // pair-mode roles
+  const int kgroupsP = blockDim.x / dpairs;
+  const int keysP = DEC_TILE / kgroupsP;
   const int dp = tid % dpairs;
-  const int kg = tid / dpairs;
+  const int kgP = tid / dpairs;
 
   float m[DEC_GMAX], l[DEC_GMAX], sc[DEC_GMAX];
-  float o0[DEC_GMAX], o1[DEC_GMAX];
+  float oa[DEC_GMAX][8];
   #pragma unroll
     for (int g = 0; g < G; ++g) {
-    m[g] = -1.0f / 0.0f; l[g] = 0.f; o0[g] = 0.f; o1[g] = 0.f;
+    m[g] = -1.0f / 0.0f; l[g] = 0.f;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) oa[g][e] = 0.f;
   }
 
   for (int tile = start; tile < end; tile += DEC_TILE) {
@@ -444,29 +460,53 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
     #pragma unroll
     for (int g = 0; g < G; ++g) l[g] = l[g] * alpha[g] + tile_sum[g];
 
-    // P*V: V row read once, used by every head
+    // P*V: V row read once (16 B/lane), used by every head
     #pragma unroll
-    for (int g = 0; g < G; ++g) { o0[g] *= alpha[g]; o1[g] *= alpha[g]; }
+    for (int g = 0; g < G; ++g) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) oa[g][e] *= alpha[g];
+    }
     // Break-free trip count so hipcc can unroll and keep several V loads
     // in flight (a data-dependent break serialises the loop into dependent
     // L2 round trips — measured 41 us/kernel at seq 640).
-    const int kbase_local = kg * keys_per_group;
     const int kmax = min(DEC_TILE, end - tile);
-    const int iters = min(keys_per_group, max(0, kmax - kbase_local));
-#pragma unroll 4
-    for (int j = 0; j < iters; ++j) {
-      const int kl = kbase_local + j;
-      const u16* vrow = vbase + (long)(tile + kl) * D + dp * 2;
-      const u32 vpair = KVNT
-          ? __builtin_nontemporal_load((const u32*)vrow)
-          : *(const u32*)vrow;
-      const float v0 = bf2f((u16)(vpair & 0xffff));
-      const float v1 = bf2f((u16)(vpair >> 16));
-      #pragma unroll
+    if (wide) {
+      const int kbase_local = kg * keys_per_group;
+      const int iters = min(keys_per_group, max(0, kmax - kbase_local));
+#pragma unroll 2
+      for (int j = 0; j < iters; ++j) {
+        const int kl = kbase_local + j;
+        const s16x8* vrow =
+            (const s16x8*)(vbase + (long)(tile + kl) * D) + dv;
+        const s16x8 v8 = KVNT ? __builtin_nontemporal_load(vrow) : *vrow;
+        float vf[8];
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) vf[e] = bf2f((u16)v8[e]);
+        #pragma unroll
     for (int g = 0; g < G; ++g) {
-        const float p = pl[g][kl];
-        o0[g] = fmaf(p, v0, o0[g]);
-        o1[g] = fmaf(p, v1, o1[g]);
+          const float p = pl[g][kl];
+          #pragma unroll
+          for (int e = 0; e < 8; ++e) oa[g][e] = fmaf(p, vf[e], oa[g][e]);
+        }
+      }
+    } else {
+      const int kbase_local = kgP * keysP;
+      const int iters = min(keysP, max(0, kmax - kbase_local));
+#pragma unroll 4
+      for (int j = 0; j < iters; ++j) {
+        const int kl = kbase_local + j;
+        const u16* vrow = vbase + (long)(tile + kl) * D + dp * 2;
+        const u32 vpair = KVNT
+            ? __builtin_nontemporal_load((const u32*)vrow)
+            : *(const u32*)vrow;
+        const float v0 = bf2f((u16)(vpair & 0xffff));
+        const float v1 = bf2f((u16)(vpair >> 16));
+        #pragma unroll
+    for (int g = 0; g < G; ++g) {
+          const float p = pl[g][kl];
+          oa[g][0] = fmaf(p, v0, oa[g][0]);
+          oa[g][1] = fmaf(p, v1, oa[g][1]);
+        }
       }
     }
     __syncthreads();                               // pl reuse next tile
@@ -480,17 +520,40 @@ k_attn_decode(const u16* __restrict__ q, u16* __restrict__ kc,
     const int hq = hkv * G + g;
     float* po = part_o + (((long)b * Hq + hq) * splits + split) * D;
     float* pml = part_ml + (((long)b * Hq + hq) * splits + split) * 2;
-    osh[kg][dp][0] = o0[g];
-    osh[kg][dp][1] = o1[g];
-    __syncthreads();
-    if (kg == 0) {
-      float s0 = osh[0][dp][0], s1 = osh[0][dp][1];
-      for (int gg = 1; gg < kgroups; ++gg) {
-        s0 += osh[gg][dp][0];
-        s1 += osh[gg][dp][1];
+    if (wide) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) osh[kg * dvecs + dv][e] = oa[g][e];
+      __syncthreads();
+      if (kg == 0) {
+        float s[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) s[e] = osh[dv][e];
+        for (int gg = 1; gg < kgroups; ++gg) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) s[e] += osh[gg * dvecs + dv][e];
+        }
+        if (arrive != nullptr) {
+#pragma unroll
+          for (int e = 0; e < 8; e += 2)
+            store2_sc1(&po[dv * 8 + e], s[e], s[e + 1]);
+        } else {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) po[dv * 8 + e] = s[e];
+        }
       }
-      if (arrive != nullptr) store2_sc1(&po[dp * 2], s0, s1);
-      else { po[dp * 2] = s0; po[dp * 2 + 1] = s1; }
+    } else {
+      osh[kgP * dpairs + dp][0] = oa[g][0];
+      osh[kgP * dpairs + dp][1] = oa[g][1];
+      __syncthreads();
+      if (kgP == 0) {
+        float s0 = osh[dp][0], s1 = osh[dp][1];
+        for (int gg = 1; gg < kgroupsP; ++gg) {
+          s0 += osh[gg * dpairs + dp][0];
+          s1 += osh[gg * dpairs + dp][1];
+        }
+        if (arrive != nullptr) store2_sc1(&po[dp * 2], s0, s1);
+        else { po[dp * 2] = s0; po[dp * 2 + 1] = s1; }
+      }
     }
     if (tid == 0) {
       if (arrive != nullptr) store2_sc1(pml, m[g], l[g]);
