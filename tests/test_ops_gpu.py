@@ -120,6 +120,39 @@ def test_attn_decode(lib, n, splits):
     assert err < 2e-2, f"max err {err}"
 
 
+@pytest.mark.parametrize("n,splits", [(2100, 32), (8100, 32), (2100, 16),
+                                      (4100, 64)])
+def test_attn_decode_long_context(lib, n, splits):
+    """chunk = n/splits >= 64 selects the WIDE V lane map (16 B/lane);
+    these lengths pin its numerics (the short-n cases all take the pair
+    path). chunk derives from each sequence's own pos, so the mixed pos
+    here runs seq 0 on the wide path and seq 1 (n//3) on the pair path
+    in the same launch — both are checked against the reference."""
+    from fei_amd.ops import reference as ref
+    B, Hq, Hkv, D, MS = 2, 8, 2, 128, 8192
+    q = randbf(B, Hq, D, seed=70 + n)
+    kc = _mk_cache(B, Hkv, MS, D, seed=71 + n)
+    vc = _mk_cache(B, Hkv, MS, D, seed=72 + n)
+    pos = torch.tensor([n - 1, n // 3], dtype=torch.int32, device=DEV)
+    out = lib.attn_decode(q, kc, vc, pos, splits=splits)
+    expected = ref.attn_decode(q, kc, vc, pos + 1)
+    err = (out.float() - expected.float()).abs().max().item()
+    assert err < 2e-2, f"max err {err}"
+    # fused-combine variant over the same long context
+    ws = (torch.zeros(B, Hq, splits, D, dtype=torch.float32, device=DEV),
+          torch.zeros(B, Hq, splits, 2, dtype=torch.float32, device=DEV),
+          torch.zeros(B * Hkv * splits, dtype=torch.int32, device=DEV))
+    import fei_amd.ops as ops_mod
+    saved = ops_mod._FUSED_CMB
+    ops_mod._FUSED_CMB = True
+    try:
+        out_f = lib.attn_decode(q, kc, vc, pos, splits=splits, workspace=ws,
+                                layer=7)
+    finally:
+        ops_mod._FUSED_CMB = saved
+    assert (out_f.float() - expected.float()).abs().max().item() < 2e-2
+
+
 @pytest.mark.parametrize("n,splits", [(1, 1), (7, 4), (300, 16), (1000, 32),
                                       (20, 32)])
 def test_attn_decode_fused_combine(lib, n, splits, monkeypatch):
