@@ -1029,7 +1029,7 @@ k_attn_decode_paged(const u16* __restrict__ q, const u16* __restrict__ kp,
   __shared__ float qs[DEC_GMAX][DEC_DMAX];
   __shared__ float pl[DEC_GMAX][DEC_TILE];
   __shared__ float red[DEC_GMAX][4];
-  __shared__ float osh[8][DEC_DMAX / 2][2];
+  __shared__ float osh[DEC_TILE][8];   // [kg * dvecs + dv][element]
 
   const int n = pos[b] + 1;
   const int chunk = (n + splits - 1) / splits;
@@ -1062,17 +1062,27 @@ k_attn_decode_paged(const u16* __restrict__ q, const u16* __restrict__ kp,
     return pool + (((long)blk * Hkv + hkv) * BS + within) * D;
   };
 
-  const int dpairs = D / 2;
-  const int kgroups = blockDim.x / dpairs;
+  // chunk-adaptive V lane maps — same scheme as k_attn_decode (wide
+  // 16 B/lane once a split's chunk covers >= 64 keys; pair otherwise)
+  const int dvecs = D / 8;
+  const int kgroups = blockDim.x / dvecs;
   const int keys_per_group = DEC_TILE / kgroups;
+  const int dv = tid % dvecs;
+  const int kg = tid / dvecs;
+  const bool wide = chunk >= 64;
+  const int dpairs = D / 2;
+  const int kgroupsP = blockDim.x / dpairs;
+  const int keysP = DEC_TILE / kgroupsP;
   const int dp = tid % dpairs;
-  const int kg = tid / dpairs;
+  const int kgP = tid / dpairs;
 
   float m[DEC_GMAX], l[DEC_GMAX], sc[DEC_GMAX];
-  float o0[DEC_GMAX], o1[DEC_GMAX];
+  float oa[DEC_GMAX][8];
 #pragma unroll
   for (int g = 0; g < G; ++g) {
-    m[g] = -1.0f / 0.0f; l[g] = 0.f; o0[g] = 0.f; o1[g] = 0.f;
+    m[g] = -1.0f / 0.0f; l[g] = 0.f;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) oa[g][e] = 0.f;
   }
 
   for (int tile = start; tile < end; tile += DEC_TILE) {
@@ -1117,21 +1127,43 @@ k_attn_decode_paged(const u16* __restrict__ q, const u16* __restrict__ kp,
     for (int g = 0; g < G; ++g) l[g] = l[g] * alpha[g] + tile_sum[g];
 
 #pragma unroll
-    for (int g = 0; g < G; ++g) { o0[g] *= alpha[g]; o1[g] *= alpha[g]; }
-    const int kbase_local = kg * keys_per_group;
-    const int kmax = min(DEC_TILE, end - tile);
-    const int iters = min(keys_per_group, max(0, kmax - kbase_local));
-#pragma unroll 4
-    for (int j = 0; j < iters; ++j) {
-      const int kl = kbase_local + j;
-      const u16* vrow = row_ptr(vp, tile + kl) + dp * 2;
-      const float v0 = bf2f(vrow[0]);
-      const float v1 = bf2f(vrow[1]);
+    for (int g = 0; g < G; ++g) {
 #pragma unroll
-      for (int g = 0; g < G; ++g) {
-        const float p = pl[g][kl];
-        o0[g] = fmaf(p, v0, o0[g]);
-        o1[g] = fmaf(p, v1, o1[g]);
+      for (int e = 0; e < 8; ++e) oa[g][e] *= alpha[g];
+    }
+    const int kmax = min(DEC_TILE, end - tile);
+    if (wide) {
+      const int kbase_local = kg * keys_per_group;
+      const int iters = min(keys_per_group, max(0, kmax - kbase_local));
+#pragma unroll 2
+      for (int j = 0; j < iters; ++j) {
+        const int kl = kbase_local + j;
+        const s16x8 v8 = *((const s16x8*)row_ptr(vp, tile + kl) + dv);
+        float vf[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) vf[e] = bf2f((u16)v8[e]);
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          const float p = pl[g][kl];
+#pragma unroll
+          for (int e = 0; e < 8; ++e) oa[g][e] = fmaf(p, vf[e], oa[g][e]);
+        }
+      }
+    } else {
+      const int kbase_local = kgP * keysP;
+      const int iters = min(keysP, max(0, kmax - kbase_local));
+#pragma unroll 4
+      for (int j = 0; j < iters; ++j) {
+        const int kl = kbase_local + j;
+        const u16* vrow = row_ptr(vp, tile + kl) + dp * 2;
+        const float v0 = bf2f(vrow[0]);
+        const float v1 = bf2f(vrow[1]);
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          const float p = pl[g][kl];
+          oa[g][0] = fmaf(p, v0, oa[g][0]);
+          oa[g][1] = fmaf(p, v1, oa[g][1]);
+        }
       }
     }
     __syncthreads();
@@ -1142,17 +1174,34 @@ k_attn_decode_paged(const u16* __restrict__ q, const u16* __restrict__ kp,
     const int hq = hkv * G + g;
     float* po = part_o + (((long)b * Hq + hq) * splits + split) * D;
     float* pml = part_ml + (((long)b * Hq + hq) * splits + split) * 2;
-    osh[kg][dp][0] = o0[g];
-    osh[kg][dp][1] = o1[g];
-    __syncthreads();
-    if (kg == 0) {
-      float s0 = osh[0][dp][0], s1 = osh[0][dp][1];
-      for (int gg = 1; gg < kgroups; ++gg) {
-        s0 += osh[gg][dp][0];
-        s1 += osh[gg][dp][1];
+    if (wide) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) osh[kg * dvecs + dv][e] = oa[g][e];
+      __syncthreads();
+      if (kg == 0) {
+        float s[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) s[e] = osh[dv][e];
+        for (int gg = 1; gg < kgroups; ++gg) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) s[e] += osh[gg * dvecs + dv][e];
+        }
+#pragma unroll
+        for (int e = 0; e < 8; ++e) po[dv * 8 + e] = s[e];
       }
-      po[dp * 2] = s0;
-      po[dp * 2 + 1] = s1;
+    } else {
+      osh[kgP * dpairs + dp][0] = oa[g][0];
+      osh[kgP * dpairs + dp][1] = oa[g][1];
+      __syncthreads();
+      if (kgP == 0) {
+        float s0 = osh[dp][0], s1 = osh[dp][1];
+        for (int gg = 1; gg < kgroupsP; ++gg) {
+          s0 += osh[gg * dpairs + dp][0];
+          s1 += osh[gg * dpairs + dp][1];
+        }
+        po[dp * 2] = s0;
+        po[dp * 2 + 1] = s1;
+      }
     }
     if (tid == 0) { pml[0] = m[g]; pml[1] = l[g]; }
     __syncthreads();

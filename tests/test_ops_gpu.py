@@ -440,9 +440,11 @@ def test_gemv_swiglu_norm_matches(lib):
     assert err < 2e-2, f"max rel err {err}"
 
 
-@pytest.mark.parametrize("n", [5, 100, 500])
+@pytest.mark.parametrize("n", [5, 100, 500, 1000])
 def test_attn_decode_paged(lib, n):
-    """Paged attention over a scrambled block table == contiguous reference."""
+    """Paged attention over a scrambled block table == contiguous reference.
+    n=500/1000 at splits=4 -> chunk >= 125 exercises the WIDE V lane map;
+    the short n cases take the pair map."""
     from fei_amd.ops import reference as ref
     B, Hq, Hkv, D, BS = 2, 8, 2, 128, 16
     max_blocks = 64
