@@ -207,9 +207,13 @@ class LlamaModel:
         5 kernels per layer instead of 9."""
         # the fused-norm chain's per-wave norm prologue scales with B
         # (recomputes all B rows' sumsq per wave): batch 1 only — B=8
-        # measured 572 vs 1130 tok/s on the plain path (docs/BENCHMARKS.md)
-        if fused_norm and self.tp_size == 1 and \
-                token.shape[0] * self.spec.hidden_size * 2 <= 8 * 1024:
+        # measured 572 vs 1130 tok/s on the plain path (docs/BENCHMARKS.md).
+        # The r02 byte-form of that gate (B*C*2 <= 8 KB) also excluded
+        # hidden > 4096 at B=1, silently dropping 70B fp8 decode to the
+        # bf16 plain chain (measured 40.4 == bf16 vs r01's 41.3 with fp8
+        # engaged); the batch gate is B==1 with a separate LDS-budget cap.
+        if fused_norm and self.tp_size == 1 and token.shape[0] == 1 and \
+                self.spec.hidden_size * 2 <= 32 * 1024:
             return self._forward_decode_fused_norm(
                 token, pos, k_caches, v_caches, attn_splits, workspace,
                 fused_attn, attn_out)
@@ -286,8 +290,10 @@ class LlamaModel:
         # fp8 norm-GEMVs stage B*C bf16 activations in LDS and recompute
         # all B norms per wave — measured slower than bf16 already at B=2
         # (386 vs 410 tok/s) and 3.5x slower at B=8. fp8 pays for B=1
-        # agent decode only.
-        if fp8 is not None and B * s.hidden_size * 2 > 8 * 1024:
+        # agent decode only (any hidden size whose row fits the 32 KB
+        # dynamic-LDS budget — the old 8 KB byte gate wrongly excluded
+        # 70B's hidden=8192 at B=1).
+        if fp8 is not None and (B != 1 or s.hidden_size * 2 > 32 * 1024):
             fp8 = None
         h = F.embedding(token.long(), self.emb).contiguous()
         scale = 1.0 / math.sqrt(self.D)
