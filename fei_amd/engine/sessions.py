@@ -124,15 +124,26 @@ class PagedSessionManager:
         return [s for s in self.sessions.values() if not s.done]
 
     def step(self) -> int:
-        """One greedy decode step for every active session (one batched
-        forward over the paged pool). Returns the number of sessions that
-        advanced."""
+        """One greedy decode step for every active session. Returns the
+        number of sessions that advanced."""
+        return self.step_chunk(1)
+
+    def step_chunk(self, chunk: int = 8) -> int:
+        """Up to ``chunk`` greedy decode steps for the CURRENT active set
+        with device-resident token feedback — ONE host sync (the chunk's
+        token download) instead of one per step; block capacity for the
+        whole chunk is reserved up front and the table tensor is built
+        once. Admission/retirement happens between chunks. A session that
+        hits EOS mid-chunk keeps decoding garbage rows into its reserved
+        blocks until the chunk ends (its `generated`/`pos` stop at the
+        EOS, so the extra rows are never attended and the blocks are
+        freed on close). Returns sessions advanced (0 = none active)."""
         act = self.active
         if not act:
             return 0
         m, dev = self.model, self.model.device
         for s in act:
-            self.pool.ensure_capacity(s.sid, s.pos + 1)
+            self.pool.ensure_capacity(s.sid, s.pos + chunk)
         max_blocks = max(len(self.pool.table(s.sid).blocks) for s in act)
         table = torch.full((len(act), max_blocks), -1, dtype=torch.int32)
         for i, s in enumerate(act):
@@ -142,20 +153,30 @@ class PagedSessionManager:
         token = torch.tensor([s.generated[-1] for s in act],
                              dtype=torch.int64, device=dev)
         pos = torch.tensor([s.pos for s in act], dtype=torch.int32, device=dev)
-        logits = self._forward_paged(token, pos, table,
-                                     [s.pos for s in act])
-        nxt = logits.argmax(dim=-1).tolist()
-        for s, t in zip(act, nxt):
-            s.generated.append(int(t))
-            s.pos += 1
-            if int(t) == self.eos or len(s.generated) >= s.max_new_tokens:
-                s.done = True
+        steps = []
+        for _ in range(chunk):
+            logits = self._forward_paged(token, pos, table)
+            token = logits.argmax(dim=-1)
+            steps.append(token)
+            pos = pos + 1
+        allt = torch.stack(steps, dim=1).tolist()          # the one sync
+        for i, s in enumerate(act):
+            for t in allt[i]:
+                s.generated.append(int(t))
+                s.pos += 1
+                if int(t) == self.eos or \
+                        len(s.generated) >= s.max_new_tokens:
+                    s.done = True
+                    break
         return len(act)
 
-    def run(self, max_steps: int = 4096) -> None:
-        for _ in range(max_steps):
-            if self.step() == 0:
+    def run(self, max_steps: int = 4096, chunk: int = 8) -> None:
+        done = 0
+        while done < max_steps:
+            k = min(chunk, max_steps - done)
+            if self.step_chunk(k) == 0:
                 break
+            done += k
 
     def result(self, sid: int) -> Dict[str, object]:
         s = self.sessions[sid]
@@ -169,12 +190,14 @@ class PagedSessionManager:
     # -- model forward over the pool ------------------------------------------
 
     def _forward_paged(self, token: torch.Tensor, pos: torch.Tensor,
-                       table: torch.Tensor,
-                       pos_host: List[int]) -> torch.Tensor:
+                       table: torch.Tensor) -> torch.Tensor:
         """Plain decode forward with block-table attention. Same GEMV /
         norm kernels as LlamaModel.forward_decode (llama.py); RoPE + the
         single-row KV append are torch-side (control plane — O(Hkv*D) per
-        session, independent of context length)."""
+        session, independent of context length) and fully device-resident:
+        the block/offset of the appended row derive from `pos` and `table`
+        with gather/advanced indexing, so a whole step chain runs without
+        a host sync (step_chunk)."""
         m = self.model
         s = m.spec
         B = token.shape[0]
@@ -184,17 +207,16 @@ class PagedSessionManager:
         n_layers = len(m.layers)
         x = ops.rmsnorm(h, m.layers[0].norm_attn, s.norm_eps)
         pos_l = pos.long()
-        blk_of = [int(table[i, p // BS]) for i, p in enumerate(pos_host)]
-        off_of = [p % BS for p in pos_host]
+        blk = table.gather(1, (pos_l // BS).unsqueeze(1)).squeeze(1).long()
+        off = pos_l % BS
         for li, lw in enumerate(m.layers):
             qkv = ops.linear_decode(x, lw.wqkv)
             q, k, v = m._qkv_views(qkv, B)
             q_r = ref.apply_rope(q, pos_l, m.rope).contiguous()
             k_r = ref.apply_rope(k, pos_l, m.rope)
             kp, vp = self.pool.k[li], self.pool.v[li]
-            for i in range(B):
-                kp[blk_of[i], :, off_of[i], :] = k_r[i]
-                vp[blk_of[i], :, off_of[i], :] = v[i]
+            kp[blk, :, off, :] = k_r
+            vp[blk, :, off, :] = v
             att = ops.attn_decode_paged(q_r, kp, vp, table, pos,
                                         splits=self._splits(B),
                                         scale=scale,

@@ -27,9 +27,11 @@ if torch.cuda.is_available():
     torch.cuda.synchronize()
 t0 = time.perf_counter()
 steps = 0
+CH = int(os.environ.get("SESS_CHUNK", "8"))
 while mgr.active and steps < NEW + 8:
-    mgr.step()
-    steps += 1
+    k = min(CH, NEW + 8 - steps)
+    mgr.step_chunk(k)
+    steps += k
 if torch.cuda.is_available():
     torch.cuda.synchronize()
 dt = time.perf_counter() - t0

@@ -71,3 +71,21 @@ def test_result_stable_after_done(engine):
     assert a == b and a["done"]
     mgr.step()                       # no active sessions: must be a no-op
     assert mgr.result(sid) == a
+
+
+def test_step_chunk_matches_single_steps(engine):
+    """Chunked stepping (device-resident token feedback, ONE sync per
+    chunk) must emit the exact token streams of step-by-step decode,
+    including mid-chunk EOS/budget stops and a chunk spanning a block
+    boundary."""
+    mgr1 = PagedSessionManager(engine, block_size=16, num_blocks=64)
+    mgr2 = PagedSessionManager(engine, block_size=16, num_blocks=64)
+    sids1 = {p: mgr1.open(p, max_new_tokens=13) for p in PROMPTS}
+    sids2 = {p: mgr2.open(p, max_new_tokens=13) for p in PROMPTS}
+    while mgr1.active:
+        mgr1.step()
+    while mgr2.active:
+        mgr2.step_chunk(8)
+    for p in PROMPTS:
+        assert (mgr1.result(sids1[p])["token_ids"]
+                == mgr2.result(sids2[p])["token_ids"]), p
