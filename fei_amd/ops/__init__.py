@@ -71,7 +71,7 @@ def _try_load() -> None:
                                      _i, _i, _f, _f, _vp]
     lib.fei_attn_decode_paged.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp,
                                           _i, _i, _i, _i, _i, _i, _i, _f,
-                                          _l, _vp]
+                                          _l, _vp, _vp, _vp, _l, _vp]
     lib.fei_add_layernorm.argtypes = [_vp, _vp, _vp, _vp, _vp, _i, _i, _f,
                                       _i, _vp]
     lib.fei_gelu.argtypes = [_vp, _vp, _l, _vp]
@@ -631,16 +631,31 @@ def prefetch(w: torch.Tensor, max_bytes: int, sink: torch.Tensor,
 def attn_decode_paged(q, k_pool, v_pool, block_table, pos, splits: int = 32,
                       scale: Optional[float] = None,
                       workspace: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
-                      out: Optional[torch.Tensor] = None) -> torch.Tensor:
+                      out: Optional[torch.Tensor] = None,
+                      k: Optional[torch.Tensor] = None,
+                      v: Optional[torch.Tensor] = None,
+                      table: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Decode attention over a PAGED KV pool (engine/kv_cache.PagedKVPool):
     pools [num_blocks, Hkv, BS, D], block_table [B, max_blocks] int32 maps
-    logical key blocks to physical pool blocks. n = pos[b]+1 keys."""
+    logical key blocks to physical pool blocks. n = pos[b]+1 keys.
+    When (k, v, table) are given the kernel also fuses the step's RoPE +
+    paged KV-append: q is the RAW qkv view and the pool row for position
+    pos[b] is written in-kernel (mirrors attn_decode's fused form)."""
     B, Hq, D = q.shape
     n_blocks, Hkv, BS, _ = k_pool.shape
     assert BS & (BS - 1) == 0, "block size must be a power of two"
     bs_log = BS.bit_length() - 1
     scale = scale if scale is not None else 1.0 / math.sqrt(D)
     if not q.is_cuda:
+        if table is not None:
+            # reference path: rope q/k and append into the POOL rows
+            q = ref.apply_rope(q, pos.long(), table).to(q.dtype)
+            k_r = ref.apply_rope(k, pos.long(), table).to(k.dtype)
+            for b in range(B):
+                p = int(pos[b])
+                blk = int(block_table[b, p >> bs_log])
+                k_pool[blk, :, p & (BS - 1), :] = k_r[b]
+                v_pool[blk, :, p & (BS - 1), :] = v[b]
         # gather logical order into a contiguous cache, then reference
         max_len = block_table.shape[1] * BS
         kc = torch.zeros(B, Hkv, max_len, D, dtype=k_pool.dtype)
@@ -663,11 +678,18 @@ def attn_decode_paged(q, k_pool, v_pool, block_table, pos, splits: int = 32,
         out = torch.empty_like(q)
     assert q.stride(2) == 1 and q.stride(1) == D
     assert splits <= 64
+    if table is not None:
+        assert k is not None and v is not None
+        assert k.stride(1) == D and k.stride(0) == v.stride(0)
+        kin, vin, cs, kv_bs = _ptr(k), _ptr(v), _ptr(table), k.stride(0)
+    else:
+        kin = vin = cs = None
+        kv_bs = 0
     lib.fei_attn_decode_paged(_ptr(q), _ptr(k_pool), _ptr(v_pool),
                               _ptr(block_table), _ptr(part_o), _ptr(part_ml),
                               _ptr(pos), B, Hq, Hkv, D, bs_log,
                               block_table.shape[1], splits, scale,
-                              q.stride(0), _stream())
+                              q.stride(0), kin, vin, cs, kv_bs, _stream())
     lib.fei_attn_decode_combine(_ptr(out), _ptr(part_o), _ptr(part_ml),
                                 B, Hq, D, splits, _stream())
     return out

@@ -1012,15 +1012,23 @@ void fei_sample_shard(const void* logits, float* out, const int* step,
 
 namespace {
 
-template <int G>
+// ROPE=true fuses the per-step RoPE + paged KV-append exactly like the
+// plain kernel (kin/vin raw, cos_sin table; the split that OWNS key n-1
+// ropes and writes the pool row via the block table before its own
+// tile pass — no other split touches that key). Replaces the serving
+// path's ~15 torch-op launches per layer (apply_rope x2, contiguous,
+// 2 advanced-index writes), which were ~30% of the sessions step.
+template <int G, bool ROPE = false>
 __global__ void __launch_bounds__(256)
-k_attn_decode_paged(const u16* __restrict__ q, const u16* __restrict__ kp,
-                    const u16* __restrict__ vp,
+k_attn_decode_paged(const u16* __restrict__ q, u16* __restrict__ kp,
+                    u16* __restrict__ vp,
                     const int* __restrict__ block_table,
                     float* __restrict__ part_o, float* __restrict__ part_ml,
                     const int* __restrict__ pos,
                     int B, int Hq, int Hkv, int D, int bs_log,
-                    int max_blocks, int splits, float scale, long q_bs) {
+                    int max_blocks, int splits, float scale, long q_bs,
+                    const u16* __restrict__ kin, const u16* __restrict__ vin,
+                    const float* __restrict__ cos_sin, long kv_bs) {
   const int split = blockIdx.x;
   const int hkv = blockIdx.y;
   const int b = blockIdx.z;
@@ -1050,17 +1058,49 @@ k_attn_decode_paged(const u16* __restrict__ q, const u16* __restrict__ kp,
     return;
   }
 
-  for (int i = tid; i < G * D; i += blockDim.x) {
-    const int g = i / D, d = i % D;
-    qs[g][d] = bf2f(q[(long)b * q_bs + (long)(hkv * G + g) * D + d]) * scale;
-  }
-  __syncthreads();
-
   auto row_ptr = [&](const u16* pool, int kk) -> const u16* {
     const int blk = bt[kk >> bs_log];
     const int within = kk & (BS - 1);
     return pool + (((long)blk * Hkv + hkv) * BS + within) * D;
   };
+
+  const int half = D / 2;
+  const int p_new = n - 1;
+  if (ROPE) {
+    // stage roped+scaled q (same math as the plain kernel's ROPE path)
+    for (int i = tid; i < G * half; i += blockDim.x) {
+      const int g = i / half, d = i % half;
+      const float c = cos_sin[((long)p_new * half + d) * 2 + 0];
+      const float sn = cos_sin[((long)p_new * half + d) * 2 + 1];
+      const u16* qp = q + (long)b * q_bs + (long)(hkv * G + g) * D;
+      const float x1 = bf2f(qp[d]);
+      const float x2 = bf2f(qp[d + half]);
+      qs[g][d] = (x1 * c - x2 * sn) * scale;
+      qs[g][d + half] = (x2 * c + x1 * sn) * scale;
+    }
+    if (p_new >= start && p_new < end) {
+      u16* kcp = (u16*)row_ptr(kp, p_new);
+      u16* vcp = (u16*)row_ptr(vp, p_new);
+      for (int d = tid; d < half; d += blockDim.x) {
+        const float c = cos_sin[((long)p_new * half + d) * 2 + 0];
+        const float sn = cos_sin[((long)p_new * half + d) * 2 + 1];
+        const u16* kq = kin + (long)b * kv_bs + (long)hkv * D;
+        const u16* vq = vin + (long)b * kv_bs + (long)hkv * D;
+        const float x1 = bf2f(kq[d]);
+        const float x2 = bf2f(kq[d + half]);
+        kcp[d] = f2bf(x1 * c - x2 * sn);
+        kcp[d + half] = f2bf(x2 * c + x1 * sn);
+        vcp[d] = vq[d];
+        vcp[d + half] = vq[d + half];
+      }
+    }
+  } else {
+    for (int i = tid; i < G * D; i += blockDim.x) {
+      const int g = i / D, d = i % D;
+      qs[g][d] = bf2f(q[(long)b * q_bs + (long)(hkv * G + g) * D + d]) * scale;
+    }
+  }
+  __syncthreads();
 
   // chunk-adaptive V lane maps — same scheme as k_attn_decode (wide
   // 16 B/lane once a split's chunk covers >= 64 keys; pair otherwise)
@@ -1212,25 +1252,31 @@ k_attn_decode_paged(const u16* __restrict__ q, const u16* __restrict__ kp,
 
 extern "C" {
 
-void fei_attn_decode_paged(const void* q, const void* k_pool,
-                           const void* v_pool, const int* block_table,
+void fei_attn_decode_paged(const void* q, void* k_pool,
+                           void* v_pool, const int* block_table,
                            float* part_o, float* part_ml, const int* pos,
                            int B, int Hq, int Hkv, int D, int bs_log,
                            int max_blocks, int splits, float scale,
-                           long q_bs, hipStream_t stream) {
+                           long q_bs,
+                           const void* kin, const void* vin,
+                           const float* cos_sin, long kv_bs,
+                           hipStream_t stream) {
   const int G = Hq / Hkv;
   dim3 grid(splits, Hkv, B);
-#define LP(GV) hipLaunchKernelGGL(k_attn_decode_paged<GV>, grid, dim3(256), \
-    0, stream, (const u16*)q, (const u16*)k_pool, (const u16*)v_pool, \
+  const int rope = cos_sin != nullptr;
+#define LP(GV, RV) hipLaunchKernelGGL((k_attn_decode_paged<GV, RV>), grid, \
+    dim3(256), 0, stream, (const u16*)q, (u16*)k_pool, (u16*)v_pool, \
     block_table, part_o, part_ml, pos, B, Hq, Hkv, D, bs_log, max_blocks, \
-    splits, scale, q_bs)
+    splits, scale, q_bs, (const u16*)kin, (const u16*)vin, cos_sin, kv_bs)
+#define LPR(GV) do { if (rope) LP(GV, true); else LP(GV, false); } while (0)
   switch (G) {
-    case 1: LP(1); break;
-    case 2: LP(2); break;
-    case 4: LP(4); break;
-    case 8: LP(8); break;
+    case 1: LPR(1); break;
+    case 2: LPR(2); break;
+    case 4: LPR(4); break;
+    case 8: LPR(8); break;
     default: break;
   }
+#undef LPR
 #undef LP
 }
 

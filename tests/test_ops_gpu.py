@@ -590,3 +590,49 @@ def test_gemv_ssq_chain(lib, M):
     assert err < 2e-2, f"gemv_swiglu_norm ssq path max rel err {err}"
     # sanity: the chain actually changed the residual (res_pre unused warn)
     assert not torch.equal(res, res_pre)
+
+
+@pytest.mark.parametrize("n", [7, 100, 500, 1000])
+def test_attn_decode_paged_rope_fused(lib, n):
+    """Fused paged form (raw q/k/v + rope table; in-kernel RoPE + pool
+    append) must equal roping torch-side, appending, then the plain
+    paged kernel — the serving hot path (sessions._forward_paged)."""
+    from fei_amd.ops import reference as ref
+    B, Hq, Hkv, D, BS = 2, 8, 2, 128, 16
+    max_blocks, n_phys = 64, 256
+    g = torch.Generator(device=DEV).manual_seed(500 + n)
+    k_pool = (torch.randn(n_phys, Hkv, BS, D, generator=g, device=DEV) * 0.3).to(torch.bfloat16)
+    v_pool = (torch.randn(n_phys, Hkv, BS, D, generator=g, device=DEV) * 0.3).to(torch.bfloat16)
+    k_pool2, v_pool2 = k_pool.clone(), v_pool.clone()
+    perm = torch.randperm(n_phys, generator=g, device=DEV)[: B * max_blocks]
+    bt = perm.view(B, max_blocks).to(torch.int32)
+    q = randbf(B, Hq, D, seed=501 + n)
+    k = randbf(B, Hkv, D, seed=502 + n)
+    v = randbf(B, Hkv, D, seed=503 + n)
+    pos = torch.tensor([n - 1, max(n // 2 - 1, 0)], dtype=torch.int32,
+                       device=DEV)
+    rope = ref.rope_table(2048, D, device=DEV)
+
+    out_f = lib.attn_decode_paged(q.clone(), k_pool, v_pool, bt, pos,
+                                  splits=4, k=k, v=v, table=rope)
+
+    # torch-side rope + append, then the plain paged kernel
+    q_r = ref.apply_rope(q.float(), pos.long(), rope).to(torch.bfloat16)
+    k_r = ref.apply_rope(k.float(), pos.long(), rope).to(torch.bfloat16)
+    for b in range(B):
+        p = int(pos[b])
+        blk = int(bt[b, p // BS])
+        k_pool2[blk, :, p % BS, :] = k_r[b]
+        v_pool2[blk, :, p % BS, :] = v[b]
+    out_u = lib.attn_decode_paged(q_r.contiguous(), k_pool2, v_pool2, bt,
+                                  pos, splits=4)
+    assert (out_f.float() - out_u.float()).abs().max().item() < 2e-2
+    # the appended pool rows must match too
+    for b in range(B):
+        p = int(pos[b])
+        blk = int(bt[b, p // BS])
+        dk = (k_pool[blk, :, p % BS].float()
+              - k_pool2[blk, :, p % BS].float()).abs().max().item()
+        dv = (v_pool[blk, :, p % BS].float()
+              - v_pool2[blk, :, p % BS].float()).abs().max().item()
+        assert dk < 2e-2 and dv == 0.0
