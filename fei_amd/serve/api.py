@@ -278,11 +278,9 @@ def create_app(engine: Optional[LocalEngine] = None,
     def session_step(n: int = 1):
         with lock:
             mgr = _mgr()
-            advanced = 0
-            for _ in range(max(1, n)):
-                advanced = mgr.step()
-                if advanced == 0:
-                    break
+            # chunked stepping: ONE host sync for the whole request
+            # (engine/sessions.step_chunk) instead of one per step
+            advanced = mgr.step_chunk(max(1, n))
         return {"active": advanced,
                 "open": len(mgr.sessions),
                 "free_blocks": mgr.pool.free_blocks()}
