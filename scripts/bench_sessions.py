@@ -15,7 +15,8 @@ from fei_amd.engine.sessions import PagedSessionManager
 
 N = int(os.environ.get("SESS_N", "8"))
 NEW = int(os.environ.get("SESS_NEW", "256"))
-eng = LocalEngine.create("llama3-8b", max_seq_len=4096, seed=7)
+MODEL = os.environ.get("SESS_MODEL", "llama3-8b")
+eng = LocalEngine.create(MODEL, max_seq_len=4096, seed=7)
 mgr = PagedSessionManager(eng)
 rng = torch.Generator().manual_seed(5)
 sids = []
@@ -40,4 +41,5 @@ print(json.dumps({
     "metric": "serving aggregate tok/s (continuous batching, paged KV)",
     "value": round(toks / dt, 1), "unit": "tok/s", "sessions": N,
     "new_tokens_per_session": NEW, "steps": steps,
-    "ms_per_step": round(dt / steps * 1000, 3), "data": "synthetic"}))
+    "ms_per_step": round(dt / steps * 1000, 3), "model": MODEL,
+    "data": "synthetic"}))

@@ -27,6 +27,8 @@ run python bench.py --model llama3-70b --prompt-len 256 --steps 48 --warmup 8
 run python bench.py --model llama3-1b --steps 256 --warmup 32
 run python scripts/bench_spec.py llama3-8b 256
 run python scripts/bench_gemv.py
+run python scripts/bench_sessions.py
+run env SESS_N=16 python scripts/bench_sessions.py
 run python scripts/bench_agent_turn.py
 run env CORPUS_N=1000000 python scripts/bench_memdir.py
 
