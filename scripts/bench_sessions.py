@@ -22,7 +22,8 @@ rng = torch.Generator().manual_seed(5)
 sids = []
 for i in range(N):
     plen = 64 + 48 * i                      # staggered contexts
-    ids = torch.randint(4, 16000, (plen,), generator=rng).tolist()
+    hi = min(16000, eng.spec.vocab_size - 2)
+    ids = torch.randint(4, hi, (plen,), generator=rng).tolist()
     sids.append(mgr.open(ids, max_new_tokens=NEW))
 if torch.cuda.is_available():
     torch.cuda.synchronize()
