@@ -232,6 +232,10 @@ def attn_decode(q, k_cache, v_cache, pos, splits: int = 32,
         out = torch.empty_like(q)
     Hkv = k_cache.shape[1]
     assert q.stride(2) == 1 and q.stride(1) == D
+    # the kernel's V lane maps tile 256 threads by D/2 and D/8 — both must
+    # divide the block (holds for the llama head dims 64/128; D=96 would
+    # silently under-cover a tile)
+    assert D in (64, 128), f"decode attention supports D in (64,128), got {D}"
     assert splits <= 64, "combine kernel stages at most 64 split partials"
     if table is not None:
         assert k is not None and v is not None
@@ -677,6 +681,7 @@ def attn_decode_paged(q, k_pool, v_pool, block_table, pos, splits: int = 32,
     if out is None:
         out = torch.empty_like(q)
     assert q.stride(2) == 1 and q.stride(1) == D
+    assert D in (64, 128), f"paged attention supports D in (64,128), got {D}"
     assert splits <= 64
     if table is not None:
         assert k is not None and v is not None
