@@ -57,6 +57,7 @@ def _try_load() -> None:
                                      _i, _i, _i, _i, _i, _i, _f, _i, _l, _vp]
     lib.fei_mfma_probe.argtypes = [_vp, _vp, _vp, _vp]
     lib.fei_gemv.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
+    lib.fei_gemm_m8.argtypes = [_vp, _vp, _vp, _i, _i, _i, _i, _vp]
     lib.fei_attn_decode_fused.argtypes = [_vp, _vp, _vp, _vp, _vp, _vp, _vp,
                                           _vp, _i, _i, _i, _i, _i, _f,
                                           _l, _l, _vp]
@@ -506,6 +507,13 @@ _GEMV_NT_MIN_BYTES = int(os.environ.get("FEI_GEMV_NT_MIN", 0))
 _FUSED_CMB = os.environ.get("FEI_FUSED_CMB", "0") == "1"
 
 
+def _gemm_m8_ok(M: int, N: int, K: int) -> bool:
+    # MFMA GEMM tile: A-fragment rows clamp to M-1 so M <= 16 is CORRECT,
+    # but at M = 16 hipBLASLt measured faster (serving-16: 2473 vs 2337
+    # tok/s) — route only the 4..8 range where the tile wins 1.27-1.65x
+    return 4 <= M <= 8 and K % 1024 == 0 and N % 16 == 0
+
+
 def _gemv_ok(M: int, K: int) -> bool:
     return M in (1, 2, 4, 8) and K % 8 == 0
 
@@ -518,13 +526,21 @@ def linear_decode(x: torch.Tensor, w: torch.Tensor,
     M = x.numel() // x.shape[-1]
     K = x.shape[-1]
     N = w.shape[0]
-    if not x.is_cuda or not _gemv_ok(M, K):
+    if not x.is_cuda or not (_gemv_ok(M, K) or _gemm_m8_ok(M, N, K)):
         return torch.nn.functional.linear(x, w)
     lib = require_lib()
     x2 = x.contiguous().view(M, K)
     if out is None:
         out = torch.empty(*x.shape[:-1], N, dtype=x.dtype, device=x.device)
     nt = 1 if (N * K * 2 >= _GEMV_NT_MIN_BYTES) else 0
+    if _gemm_m8_ok(M, N, K):
+        # MFMA tile: beats the VALU GEMV at every llama batch shape
+        # (1.27-1.65x — profiles/r02_batch_attention.md addendum).
+        # ALWAYS plain loads: the nt hint on the 16 B/lane staged W
+        # fetch measured 6.50 vs 5.01 ms/step at batch 8.
+        lib.fei_gemm_m8(_ptr(out), _ptr(x2), _ptr(w), M, N, K, 0,
+                        _stream())
+        return out
     lib.fei_gemv(_ptr(out), _ptr(x2), _ptr(w), M, N, K, nt, _stream())
     return out
 
@@ -535,13 +551,22 @@ def gemv_swiglu(x: torch.Tensor, wgu: torch.Tensor,
     M = x.numel() // x.shape[-1]
     K = x.shape[-1]
     I = wgu.shape[0] // 2
-    if not x.is_cuda or not _gemv_ok(M, K):
+    if not x.is_cuda or not (_gemv_ok(M, K) or _gemm_m8_ok(M, 2 * I, K)):
         return swiglu(torch.nn.functional.linear(x, wgu))
     lib = require_lib()
     x2 = x.contiguous().view(M, K)
     if out is None:
         out = torch.empty(*x.shape[:-1], I, dtype=x.dtype, device=x.device)
     nt = 1 if (2 * I * K * 2 >= _GEMV_NT_MIN_BYTES) else 0
+    if _gemm_m8_ok(M, 2 * I, K):
+        # MFMA over the stacked [2I, K] weight (its [M, 2I] output is
+        # k_swiglu's packed input): 68.3 -> ~47 us on the 8B gate/up at
+        # batch 8 vs the fused VALU kernel
+        gu = torch.empty(M, 2 * I, dtype=x.dtype, device=x.device)
+        lib.fei_gemm_m8(_ptr(gu), _ptr(x2), _ptr(wgu), M, 2 * I, K, 0,
+                        _stream())
+        lib.fei_swiglu(_ptr(out), _ptr(gu), M, I, _stream())
+        return out
     lib.fei_gemv_swiglu(_ptr(out), _ptr(x2), _ptr(wgu), M, I, K, nt,
                         _stream())
     return out

@@ -431,3 +431,101 @@ void fei_gemv_swiglu_norm(void* out, const void* res, const void* wn,
 #undef DISPATCH_M
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// MFMA M<=8 GEMM: out[M, N] = x[M, K] @ W[N, K]^T for the decode batch
+// path. Four VALU-kernel variants measured slower than the pipelined
+// k_gemv<8> (profiles/r02_batch_attention.md); this MFMA tile beats it
+// on every llama shape (qkv 1.45x, o 1.43x, down 1.27x, stacked gate/up
+// 1.65x — experimental/gemv_mfma.hip, numerics exact vs the VALU form).
+// Structure: 16 n-rows per workgroup, each wave owns a K-QUARTER with
+// wave-private double-buffered LDS staging (register prefetch overlaps
+// the next tile's loads with the current tile's MFMAs; same-wave ds
+// ordering needs no barriers), one end-of-kernel cross-wave C reduce.
+// A-fragment rows clamp to M-1: the duplicate rows only produce C rows
+// >= M, which are never written. Requires K % 1024 == 0, N % 16 == 0
+// (the wrapper routes other shapes to k_gemv).
+#define GEMM_NT 16
+#define GEMM_KC 128
+#define GSWZ(row, col8) ((col8) ^ ((row) & 7))
+
+namespace {
+
+typedef __attribute__((ext_vector_type(4))) float f32x4_g;
+
+template <bool NTW>
+__global__ void __launch_bounds__(256)
+k_gemm_m8(u16* __restrict__ out, const u16* __restrict__ x,
+          const u16* __restrict__ w, int M, int N, int K) {
+  const int n0 = blockIdx.x * GEMM_NT;
+  const int tid = threadIdx.x;
+  const int wv = tid >> 6;
+  const int lane = tid & 63;
+  const int l15 = lane & 15, lg = lane >> 4;
+  __shared__ u16 wt[4][2][GEMM_NT * GEMM_KC];
+  __shared__ float cred[4][16][16];
+
+  f32x4_g acc = {0.f, 0.f, 0.f, 0.f};
+  const int kq = K / 4;
+  const int nv8 = GEMM_KC / 8;
+  const int k_lo = wv * kq, k_hi = (wv + 1) * kq;
+  const int arow = l15 < M ? l15 : M - 1;
+  const int srow = lane / 4, scol8 = (lane % 4) * 4;
+  auto fetch = [&](int k0, s16x8 r[4]) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const s16x8* p = (const s16x8*)(w + (long)(n0 + srow) * K + k0 +
+                                      (scol8 + j) * 8);
+      r[j] = NTW ? __builtin_nontemporal_load(p) : *p;
+    }
+  };
+  auto put = [&](int buf, s16x8 r[4]) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      ((s16x8*)wt[wv][buf])[srow * nv8 + GSWZ(srow, scol8 + j)] = r[j];
+  };
+  s16x8 pre[4];
+  fetch(k_lo, pre);
+  put(0, pre);
+  int buf = 0;
+  for (int k0 = k_lo; k0 < k_hi; k0 += GEMM_KC) {
+    if (k0 + GEMM_KC < k_hi) fetch(k0 + GEMM_KC, pre);
+#pragma unroll
+    for (int kb = 0; kb < GEMM_KC / 32; ++kb) {
+      const s16x8 a_frag =
+          *(const s16x8*)(x + (long)arow * K + k0 + kb * 32 + lg * 8);
+      const int col8 = kb * 4 + lg;
+      const s16x8 b_frag =
+          ((s16x8*)wt[wv][buf])[l15 * nv8 + GSWZ(l15, col8)];
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc,
+                                                    0, 0, 0);
+    }
+    if (k0 + GEMM_KC < k_hi) put(buf ^ 1, pre);
+    buf ^= 1;
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) cred[wv][lg * 4 + r][l15] = acc[r];
+  __syncthreads();
+  if (wv == 0) {
+    for (int i = lane; i < M * 16; i += 64) {
+      const int m = i / 16, n = i % 16;
+      const float v = cred[0][m][n] + cred[1][m][n] + cred[2][m][n] +
+                      cred[3][m][n];
+      out[(long)m * N + n0 + n] = f2bf(v);
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" void fei_gemm_m8(void* out, const void* x, const void* w, int M,
+                            int N, int K, int nontemporal,
+                            hipStream_t stream) {
+  dim3 grid(N / GEMM_NT);
+  if (nontemporal)
+    hipLaunchKernelGGL(k_gemm_m8<true>, grid, dim3(256), 0, stream,
+                       (u16*)out, (const u16*)x, (const u16*)w, M, N, K);
+  else
+    hipLaunchKernelGGL(k_gemm_m8<false>, grid, dim3(256), 0, stream,
+                       (u16*)out, (const u16*)x, (const u16*)w, M, N, K);
+}

@@ -302,8 +302,12 @@ def test_advance(lib):
 
 def test_gemv_matches_linear(lib):
     import torch.nn.functional as F
+    # M>=4 with K%1024==0 and N%16==0 routes to the MFMA GEMM tile
+    # (k_gemm_m8); the odd-shape cases pin the VALU fallback
     for M, N, K in [(1, 4096, 4096), (1, 6144, 4096), (2, 512, 1024),
-                    (4, 128256, 4096)]:
+                    (4, 128256, 4096), (8, 4096, 4096), (8, 6144, 4096),
+                    (8, 4096, 14336), (5, 1024, 2048), (8, 120, 4096),
+                    (8, 4096, 1536), (12, 4096, 4096), (16, 4096, 4096)]:
         x = randbf(M, K, seed=80 + M)
         w = randbf(N, K, seed=81 + N % 97, scale=0.02)
         out = lib.linear_decode(x, w)
@@ -313,10 +317,11 @@ def test_gemv_matches_linear(lib):
         assert err < tol, f"M{M} N{N} K{K}: max err {err}"
 
 
-def test_gemv_swiglu_matches(lib):
+@pytest.mark.parametrize("M", [1, 8])
+def test_gemv_swiglu_matches(lib, M):
     import torch.nn.functional as F
     from fei_amd.ops import reference as ref
-    M, I, K = 1, 14336, 4096
+    I, K = 14336, 4096
     x = randbf(M, K, seed=90)
     wgu = randbf(2 * I, K, seed=91, scale=0.02)
     out = lib.gemv_swiglu(x, wgu)
