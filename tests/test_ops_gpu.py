@@ -307,7 +307,8 @@ def test_gemv_matches_linear(lib):
     for M, N, K in [(1, 4096, 4096), (1, 6144, 4096), (2, 512, 1024),
                     (4, 128256, 4096), (8, 4096, 4096), (8, 6144, 4096),
                     (8, 4096, 14336), (5, 1024, 2048), (8, 120, 4096),
-                    (8, 4096, 1536), (12, 4096, 4096), (16, 4096, 4096)]:
+                    (8, 4096, 1536), (12, 4096, 4096), (16, 4096, 4096),
+                    (4, 1024, 8192), (8, 2048, 8192)]:
         x = randbf(M, K, seed=80 + M)
         w = randbf(N, K, seed=81 + N % 97, scale=0.02)
         out = lib.linear_decode(x, w)
